@@ -1,0 +1,206 @@
+"""Model configurations for the native model zoo.
+
+The reference delegates all model structure to HuggingFace ``AutoConfig`` /
+``AutoModelForCausalLM`` (reference ``tensorlink/ml/utils.py:890``,
+``tensorlink/ml/worker.py:1122``).  The MI355X build owns its model
+definitions, so configs are explicit dataclasses; ``from_hf_config`` maps a
+HuggingFace ``config.json`` (dict or transformers object) onto them so the
+same checkpoints/skeletons remain loadable.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import json
+import os
+from dataclasses import dataclass, field
+from typing import Optional
+
+
+@dataclass
+class ModelConfig:
+    name: str = "custom"
+    vocab_size: int = 32000
+    hidden_size: int = 4096
+    intermediate_size: int = 11008
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 32
+    head_dim: Optional[int] = None          # defaults to hidden/heads
+    max_position_embeddings: int = 4096
+    rope_theta: float = 10000.0
+    rms_norm_eps: float = 1e-6
+    tie_word_embeddings: bool = False
+    attention_bias: bool = False            # Qwen2 uses qkv bias
+    qkv_bias: bool = False
+    hidden_act: str = "silu"
+    dtype: str = "bfloat16"
+    # MoE (Mixtral-style); num_local_experts == 0 => dense MLP
+    num_local_experts: int = 0
+    num_experts_per_tok: int = 2
+    # architecture tag for checkpoint key mapping
+    architecture: str = "llama"
+
+    def __post_init__(self):
+        if self.head_dim is None:
+            self.head_dim = self.hidden_size // self.num_attention_heads
+
+    @property
+    def is_moe(self) -> bool:
+        return self.num_local_experts > 0
+
+    @property
+    def q_size(self) -> int:
+        return self.num_attention_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_key_value_heads * self.head_dim
+
+    def param_count(self, include_embeddings: bool = True) -> int:
+        """Exact parameter count for this architecture (used by the planner)."""
+        h, hd = self.hidden_size, self.head_dim
+        q, kv = self.q_size, self.kv_size
+        attn = h * q + 2 * h * kv + q * h
+        if self.qkv_bias:
+            attn += q + 2 * kv
+        if self.is_moe:
+            mlp = self.num_local_experts * 3 * h * self.intermediate_size
+            mlp += h * self.num_local_experts  # router
+        else:
+            mlp = 3 * h * self.intermediate_size
+        norms = 2 * h
+        per_layer = attn + mlp + norms
+        total = self.num_hidden_layers * per_layer + h  # final norm
+        if include_embeddings:
+            total += self.vocab_size * h
+            if not self.tie_word_embeddings:
+                total += self.vocab_size * h
+        return total
+
+    def to_json(self) -> str:
+        return json.dumps(dataclasses.asdict(self))
+
+    @classmethod
+    def from_json(cls, s: str) -> "ModelConfig":
+        return cls(**json.loads(s))
+
+    @classmethod
+    def from_hf_config(cls, cfg, name: str = "custom") -> "ModelConfig":
+        """Build from a HF config object / dict (see reference
+        ``ml/utils.py:890-916`` which uses AutoConfig the same way)."""
+        if not isinstance(cfg, dict):
+            cfg = {k: getattr(cfg, k) for k in dir(cfg) if not k.startswith("_")
+                   if not callable(getattr(cfg, k, None))}
+        archs = cfg.get("architectures") or []
+        arch = "llama"
+        if any("Qwen" in a for a in archs):
+            arch = "qwen2"
+        if any("Mixtral" in a for a in archs):
+            arch = "mixtral"
+        get = cfg.get
+        return cls(
+            name=name,
+            vocab_size=get("vocab_size", 32000),
+            hidden_size=get("hidden_size", 4096),
+            intermediate_size=get("intermediate_size", 11008),
+            num_hidden_layers=get("num_hidden_layers", 32),
+            num_attention_heads=get("num_attention_heads", 32),
+            num_key_value_heads=get("num_key_value_heads",
+                                    get("num_attention_heads", 32)),
+            head_dim=get("head_dim", None),
+            max_position_embeddings=get("max_position_embeddings", 4096),
+            rope_theta=get("rope_theta", 10000.0),
+            rms_norm_eps=get("rms_norm_eps", 1e-6),
+            tie_word_embeddings=get("tie_word_embeddings", False),
+            qkv_bias=(arch == "qwen2" and "Qwen3" not in str(archs)),
+            num_local_experts=get("num_local_experts", 0) or 0,
+            num_experts_per_tok=get("num_experts_per_tok", 2) or 2,
+            architecture=arch,
+        )
+
+
+def _qwen25_7b() -> ModelConfig:
+    return ModelConfig(
+        name="Qwen/Qwen2.5-7B-Instruct", vocab_size=152064, hidden_size=3584,
+        intermediate_size=18944, num_hidden_layers=28, num_attention_heads=28,
+        num_key_value_heads=4, max_position_embeddings=32768,
+        rope_theta=1000000.0, rms_norm_eps=1e-6, tie_word_embeddings=False,
+        qkv_bias=True, architecture="qwen2")
+
+
+def _qwen3_8b() -> ModelConfig:
+    return ModelConfig(
+        name="Qwen/Qwen3-8B", vocab_size=151936, hidden_size=4096,
+        intermediate_size=12288, num_hidden_layers=36, num_attention_heads=32,
+        num_key_value_heads=8, head_dim=128, max_position_embeddings=40960,
+        rope_theta=1000000.0, rms_norm_eps=1e-6, tie_word_embeddings=False,
+        qkv_bias=False, architecture="qwen3")
+
+
+def _llama3_70b() -> ModelConfig:
+    return ModelConfig(
+        name="meta-llama/Llama-3-70B", vocab_size=128256, hidden_size=8192,
+        intermediate_size=28672, num_hidden_layers=80, num_attention_heads=64,
+        num_key_value_heads=8, max_position_embeddings=8192,
+        rope_theta=500000.0, rms_norm_eps=1e-5, tie_word_embeddings=False,
+        architecture="llama")
+
+
+def _llama3_8b() -> ModelConfig:
+    return ModelConfig(
+        name="meta-llama/Llama-3-8B", vocab_size=128256, hidden_size=4096,
+        intermediate_size=14336, num_hidden_layers=32, num_attention_heads=32,
+        num_key_value_heads=8, max_position_embeddings=8192,
+        rope_theta=500000.0, rms_norm_eps=1e-5, architecture="llama")
+
+
+def _mixtral_8x7b() -> ModelConfig:
+    return ModelConfig(
+        name="mistralai/Mixtral-8x7B-v0.1", vocab_size=32000, hidden_size=4096,
+        intermediate_size=14336, num_hidden_layers=32, num_attention_heads=32,
+        num_key_value_heads=8, max_position_embeddings=32768,
+        rope_theta=1000000.0, rms_norm_eps=1e-5, num_local_experts=8,
+        num_experts_per_tok=2, architecture="mixtral")
+
+
+def _tiny() -> ModelConfig:
+    # CI-scale stand-in for sshleifer/tiny-gpt2-class smoke models
+    # (reference tests/test_distributed_model.py:24).
+    return ModelConfig(
+        name="tiny", vocab_size=1024, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=512, rope_theta=10000.0, architecture="llama")
+
+
+def _tiny_moe() -> ModelConfig:
+    return ModelConfig(
+        name="tiny-moe", vocab_size=1024, hidden_size=256,
+        intermediate_size=384, num_hidden_layers=4, num_attention_heads=4,
+        num_key_value_heads=2, max_position_embeddings=512,
+        num_local_experts=4, num_experts_per_tok=2, architecture="mixtral")
+
+
+PRESETS = {
+    "Qwen/Qwen2.5-7B-Instruct": _qwen25_7b,
+    "Qwen/Qwen2.5-7B": _qwen25_7b,
+    "Qwen/Qwen3-8B": _qwen3_8b,
+    "meta-llama/Llama-3-70B": _llama3_70b,
+    "meta-llama/Llama-3-8B": _llama3_8b,
+    "mistralai/Mixtral-8x7B-v0.1": _mixtral_8x7b,
+    "tiny": _tiny,
+    "tiny-moe": _tiny_moe,
+}
+
+
+def get_config(name: str) -> ModelConfig:
+    """Resolve a model name to a config: preset table, then a local HF
+    checkpoint directory with a config.json (no hub access on this node)."""
+    if name in PRESETS:
+        return PRESETS[name]()
+    if os.path.isdir(name) and os.path.exists(os.path.join(name, "config.json")):
+        with open(os.path.join(name, "config.json")) as f:
+            return ModelConfig.from_hf_config(json.load(f), name=name)
+    raise KeyError(
+        f"unknown model {name!r}: not a preset and no local checkpoint dir "
+        f"(no network access for hub downloads)")

@@ -1,0 +1,273 @@
+"""Native decoder model zoo (Llama / Qwen2 / Qwen3 dense families).
+
+The reference runs HuggingFace ``transformers`` modules eagerly per stage
+(``tensorlink/ml/worker.py:326-335``) and ships rotary buffers / KV caches
+over the network. Here the decoder is defined natively on top of the
+CDNA4 op library (:mod:`tensorlink_amd.ops`): fused residual+RMSNorm, RoPE
+computed on device, flash prefill + KV-cache decode attention, fused
+SwiGLU — GEMMs go through torch.matmul (hipBLASLt on ROCm).
+
+A model instance holds a contiguous *slice* of layers plus optionally the
+embedding (first stage) and final-norm/lm_head (last stage) — the unit the
+planner assigns to one GPU rank (cf. reference grouped entries
+``model.layers.N-M``, ``ml/graphing.py:64-128``).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from tensorlink_amd import ops
+from tensorlink_amd.models.configs import ModelConfig
+
+
+class KVCache:
+    """Device-resident contiguous KV cache for one stage.
+
+    Layout [B, Hkv, Smax, D] per layer (head-major so decode streams a
+    contiguous [L, D] slab per (batch, head) — see decode_attn.hip).
+    The reference serializes its HF DynamicCache over TCP every step
+    (``ml/utils.py:210-221``); this cache never leaves the GPU.
+    """
+
+    def __init__(self, n_layers: int, batch: int, max_seq: int, config,
+                 device, dtype=torch.bfloat16):
+        self.k = [torch.zeros(batch, config.num_key_value_heads, max_seq,
+                              config.head_dim, device=device, dtype=dtype)
+                  for _ in range(n_layers)]
+        self.v = [torch.zeros_like(self.k[0]) for _ in range(n_layers)]
+        self.seq_lens = torch.zeros(batch, device=device, dtype=torch.int32)
+        self.max_seq = max_seq
+        self.batch = batch
+
+    def reset(self):
+        self.seq_lens.zero_()
+
+    def append(self, layer: int, k_new: torch.Tensor, v_new: torch.Tensor,
+               positions: torch.Tensor):
+        """Write new K/V at given positions. k_new [B, S_new, Hkv, D];
+        positions [B, S_new] int."""
+        B, S_new, Hkv, D = k_new.shape
+        idx = positions.to(self.k[layer].device, torch.long)
+        idx = idx.view(B, 1, S_new, 1).expand(B, Hkv, S_new, D)
+        self.k[layer].scatter_(2, idx, k_new.transpose(1, 2))
+        self.v[layer].scatter_(2, idx, v_new.transpose(1, 2))
+
+    def advance(self, n: int):
+        self.seq_lens += n
+
+
+class Attention(nn.Module):
+    def __init__(self, config: ModelConfig):
+        super().__init__()
+        self.config = config
+        h, q, kv = config.hidden_size, config.q_size, config.kv_size
+        bias = config.qkv_bias
+        self.q_proj = nn.Linear(h, q, bias=bias)
+        self.k_proj = nn.Linear(h, kv, bias=bias)
+        self.v_proj = nn.Linear(h, kv, bias=bias)
+        self.o_proj = nn.Linear(q, h, bias=False)
+        self.n_heads = config.num_attention_heads
+        self.n_kv = config.num_key_value_heads
+        self.head_dim = config.head_dim
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+        # Qwen3 applies RMSNorm to q/k per head
+        self.use_qk_norm = config.architecture == "qwen3"
+        if self.use_qk_norm:
+            self.q_norm = nn.Parameter(torch.ones(self.head_dim))
+            self.k_norm = nn.Parameter(torch.ones(self.head_dim))
+        inv_freq = 1.0 / (config.rope_theta ** (
+            torch.arange(0, self.head_dim, 2, dtype=torch.float32)
+            / self.head_dim))
+        self.register_buffer("inv_freq", inv_freq, persistent=False)
+
+    def forward(self, x: torch.Tensor, positions: torch.Tensor,
+                kv_cache: Optional[KVCache] = None, layer_idx: int = 0,
+                training: bool = False) -> torch.Tensor:
+        B, S, H = x.shape
+        q = self.q_proj(x).view(B, S, self.n_heads, self.head_dim)
+        k = self.k_proj(x).view(B, S, self.n_kv, self.head_dim)
+        v = self.v_proj(x).view(B, S, self.n_kv, self.head_dim)
+        if self.use_qk_norm:
+            q = ops.rmsnorm(q, self.q_norm.to(q.dtype),
+                            self.config.rms_norm_eps)
+            k = ops.rmsnorm(k, self.k_norm.to(k.dtype),
+                            self.config.rms_norm_eps)
+
+        flat_pos = positions.reshape(-1)
+        if training:
+            q2, k2 = ops.apply_rope(q.reshape(B * S, self.n_heads, -1),
+                                    k.reshape(B * S, self.n_kv, -1),
+                                    flat_pos, self.inv_freq)
+            q = q2.view(B, S, self.n_heads, -1)
+            k = k2.view(B, S, self.n_kv, -1)
+            out = ops.attention_train(q, k, v, causal=True, scale=self.scale)
+        else:
+            q = q.contiguous()
+            k = k.contiguous()
+            ops.apply_rope_(q.view(B * S, self.n_heads, -1),
+                            k.view(B * S, self.n_kv, -1), flat_pos,
+                            self.inv_freq)
+            if kv_cache is None:
+                out = ops.attention_prefill(q, k, v, causal=True,
+                                            scale=self.scale)
+            else:
+                kv_cache.append(layer_idx, k, v, positions)
+                if S == 1:
+                    out = ops.attention_decode(
+                        q, kv_cache.k[layer_idx], kv_cache.v[layer_idx],
+                        kv_cache.seq_lens + 1, scale=self.scale)
+                else:
+                    # prefill with cache write (fresh prompt: cache was empty)
+                    out = ops.attention_prefill(q, k, v, causal=True,
+                                                scale=self.scale)
+        return self.o_proj(out.reshape(B, S, -1))
+
+
+class MLP(nn.Module):
+    def __init__(self, config: ModelConfig):
+        super().__init__()
+        h, i = config.hidden_size, config.intermediate_size
+        self.gate_proj = nn.Linear(h, i, bias=False)
+        self.up_proj = nn.Linear(h, i, bias=False)
+        self.down_proj = nn.Linear(i, h, bias=False)
+
+    def forward(self, x):
+        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class MoEMLP(nn.Module):
+    """Mixtral-style sparse MoE block (new capability — the reference has no
+    MoE-aware code)."""
+
+    def __init__(self, config: ModelConfig):
+        super().__init__()
+        h, i = config.hidden_size, config.intermediate_size
+        self.num_experts = config.num_local_experts
+        self.top_k = config.num_experts_per_tok
+        self.gate = nn.Linear(h, self.num_experts, bias=False)
+        self.experts = nn.ModuleList([MLP(config) for _ in range(self.num_experts)])
+
+    def forward(self, x):
+        B, S, H = x.shape
+        flat = x.reshape(-1, H)
+        weights, idx = ops.moe_topk_router(self.gate(flat), self.top_k)
+        weights = weights.to(x.dtype)
+        out = torch.zeros_like(flat)
+        for e in range(self.num_experts):
+            mask = (idx == e)
+            tok, slot = mask.nonzero(as_tuple=True)
+            if tok.numel() == 0:
+                continue
+            out.index_add_(0, tok,
+                           self.experts[e](flat[tok]) * weights[tok, slot, None])
+        return out.reshape(B, S, H)
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, config: ModelConfig):
+        super().__init__()
+        self.config = config
+        self.input_layernorm = nn.Parameter(torch.ones(config.hidden_size))
+        self.post_attention_layernorm = nn.Parameter(
+            torch.ones(config.hidden_size))
+        self.self_attn = Attention(config)
+        self.mlp = MoEMLP(config) if config.is_moe else MLP(config)
+
+    def forward(self, hidden, positions, kv_cache=None, layer_idx=0,
+                training=False):
+        eps = self.config.rms_norm_eps
+        w_in = self.input_layernorm.to(hidden.dtype)
+        w_post = self.post_attention_layernorm.to(hidden.dtype)
+        if training or not hidden.is_cuda:
+            h = ops.rmsnorm(hidden, w_in, eps)
+            hidden = hidden + self.self_attn(h, positions, kv_cache,
+                                             layer_idx, training)
+            h = ops.rmsnorm(hidden, w_post, eps)
+            hidden = hidden + self.mlp(h)
+            return hidden
+        # inference: fused residual + norm (one HBM round trip)
+        h = ops.rmsnorm(hidden, w_in, eps)
+        attn_out = self.self_attn(h, positions, kv_cache, layer_idx, False)
+        h, hidden = ops.rmsnorm_residual(attn_out, hidden, w_post, eps)
+        return hidden + self.mlp(h)
+
+
+class StageModel(nn.Module):
+    """A contiguous slice of the decoder living on one rank."""
+
+    def __init__(self, config: ModelConfig, layer_start: int, layer_end: int,
+                 has_embedding: bool, has_head: bool):
+        super().__init__()
+        self.config = config
+        self.layer_start = layer_start
+        self.layer_end = layer_end
+        self.has_embedding = has_embedding
+        self.has_head = has_head
+        if has_embedding:
+            self.embed_tokens = nn.Embedding(config.vocab_size,
+                                             config.hidden_size)
+        self.layers = nn.ModuleList(
+            [DecoderLayer(config) for _ in range(layer_end - layer_start)])
+        if has_head:
+            self.norm = nn.Parameter(torch.ones(config.hidden_size))
+            # Tied embeddings share storage only when the embedding lives on
+            # this same stage; with PP>1 the last stage materializes its own
+            # copy of the tied weight (the reference instead pins tied
+            # modules to the host — graphing.py:532-537).
+            if not (config.tie_word_embeddings and has_embedding):
+                self.lm_head = nn.Linear(config.hidden_size,
+                                         config.vocab_size, bias=False)
+
+    @property
+    def num_layers(self):
+        return len(self.layers)
+
+    def make_kv_cache(self, batch: int, max_seq: int, device,
+                      dtype=None) -> KVCache:
+        if dtype is None:
+            dtype = next(self.parameters()).dtype
+        return KVCache(self.num_layers, batch, max_seq, self.config, device,
+                       dtype)
+
+    def embed(self, input_ids: torch.Tensor) -> torch.Tensor:
+        w = self.embed_tokens.weight
+        return self.embed_tokens(input_ids).to(w.dtype)
+
+    def head(self, hidden: torch.Tensor) -> torch.Tensor:
+        h = ops.rmsnorm(hidden, self.norm.to(hidden.dtype),
+                        self.config.rms_norm_eps)
+        if self.config.tie_word_embeddings and self.has_embedding:
+            return h @ self.embed_tokens.weight.t()
+        return self.lm_head(h)
+
+    def forward(self, hidden_or_ids, positions, kv_cache: Optional[KVCache]
+                = None, training: bool = False,
+                return_logits: bool = True) -> torch.Tensor:
+        if self.has_embedding and hidden_or_ids.dtype in (torch.int32,
+                                                          torch.int64):
+            hidden = self.embed(hidden_or_ids)
+        else:
+            hidden = hidden_or_ids
+        for i, layer in enumerate(self.layers):
+            hidden = layer(hidden, positions, kv_cache, i, training)
+        if kv_cache is not None:
+            kv_cache.advance(hidden.shape[1])
+        if self.has_head and return_logits:
+            return self.head(hidden)
+        return hidden
+
+
+def build_full_model(config: ModelConfig) -> StageModel:
+    """Whole model as a single stage (PP=1 / whole-model offload — the
+    reference's ``entire_model`` path, ``ml/module.py:894-897``)."""
+    return StageModel(config, 0, config.num_hidden_layers, True, True)
+
+
+def build_stage(config: ModelConfig, spec) -> StageModel:
+    return StageModel(config, spec.layer_start, spec.layer_end,
+                      spec.has_embedding, spec.has_head)

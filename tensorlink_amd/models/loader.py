@@ -1,0 +1,163 @@
+"""Per-stage checkpoint loading + random init.
+
+Keeps the reference's checkpoint story (HF layout: config.json +
+safetensors shards) and its per-stage shard scan with prefix remap
+``model.layers.N.* -> layers.i.*`` (reference ``ml/worker.py:542-638,
+589-616``) — but loads from a *local* directory (no hub on this node) and
+maps onto the native :class:`StageModel` parameter names.
+
+Random init (`init_random_stage`) builds the synthetic-weight models used by
+bench.py / smoke (BASELINE.json: "synthetic data / random-init weights").
+"""
+
+from __future__ import annotations
+
+import json
+import math
+import os
+from typing import Dict, Optional
+
+import torch
+
+from tensorlink_amd.models.configs import ModelConfig
+from tensorlink_amd.models.dense import StageModel, build_stage
+
+
+# HF parameter name -> native name, relative to one decoder layer
+_HF_LAYER_MAP = {
+    "input_layernorm.weight": "input_layernorm",
+    "post_attention_layernorm.weight": "post_attention_layernorm",
+    "self_attn.q_proj.weight": "self_attn.q_proj.weight",
+    "self_attn.q_proj.bias": "self_attn.q_proj.bias",
+    "self_attn.k_proj.weight": "self_attn.k_proj.weight",
+    "self_attn.k_proj.bias": "self_attn.k_proj.bias",
+    "self_attn.v_proj.weight": "self_attn.v_proj.weight",
+    "self_attn.v_proj.bias": "self_attn.v_proj.bias",
+    "self_attn.o_proj.weight": "self_attn.o_proj.weight",
+    "self_attn.q_norm.weight": "self_attn.q_norm",
+    "self_attn.k_norm.weight": "self_attn.k_norm",
+    "mlp.gate_proj.weight": "mlp.gate_proj.weight",
+    "mlp.up_proj.weight": "mlp.up_proj.weight",
+    "mlp.down_proj.weight": "mlp.down_proj.weight",
+    # Mixtral MoE
+    "block_sparse_moe.gate.weight": "mlp.gate.weight",
+}
+
+
+def _map_hf_key(key: str, layer_start: int, layer_end: int,
+                stage: StageModel) -> Optional[str]:
+    """Map an HF checkpoint key to a native stage parameter name (or None if
+    the key belongs to another stage)."""
+    if key.startswith("model.embed_tokens.weight"):
+        if stage.has_embedding:
+            return "embed_tokens.weight"
+        if stage.has_head and stage.config.tie_word_embeddings:
+            return "lm_head.weight"
+        return None
+    if key == "model.norm.weight":
+        return "norm" if stage.has_head else None
+    if key == "lm_head.weight":
+        return "lm_head.weight" if (stage.has_head and
+                                    hasattr(stage, "lm_head")) else None
+    if key.startswith("model.layers."):
+        rest = key[len("model.layers."):]
+        idx_str, _, sub = rest.partition(".")
+        idx = int(idx_str)
+        if not (layer_start <= idx < layer_end):
+            return None
+        local = idx - layer_start
+        if sub in _HF_LAYER_MAP:
+            return f"layers.{local}.{_HF_LAYER_MAP[sub]}"
+        # Mixtral experts: block_sparse_moe.experts.E.w1/w3/w2
+        if sub.startswith("block_sparse_moe.experts."):
+            parts = sub.split(".")
+            e = parts[2]
+            wname = parts[3]
+            proj = {"w1": "gate_proj", "w3": "up_proj", "w2": "down_proj"}[wname]
+            return f"layers.{local}.mlp.experts.{e}.{proj}.weight"
+    return None
+
+
+def load_stage_from_checkpoint(stage: StageModel, ckpt_dir: str,
+                               device="cpu", dtype=torch.bfloat16) -> int:
+    """Load only this stage's parameters from a local HF checkpoint dir.
+    Returns the number of tensors loaded."""
+    from safetensors import safe_open
+
+    index_path = os.path.join(ckpt_dir, "model.safetensors.index.json")
+    shards: Dict[str, list] = {}
+    if os.path.exists(index_path):
+        with open(index_path) as f:
+            weight_map = json.load(f)["weight_map"]
+        for key, shard in weight_map.items():
+            shards.setdefault(shard, []).append(key)
+    else:
+        single = os.path.join(ckpt_dir, "model.safetensors")
+        if not os.path.exists(single):
+            raise FileNotFoundError(f"no safetensors in {ckpt_dir}")
+        shards = {"model.safetensors": None}
+
+    params = dict(stage.named_parameters())
+    loaded = 0
+    for shard, keys in shards.items():
+        path = os.path.join(ckpt_dir, shard)
+        with safe_open(path, framework="pt", device="cpu") as f:
+            shard_keys = keys if keys is not None else f.keys()
+            for key in shard_keys:
+                native = _map_hf_key(key, stage.layer_start, stage.layer_end,
+                                     stage)
+                if native is None or native not in params:
+                    continue
+                t = f.get_tensor(key).to(dtype)
+                with torch.no_grad():
+                    params[native].copy_(t)
+                loaded += 1
+    stage.to(device=device, dtype=dtype)
+    return loaded
+
+
+@torch.no_grad()
+def init_random_stage(stage: StageModel, device="cpu", dtype=torch.bfloat16,
+                      seed: int = 0) -> StageModel:
+    """Deterministic scaled-normal init directly on the target device."""
+    gen_dev = device if str(device).startswith("cuda") else "cpu"
+    g = torch.Generator(device=gen_dev)
+    g.manual_seed(seed)
+    stage.to(device=device, dtype=dtype)
+    std = 1.0 / math.sqrt(stage.config.hidden_size)
+    for name, p in stage.named_parameters():
+        if "layernorm" in name or name.endswith(("norm", "q_norm", "k_norm")):
+            p.fill_(1.0)
+        elif name.endswith(".bias"):
+            p.zero_()
+        else:
+            tmp = torch.empty(p.shape, device=gen_dev, dtype=torch.float32)
+            tmp.normal_(0.0, std, generator=g)
+            p.copy_(tmp.to(dtype))
+    return stage
+
+
+def save_stage_to_safetensors(stage: StageModel, out_dir: str,
+                              rank: int) -> str:
+    """Checkpoint dump (reference parity: parameter retrieval →
+    ``models/<name>/`` safetensors, ``ml/module.py:577-670``)."""
+    from safetensors.torch import save_file
+    os.makedirs(out_dir, exist_ok=True)
+    path = os.path.join(out_dir, f"stage_{rank}.safetensors")
+    state = {k: v.detach().cpu().contiguous()
+             for k, v in stage.state_dict().items()}
+    save_file(state, path)
+    with open(os.path.join(out_dir, f"stage_{rank}.json"), "w") as f:
+        json.dump({"layer_start": stage.layer_start,
+                   "layer_end": stage.layer_end,
+                   "has_embedding": stage.has_embedding,
+                   "has_head": stage.has_head,
+                   "config": stage.config.to_json()}, f)
+    return path
+
+
+def load_stage_from_safetensors(stage: StageModel, out_dir: str,
+                                rank: int) -> None:
+    from safetensors.torch import load_file
+    state = load_file(os.path.join(out_dir, f"stage_{rank}.safetensors"))
+    stage.load_state_dict(state)

@@ -1,0 +1,234 @@
+"""Op dispatch: hand-written CDNA4 HIP kernels on GPU, PyTorch reference on CPU.
+
+Policy (required by the build contract): on a GPU box the HIP extension MUST
+be loadable — a missing extension raises instead of silently falling back to
+eager PyTorch. On CPU-only hosts (the test tier) the fp32 reference
+implementations in :mod:`tensorlink_amd.ops.reference` run instead.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from tensorlink_amd.ops import reference as ref
+
+_C = None
+_IMPORT_ERROR: Optional[BaseException] = None
+try:
+    from tensorlink_amd import _C  # type: ignore  # built by setup.py
+except Exception as e:  # pragma: no cover - absent on CPU CI before build
+    _IMPORT_ERROR = e
+
+
+def extension_loaded() -> bool:
+    return _C is not None
+
+
+def _require_ext():
+    if _C is None:
+        raise RuntimeError(
+            "tensorlink_amd._C HIP extension is not built but a CUDA/HIP "
+            "tensor reached the op layer. Build it in-tree with "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace` "
+            f"(import error: {_IMPORT_ERROR!r})")
+    return _C
+
+
+def _on_gpu(*tensors: torch.Tensor) -> bool:
+    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm (autograd-aware)
+# ---------------------------------------------------------------------------
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        C = _require_ext()
+        need_grad = x.requires_grad or weight.requires_grad
+        outs = C.rmsnorm_fwd(x.contiguous(), None, weight.contiguous(), eps,
+                             need_grad)
+        y = outs[0]
+        if need_grad:
+            ctx.save_for_backward(x, weight, outs[1])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = _require_ext()
+        x, weight, rstd = ctx.saved_tensors
+        dx, dw = C.rmsnorm_bwd(dy.contiguous(), x.contiguous(),
+                               weight.contiguous(), rstd)
+        return dx, dw, None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6):
+    if not _on_gpu(x):
+        return ref.rmsnorm(x, weight, eps)
+    if torch.is_grad_enabled() and (x.requires_grad or weight.requires_grad):
+        return _RMSNormFn.apply(x, weight, eps)
+    return _require_ext().rmsnorm_fwd(x.contiguous(), None,
+                                      weight.contiguous(), eps, False)[0]
+
+
+def rmsnorm_residual(x: torch.Tensor, residual: torch.Tensor,
+                     weight: torch.Tensor, eps: float = 1e-6):
+    """Fused residual add + RMSNorm (inference path). Returns (y, r)."""
+    if not _on_gpu(x):
+        return ref.rmsnorm_residual(x, residual, weight, eps)
+    outs = _require_ext().rmsnorm_fwd(x.contiguous(), residual.contiguous(),
+                                      weight.contiguous(), eps, False)
+    return outs[0], outs[1]
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+def apply_rope_(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
+                inv_freq: torch.Tensor, sign: float = 1.0) -> None:
+    """In-place RoPE on q [T,Hq,D] / k [T,Hkv,D] with positions [T].
+
+    Inference path (no autograd). sign=-1 applies the inverse rotation.
+    """
+    if not _on_gpu(q):
+        T, Hq, D = q.shape
+        cos, sin = ref.rope_cos_sin(D, positions, device=q.device)
+        # reference expects [B,S,H,D]
+        qq, kk = ref.apply_rope(q.unsqueeze(0), k.unsqueeze(0), cos,
+                                sin * sign)
+        q.copy_(qq[0])
+        k.copy_(kk[0])
+        return
+    _require_ext().rope_(q, k, positions.int(), inv_freq.float(), sign)
+
+
+class _RopeFn(torch.autograd.Function):
+    """Autograd RoPE for the training path (operates out-of-place)."""
+
+    @staticmethod
+    def forward(ctx, q, k, positions, inv_freq):
+        q2, k2 = q.contiguous().clone(), k.contiguous().clone()
+        apply_rope_(q2, k2, positions, inv_freq, 1.0)
+        ctx.save_for_backward(positions, inv_freq)
+        return q2, k2
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        positions, inv_freq = ctx.saved_tensors
+        dq2, dk2 = dq.contiguous().clone(), dk.contiguous().clone()
+        apply_rope_(dq2, dk2, positions, inv_freq, -1.0)
+        return dq2, dk2, None, None
+
+
+def apply_rope(q, k, positions, inv_freq):
+    return _RopeFn.apply(q, k, positions, inv_freq)
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        C = _require_ext()
+        ctx.save_for_backward(gate, up)
+        return C.swiglu_fwd(gate.contiguous(), up.contiguous())
+
+    @staticmethod
+    def backward(ctx, dout):
+        C = _require_ext()
+        gate, up = ctx.saved_tensors
+        dg, du = C.swiglu_bwd(dout.contiguous(), gate.contiguous(),
+                              up.contiguous())
+        return dg, du
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if not _on_gpu(gate):
+        return ref.swiglu(gate, up)
+    if torch.is_grad_enabled() and (gate.requires_grad or up.requires_grad):
+        return _SwiGLUFn.apply(gate, up)
+    return _require_ext().swiglu_fwd(gate.contiguous(), up.contiguous())
+
+
+# ---------------------------------------------------------------------------
+# Attention
+# ---------------------------------------------------------------------------
+def attention_prefill(q, k, v, causal: bool = True,
+                      scale: Optional[float] = None) -> torch.Tensor:
+    """q [B,S,Hq,D], k/v [B,S,Hkv,D] -> [B,S,Hq,D]. Inference path."""
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if not _on_gpu(q):
+        return ref.attention_prefill(q, k, v, causal, scale)
+    return _require_ext().prefill_attn(q.contiguous(), k.contiguous(),
+                                       v.contiguous(), scale, causal)
+
+
+def attention_train(q, k, v, causal: bool = True,
+                    scale: Optional[float] = None) -> torch.Tensor:
+    """Training attention with autograd: composed torch SDPA (backed by
+    hipBLASLt GEMMs + ROCm flash backward). Hand-written backward kernels
+    are a later milestone; forward inference never uses this."""
+    B, S, Hq, D = q.shape
+    Hkv = k.shape[2]
+    rep = Hq // Hkv
+    qt = q.transpose(1, 2)
+    kt = k.transpose(1, 2).repeat_interleave(rep, dim=1)
+    vt = v.transpose(1, 2).repeat_interleave(rep, dim=1)
+    out = torch.nn.functional.scaled_dot_product_attention(
+        qt, kt, vt, is_causal=causal, scale=scale)
+    return out.transpose(1, 2)
+
+
+def attention_decode(q, k_cache, v_cache, seq_lens,
+                     scale: Optional[float] = None) -> torch.Tensor:
+    """q [B,1,Hq,D] or [B,Hq,D]; caches [B,Hkv,Smax,D]; seq_lens [B]."""
+    squeeze = q.dim() == 4
+    if squeeze:
+        q3 = q.squeeze(1)
+    else:
+        q3 = q
+    scale = scale if scale is not None else 1.0 / math.sqrt(q3.shape[-1])
+    if not _on_gpu(q3):
+        # reference wants [B,Smax,Hkv,D]
+        out = ref.attention_decode(q3.unsqueeze(1),
+                                   k_cache.permute(0, 2, 1, 3),
+                                   v_cache.permute(0, 2, 1, 3), seq_lens,
+                                   scale)
+        return out if squeeze else out.squeeze(1)
+    out = _require_ext().decode_attn(q3.contiguous(), k_cache, v_cache,
+                                     seq_lens.int(), scale)
+    return out.unsqueeze(1) if squeeze else out
+
+
+# ---------------------------------------------------------------------------
+# AdamW
+# ---------------------------------------------------------------------------
+def adamw_(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,
+           exp_avg_sq: torch.Tensor, *, lr: float, beta1: float = 0.9,
+           beta2: float = 0.999, eps: float = 1e-8,
+           weight_decay: float = 0.0, step: int = 1) -> None:
+    if not _on_gpu(param):
+        ref.adamw_step(param, grad, exp_avg, exp_avg_sq, lr=lr, beta1=beta1,
+                       beta2=beta2, eps=eps, weight_decay=weight_decay,
+                       step=step)
+        return
+    _require_ext().adamw_(param, grad.contiguous(), exp_avg, exp_avg_sq, lr,
+                          beta1, beta2, eps, weight_decay, step)
+
+
+# ---------------------------------------------------------------------------
+# Sampling (torch-composed on GPU for now; fused kernel is a later milestone)
+# ---------------------------------------------------------------------------
+def sample_token(logits, *, temperature=1.0, top_p=1.0, top_k=0,
+                 generator=None):
+    return ref.sample_token(logits, temperature=temperature, top_p=top_p,
+                            top_k=top_k, generator=generator)
+
+
+moe_topk_router = ref.moe_topk_router
+causal_lm_loss = ref.causal_lm_loss
+rope_cos_sin = ref.rope_cos_sin

@@ -1,0 +1,184 @@
+// Python bindings for the tensorlink_amd CDNA4 kernel library.
+// Host-side glue only — every kernel lives in the sibling .hip files and is
+// reached through an extern "C" launcher taking raw pointers + hipStream_t.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void tl_rmsnorm_fwd(const void* x, const void* residual, const void* w,
+                    void* y, void* r_out, void* rstd, int64_t N, int H,
+                    float eps, hipStream_t stream);
+void tl_rmsnorm_bwd(const void* dy, const void* x, const void* w,
+                    const void* rstd, void* dx, void* dw, void* dw_partial,
+                    int n_blocks, int64_t N, int H, hipStream_t stream);
+void tl_rope(void* q, void* k, const void* positions, const void* inv_freq,
+             int64_t T, int Hq, int Hkv, int D, float sign,
+             hipStream_t stream);
+void tl_swiglu_fwd(const void* gate, const void* up, void* out, int64_t n,
+                   hipStream_t stream);
+void tl_swiglu_bwd(const void* dout, const void* gate, const void* up,
+                   void* dgate, void* dup, int64_t n, hipStream_t stream);
+void tl_adamw(void* param, const void* grad, void* m, void* v, int64_t n,
+              int is_bf16, float lr, float beta1, float beta2, float eps,
+              float weight_decay, int step, hipStream_t stream);
+void tl_decode_attn(const void* q, const void* k_cache, const void* v_cache,
+                    const void* seq_lens, void* out, int B, int Hq, int Hkv,
+                    int Smax, int D, float scale, hipStream_t stream);
+void tl_prefill_attn(const void* q, const void* k, const void* v, void* out,
+                     int B, int S, int Hq, int Hkv, int D, float scale,
+                     int causal, hipStream_t stream);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_IN(t, d)                                              \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                 \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");       \
+  TORCH_CHECK((t).scalar_type() == (d), #t " wrong dtype")
+
+using torch::Tensor;
+
+std::vector<Tensor> rmsnorm_fwd(Tensor x, c10::optional<Tensor> residual,
+                                Tensor w, double eps, bool save_rstd) {
+  CHECK_IN(x, torch::kBFloat16);
+  CHECK_IN(w, torch::kBFloat16);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "H must be divisible by 8");
+  const int64_t N = x.numel() / H;
+  auto y = torch::empty_like(x);
+  Tensor rstd, r_out;
+  if (save_rstd) rstd = torch::empty({N}, x.options().dtype(torch::kFloat));
+  const void* res_ptr = nullptr;
+  void* rout_ptr = nullptr;
+  if (residual.has_value()) {
+    CHECK_IN(residual.value(), torch::kBFloat16);
+    r_out = torch::empty_like(x);
+    res_ptr = residual->data_ptr();
+    rout_ptr = r_out.data_ptr();
+  }
+  tl_rmsnorm_fwd(x.data_ptr(), res_ptr, w.data_ptr(), y.data_ptr(), rout_ptr,
+                 save_rstd ? rstd.data_ptr() : nullptr, N, H, (float)eps,
+                 cur_stream());
+  std::vector<Tensor> outs = {y};
+  if (residual.has_value()) outs.push_back(r_out);
+  if (save_rstd) outs.push_back(rstd);
+  return outs;
+}
+
+std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor rstd) {
+  CHECK_IN(dy, torch::kBFloat16);
+  CHECK_IN(x, torch::kBFloat16);
+  CHECK_IN(w, torch::kBFloat16);
+  CHECK_IN(rstd, torch::kFloat);
+  const int H = x.size(-1);
+  const int64_t N = x.numel() / H;
+  const int n_blocks = (int)std::min<int64_t>(N, 512);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::empty_like(w);
+  auto partial = torch::empty({n_blocks, (int64_t)H},
+                              x.options().dtype(torch::kFloat));
+  tl_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(), rstd.data_ptr(),
+                 dx.data_ptr(), dw.data_ptr(), partial.data_ptr(), n_blocks,
+                 N, H, cur_stream());
+  return {dx, dw};
+}
+
+void rope_(Tensor q, Tensor k, Tensor positions, Tensor inv_freq,
+           double sign) {
+  CHECK_IN(q, torch::kBFloat16);
+  CHECK_IN(k, torch::kBFloat16);
+  CHECK_IN(positions, torch::kInt);
+  CHECK_IN(inv_freq, torch::kFloat);
+  const int D = q.size(-1);
+  const int Hq = q.size(-2);
+  const int Hkv = k.size(-2);
+  const int64_t T = q.numel() / ((int64_t)Hq * D);
+  TORCH_CHECK(k.numel() / ((int64_t)Hkv * D) == T, "q/k token mismatch");
+  TORCH_CHECK(positions.numel() == T, "positions size mismatch");
+  tl_rope(q.data_ptr(), k.data_ptr(), positions.data_ptr(),
+          inv_freq.data_ptr(), T, Hq, Hkv, D, (float)sign, cur_stream());
+}
+
+Tensor swiglu_fwd(Tensor gate, Tensor up) {
+  CHECK_IN(gate, torch::kBFloat16);
+  CHECK_IN(up, torch::kBFloat16);
+  TORCH_CHECK(gate.numel() % 8 == 0, "numel must be divisible by 8");
+  auto out = torch::empty_like(gate);
+  tl_swiglu_fwd(gate.data_ptr(), up.data_ptr(), out.data_ptr(), gate.numel(),
+                cur_stream());
+  return out;
+}
+
+std::vector<Tensor> swiglu_bwd(Tensor dout, Tensor gate, Tensor up) {
+  CHECK_IN(dout, torch::kBFloat16);
+  auto dg = torch::empty_like(gate);
+  auto du = torch::empty_like(up);
+  tl_swiglu_bwd(dout.data_ptr(), gate.data_ptr(), up.data_ptr(), dg.data_ptr(),
+                du.data_ptr(), gate.numel(), cur_stream());
+  return {dg, du};
+}
+
+void adamw_(Tensor param, Tensor grad, Tensor m, Tensor v, double lr,
+            double beta1, double beta2, double eps, double weight_decay,
+            int64_t step) {
+  TORCH_CHECK(param.is_cuda() && param.is_contiguous());
+  TORCH_CHECK(m.scalar_type() == torch::kFloat &&
+              v.scalar_type() == torch::kFloat);
+  TORCH_CHECK(grad.scalar_type() == param.scalar_type());
+  const bool is_bf16 = param.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(is_bf16 || param.scalar_type() == torch::kFloat);
+  tl_adamw(param.data_ptr(), grad.data_ptr(), m.data_ptr(), v.data_ptr(),
+           param.numel(), is_bf16 ? 1 : 0, (float)lr, (float)beta1,
+           (float)beta2, (float)eps, (float)weight_decay, (int)step,
+           cur_stream());
+}
+
+Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
+                   double scale) {
+  CHECK_IN(q, torch::kBFloat16);
+  CHECK_IN(k_cache, torch::kBFloat16);
+  CHECK_IN(v_cache, torch::kBFloat16);
+  CHECK_IN(seq_lens, torch::kInt);
+  const int B = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Hkv = k_cache.size(1), Smax = k_cache.size(2);
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  auto out = torch::empty_like(q);
+  tl_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+                 seq_lens.data_ptr(), out.data_ptr(), B, Hq, Hkv, Smax, D,
+                 (float)scale, cur_stream());
+  return out;
+}
+
+Tensor prefill_attn(Tensor q, Tensor k, Tensor v, double scale, bool causal) {
+  CHECK_IN(q, torch::kBFloat16);
+  CHECK_IN(k, torch::kBFloat16);
+  CHECK_IN(v, torch::kBFloat16);
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  auto out = torch::empty_like(q);
+  tl_prefill_attn(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(), B,
+                  S, Hq, Hkv, D, (float)scale, causal ? 1 : 0, cur_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm fwd (+residual)");
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm bwd");
+  mod.def("rope_", &rope_, "in-place RoPE apply");
+  mod.def("swiglu_fwd", &swiglu_fwd, "silu(gate)*up");
+  mod.def("swiglu_bwd", &swiglu_bwd, "SwiGLU bwd");
+  mod.def("adamw_", &adamw_, "fused AdamW step");
+  mod.def("decode_attn", &decode_attn, "GQA decode attention over KV cache");
+  mod.def("prefill_attn", &prefill_attn, "causal GQA prefill attention");
+}

@@ -1,0 +1,196 @@
+// Single-token (decode) GQA attention over a device-resident KV cache.
+//
+// This is THE algorithmic win over the reference: tensorlink serializes the
+// HF DynamicCache and ships it over TCP on every pipelined forward
+// (tensorlink/ml/utils.py:210-221,599-605; ml/worker.py:100-120). Here the
+// cache never leaves the GPU: layout [B, Hkv, Smax, D] bf16, head-major so
+// each (b, h) reads a contiguous [L, D] slab at HBM streaming rate.
+//
+// Work decomposition (bandwidth-first):
+//   grid = (B, Hkv, ceil(G / GMAX)) where G = Hq/Hkv. One block services up
+//   to GMAX=4 query heads of one kv head, so the K/V slab is read ONCE per
+//   4 query heads instead of once per query head (GQA bandwidth saving).
+//   Block = 256 threads = 4 waves; waves split the key range; within a
+//   wave, 8 lanes cooperate per key (lane covers D/8 elements = 32 B
+//   contiguous at D=128 — a wave streams 8 keys x 256 B contiguous K).
+//   Online softmax per (wave, head); wave partials merged through LDS.
+
+#include "common.hpp"
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr int NWAVE = BLOCK / WAVE_SIZE;  // 4
+constexpr int LPK = 8;                    // lanes per key
+constexpr int KPW = WAVE_SIZE / LPK;      // keys per wave per iter = 8
+constexpr int GMAX = 4;                   // query heads per block
+
+template <int D>
+__global__ __launch_bounds__(BLOCK) void decode_attn_kernel(
+    const bf16* __restrict__ q,        // [B, Hq, D]
+    const bf16* __restrict__ k_cache,  // [B, Hkv, Smax, D]
+    const bf16* __restrict__ v_cache,  // [B, Hkv, Smax, D]
+    const int* __restrict__ seq_lens,  // [B]
+    bf16* __restrict__ out,            // [B, Hq, D]
+    int Hq, int Hkv, int Smax, float scale) {
+  constexpr int EPL = D / LPK;  // elements per lane (16 at D=128)
+  const int b = blockIdx.x;
+  const int hkv = blockIdx.y;
+  const int G = Hq / Hkv;
+  const int g0 = blockIdx.z * GMAX;          // first query head in group
+  const int gq = min(GMAX, G - g0);          // heads this block handles
+  const int h0 = hkv * G + g0;               // absolute first q head
+  const int L = seq_lens[b];
+
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int kg = lane / LPK;    // key slot within wave [0,8)
+  const int li = lane % LPK;    // lane within key [0,8)
+
+  __shared__ float q_lds[GMAX][D];
+  __shared__ float red_m[NWAVE][GMAX];
+  __shared__ float red_l[NWAVE][GMAX];
+  __shared__ float red_o[NWAVE][GMAX][D];
+
+  // load q for the block's heads into LDS (fp32)
+  for (int i = threadIdx.x; i < gq * D; i += BLOCK) {
+    const int g = i / D, d = i - g * D;
+    q_lds[g][d] = bf2f(q[((int64_t)b * Hq + h0 + g) * D + d]);
+  }
+  __syncthreads();
+
+  const bf16* kbase = k_cache + ((int64_t)b * Hkv + hkv) * Smax * D;
+  const bf16* vbase = v_cache + ((int64_t)b * Hkv + hkv) * Smax * D;
+
+  float m[GMAX], l[GMAX], acc[GMAX][EPL];
+#pragma unroll
+  for (int g = 0; g < GMAX; ++g) {
+    m[g] = -1e30f;
+    l[g] = 0.f;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) acc[g][e] = 0.f;
+  }
+
+  // each wave strides over keys: wave w handles keys w*KPW + kg + it*NWAVE*KPW
+  for (int k0 = wave * KPW; k0 < L; k0 += NWAVE * KPW) {
+    const int key = k0 + kg;
+    const bool valid = key < L;
+    // K load: 32 B per lane, contiguous per key
+    float kv_elems[EPL];
+    {
+      const bf16* kp = kbase + (int64_t)(valid ? key : 0) * D + li * EPL;
+      const bf16x8* kp8 = reinterpret_cast<const bf16x8*>(kp);
+#pragma unroll
+      for (int c = 0; c < EPL / 8; ++c) {
+        bf16x8 chunk = kp8[c];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kv_elems[c * 8 + j] = bf2f(chunk.v[j]);
+      }
+    }
+    // scores + online-softmax update per head
+    float p[GMAX];
+#pragma unroll
+    for (int g = 0; g < GMAX; ++g) {
+      if (g >= gq) break;
+      float dot = 0.f;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) dot += q_lds[g][li * EPL + e] * kv_elems[e];
+      dot = group_reduce_sum<LPK>(dot);  // full dot, all LPK lanes hold it
+      float score = valid ? dot * scale : -1e30f;
+      const float tile_max = wave_reduce_max(score);
+      const float m_new = fmaxf(m[g], tile_max);
+      const float alpha = __expf(m[g] - m_new);
+      p[g] = valid ? __expf(score - m_new) : 0.f;
+      // sum each key's p once (li == 0 representative)
+      const float psum = wave_reduce_sum(li == 0 ? p[g] : 0.f);
+      l[g] = l[g] * alpha + psum;
+      m[g] = m_new;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) acc[g][e] *= alpha;
+    }
+    // V load + accumulate
+    {
+      const bf16* vp = vbase + (int64_t)(valid ? key : 0) * D + li * EPL;
+      const bf16x8* vp8 = reinterpret_cast<const bf16x8*>(vp);
+#pragma unroll
+      for (int c = 0; c < EPL / 8; ++c) {
+        bf16x8 chunk = vp8[c];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float ve = bf2f(chunk.v[j]);
+#pragma unroll
+          for (int g = 0; g < GMAX; ++g) {
+            if (g >= gq) break;
+            acc[g][c * 8 + j] += p[g] * ve;
+          }
+        }
+      }
+    }
+  }
+
+  // reduce acc across the 8 key-groups (lanes sharing li): bits 3..5 of lane
+#pragma unroll
+  for (int g = 0; g < GMAX; ++g) {
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) {
+#pragma unroll
+      for (int off = LPK; off < WAVE_SIZE; off <<= 1)
+        acc[g][e] += __shfl_xor(acc[g][e], off, WAVE_SIZE);
+    }
+  }
+
+  // wave partials -> LDS (lanes with kg == 0 carry the result)
+  if (kg == 0) {
+#pragma unroll
+    for (int g = 0; g < GMAX; ++g) {
+      if (g < gq) {
+#pragma unroll
+        for (int e = 0; e < EPL; ++e) red_o[wave][g][li * EPL + e] = acc[g][e];
+        if (li == 0) {
+          red_m[wave][g] = m[g];
+          red_l[wave][g] = l[g];
+        }
+      }
+    }
+  }
+  __syncthreads();
+
+  // final merge across waves + store: thread d of [0, D) handles dim d
+  for (int i = threadIdx.x; i < gq * D; i += BLOCK) {
+    const int g = i / D, d = i - g * D;
+    float m_tot = -1e30f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) m_tot = fmaxf(m_tot, red_m[w][g]);
+    float l_tot = 0.f, o = 0.f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) {
+      const float f = __expf(red_m[w][g] - m_tot);
+      l_tot += red_l[w][g] * f;
+      o += red_o[w][g][d] * f;
+    }
+    out[((int64_t)b * Hq + h0 + g) * D + d] = f2bf(o / fmaxf(l_tot, 1e-30f));
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void tl_decode_attn(const void* q, const void* k_cache, const void* v_cache,
+                    const void* seq_lens, void* out, int B, int Hq, int Hkv,
+                    int Smax, int D, float scale, hipStream_t stream) {
+  const int G = Hq / Hkv;
+  dim3 grid(B, Hkv, (G + GMAX - 1) / GMAX), block(BLOCK);
+  if (D == 128)
+    hipLaunchKernelGGL((decode_attn_kernel<128>), grid, block, 0, stream,
+                       (const bf16*)q, (const bf16*)k_cache,
+                       (const bf16*)v_cache, (const int*)seq_lens, (bf16*)out,
+                       Hq, Hkv, Smax, scale);
+  else if (D == 64)
+    hipLaunchKernelGGL((decode_attn_kernel<64>), grid, block, 0, stream,
+                       (const bf16*)q, (const bf16*)k_cache,
+                       (const bf16*)v_cache, (const int*)seq_lens, (bf16*)out,
+                       Hq, Hkv, Smax, scale);
+}
+
+}  // extern "C"

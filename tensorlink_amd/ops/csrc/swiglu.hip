@@ -1,0 +1,75 @@
+// Fused SwiGLU activation: out = silu(gate) * up, plus backward.
+//
+// In the reference this lives inside HF MLP modules run eagerly per stage
+// (tensorlink/ml/worker.py:330-335). Fusing the silu-mul into one
+// elementwise kernel removes a full intermediate round trip to HBM.
+// Flat over n elements, bf16x8 vectorized (n % 8 == 0, wrapper pads).
+
+#include "common.hpp"
+
+namespace {
+
+constexpr int BLOCK = 256;
+
+DEVINLINE float silu(float x) { return x / (1.f + __expf(-x)); }
+
+__global__ __launch_bounds__(BLOCK) void swiglu_fwd_kernel(
+    const bf16* __restrict__ gate, const bf16* __restrict__ up,
+    bf16* __restrict__ out, int64_t n8) {
+  const int64_t i = ((int64_t)blockIdx.x * BLOCK + threadIdx.x);
+  if (i >= n8) return;
+  bf16x8 g = reinterpret_cast<const bf16x8*>(gate)[i];
+  bf16x8 u = reinterpret_cast<const bf16x8*>(up)[i];
+  bf16x8 o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    o.v[j] = f2bf(silu(bf2f(g.v[j])) * bf2f(u.v[j]));
+  reinterpret_cast<bf16x8*>(out)[i] = o;
+}
+
+__global__ __launch_bounds__(BLOCK) void swiglu_bwd_kernel(
+    const bf16* __restrict__ dout, const bf16* __restrict__ gate,
+    const bf16* __restrict__ up, bf16* __restrict__ dgate,
+    bf16* __restrict__ dup, int64_t n8) {
+  const int64_t i = ((int64_t)blockIdx.x * BLOCK + threadIdx.x);
+  if (i >= n8) return;
+  bf16x8 dO = reinterpret_cast<const bf16x8*>(dout)[i];
+  bf16x8 g = reinterpret_cast<const bf16x8*>(gate)[i];
+  bf16x8 u = reinterpret_cast<const bf16x8*>(up)[i];
+  bf16x8 dg, du;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const float gv = bf2f(g.v[j]);
+    const float sig = 1.f / (1.f + __expf(-gv));
+    const float si = gv * sig;
+    const float dsilu = sig * (1.f + gv * (1.f - sig));
+    const float dov = bf2f(dO.v[j]);
+    dg.v[j] = f2bf(dov * bf2f(u.v[j]) * dsilu);
+    du.v[j] = f2bf(dov * si);
+  }
+  reinterpret_cast<bf16x8*>(dgate)[i] = dg;
+  reinterpret_cast<bf16x8*>(dup)[i] = du;
+}
+
+}  // namespace
+
+extern "C" {
+
+void tl_swiglu_fwd(const void* gate, const void* up, void* out, int64_t n,
+                   hipStream_t stream) {
+  const int64_t n8 = n / 8;
+  dim3 grid((uint32_t)cdiv(n8, BLOCK)), block(BLOCK);
+  hipLaunchKernelGGL(swiglu_fwd_kernel, grid, block, 0, stream,
+                     (const bf16*)gate, (const bf16*)up, (bf16*)out, n8);
+}
+
+void tl_swiglu_bwd(const void* dout, const void* gate, const void* up,
+                   void* dgate, void* dup, int64_t n, hipStream_t stream) {
+  const int64_t n8 = n / 8;
+  dim3 grid((uint32_t)cdiv(n8, BLOCK)), block(BLOCK);
+  hipLaunchKernelGGL(swiglu_bwd_kernel, grid, block, 0, stream,
+                     (const bf16*)dout, (const bf16*)gate, (const bf16*)up,
+                     (bf16*)dgate, (bf16*)dup, n8);
+}
+
+}  // extern "C"
