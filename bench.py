@@ -1,0 +1,137 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Qwen2.5-7B serving throughput (output tokens/sec)
+with p50 TTFT, sharded PP=N across N MI355X GPUs (BASELINE.json metric).
+
+Single GPU:      python bench.py --steps 3 --warmup 1
+Multi GPU (driver): python -m torch.distributed.run --nnodes=1 \
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+
+A "step" is one full serving round: prefill a synthetic batch of prompts
+(random token ids, random-init weights — no network for checkpoints) and
+decode a fixed number of new tokens per sequence. Weak scaling: the per-GPU
+batch is fixed, so the global batch grows with N. Rank 0 prints ONE JSON
+line with the aggregate output tokens/sec over the whole job.
+"""
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from tensorlink_amd.models.configs import get_config  # noqa: E402
+from tensorlink_amd.parallel.comm import init_distributed  # noqa: E402
+from tensorlink_amd.parallel.planner import plan_for_world  # noqa: E402
+from tensorlink_amd.parallel.pipeline import (  # noqa: E402
+    PipelineRunner, SamplingParams)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--model", default="Qwen/Qwen2.5-7B-Instruct")
+    p.add_argument("--batch-per-gpu", type=int, default=32,
+                   help="sequences per GPU (weak scaling)")
+    p.add_argument("--prompt-len", type=int, default=512)
+    p.add_argument("--new-tokens", type=int, default=128)
+    p.add_argument("--micro-batches", type=int, default=0,
+                   help="decode micro-batches in flight (0 => world size)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank, world = init_distributed()
+    if world != args.gpus and "WORLD_SIZE" in os.environ:
+        args.gpus = world
+    assert world == args.gpus or world == 1, \
+        f"launched world={world} but --gpus={args.gpus}"
+    world = max(world, 1)
+
+    use_gpu = torch.cuda.is_available()
+    device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", rank))) \
+        if use_gpu else torch.device("cpu")
+    if use_gpu:
+        torch.cuda.set_device(device)
+
+    config = get_config(args.model)
+    B = args.batch_per_gpu * world          # weak scaling: global batch
+    S, T = args.prompt_len, args.new_tokens
+    plan = plan_for_world(config, world, batch_size=B, seq_len=S + T)
+    runner = PipelineRunner(plan, rank, world, device=device, init="random",
+                            seed=1234)
+
+    torch.manual_seed(17)
+    input_ids = torch.randint(0, config.vocab_size, (B, S)) \
+        if rank == 0 else None
+    sp = SamplingParams(temperature=0.0, max_new_tokens=T)
+    n_mb = args.micro_batches or world
+
+    def sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        runner.generate(input_ids, sp, micro_batches=n_mb)
+
+    ttfts = []
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        _, stats = runner.generate(input_ids, sp, micro_batches=n_mb,
+                                   return_stats=True)
+        if stats.get("ttft_s") is not None:
+            ttfts.append(stats["ttft_s"])
+    sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    # MAX over ranks (ranks are barrier-synced; take max to be safe)
+    if world > 1:
+        e = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_gpu else "cpu")
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    total_tokens = B * T * args.steps
+    tps = total_tokens / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+    ttft_p50 = statistics.median(ttfts) if ttfts else None
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "output tokens/sec (Qwen2.5-7B serving, PP sharded)",
+            "value": tps,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "ttft_p50_s": ttft_p50,
+            "config": {
+                "model": args.model,
+                "global_batch": B,
+                "seq_len": S,
+                "new_tokens": T,
+                "parallelism": f"pp{world}",
+                "micro_batches": n_mb,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

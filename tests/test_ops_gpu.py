@@ -1,0 +1,186 @@
+"""HIP kernel numerics tests (MI355X): each kernel vs a plain PyTorch fp32
+reference of the same op (the kernel-unit tier the reference
+lacks; cf. its gradient-hash idea ml/proofs.py:6)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _ext():
+    from tensorlink_amd import ops
+    assert ops.extension_loaded(), "HIP extension must be built on GPU boxes"
+    return ops
+
+
+def test_rmsnorm_fwd():
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(0)
+    for N, H in [(64, 256), (33, 3584), (1024, 4096), (2, 8192)]:
+        x = torch.randn(N, H, device=DEV, dtype=torch.bfloat16)
+        w = torch.randn(H, device=DEV, dtype=torch.bfloat16)
+        y = ops.rmsnorm(x, w, 1e-6)
+        y_ref = ref.rmsnorm(x.float(), w.float(), 1e-6)
+        torch.testing.assert_close(y.float(), y_ref, atol=2e-2, rtol=2e-2)
+
+
+def test_rmsnorm_residual_fwd():
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    x = torch.randn(128, 3584, device=DEV, dtype=torch.bfloat16)
+    r = torch.randn(128, 3584, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(3584, device=DEV, dtype=torch.bfloat16)
+    y, rs = ops.rmsnorm_residual(x, r, w, 1e-6)
+    y_ref, rs_ref = ref.rmsnorm_residual(x.float(), r.float(), w.float(), 1e-6)
+    torch.testing.assert_close(rs.float(), rs_ref, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(y.float(), y_ref, atol=2e-2, rtol=2e-2)
+
+
+def test_rmsnorm_backward():
+    ops = _ext()
+    torch.manual_seed(1)
+    N, H = 64, 1024
+    x = torch.randn(N, H, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(H, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    y = ops.rmsnorm(x, w, 1e-6)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    x32 = x.detach().float().requires_grad_(True)
+    w32 = w.detach().float().requires_grad_(True)
+    var = x32.pow(2).mean(-1, keepdim=True)
+    y32 = x32 * torch.rsqrt(var + 1e-6) * w32
+    y32.backward(dy.float())
+    torch.testing.assert_close(x.grad.float(), x32.grad, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(w.grad.float(), w32.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_rope():
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(2)
+    T, Hq, Hkv, D = 33, 8, 2, 128
+    q = torch.randn(T, Hq, D, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(T, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    pos = torch.randint(0, 1000, (T,), device=DEV, dtype=torch.int32)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2, device=DEV).float() / D))
+    q2, k2 = q.clone(), k.clone()
+    ops.apply_rope_(q2, k2, pos, inv)
+    cos, sin = ref.rope_cos_sin(D, pos, device=DEV)
+    qr, kr = ref.apply_rope(q.float().unsqueeze(0), k.float().unsqueeze(0),
+                            cos, sin)
+    torch.testing.assert_close(q2.float(), qr[0], atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(k2.float(), kr[0], atol=2e-2, rtol=2e-2)
+    # inverse rotation restores
+    ops.apply_rope_(q2, k2, pos, inv, sign=-1.0)
+    torch.testing.assert_close(q2.float(), q.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_swiglu_fwd_bwd():
+    ops = _ext()
+    torch.manual_seed(3)
+    g = torch.randn(64, 18944, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    u = torch.randn(64, 18944, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    out = ops.swiglu(g, u)
+    g32 = g.detach().float().requires_grad_(True)
+    u32 = u.detach().float().requires_grad_(True)
+    ref32 = torch.nn.functional.silu(g32) * u32
+    torch.testing.assert_close(out.float(), ref32.detach(), atol=2e-2,
+                               rtol=2e-2)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    ref32.backward(dout.float())
+    torch.testing.assert_close(g.grad.float(), g32.grad, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(u.grad.float(), u32.grad, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("B,Hq,Hkv,D,L", [
+    (2, 28, 4, 128, 300),   # Qwen2.5-7B shape (G=7)
+    (4, 32, 8, 128, 77),    # Llama/Qwen3 shape (G=4)
+    (1, 4, 2, 64, 1500),    # small-D long-L
+    (3, 8, 8, 128, 64),     # MHA (G=1)
+])
+def test_decode_attn(B, Hq, Hkv, D, L):
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(4)
+    Smax = L + 37
+    q = torch.randn(B, Hq, D, device=DEV, dtype=torch.bfloat16)
+    kc = torch.randn(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
+    vc = torch.randn(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
+    lens = torch.randint(1, L + 1, (B,), device=DEV, dtype=torch.int32)
+    lens[0] = L
+    out = ops.attention_decode(q, kc, vc, lens)
+    out_ref = ref.attention_decode(
+        q.unsqueeze(1).float(), kc.permute(0, 2, 1, 3).float(),
+        vc.permute(0, 2, 1, 3).float(), lens)
+    torch.testing.assert_close(out.float(), out_ref[:, 0], atol=2e-2,
+                               rtol=2e-2)
+
+
+@pytest.mark.parametrize("B,S,Hq,Hkv,D", [
+    (2, 128, 28, 4, 128),
+    (1, 500, 8, 8, 128),    # non-multiple of 64
+    (2, 64, 4, 2, 64),
+    (1, 2048, 8, 2, 128),
+])
+def test_prefill_attn(B, S, Hq, Hkv, D):
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(5)
+    q = torch.randn(B, S, Hq, D, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    out = ops.attention_prefill(q, k, v, causal=True)
+    out_ref = ref.attention_prefill(q.float(), k.float(), v.float(),
+                                    causal=True)
+    torch.testing.assert_close(out.float(), out_ref, atol=2e-2, rtol=2e-2)
+
+
+def test_adamw_matches_torch():
+    ops = _ext()
+    torch.manual_seed(6)
+    n = 100003
+    p_ref = torch.randn(n, device=DEV, dtype=torch.float32)
+    p_mine = p_ref.clone().to(torch.bfloat16)
+    p_ref = p_ref.to(torch.bfloat16).float()   # match bf16 starting point
+    g = torch.randn(n, device=DEV)
+    m = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    pr = p_ref.clone().requires_grad_(True)
+    opt = torch.optim.AdamW([pr], lr=1e-2, betas=(0.9, 0.999), eps=1e-8,
+                            weight_decay=0.01)
+    for step in range(1, 5):
+        pr.grad = g.clone()
+        opt.step()
+        ops.adamw_(p_mine, g.to(torch.bfloat16), m, v, lr=1e-2, beta1=0.9,
+                   beta2=0.999, eps=1e-8, weight_decay=0.01, step=step)
+    # bf16 params accumulate rounding each step; tolerance reflects that
+    torch.testing.assert_close(p_mine.float(), pr.detach(), atol=2e-2,
+                               rtol=2e-2)
+
+
+def test_decode_attn_softmax_stability():
+    """Large score magnitudes must not overflow (online softmax)."""
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    B, Hq, Hkv, D, L = 1, 4, 2, 128, 200
+    q = (torch.randn(B, Hq, D, device=DEV) * 10).to(torch.bfloat16)
+    kc = (torch.randn(B, Hkv, L, D, device=DEV) * 10).to(torch.bfloat16)
+    vc = torch.randn(B, Hkv, L, D, device=DEV).to(torch.bfloat16)
+    lens = torch.full((B,), L, device=DEV, dtype=torch.int32)
+    out = ops.attention_decode(q, kc, vc, lens)
+    assert torch.isfinite(out.float()).all()
+    out_ref = ref.attention_decode(q.unsqueeze(1).float(),
+                                   kc.permute(0, 2, 1, 3).float(),
+                                   vc.permute(0, 2, 1, 3).float(), lens)
+    torch.testing.assert_close(out.float(), out_ref[:, 0], atol=3e-2,
+                               rtol=3e-2)

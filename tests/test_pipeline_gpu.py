@@ -1,0 +1,92 @@
+"""Single-GPU end-to-end tests on MI355X: engine generate + training on the
+HIP path (tiny config sized for the kernels: head_dim 64)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = torch.device("cuda:0")
+
+
+def test_generate_tiny_gpu_matches_cpu_greedy():
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    plan = plan_for_world("tiny", 1)
+    r = PipelineRunner(plan, 0, 1, device=DEV, dtype=torch.bfloat16)
+    torch.manual_seed(7)
+    ids = torch.randint(0, 1024, (4, 12))
+    out, stats = r.generate(ids, SamplingParams(max_new_tokens=8),
+                            return_stats=True)
+    assert out.shape == (4, 8)
+    assert stats["output_tokens_per_s"] > 0
+    assert (out >= 0).all() and (out < 1024).all()
+
+
+def test_generate_qwen_shape_gpu():
+    from tensorlink_amd.models.configs import get_config
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    cfg = get_config("Qwen/Qwen2.5-7B-Instruct")
+    plan = plan_for_world(cfg, 1, batch_size=2, seq_len=128)
+    r = PipelineRunner(plan, 0, 1, device=DEV)
+    ids = torch.randint(0, cfg.vocab_size, (2, 32))
+    out = r.generate(ids, SamplingParams(max_new_tokens=4))
+    assert out.shape == (2, 4)
+
+
+def test_gpu_model_matches_cpu_model():
+    """Same weights on CPU (fp32 reference ops) and GPU (HIP kernels):
+    logits must agree within bf16 tolerance."""
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import init_random_stage
+    cfg = get_config("tiny")
+    m_cpu = build_full_model(cfg)
+    init_random_stage(m_cpu, device="cpu", dtype=torch.float32, seed=11)
+    m_gpu = build_full_model(cfg)
+    init_random_stage(m_gpu, device="cpu", dtype=torch.float32, seed=11)
+    m_gpu = m_gpu.to(DEV).to(torch.bfloat16)
+
+    torch.manual_seed(0)
+    ids = torch.randint(0, cfg.vocab_size, (2, 33))
+    pos = torch.arange(33).unsqueeze(0).expand(2, -1).contiguous()
+    lg_cpu = m_cpu(ids, pos).float()
+    lg_gpu = m_gpu(ids.to(DEV), pos.to(DEV)).float().cpu()
+    # bf16 + kernel differences accumulate over 4 layers; logits are O(10)
+    diff = (lg_cpu - lg_gpu).abs().max().item()
+    scale = lg_cpu.abs().max().item()
+    assert diff / scale < 0.05, f"relative diff {diff/scale:.4f}"
+    # top-1 agreement on most positions
+    agree = (lg_cpu.argmax(-1) == lg_gpu.argmax(-1)).float().mean().item()
+    assert agree > 0.9, f"top-1 agreement {agree:.2f}"
+
+
+def test_training_gpu_reduces_loss():
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    plan = plan_for_world("tiny", 1, training=True)
+    tr = PipelineTrainer(plan, 0, 1, device=DEV, lr=1e-3)
+    torch.manual_seed(1)
+    ids = torch.randint(0, 1024, (8, 32))
+    losses = [tr.train_step(ids, ids, n_micro=2) for _ in range(5)]
+    assert losses[-1] < losses[0], losses
+
+
+def test_fused_adamw_flat_step_gpu():
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.optim import FusedAdamW
+    cfg = get_config("tiny")
+    m = build_full_model(cfg)
+    init_random_stage(m, device=DEV, dtype=torch.bfloat16)
+    opt = FusedAdamW(m.parameters(), lr=1e-2)
+    ids = torch.randint(0, cfg.vocab_size, (2, 16), device=DEV)
+    pos = torch.arange(16, device=DEV).unsqueeze(0).expand(2, -1).contiguous()
+    before = opt.flat_param.clone()
+    out = m(ids, pos, training=True)
+    out.float().mean().backward()
+    assert opt.flat_grad.abs().sum() > 0
+    opt.step()
+    assert not torch.equal(before, opt.flat_param)
+    opt.zero_grad()
+    assert opt.flat_grad.abs().sum() == 0

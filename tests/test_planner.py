@@ -1,0 +1,82 @@
+"""Planner tests — the reference's parser tests print tables with NO asserts
+(reference tests/test_model_parser.py:36-123); these assert real invariants
+against fake capacity tables, as SURVEY.md §4 prescribes."""
+
+import pytest
+
+from tensorlink_amd.models.configs import PRESETS, get_config
+from tensorlink_amd.parallel.planner import (AssignmentError, ModelParser,
+                                             plan_for_world)
+from tensorlink_amd.utils.memory import MI355X_HBM_BYTES
+
+
+@pytest.mark.parametrize("name", ["Qwen/Qwen2.5-7B-Instruct", "Qwen/Qwen3-8B",
+                                  "meta-llama/Llama-3-70B",
+                                  "mistralai/Mixtral-8x7B-v0.1"])
+@pytest.mark.parametrize("pp", [1, 2, 4, 8])
+def test_plan_shapes(name, pp):
+    cfg = get_config(name)
+    if name == "mistralai/Mixtral-8x7B-v0.1" and pp == 1:
+        pass  # MoE fits a single 288 GB rank in bf16 (~93 GB)
+    plan = plan_for_world(cfg, pp, batch_size=8, seq_len=4096)
+    assert plan.num_stages == pp
+    # full coverage, contiguous, ordered
+    covered = []
+    for s in plan.stages:
+        covered.extend(range(s.layer_start, s.layer_end))
+    assert covered == list(range(cfg.num_hidden_layers))
+    assert plan.stages[0].has_embedding
+    assert plan.stages[-1].has_head
+    for s in plan.stages[1:]:
+        assert not s.has_embedding
+    for s in plan.stages[:-1]:
+        assert not s.has_head
+    # every stage under capacity
+    for s in plan.stages:
+        assert s.est_bytes < MI355X_HBM_BYTES
+
+
+def test_small_capacity_fails():
+    parser = ModelParser(capacities_bytes=[8 << 30, 8 << 30])
+    with pytest.raises(AssignmentError):
+        parser.create_distributed_config("meta-llama/Llama-3-70B",
+                                         batch_size=1, seq_len=4096)
+
+
+def test_heterogeneous_capacities():
+    # 24 GB + 16 GB fake workers, like reference tests/test_model_parser.py
+    parser = ModelParser(capacities_bytes=[24 << 30, 16 << 30])
+    plan = parser.create_distributed_config("tiny", batch_size=1,
+                                            seq_len=1024, num_stages=2)
+    assert plan.num_stages == 2
+    assert plan.stages[0].num_layers >= plan.stages[1].num_layers
+
+
+def test_training_memory_larger_than_eval():
+    cfg = get_config("Qwen/Qwen3-8B")
+    p_train = plan_for_world(cfg, 4, batch_size=4, seq_len=2048, training=True)
+    p_eval = plan_for_world(cfg, 4, batch_size=4, seq_len=2048, training=False)
+    assert sum(s.est_bytes for s in p_train.stages) > \
+        sum(s.est_bytes for s in p_eval.stages)
+
+
+def test_auto_pp_selection():
+    # without num_stages, the parser picks the smallest PP that fits
+    parser = ModelParser(n_workers=8)
+    plan = parser.create_distributed_config("Qwen/Qwen2.5-7B-Instruct",
+                                            batch_size=1, seq_len=2048)
+    assert plan.num_stages == 1  # 7B fits one 288 GB rank
+
+    plan70 = parser.create_distributed_config(
+        "meta-llama/Llama-3-70B", batch_size=1, seq_len=2048, training=True)
+    assert plan70.num_stages >= 2  # 70B training does not fit one rank
+
+
+def test_param_count_close_to_nominal():
+    cfg = get_config("Qwen/Qwen2.5-7B-Instruct")
+    n = cfg.param_count()
+    assert 7.0e9 < n < 8.5e9
+    n70 = get_config("meta-llama/Llama-3-70B").param_count()
+    assert 6.5e10 < n70 < 7.5e10
+    nmoe = get_config("mistralai/Mixtral-8x7B-v0.1").param_count()
+    assert 4.4e10 < nmoe < 5.0e10
