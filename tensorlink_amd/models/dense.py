@@ -109,14 +109,20 @@ class Attention(nn.Module):
         else:
             q = q.contiguous()
             k = k.contiguous()
-            ops.apply_rope_(q.view(B * S, self.n_heads, -1),
-                            k.view(B * S, self.n_kv, -1), flat_pos,
-                            self.inv_freq)
             if kv_cache is None:
+                ops.apply_rope_(q.view(B * S, self.n_heads, -1),
+                                k.view(B * S, self.n_kv, -1), flat_pos,
+                                self.inv_freq)
                 out = ops.attention_prefill(q, k, v, causal=True,
                                             scale=self.scale)
             else:
-                kv_cache.append(layer_idx, k, v, positions)
+                v = v.contiguous()
+                ops.rope_append_(q.view(B * S, self.n_heads, -1),
+                                 k.view(B * S, self.n_kv, -1),
+                                 v.view(B * S, self.n_kv, -1),
+                                 kv_cache.k[layer_idx],
+                                 kv_cache.v[layer_idx], flat_pos,
+                                 self.inv_freq, S)
                 if S == 1:
                     out = ops.attention_decode(
                         q, kv_cache.k[layer_idx], kv_cache.v[layer_idx],

@@ -105,6 +105,28 @@ def apply_rope_(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
     _require_ext().rope_(q, k, positions.int(), inv_freq.float(), sign)
 
 
+def rope_append_(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 k_cache: torch.Tensor, v_cache: torch.Tensor,
+                 positions: torch.Tensor, inv_freq: torch.Tensor,
+                 S: int) -> None:
+    """Fused in-place RoPE on q/k + KV-cache append (inference hot path).
+
+    q [T,Hq,D], k/v [T,Hkv,D] with T = B*S; caches [B,Hkv,Smax,D];
+    positions [T] gives the RoPE angle and the cache slot per token.
+    """
+    if _on_gpu(q):
+        _require_ext().rope_append_(q, k, v, k_cache, v_cache,
+                                    positions.int(), inv_freq.float(), S)
+        return
+    apply_rope_(q, k, positions, inv_freq)
+    T, Hkv, D = k.shape
+    B = T // S
+    pos = positions.view(B, S).long()
+    for b in range(B):
+        k_cache[b, :, pos[b]] = k.view(B, S, Hkv, D)[b].transpose(0, 1)
+        v_cache[b, :, pos[b]] = v.view(B, S, Hkv, D)[b].transpose(0, 1)
+
+
 class _RopeFn(torch.autograd.Function):
     """Autograd RoPE for the training path (operates out-of-place)."""
 

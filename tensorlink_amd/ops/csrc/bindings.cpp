@@ -30,6 +30,10 @@ void tl_decode_attn(const void* q, const void* k_cache, const void* v_cache,
 void tl_prefill_attn(const void* q, const void* k, const void* v, void* out,
                      int B, int S, int Hq, int Hkv, int D, float scale,
                      int causal, hipStream_t stream);
+void tl_rope_append(void* q, void* k, const void* v, void* k_cache,
+                    void* v_cache, const void* positions,
+                    const void* inv_freq, int64_t T, int S, int Hq, int Hkv,
+                    int D, int Smax, hipStream_t stream);
 }
 
 namespace {
@@ -170,9 +174,35 @@ Tensor prefill_attn(Tensor q, Tensor k, Tensor v, double scale, bool causal) {
   return out;
 }
 
+void rope_append_(Tensor q, Tensor k, Tensor v, Tensor k_cache,
+                  Tensor v_cache, Tensor positions, Tensor inv_freq,
+                  int64_t S) {
+  CHECK_IN(q, torch::kBFloat16);
+  CHECK_IN(k, torch::kBFloat16);
+  CHECK_IN(v, torch::kBFloat16);
+  CHECK_IN(k_cache, torch::kBFloat16);
+  CHECK_IN(v_cache, torch::kBFloat16);
+  CHECK_IN(positions, torch::kInt);
+  CHECK_IN(inv_freq, torch::kFloat);
+  const int D = q.size(-1);
+  const int Hq = q.size(-2);
+  const int Hkv = k.size(-2);
+  const int64_t T = q.numel() / ((int64_t)Hq * D);
+  const int Smax = k_cache.size(2);
+  TORCH_CHECK(T % S == 0, "T must be divisible by S");
+  TORCH_CHECK(k_cache.size(0) * S == T, "cache batch mismatch");
+  TORCH_CHECK(D % 16 == 0, "D must be divisible by 16");
+  tl_rope_append(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                 k_cache.data_ptr(), v_cache.data_ptr(),
+                 positions.data_ptr(), inv_freq.data_ptr(), T, (int)S, Hq,
+                 Hkv, D, Smax, cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("rope_append_", &rope_append_,
+          "fused in-place RoPE + KV-cache append");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm fwd (+residual)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm bwd");
   mod.def("rope_", &rope_, "in-place RoPE apply");

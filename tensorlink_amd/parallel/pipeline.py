@@ -82,6 +82,7 @@ class PipelineRunner:
         self.H = self.config.hidden_size
         self.kv_cache: Optional[KVCache] = None
         self._gen = None
+        self._decode_graph = None       # (graph, tok_buf, pos_buf)
 
     # ------------------------------------------------------------------
     def _sample(self, logits: torch.Tensor, sp: SamplingParams) -> torch.Tensor:
@@ -94,8 +95,15 @@ class PipelineRunner:
                                 generator=self._gen)
 
     def alloc_cache(self, batch: int, max_seq: int):
+        # reuse the allocation when shapes match — keeps hipGraph-captured
+        # decode valid across generate() calls (pointers must not move)
+        c = self.kv_cache
+        if (c is not None and c.batch == batch and c.max_seq >= max_seq):
+            c.reset()
+            return c
         self.kv_cache = self.stage.make_kv_cache(batch, max_seq, self.device,
                                                  self.dtype)
+        self._decode_graph = None
         return self.kv_cache
 
     # ------------------------------------------------------------------
@@ -170,15 +178,21 @@ class PipelineRunner:
                 out_tokens[mb * b:(mb + 1) * b, 0] = first_tokens[mb]
 
         if self.world == 1:
-            # single stage: straight decode loop
             cur = first_tokens[0] if n_mb == 1 else torch.cat(first_tokens)
-            for t in range(1, T):
-                pos = positions.unsqueeze(1)
-                logits = self.stage(cur.unsqueeze(1), pos.int(),
-                                    kv_cache=self.kv_cache).squeeze(1)
-                cur = self._sample(logits, sp).to(torch.int64)
-                out_tokens[:, t] = cur
-                positions += 1
+            if (self.device.type == "cuda" and sp.temperature <= 0
+                    and T > 4):
+                # hipGraph-captured decode: the whole per-token step (all
+                # layers + head + argmax) replays as one graph launch,
+                # eliminating ~10 kernel-launch gaps per layer.
+                self._graph_decode(cur, positions, out_tokens, T)
+            else:
+                for t in range(1, T):
+                    pos = positions.unsqueeze(1)
+                    logits = self.stage(cur.unsqueeze(1), pos.int(),
+                                        kv_cache=self.kv_cache).squeeze(1)
+                    cur = self._sample(logits, sp).to(torch.int64)
+                    out_tokens[:, t] = cur
+                    positions += 1
             stats = self._finish_stats(B, S, T, t_start, t_first)
             return (out_tokens, stats) if return_stats else out_tokens
 
@@ -237,6 +251,47 @@ class PipelineRunner:
         if return_stats:
             return result, stats
         return result
+
+    def _graph_decode(self, cur: torch.Tensor, positions: torch.Tensor,
+                      out_tokens: torch.Tensor, T: int) -> None:
+        """Greedy decode via a captured hipGraph (world==1).
+
+        The KV cache must already hold the prefill; `positions` holds the
+        next position per sequence. State mutated by the captured step:
+        tok/pos static buffers, cache contents and seq_lens (all by device
+        pointer, so replays see fresh values).
+        """
+        B = cur.shape[0]
+        if self._decode_graph is None:
+            tok_buf = cur.clone()
+            pos_buf = positions.clone()
+            cache = self.kv_cache
+            saved_lens = cache.seq_lens.clone()
+
+            def step():
+                logits = self.stage(tok_buf.unsqueeze(1),
+                                    pos_buf.unsqueeze(1), kv_cache=cache)
+                tok_buf.copy_(logits.squeeze(1).argmax(-1))
+                pos_buf += 1
+
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                step()  # warmup (writes garbage at pos S; replay 1 rewrites)
+            torch.cuda.current_stream().wait_stream(s)
+            cache.seq_lens.copy_(saved_lens)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                step()
+            cache.seq_lens.copy_(saved_lens)
+            self._decode_graph = (graph, tok_buf, pos_buf)
+        graph, tok_buf, pos_buf = self._decode_graph
+        tok_buf.copy_(cur)
+        pos_buf.copy_(positions)
+        for t in range(1, T):
+            graph.replay()
+            out_tokens[:, t].copy_(tok_buf)
+        positions.copy_(pos_buf)
 
     def _finish_stats(self, B, S, T, t_start, t_first):
         if self.device.type == "cuda":

@@ -184,3 +184,35 @@ def test_decode_attn_softmax_stability():
                                    vc.permute(0, 2, 1, 3).float(), lens)
     torch.testing.assert_close(out.float(), out_ref[:, 0], atol=3e-2,
                                rtol=3e-2)
+
+
+def test_rope_append_fused():
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(7)
+    B, S, Hq, Hkv, D, Smax = 3, 5, 8, 2, 128, 64
+    T = B * S
+    q = torch.randn(T, Hq, D, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(T, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(T, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    kc = torch.zeros(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    pos = torch.arange(S, device=DEV, dtype=torch.int32).repeat(B) + 3
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2, device=DEV).float() / D))
+    q2, k2 = q.clone(), k.clone()
+    ops.rope_append_(q2, k2, v, kc, vc, pos, inv, S)
+    # reference: rotate then place
+    cos, sin = ref.rope_cos_sin(D, pos, device=DEV)
+    qr, kr = ref.apply_rope(q.float().unsqueeze(0), k.float().unsqueeze(0),
+                            cos, sin)
+    torch.testing.assert_close(q2.float(), qr[0], atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(k2.float(), kr[0], atol=2e-2, rtol=2e-2)
+    krB = kr[0].view(B, S, Hkv, D)
+    vB = v.view(B, S, Hkv, D).float()
+    for b in range(B):
+        for s in range(S):
+            p = int(pos[b * S + s])
+            torch.testing.assert_close(kc[b, :, p].float(), krB[b, s],
+                                       atol=2e-2, rtol=2e-2)
+            torch.testing.assert_close(vc[b, :, p].float(), vB[b, s],
+                                       atol=1e-3, rtol=1e-3)
