@@ -272,7 +272,7 @@ class PipelineRunner:
                 logits = self.stage(tok_buf.unsqueeze(1),
                                     pos_buf.unsqueeze(1), kv_cache=cache)
                 tok_buf.copy_(logits.squeeze(1).argmax(-1))
-                pos_buf += 1
+                pos_buf.add_(1)
 
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
