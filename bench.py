@@ -89,6 +89,8 @@ def main():
     for _ in range(args.steps):
         _, stats = runner.generate(input_ids, sp, micro_batches=n_mb,
                                    return_stats=True)
+        if rank == 0:
+            print(f"[step] {stats}", file=sys.stderr)
         if stats.get("ttft_s") is not None:
             ttfts.append(stats["ttft_s"])
     sync()

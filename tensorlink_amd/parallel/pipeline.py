@@ -164,6 +164,8 @@ class PipelineRunner:
                 logits = self.stage.head(hidden[:, -1:]).squeeze(1)
                 tok = self._sample(logits, sp).to(torch.int64)
                 if t_first is None:
+                    if self.device.type == "cuda":
+                        torch.cuda.synchronize(self.device)
                     t_first = time.perf_counter()
                 first_tokens.append(tok)
 
@@ -178,9 +180,11 @@ class PipelineRunner:
                 out_tokens[mb * b:(mb + 1) * b, 0] = first_tokens[mb]
 
         if self.world == 1:
+            t_prefill_end = t_first
             cur = first_tokens[0] if n_mb == 1 else torch.cat(first_tokens)
+            import os as _os
             if (self.device.type == "cuda" and sp.temperature <= 0
-                    and T > 4):
+                    and T > 4 and not _os.environ.get("TL_NO_GRAPH")):
                 # hipGraph-captured decode: the whole per-token step (all
                 # layers + head + argmax) replays as one graph launch,
                 # eliminating ~10 kernel-launch gaps per layer.
@@ -194,6 +198,11 @@ class PipelineRunner:
                     out_tokens[:, t] = cur
                     positions += 1
             stats = self._finish_stats(B, S, T, t_start, t_first)
+            if t_prefill_end is not None and stats is not None:
+                dec = stats["total_s"] - (t_prefill_end - t_start)
+                stats["prefill_s"] = t_prefill_end - t_start
+                stats["decode_s"] = dec
+                stats["decode_tokens_per_s"] = B * (T - 1) / dec if dec > 0 else None
             return (out_tokens, stats) if return_stats else out_tokens
 
         # world > 1: step-major ring. In round `step`, rank 0 feeds each
