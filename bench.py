@@ -61,6 +61,9 @@ def main():
     if use_gpu:
         torch.cuda.set_device(device)
 
+    from tensorlink_amd.utils.tunable import setup_tunableop
+    setup_tunableop(tune=bool(os.environ.get("TL_TUNE")))
+
     config = get_config(args.model)
     B = args.batch_per_gpu * world          # weak scaling: global batch
     S, T = args.prompt_len, args.new_tokens
@@ -108,6 +111,10 @@ def main():
     tps = total_tokens / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
     ttft_p50 = statistics.median(ttfts) if ttfts else None
+
+    if os.environ.get("TL_TUNE") and rank == 0:
+        from tensorlink_amd.utils.tunable import save_tunableop
+        save_tunableop()
 
     if rank == 0:
         print(json.dumps({

@@ -23,9 +23,8 @@ constexpr int BLOCK = 256;
 constexpr int NWAVE = BLOCK / WAVE_SIZE;  // 4
 constexpr int LPK = 8;                    // lanes per key
 constexpr int KPW = WAVE_SIZE / LPK;      // keys per wave per iter = 8
-constexpr int GMAX = 4;                   // query heads per block
 
-template <int D>
+template <int D, int GMAX>
 __global__ __launch_bounds__(BLOCK) void decode_attn_kernel(
     const bf16* __restrict__ q,        // [B, Hq, D]
     const bf16* __restrict__ k_cache,  // [B, Hkv, Smax, D]
@@ -180,17 +179,28 @@ void tl_decode_attn(const void* q, const void* k_cache, const void* v_cache,
                     const void* seq_lens, void* out, int B, int Hq, int Hkv,
                     int Smax, int D, float scale, hipStream_t stream) {
   const int G = Hq / Hkv;
-  dim3 grid(B, Hkv, (G + GMAX - 1) / GMAX), block(BLOCK);
-  if (D == 128)
-    hipLaunchKernelGGL((decode_attn_kernel<128>), grid, block, 0, stream,
-                       (const bf16*)q, (const bf16*)k_cache,
-                       (const bf16*)v_cache, (const int*)seq_lens, (bf16*)out,
-                       Hq, Hkv, Smax, scale);
-  else if (D == 64)
-    hipLaunchKernelGGL((decode_attn_kernel<64>), grid, block, 0, stream,
-                       (const bf16*)q, (const bf16*)k_cache,
-                       (const bf16*)v_cache, (const int*)seq_lens, (bf16*)out,
-                       Hq, Hkv, Smax, scale);
+  // One block covers up to GMAX query heads of one kv head; pick the
+  // smallest GMAX >= G (<= 8) so the K/V slab is streamed ONCE per group
+  // (G=7 for Qwen2.5-7B, 8 for Llama-3-70B, 4 for Qwen3-8B). GMAX=8 costs
+  // occupancy (256 VGPR, 1 wave/SIMD) — overridable for A/B measurement.
+  int gmax = G <= 2 ? 2 : (G <= 4 ? 4 : 8);
+  if (const char* e = getenv("TL_DECODE_GMAX")) gmax = atoi(e);
+  dim3 grid(B, Hkv, (G + gmax - 1) / gmax), block(BLOCK);
+#define LAUNCH(DD, GG)                                                     \
+  hipLaunchKernelGGL((decode_attn_kernel<DD, GG>), grid, block, 0, stream, \
+                     (const bf16*)q, (const bf16*)k_cache,                 \
+                     (const bf16*)v_cache, (const int*)seq_lens,           \
+                     (bf16*)out, Hq, Hkv, Smax, scale)
+  if (D == 128) {
+    if (gmax == 2) LAUNCH(128, 2);
+    else if (gmax == 4) LAUNCH(128, 4);
+    else LAUNCH(128, 8);
+  } else if (D == 64) {
+    if (gmax == 2) LAUNCH(64, 2);
+    else if (gmax == 4) LAUNCH(64, 4);
+    else LAUNCH(64, 8);
+  }
+#undef LAUNCH
 }
 
 }  // extern "C"
