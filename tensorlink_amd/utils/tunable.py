@@ -47,5 +47,10 @@ def setup_tunableop(tune: bool = False, filename: str | None = None) -> bool:
 
 
 def save_tunableop(filename: str | None = None) -> None:
+    # torch 2.10 writes the results file automatically at process exit
+    # (set_filename + tuning_enable); write_file only exists on older
+    # versions, so this is best-effort.
     fn = filename or TUNED_FILE
-    torch.cuda.tunable.write_file(fn)
+    wf = getattr(torch.cuda.tunable, "write_file", None)
+    if wf is not None:
+        wf(fn)
