@@ -181,7 +181,19 @@ def _tiny_moe() -> ModelConfig:
         num_local_experts=4, num_experts_per_tok=2, architecture="mixtral")
 
 
+def _gpt2(name):
+    def make():
+        from tensorlink_amd.models.gpt2 import gpt2_config
+        return gpt2_config(name)
+    return make
+
+
 PRESETS = {
+    "gpt2-small": _gpt2("gpt2-small"),
+    "gpt2": _gpt2("gpt2-small"),
+    "gpt2-medium": _gpt2("gpt2-medium"),
+    "tiny-gpt2": _gpt2("tiny-gpt2"),
+    "sshleifer/tiny-gpt2": _gpt2("tiny-gpt2"),
     "Qwen/Qwen2.5-7B-Instruct": _qwen25_7b,
     "Qwen/Qwen2.5-7B": _qwen25_7b,
     "Qwen/Qwen3-8B": _qwen3_8b,

@@ -268,12 +268,21 @@ class StageModel(nn.Module):
         return hidden
 
 
-def build_full_model(config: ModelConfig) -> StageModel:
+def build_full_model(config: ModelConfig):
     """Whole model as a single stage (PP=1 / whole-model offload — the
     reference's ``entire_model`` path, ``ml/module.py:894-897``)."""
-    return StageModel(config, 0, config.num_hidden_layers, True, True)
+    cls = _stage_class(config)
+    return cls(config, 0, config.num_hidden_layers, True, True)
 
 
-def build_stage(config: ModelConfig, spec) -> StageModel:
-    return StageModel(config, spec.layer_start, spec.layer_end,
-                      spec.has_embedding, spec.has_head)
+def build_stage(config: ModelConfig, spec):
+    cls = _stage_class(config)
+    return cls(config, spec.layer_start, spec.layer_end,
+               spec.has_embedding, spec.has_head)
+
+
+def _stage_class(config: ModelConfig):
+    if config.architecture == "gpt2":
+        from tensorlink_amd.models.gpt2 import Gpt2StageModel
+        return Gpt2StageModel
+    return StageModel

@@ -126,7 +126,9 @@ def init_random_stage(stage: StageModel, device="cpu", dtype=torch.bfloat16,
     stage.to(device=device, dtype=dtype)
     std = 1.0 / math.sqrt(stage.config.hidden_size)
     for name, p in stage.named_parameters():
-        if "layernorm" in name or name.endswith(("norm", "q_norm", "k_norm")):
+        if ("layernorm" in name or name.endswith(("norm", "q_norm", "k_norm"))
+                or (".ln_" in name and name.endswith(".weight"))
+                or (name.startswith("ln_") and name.endswith(".weight"))):
             p.fill_(1.0)
         elif name.endswith(".bias"):
             p.zero_()

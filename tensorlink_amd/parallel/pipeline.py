@@ -111,18 +111,19 @@ class PipelineRunner:
     def generate(self, input_ids: Optional[torch.Tensor] = None,
                  sampling: Optional[SamplingParams] = None,
                  micro_batches: Optional[int] = None,
-                 return_stats: bool = False):
+                 return_stats: bool = False,
+                 on_token=None):
         """SPMD generate. input_ids [B, S] significant on rank 0 only.
         Returns generated tokens [B, max_new] on the FIRST rank, else None.
         """
         sp = sampling or SamplingParams()
-        # agree on shapes
+        # agree on shapes AND micro-batch split (rank 0 is authoritative)
         if self.world > 1:
             meta = None
             if self.is_first:
-                meta = (tuple(input_ids.shape), sp)
+                meta = (tuple(input_ids.shape), sp, micro_batches)
             meta = self.p2p.broadcast_obj(meta, src=0)
-            (B, S), sp = meta
+            (B, S), sp, micro_batches = meta
         else:
             B, S = input_ids.shape
 
@@ -182,9 +183,15 @@ class PipelineRunner:
         if self.world == 1:
             t_prefill_end = t_first
             cur = first_tokens[0] if n_mb == 1 else torch.cat(first_tokens)
+            if on_token is not None:
+                on_token(0, cur)
+            eos = sp.eos_token_id
+            finished = (cur == eos) if eos is not None else None
+            n_out = T
             import os as _os
             if (self.device.type == "cuda" and sp.temperature <= 0
-                    and T > 4 and not _os.environ.get("TL_NO_GRAPH")):
+                    and T > 4 and not _os.environ.get("TL_NO_GRAPH")
+                    and on_token is None and eos is None):
                 # hipGraph-captured decode: the whole per-token step (all
                 # layers + head + argmax) replays as one graph launch,
                 # eliminating ~10 kernel-launch gaps per layer.
@@ -197,13 +204,21 @@ class PipelineRunner:
                     cur = self._sample(logits, sp).to(torch.int64)
                     out_tokens[:, t] = cur
                     positions += 1
-            stats = self._finish_stats(B, S, T, t_start, t_first)
+                    if on_token is not None:
+                        on_token(t, cur)
+                    if finished is not None:
+                        finished |= (cur == eos)
+                        if bool(finished.all()):
+                            n_out = t + 1
+                            break
+            out = out_tokens[:, :n_out]
+            stats = self._finish_stats(B, S, n_out, t_start, t_first)
             if t_prefill_end is not None and stats is not None:
                 dec = stats["total_s"] - (t_prefill_end - t_start)
                 stats["prefill_s"] = t_prefill_end - t_start
                 stats["decode_s"] = dec
-                stats["decode_tokens_per_s"] = B * (T - 1) / dec if dec > 0 else None
-            return (out_tokens, stats) if return_stats else out_tokens
+                stats["decode_tokens_per_s"] = B * (n_out - 1) / dec if dec > 0 else None
+            return (out, stats) if return_stats else out
 
         # world > 1: step-major ring. In round `step`, rank 0 feeds each
         # micro-batch its token (mb, step); the hidden flows down the
@@ -228,6 +243,8 @@ class PipelineRunner:
                 if self.is_first:
                     tok = self.p2p.recv((b,), torch.int64, self.world - 1,
                                         self.device)
+                    if on_token is not None:
+                        on_token(step, tok)
                     hidden = self.stage(tok.unsqueeze(1), pos, kv_cache=cache,
                                         return_logits=False)
                     _isend(hidden, self.next_rank)
@@ -444,3 +461,66 @@ class PipelineTrainer:
         if P > 1:
             loss_val = self.p2p.broadcast_obj(loss_val, src=P - 1)
         return loss_val
+
+    # ------------------------------------------------------------------
+    # Split forward/backward for the DistributedModel user API: the user
+    # computes an arbitrary loss on rank 0's returned logits and calls
+    # loss.backward(), which routes here (reference CustomAutogradRouter,
+    # ml/module.py:126-144).
+    # ------------------------------------------------------------------
+    def spmd_forward(self, input_ids: Optional[torch.Tensor] = None):
+        """Single-micro-batch pipeline forward with autograd stashes.
+        Returns logits [B,S,V] on rank 0 (detached), None elsewhere."""
+        P, r = self.world, self.rank
+        is_first, is_last = r == 0, r == P - 1
+        if P > 1:
+            meta = tuple(input_ids.shape) if is_first else None
+            B, S = self.p2p.broadcast_obj(meta, src=0)
+        else:
+            B, S = input_ids.shape
+        pos = torch.arange(S, device=self.device, dtype=torch.int32) \
+            .unsqueeze(0).expand(B, -1).contiguous()
+        if is_first:
+            input_ids = input_ids.to(self.device)
+            inp = None
+            out = self.stage(input_ids, pos, training=True,
+                             return_logits=is_last)
+        else:
+            inp = self.p2p.recv((B, S, self.H), self.dtype, r - 1,
+                                self.device).requires_grad_(True)
+            out = self.stage(inp, pos, training=True, return_logits=is_last)
+        self._fb_stash = (inp, out, (B, S))
+        if P == 1:
+            return out          # local graph intact: user autograd works
+        if is_last:
+            self.p2p.send(out.detach().to(self.dtype), 0)
+            return None
+        self.p2p.send(out.detach(), r + 1)
+        if is_first:
+            V = self.runner.config.vocab_size
+            logits = self.p2p.recv((B, S, V), self.dtype, P - 1, self.device)
+            return logits
+        return None
+
+    def spmd_backward(self, grad_logits: Optional[torch.Tensor] = None):
+        """Backward for the stashed spmd_forward."""
+        P, r = self.world, self.rank
+        is_first, is_last = r == 0, r == P - 1
+        inp, out, (B, S) = self._fb_stash
+        if P == 1:
+            raise RuntimeError("PP=1 uses local autograd directly")
+        if is_last:
+            V = self.runner.config.vocab_size
+            g = self.p2p.recv((B, S, V), self.dtype, 0, self.device)
+            out.backward(g)
+            self.p2p.send(inp.grad, r - 1)
+        elif not is_first:
+            g = self.p2p.recv((B, S, self.H), self.dtype, r + 1, self.device)
+            out.backward(g)
+            self.p2p.send(inp.grad, r - 1)
+        else:
+            if grad_logits is not None:
+                self.p2p.send(grad_logits.to(self.device, self.dtype), P - 1)
+            g = self.p2p.recv((B, S, self.H), self.dtype, r + 1, self.device)
+            out.backward(g)
+        self._fb_stash = None

@@ -1,0 +1,101 @@
+"""REST/SSE schema tests (reference tier: tests/test_model_api.py — same
+endpoints and response-shape asserts, CPU, tiny model, byte tokenizer)."""
+
+import json
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from tensorlink_amd.api.server import create_app
+from tensorlink_amd.engine.engine import InferenceEngine
+
+
+@pytest.fixture(scope="module")
+def client():
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    return TestClient(create_app(eng))
+
+
+def test_request_model_and_status(client):
+    r = client.post("/request-model", json={"hf_name": "tiny"})
+    assert r.status_code == 200
+    assert r.json()["status"] == "ready"
+    r = client.get("/model-status", params={"hf_name": "tiny"})
+    assert r.json()["status"] == "ready"
+    r = client.get("/model-status", params={"hf_name": "nope"})
+    assert r.json()["status"] == "not_loaded"
+
+
+def test_generate_simple_format(client):
+    r = client.post("/v1/generate", json={
+        "hf_name": "tiny", "message": "hello world",
+        "max_new_tokens": 8, "do_sample": False,
+        "output_format": "simple"})
+    assert r.status_code == 200
+    body = r.json()
+    assert "response" in body and body["model"] == "tiny"
+
+
+def test_generate_openai_format(client):
+    r = client.post("/v1/generate", json={
+        "hf_name": "tiny", "message": "hi", "max_new_tokens": 6,
+        "do_sample": False, "output_format": "openai"})
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    assert body["usage"]["completion_tokens"] > 0
+
+
+def test_generate_unknown_model(client):
+    r = client.post("/v1/generate", json={
+        "hf_name": "not-a-model", "message": "x"})
+    assert "error" in r.json()
+
+
+def test_generate_stream_sse(client):
+    with client.stream("POST", "/v1/generate", json={
+            "hf_name": "tiny", "message": "abc", "max_new_tokens": 5,
+            "do_sample": False, "stream": True,
+            "output_format": "simple"}) as r:
+        assert r.status_code == 200
+        assert "text/event-stream" in r.headers["content-type"]
+        lines = [ln for ln in r.iter_lines() if ln]
+    assert lines[-1] == "data: [DONE]"
+    chunks = [json.loads(ln[6:]) for ln in lines[:-1]]
+    assert all("token" in c for c in chunks)
+
+
+def test_generate_stream_openai_sse(client):
+    with client.stream("POST", "/v1/generate", json={
+            "hf_name": "tiny", "message": "abc", "max_new_tokens": 5,
+            "do_sample": False, "stream": True,
+            "output_format": "openai"}) as r:
+        lines = [ln for ln in r.iter_lines() if ln]
+    assert lines[-1] == "data: [DONE]"
+    chunks = [json.loads(ln[6:]) for ln in lines[:-1]]
+    assert chunks[0]["object"] == "chat.completion.chunk"
+    assert chunks[-1]["choices"][0]["finish_reason"] == "stop"
+
+
+def test_chat_completions(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny",
+        "messages": [{"role": "system", "content": "be nice"},
+                     {"role": "user", "content": "hello"}],
+        "max_tokens": 6, "temperature": 0})
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert isinstance(body["choices"][0]["message"]["content"], str)
+
+
+def test_stats_and_info(client):
+    s = client.get("/stats").json()
+    assert s["requests_total"] >= 1
+    assert any(m["model"] == "tiny" for m in s["models"])
+    info = client.get("/node-info").json()
+    assert info["world_size"] == 1
+    assert client.get("/models").json()["models"]
+    assert isinstance(client.get("/model-demand").json(), dict)
+    assert "history" in client.get("/network-history").json()
