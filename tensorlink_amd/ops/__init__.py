@@ -206,8 +206,10 @@ def attention_train(q, k, v, causal: bool = True,
 
 
 def attention_decode(q, k_cache, v_cache, seq_lens,
-                     scale: Optional[float] = None) -> torch.Tensor:
-    """q [B,1,Hq,D] or [B,Hq,D]; caches [B,Hkv,Smax,D]; seq_lens [B]."""
+                     scale: Optional[float] = None,
+                     n_split: int = 0) -> torch.Tensor:
+    """q [B,1,Hq,D] or [B,Hq,D]; caches [B,Hkv,Smax,D]; seq_lens [B].
+    n_split: flash-decode seq splits (0 = auto heuristic)."""
     squeeze = q.dim() == 4
     if squeeze:
         q3 = q.squeeze(1)
@@ -222,7 +224,7 @@ def attention_decode(q, k_cache, v_cache, seq_lens,
                                    scale)
         return out if squeeze else out.squeeze(1)
     out = _require_ext().decode_attn(q3.contiguous(), k_cache, v_cache,
-                                     seq_lens.int(), scale)
+                                     seq_lens.int(), scale, n_split)
     return out.unsqueeze(1) if squeeze else out
 
 

@@ -34,6 +34,11 @@ void tl_rope_append(void* q, void* k, const void* v, void* k_cache,
                     void* v_cache, const void* positions,
                     const void* inv_freq, int64_t T, int S, int Hq, int Hkv,
                     int D, int Smax, hipStream_t stream);
+void tl_decode_attn_mfma(const void* q, const void* k_cache,
+                         const void* v_cache, const void* seq_lens, void* out,
+                         void* partial, void* partial_ml, int B, int Hq,
+                         int Hkv, int Smax, int D, float scale, int n_split,
+                         hipStream_t stream);
 }
 
 namespace {
@@ -145,19 +150,45 @@ void adamw_(Tensor param, Tensor grad, Tensor m, Tensor v, double lr,
 }
 
 Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
-                   double scale) {
+                   double scale, int64_t n_split) {
   CHECK_IN(q, torch::kBFloat16);
   CHECK_IN(k_cache, torch::kBFloat16);
   CHECK_IN(v_cache, torch::kBFloat16);
   CHECK_IN(seq_lens, torch::kInt);
   const int B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int Hkv = k_cache.size(1), Smax = k_cache.size(2);
+  const int G = Hq / Hkv;
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   auto out = torch::empty_like(q);
-  tl_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-                 seq_lens.data_ptr(), out.data_ptr(), B, Hq, Hkv, Smax, D,
-                 (float)scale, cur_stream());
+  const bool legacy = getenv("TL_DECODE_LEGACY") != nullptr || G > 16;
+  if (legacy) {
+    tl_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+                   seq_lens.data_ptr(), out.data_ptr(), B, Hq, Hkv, Smax, D,
+                   (float)scale, cur_stream());
+    return out;
+  }
+  if (n_split <= 0) {
+    // flash-decode split heuristic: target >= ~512 blocks (2 per CU)
+    int ns = 1;
+    while (B * Hkv * ns < 512 && ns < 16) ns <<= 1;
+    n_split = ns;
+  }
+  TORCH_CHECK(n_split <= 64, "n_split too large");
+  Tensor partial, partial_ml;
+  void *pp = nullptr, *pml = nullptr;
+  if (n_split > 1) {
+    partial = torch::empty({(int64_t)B * Hq * n_split * D},
+                           q.options().dtype(torch::kFloat));
+    partial_ml = torch::empty({(int64_t)B * Hq * n_split * 2},
+                              q.options().dtype(torch::kFloat));
+    pp = partial.data_ptr();
+    pml = partial_ml.data_ptr();
+  }
+  tl_decode_attn_mfma(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+                      seq_lens.data_ptr(), out.data_ptr(), pp, pml, B, Hq,
+                      Hkv, Smax, D, (float)scale, (int)n_split,
+                      cur_stream());
   return out;
 }
 

@@ -1,0 +1,304 @@
+// MFMA decode attention (v2) for gfx950.
+//
+// The v1 kernel (decode_attn.hip) is VALU-bound: ~85 VALU ops per 64 B of
+// KV streamed caps it at ~2.6 TB/s effective (profiles/). Here the score
+// and PV math run on the matrix cores instead:
+//
+//   grid (B, Hkv, SPLIT): ONE block covers ALL G <= 16 query heads of one
+//   kv head (single pass over the K/V slab — v1 needed ceil(G/4) passes),
+//   with the MFMA M-dimension = query heads.
+//   Per 16-key tile: QK^T = mfma_f32_16x16x32_bf16 over D/32 k-chunks,
+//   where the B-fragment IS the natural K layout (lane l reads 16 B of key
+//   l&15 at d-offset (l>>4)*8 — no LDS staging for K). V is staged
+//   transposed through LDS (pad stride 40 elems: 16-B aligned, 20-dword
+//   row stride is conflict-free over 16 rows) so the PV B-fragment is a
+//   contiguous ds_read_b128. P round-trips through a small LDS tile to
+//   convert C-layout -> A-layout.
+//
+// Flash-decode SPLIT: for small B*Hkv the seq dim is partitioned across
+// SPLIT blocks writing (m, l, o) partials; a tiny combine kernel merges.
+
+#include "common.hpp"
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr int NWAVE = 4;
+constexpr int KT = 16;            // keys per MFMA tile
+constexpr int VPAD = 8;           // vt row pad (stride 40 elems = 80 B)
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+
+// out layout when SPLIT > 1: partial [B, Hq, SPLIT, D] float, ml [B, Hq,
+// SPLIT, 2] float (m, l). SPLIT == 1 writes bf16 out directly.
+template <int D, int SPLIT_MODE>
+__global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
+    const bf16* __restrict__ q,        // [B, Hq, D]
+    const bf16* __restrict__ k_cache,  // [B, Hkv, Smax, D]
+    const bf16* __restrict__ v_cache,
+    const int* __restrict__ seq_lens,  // [B]
+    bf16* __restrict__ out,            // [B, Hq, D]
+    float* __restrict__ partial,       // [B, Hq, SPLIT, D]
+    float* __restrict__ partial_ml,    // [B, Hq, SPLIT, 2]
+    int Hq, int Hkv, int Smax, float scale, int n_split) {
+  constexpr int KCH = D / 32;        // MFMA k-chunks for QK^T
+  constexpr int NS = D / 16;         // PV output col tiles
+  const int b = blockIdx.x;
+  const int hkv = blockIdx.y;
+  const int split = blockIdx.z;
+  const int G = Hq / Hkv;
+  const int h0 = hkv * G;
+  const int L = seq_lens[b];
+
+  // this split's key range
+  const int per_split = (L + n_split - 1) / n_split;
+  const int k_begin = split * per_split;
+  const int k_end = min(L, k_begin + per_split);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int col = lane & 15;         // key-in-tile / head-col
+  const int quad = lane >> 4;        // 0..3
+
+  __shared__ bf16 vt_lds[NWAVE][D][KT + VPAD];
+  __shared__ bf16 p_lds[NWAVE][16][KT];
+  __shared__ float red_m[NWAVE][16], red_l[NWAVE][16];
+  __shared__ float red_o[NWAVE][16][D];
+
+  const bf16* kbase = k_cache + ((int64_t)b * Hkv + hkv) * Smax * D;
+  const bf16* vbase = v_cache + ((int64_t)b * Hkv + hkv) * Smax * D;
+
+  // ---- A fragments: q for this head group; rows >= G zeroed ----
+  bf16x8_t afrag[KCH];
+  {
+    const bool valid = col < G;
+    const bf16* qrow = q + ((int64_t)b * Hq + h0 + (valid ? col : 0)) * D;
+#pragma unroll
+    for (int c = 0; c < KCH; ++c) {
+      bf16x8_t f = *reinterpret_cast<const bf16x8_t*>(
+          qrow + c * 32 + quad * 8);
+      if (!valid)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) f[j] = (__bf16)0.f;
+      afrag[c] = f;
+    }
+  }
+
+  float m_run[4], l_run[4];
+  f32x4 oacc[NS];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+#pragma unroll
+  for (int n = 0; n < NS; ++n) oacc[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  // ---- main loop: each wave takes every NWAVE'th 16-key tile ----
+  for (int t0 = k_begin + wave * KT; t0 < k_end; t0 += NWAVE * KT) {
+    const int key = t0 + col;
+    const bool kv_ok = key < k_end;
+    const int safe_key = kv_ok ? key : (k_end - 1);
+
+    // B fragments straight from global K + stage V transposed into LDS
+    f32x4 sacc = (f32x4){0.f, 0.f, 0.f, 0.f};
+    {
+      const bf16* krow = kbase + (int64_t)safe_key * D;
+      const bf16* vrow = vbase + (int64_t)safe_key * D;
+#pragma unroll
+      for (int c = 0; c < KCH; ++c) {
+        bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
+            krow + c * 32 + quad * 8);
+        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[c], kf, sacc,
+                                                       0, 0, 0);
+        // V: same addressing; scatter-transpose into vt[d][key]
+        bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+            vrow + c * 32 + quad * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<__bf16*>(
+              &vt_lds[wave][c * 32 + quad * 8 + j][col]) = vf[j];
+      }
+    }
+
+    // ---- online softmax on the 16-key score tile ----
+    // lane holds S[row=quad*4+r][col] in sacc[r]
+    float p[4];
+    float tile_max = -1e30f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      sacc[r] = kv_ok ? sacc[r] * scale : -1e30f;
+      tile_max = fmaxf(tile_max, sacc[r]);
+    }
+    // max over the 16 cols (row set identical within 16-lane group)
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1)
+      tile_max = fmaxf(tile_max, __shfl_xor(tile_max, off, WAVE_SIZE));
+    // NOTE: tile_max is max over the lane's 4 rows and 16 cols; per-row
+    // max would be tighter but needs 4 separate reduces — exp headroom in
+    // fp32 is ample (scores are O(10)), shared max is safe and cheaper.
+    float m_new[4], alpha[4], psum[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_new[r] = fmaxf(m_run[r], tile_max);
+      alpha[r] = __expf(m_run[r] - m_new[r]);
+      p[r] = (kv_ok) ? __expf(sacc[r] - m_new[r]) : 0.f;
+      psum[r] = p[r];
+      m_run[r] = m_new[r];
+    }
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        psum[r] += __shfl_xor(psum[r], off, WAVE_SIZE);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) l_run[r] = l_run[r] * alpha[r] + psum[r];
+
+    // P -> LDS (C-layout write), then A-fragment read. NO barriers here:
+    // each wave owns its own LDS slabs ([wave] index) and same-wave
+    // ds_write -> ds_read ordering is enforced by lgkmcnt; a block-wide
+    // barrier would deadlock (waves run different trip counts).
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      p_lds[wave][quad * 4 + r][col] = f2bf(p[r]);
+    // rescale O accumulators by alpha (row = quad*4+r)
+#pragma unroll
+    for (int n = 0; n < NS; ++n)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) oacc[n][r] *= alpha[r];
+
+    // ---- PV: K=32 MFMA with 16 valid keys + 16 zero keys ----
+    // A: lane holds P[row=col][key=quad*8+j]; keys 16..31 read zeros
+    bf16x8_t pa;
+    {
+      // quad 0,1 -> keys 0..15 from p_lds; quad 2,3 -> zero
+      if (quad < 2) {
+        pa = *reinterpret_cast<const bf16x8_t*>(&p_lds[wave][col][quad * 8]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) pa[j] = (__bf16)0.f;
+      }
+    }
+#pragma unroll
+    for (int n = 0; n < NS; ++n) {
+      // B: lane holds V[key=quad*8+j][dv=n*16+col] = vt[dv][key]
+      bf16x8_t vb;
+      if (quad < 2) {
+        vb = *reinterpret_cast<const bf16x8_t*>(
+            &vt_lds[wave][n * 16 + col][quad * 8]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vb[j] = (__bf16)0.f;
+      }
+      oacc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, oacc[n],
+                                                        0, 0, 0);
+    }
+  }
+
+  // ---- wave partials -> LDS ----
+  // lane holds O[row=quad*4+r][col=n*16+col]; every 16-lane group holds the
+  // full row set, so groups 1..3 duplicate group 0 -> write from quad 0..3
+  // cover distinct (row, col) pairs already: row=quad*4+r col covers only
+  // cols where lane&15 == col. Write all lanes (distinct rows per quad).
+#pragma unroll
+  for (int n = 0; n < NS; ++n)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      red_o[wave][quad * 4 + r][n * 16 + col] = oacc[n][r];
+  if (col == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      red_m[wave][quad * 4 + r] = m_run[r];
+      red_l[wave][quad * 4 + r] = l_run[r];
+    }
+  }
+  __syncthreads();
+
+  // ---- final merge across waves; write out ----
+  const int g_count = min(G, 16);
+  for (int i = threadIdx.x; i < g_count * D; i += BLOCK) {
+    const int g = i / D, d = i - g * D;
+    float m_tot = -1e30f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) m_tot = fmaxf(m_tot, red_m[w][g]);
+    float l_tot = 0.f, o = 0.f;
+#pragma unroll
+    for (int w = 0; w < NWAVE; ++w) {
+      const float f = __expf(red_m[w][g] - m_tot);
+      l_tot += red_l[w][g] * f;
+      o += red_o[w][g][d] * f;
+    }
+    if (SPLIT_MODE) {
+      const int64_t base = (((int64_t)b * Hq + h0 + g) * n_split + split);
+      partial[base * D + d] = o;
+      if (d == 0) {
+        partial_ml[base * 2] = m_tot;
+        partial_ml[base * 2 + 1] = l_tot;
+      }
+    } else {
+      out[((int64_t)b * Hq + h0 + g) * D + d] =
+          f2bf(o / fmaxf(l_tot, 1e-30f));
+    }
+  }
+}
+
+// combine kernel: merge SPLIT partials per (b, h)
+template <int D>
+__global__ __launch_bounds__(128) void decode_attn_combine_kernel(
+    const float* __restrict__ partial, const float* __restrict__ partial_ml,
+    bf16* __restrict__ out, int Hq, int n_split) {
+  const int bh = blockIdx.x;                    // b * Hq + h
+  const int d = threadIdx.x;                    // [0, D)
+  __shared__ float sm[64], sl[64];
+  if (threadIdx.x < n_split) {
+    sm[threadIdx.x] = partial_ml[((int64_t)bh * n_split + threadIdx.x) * 2];
+    sl[threadIdx.x] = partial_ml[((int64_t)bh * n_split + threadIdx.x) * 2 + 1];
+  }
+  __syncthreads();
+  float m_tot = -1e30f;
+  for (int s = 0; s < n_split; ++s) m_tot = fmaxf(m_tot, sm[s]);
+  float l_tot = 0.f, o = 0.f;
+  for (int s = 0; s < n_split; ++s) {
+    const float f = __expf(sm[s] - m_tot);
+    l_tot += sl[s] * f;
+    o += partial[((int64_t)bh * n_split + s) * D + d] * f;
+  }
+  out[(int64_t)bh * D + d] = f2bf(o / fmaxf(l_tot, 1e-30f));
+}
+
+}  // namespace
+
+extern "C" {
+
+// scratch: partial floats [B*Hq*n_split*D] + ml [B*Hq*n_split*2]; pass
+// nullptr when n_split == 1.
+void tl_decode_attn_mfma(const void* q, const void* k_cache,
+                         const void* v_cache, const void* seq_lens, void* out,
+                         void* partial, void* partial_ml, int B, int Hq,
+                         int Hkv, int Smax, int D, float scale, int n_split,
+                         hipStream_t stream) {
+  dim3 grid(B, Hkv, n_split), block(BLOCK);
+#define LAUNCH(DD, SM)                                                       \
+  hipLaunchKernelGGL((decode_attn_mfma_kernel<DD, SM>), grid, block, 0,      \
+                     stream, (const bf16*)q, (const bf16*)k_cache,           \
+                     (const bf16*)v_cache, (const int*)seq_lens, (bf16*)out, \
+                     (float*)partial, (float*)partial_ml, Hq, Hkv, Smax,     \
+                     scale, n_split)
+  if (D == 128) {
+    if (n_split == 1) LAUNCH(128, 0); else LAUNCH(128, 1);
+  } else if (D == 64) {
+    if (n_split == 1) LAUNCH(64, 0); else LAUNCH(64, 1);
+  }
+#undef LAUNCH
+  if (n_split > 1) {
+    dim3 cgrid(B * Hq), cblock(D);
+    if (D == 128)
+      hipLaunchKernelGGL((decode_attn_combine_kernel<128>), cgrid, cblock, 0,
+                         stream, (const float*)partial,
+                         (const float*)partial_ml, (bf16*)out, Hq, n_split);
+    else
+      hipLaunchKernelGGL((decode_attn_combine_kernel<64>), cgrid, cblock, 0,
+                         stream, (const float*)partial,
+                         (const float*)partial_ml, (bf16*)out, Hq, n_split);
+  }
+}
+
+}  // extern "C"

@@ -216,3 +216,31 @@ def test_rope_append_fused():
                                        atol=2e-2, rtol=2e-2)
             torch.testing.assert_close(vc[b, :, p].float(), vB[b, s],
                                        atol=1e-3, rtol=1e-3)
+
+
+@pytest.mark.parametrize("B,Hq,Hkv,D,L,ns", [
+    (2, 28, 4, 128, 300, 1),    # Qwen2.5-7B, no split
+    (2, 28, 4, 128, 300, 4),    # with flash-decode split
+    (1, 32, 8, 128, 1000, 8),   # Llama/Qwen3 G=4, long L, split
+    (4, 32, 8, 128, 77, 1),
+    (1, 4, 2, 64, 333, 2),      # D=64
+    (3, 8, 8, 128, 64, 1),      # MHA G=1
+    (2, 64, 8, 128, 129, 2),    # G=8 (Llama-3-70B shape)
+])
+def test_decode_attn_mfma(B, Hq, Hkv, D, L, ns):
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(11)
+    Smax = L + 19
+    q = torch.randn(B, Hq, D, device=DEV, dtype=torch.bfloat16)
+    kc = torch.randn(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
+    vc = torch.randn(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
+    lens = torch.randint(max(1, L - 40), L + 1, (B,), device=DEV,
+                         dtype=torch.int32)
+    lens[0] = L
+    out = ops.attention_decode(q, kc, vc, lens, n_split=ns)
+    out_ref = ref.attention_decode(
+        q.unsqueeze(1).float(), kc.permute(0, 2, 1, 3).float(),
+        vc.permute(0, 2, 1, 3).float(), lens)
+    torch.testing.assert_close(out.float(), out_ref[:, 0], atol=2e-2,
+                               rtol=2e-2)
