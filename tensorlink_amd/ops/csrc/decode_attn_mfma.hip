@@ -120,25 +120,24 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
     }
 
     // ---- online softmax on the 16-key score tile ----
-    // lane holds S[row=quad*4+r][col] in sacc[r]
-    float p[4];
-    float tile_max = -1e30f;
+    // lane holds S[row=quad*4+r][col] in sacc[r]; per-ROW max (a shared
+    // 4-row max underflows exp when head score ranges differ by >88)
+    float p[4], row_max[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       sacc[r] = kv_ok ? sacc[r] * scale : -1e30f;
-      tile_max = fmaxf(tile_max, sacc[r]);
+      row_max[r] = sacc[r];
     }
-    // max over the 16 cols (row set identical within 16-lane group)
 #pragma unroll
     for (int off = 1; off < 16; off <<= 1)
-      tile_max = fmaxf(tile_max, __shfl_xor(tile_max, off, WAVE_SIZE));
-    // NOTE: tile_max is max over the lane's 4 rows and 16 cols; per-row
-    // max would be tighter but needs 4 separate reduces — exp headroom in
-    // fp32 is ample (scores are O(10)), shared max is safe and cheaper.
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        row_max[r] = fmaxf(row_max[r], __shfl_xor(row_max[r], off,
+                                                  WAVE_SIZE));
     float m_new[4], alpha[4], psum[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      m_new[r] = fmaxf(m_run[r], tile_max);
+      m_new[r] = fmaxf(m_run[r], row_max[r]);
       alpha[r] = __expf(m_run[r] - m_new[r]);
       p[r] = (kv_ok) ? __expf(sacc[r] - m_new[r]) : 0.f;
       psum[r] = p[r];
