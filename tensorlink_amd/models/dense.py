@@ -61,16 +61,24 @@ class KVCache:
         self.seq_lens += n
 
 
+class TLLinear(nn.Linear):
+    """nn.Linear whose inference forward routes decode-M shapes to the
+    hand-written MFMA skinny GEMM (ops/csrc/skinny_gemm.hip)."""
+
+    def forward(self, x):
+        return ops.linear(x, self.weight, self.bias)
+
+
 class Attention(nn.Module):
     def __init__(self, config: ModelConfig):
         super().__init__()
         self.config = config
         h, q, kv = config.hidden_size, config.q_size, config.kv_size
         bias = config.qkv_bias
-        self.q_proj = nn.Linear(h, q, bias=bias)
-        self.k_proj = nn.Linear(h, kv, bias=bias)
-        self.v_proj = nn.Linear(h, kv, bias=bias)
-        self.o_proj = nn.Linear(q, h, bias=False)
+        self.q_proj = TLLinear(h, q, bias=bias)
+        self.k_proj = TLLinear(h, kv, bias=bias)
+        self.v_proj = TLLinear(h, kv, bias=bias)
+        self.o_proj = TLLinear(q, h, bias=False)
         self.n_heads = config.num_attention_heads
         self.n_kv = config.num_key_value_heads
         self.head_dim = config.head_dim
@@ -138,9 +146,9 @@ class MLP(nn.Module):
     def __init__(self, config: ModelConfig):
         super().__init__()
         h, i = config.hidden_size, config.intermediate_size
-        self.gate_proj = nn.Linear(h, i, bias=False)
-        self.up_proj = nn.Linear(h, i, bias=False)
-        self.down_proj = nn.Linear(i, h, bias=False)
+        self.gate_proj = TLLinear(h, i, bias=False)
+        self.up_proj = TLLinear(h, i, bias=False)
+        self.down_proj = TLLinear(i, h, bias=False)
 
     def forward(self, x):
         return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
@@ -226,8 +234,8 @@ class StageModel(nn.Module):
             # copy of the tied weight (the reference instead pins tied
             # modules to the host — graphing.py:532-537).
             if not (config.tie_word_embeddings and has_embedding):
-                self.lm_head = nn.Linear(config.hidden_size,
-                                         config.vocab_size, bias=False)
+                self.lm_head = TLLinear(config.hidden_size,
+                                        config.vocab_size, bias=False)
 
     @property
     def num_layers(self):

@@ -229,6 +229,31 @@ def attention_decode(q, k_cache, v_cache, seq_lens,
 
 
 # ---------------------------------------------------------------------------
+# Linear (skinny-M decode GEMM)
+# ---------------------------------------------------------------------------
+_SKINNY_MAX_M = 512
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor,
+           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """x @ W^T + bias. Routes skinny decode shapes (M <= 512) to the
+    hand-written MFMA streaming kernel; everything else (prefill M,
+    training) to torch/hipBLASLt."""
+    import os as _os
+    if (x.is_cuda and not torch.is_grad_enabled()
+            and x.dtype == torch.bfloat16
+            and not _os.environ.get("TL_NO_SKINNY")):
+        K = x.shape[-1]
+        M = x.numel() // K
+        N = weight.shape[0]
+        if M <= _SKINNY_MAX_M and K % 32 == 0 and N % 64 == 0:
+            return _require_ext().skinny_gemm(
+                x.contiguous(), weight,
+                bias.to(torch.bfloat16) if bias is not None else None)
+    return torch.nn.functional.linear(x, weight, bias)
+
+
+# ---------------------------------------------------------------------------
 # AdamW
 # ---------------------------------------------------------------------------
 def adamw_(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,

@@ -39,6 +39,8 @@ void tl_decode_attn_mfma(const void* q, const void* k_cache,
                          void* partial, void* partial_ml, int B, int Hq,
                          int Hkv, int Smax, int D, float scale, int n_split,
                          hipStream_t stream);
+void tl_skinny_gemm(const void* x, const void* w, const void* bias,
+                    void* out, int M, int N, int K, hipStream_t stream);
 }
 
 namespace {
@@ -229,11 +231,33 @@ void rope_append_(Tensor q, Tensor k, Tensor v, Tensor k_cache,
                  Hkv, D, Smax, cur_stream());
 }
 
+Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
+  CHECK_IN(x, torch::kBFloat16);
+  CHECK_IN(w, torch::kBFloat16);
+  const int K = x.size(-1);
+  const int64_t M = x.numel() / K;
+  const int N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "unsupported shape");
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto out = torch::empty(sizes, x.options());
+  const void* bp = nullptr;
+  if (bias.has_value()) {
+    CHECK_IN(bias.value(), torch::kBFloat16);
+    bp = bias->data_ptr();
+  }
+  tl_skinny_gemm(x.data_ptr(), w.data_ptr(), bp, out.data_ptr(), (int)M, N,
+                 K, cur_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("rope_append_", &rope_append_,
           "fused in-place RoPE + KV-cache append");
+  mod.def("skinny_gemm", &skinny_gemm, "decode-M GEMM: x @ W^T + bias");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm fwd (+residual)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm bwd");
   mod.def("rope_", &rope_, "in-place RoPE apply");

@@ -244,3 +244,23 @@ def test_decode_attn_mfma(B, Hq, Hkv, D, L, ns):
         vc.permute(0, 2, 1, 3).float(), lens)
     torch.testing.assert_close(out.float(), out_ref[:, 0], atol=2e-2,
                                rtol=2e-2)
+
+
+@pytest.mark.parametrize("M,N,K,bias", [
+    (256, 3584, 3584, False),     # o_proj
+    (256, 4608, 3584, True),      # fused-qkv shape w/ bias
+    (256, 18944, 3584, False),    # gate/up
+    (256, 3584, 18944, False),    # down
+    (7, 1024, 256, True),         # tiny M + K%64!=0 tail (K=256 ok), M tail
+    (33, 512, 96, False),         # K%64==32 tail path
+])
+def test_skinny_gemm(M, N, K, bias):
+    ops = _ext()
+    torch.manual_seed(12)
+    x = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(N, K, device=DEV, dtype=torch.bfloat16) / (K ** 0.5)
+    b = torch.randn(N, device=DEV, dtype=torch.bfloat16) if bias else None
+    out = ops.linear(x, w, b)
+    ref = torch.nn.functional.linear(x.float(), w.float(),
+                                     b.float() if bias else None)
+    torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
