@@ -231,7 +231,7 @@ def attention_decode(q, k_cache, v_cache, seq_lens,
 # ---------------------------------------------------------------------------
 # Linear (skinny-M decode GEMM)
 # ---------------------------------------------------------------------------
-_SKINNY_MAX_M = 512
+_SKINNY_MAX_M = 256
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor,

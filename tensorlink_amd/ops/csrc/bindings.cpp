@@ -40,7 +40,8 @@ void tl_decode_attn_mfma(const void* q, const void* k_cache,
                          int Hkv, int Smax, int D, float scale, int n_split,
                          hipStream_t stream);
 void tl_skinny_gemm(const void* x, const void* w, const void* bias,
-                    void* out, int M, int N, int K, hipStream_t stream);
+                    void* out, void* partial, int M, int N, int K,
+                    int n_split, hipStream_t stream);
 }
 
 namespace {
@@ -239,6 +240,7 @@ Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
   const int N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "K mismatch");
   TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "unsupported shape");
+  TORCH_CHECK(M <= 256, "skinny_gemm supports M <= 256");
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto out = torch::empty(sizes, x.options());
@@ -247,8 +249,20 @@ Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
     CHECK_IN(bias.value(), torch::kBFloat16);
     bp = bias->data_ptr();
   }
-  tl_skinny_gemm(x.data_ptr(), w.data_ptr(), bp, out.data_ptr(), (int)M, N,
-                 K, cur_stream());
+  // split K so (N/64)*n_split lands near 1 block/CU (guide: ~0.5-1x CUs)
+  int n_split = 1;
+  while ((N / 64) * n_split < 256 && n_split < 8 &&
+         (K / 32) / (n_split * 2) >= 2)
+    n_split <<= 1;
+  Tensor partial;
+  void* pp = nullptr;
+  if (n_split > 1) {
+    partial = torch::empty({(int64_t)n_split * M * N},
+                           x.options().dtype(torch::kFloat));
+    pp = partial.data_ptr();
+  }
+  tl_skinny_gemm(x.data_ptr(), w.data_ptr(), bp, out.data_ptr(), pp, (int)M,
+                 N, K, n_split, cur_stream());
   return out;
 }
 
