@@ -54,19 +54,26 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
   for (int kc = k_begin + quad * 8; kc < k_end; kc += 32) {
     bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(wrow + kc);
     const bf16* xcol = x + kc;
-    for (int m = 0; m < m_tiles; ++m) {
-      const int row = min(m * 16 + col, M - 1);
-      bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-          xcol + (int64_t)row * K);
-      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[m],
-                                                       0, 0, 0);
+    // fully unrolled with compile-time m: a runtime-indexed acc[m] would
+    // allocate in scratch (guide §5.4 rule 20 — measured 5x+ slower)
+#pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      if (m < m_tiles) {
+        const int row = min(m * 16 + col, M - 1);
+        bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
+            xcol + (int64_t)row * K);
+        acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                         acc[m], 0, 0, 0);
+      }
     }
   }
 
   // epilogue: lane holds C[row=quad*4+r][col] per m-tile
   const float b = HAS_BIAS ? bf2f(bias[blockIdx.x * 64 + wave * 16 + col])
                            : 0.f;
-  for (int m = 0; m < m_tiles; ++m) {
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    if (m >= m_tiles) break;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int orow = m * 16 + quad * 4 + r;
