@@ -17,4 +17,4 @@ for M,N,K,name in shapes:
     t_lib = timeit(lambda: torch.nn.functional.linear(x,w))
     t_sk  = timeit(lambda: ops._require_ext().skinny_gemm(x,w,None))
     wbytes = N*K*2/1e9
-    print(f"{name:8s} M{M} N{N} K{K}: lib {t_lib:7.1f}us ({wbytes/t_lib*1e6:5.2f} TB/s)  skinny {t_sk:7.1f}us ({wbytes/t_sk*1e6:5.2f} TB/s)")
+    print(f"{name:8s} M{M} N{N} K{K}: lib {t_lib:7.1f}us ({wbytes/t_lib*1e3:6.2f} TB/s)  skinny {t_sk:7.1f}us ({wbytes/t_sk*1e3:6.2f} TB/s)")

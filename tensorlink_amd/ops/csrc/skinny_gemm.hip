@@ -2,29 +2,31 @@
 //
 // torch Linear stores W as [N, K] row-major, so both operands are
 // k-contiguous — the same A·B^T fragment pattern as QK^T. At decode M
-// (<= 256 rows) the GEMM is pure W-streaming; hipBLASLt measured only
-// 1.7-3.5 TB/s on these shapes (profiles/).
+// (<= 256 rows) the GEMM is pure W-streaming.
 //
-// Geometry (v2 — v1 gridded over M too, which re-streamed every W panel
-// once per M-tile through different XCD L2s: 4x HBM traffic, 7x slower
-// than hipBLASLt): one block per 64-column W panel (grid.x = N/64,
-// grid.y = SPLITK), 4 waves; wave w owns columns w*16 and loops ALL
-// M-tiles in registers (acc[16][4] fp32), so every W byte is loaded
-// exactly once chip-wide and each wave walks its 16 W rows sequentially
-// along k (DRAM-friendly streams). x is tiny (M*K bf16, L2-resident) and
-// re-read per wave. SPLITK slices K when N/64 alone can't fill the chip;
-// fp32 partials are combined by a tiny second kernel.
+// v3 geometry (measured on MI355X):
+//   v1 gridded over M: every W panel re-streamed per M-tile -> 0.4 TB/s.
+//   v2 looped M-tiles with fragment-shaped x loads straight from global:
+//   16 scatter-line loads per W load -> TA transaction-bound, 0.23-0.37
+//   TB/s (the guide's "fragment-shaped x" trap).
+//   v3 stages the x k-slab into LDS ONCE per block with full-line
+//   coalesced loads (+16 B row pad -> conflict-free ds_read_b128
+//   A-fragments), shared by all 4 waves x all M-tiles; W streams straight
+//   from HBM exactly once chip-wide.
+// Grid (N/64, SPLITK); block = 4 waves; wave w owns columns w*16; SPLITK
+// slices K when N/64 can't fill the chip (fp32 partials + reduce kernel).
 
 #include "common.hpp"
 
 namespace {
 
 constexpr int BLOCK = 256;
+constexpr int XPAD = 8;              // x_lds row stride 72 elems (144 B):
+                                     // 16-B aligned, conflict-free b128
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 
-// MT = max M-tiles (compile-time); runtime m_tiles <= MT.
 template <int MT, int HAS_BIAS, int SPLIT>
 __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
     const bf16* __restrict__ x,     // [M, K]
@@ -40,10 +42,13 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
   const int oc = blockIdx.x * 64 + wave * 16 + col;   // output column
 
   const int m_tiles = (M + 15) / 16;
+  const int rows = m_tiles * 16;
   const int split = SPLIT ? blockIdx.y : 0;
   const int k_per = SPLIT ? ((K / 32 + n_split - 1) / n_split) * 32 : K;
   const int k_begin = split * k_per;
   const int k_end = min(K, k_begin + k_per);
+
+  __shared__ bf16 x_lds[MT * 16][64 + XPAD];
 
   f32x4 acc[MT];
 #pragma unroll
@@ -51,26 +56,43 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
 
   const bf16* wrow = w + (int64_t)oc * K;   // this lane's W row
 
-  for (int kc = k_begin + quad * 8; kc < k_end; kc += 32) {
-    bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(wrow + kc);
-    const bf16* xcol = x + kc;
-    // fully unrolled with compile-time m: a runtime-indexed acc[m] would
-    // allocate in scratch (guide §5.4 rule 20 — measured 5x+ slower)
-#pragma unroll
-    for (int m = 0; m < MT; ++m) {
-      if (m < m_tiles) {
-        const int row = min(m * 16 + col, M - 1);
-        bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-            xcol + (int64_t)row * K);
-        acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                         acc[m], 0, 0, 0);
+  for (int k0 = k_begin; k0 < k_end; k0 += 64) {
+    const int tile_k = min(64, k_end - k0);            // 64 or 32
+    // ---- stage x[0:rows][k0:k0+tile_k] into LDS, full-line loads ----
+    {
+      const int vecs = rows * (tile_k / 8);            // 16-B units
+      for (int i = threadIdx.x; i < vecs; i += BLOCK) {
+        const int r = i / (tile_k / 8);
+        const int c8 = i - r * (tile_k / 8);
+        const int xr = min(r, M - 1);
+        bf16x8_t val = *reinterpret_cast<const bf16x8_t*>(
+            x + (int64_t)xr * K + k0 + c8 * 8);
+        *reinterpret_cast<bf16x8_t*>(&x_lds[r][c8 * 8]) = val;
       }
     }
+    __syncthreads();
+    // ---- compute: W streams from HBM, A-fragments from LDS ----
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      if (kk * 32 < tile_k) {
+        bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
+            wrow + k0 + kk * 32 + quad * 8);
+#pragma unroll
+        for (int m = 0; m < MT; ++m) {
+          if (m < m_tiles) {
+            bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
+                &x_lds[m * 16 + col][kk * 32 + quad * 8]);
+            acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag, bfrag, acc[m], 0, 0, 0);
+          }
+        }
+      }
+    }
+    __syncthreads();
   }
 
   // epilogue: lane holds C[row=quad*4+r][col] per m-tile
-  const float b = HAS_BIAS ? bf2f(bias[blockIdx.x * 64 + wave * 16 + col])
-                           : 0.f;
+  const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
 #pragma unroll
   for (int m = 0; m < MT; ++m) {
     if (m >= m_tiles) break;
@@ -79,11 +101,9 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
       const int orow = m * 16 + quad * 4 + r;
       if (orow < M) {
         if (SPLIT)
-          partial[((int64_t)split * M + orow) * N + blockIdx.x * 64 +
-                  wave * 16 + col] = acc[m][r];
+          partial[((int64_t)split * M + orow) * N + oc] = acc[m][r];
         else
-          out[(int64_t)orow * N + blockIdx.x * 64 + wave * 16 + col] =
-              f2bf(acc[m][r] + b);
+          out[(int64_t)orow * N + oc] = f2bf(acc[m][r] + b);
       }
     }
   }
