@@ -231,7 +231,10 @@ def attention_decode(q, k_cache, v_cache, seq_lens,
 # ---------------------------------------------------------------------------
 # Linear (skinny-M decode GEMM)
 # ---------------------------------------------------------------------------
-_SKINNY_MAX_M = 256
+# Measured on MI355X: the hand-written streaming GEMM beats hipBLASLt up
+# to M≈64 (2.2 vs 1.3 TB/s at M=16); at M=256 hipBLASLt's tuned picks win
+# (profiles/skinny_gemm_ab.md), so large decode batches route to the lib.
+_SKINNY_MAX_M = 64
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor,

@@ -1,14 +1,19 @@
 // Causal GQA flash-attention forward (prefill) on gfx950 MFMA.
 //
 // Replaces the reference's HF SDPA-inside-eager-blocks prefill
-// (tensorlink/ml/worker.py:330-335). Tiled for CDNA4: 64-query-row blocks,
-// 64-key tiles staged through LDS, QK^T and P·V on
-// v_mfma_f32_16x16x32_bf16, online softmax with fp32 running stats.
+// (tensorlink/ml/worker.py:330-335).
 //
-// v1 structure (correctness-first): scores round-trip through LDS for the
-// softmax phase; V is transposed into LDS at load time so the P·V B-fragment
-// reads are contiguous ds_read_b128. In-register softmax / tr_b16 reads are
-// later optimizations (guide T10/T12).
+// v2 structure: v1 round-tripped the fp32 score tile through LDS for a
+// separate softmax phase — 16-way bank conflicts on the [64][64] fp32 tile
+// and a serial 4-threads-per-row pass held it at 59 TF. v2 keeps scores in
+// the MFMA C-fragments and does the online softmax with 16-lane shuffles
+// (rows live on fixed (quad, reg) lanes across all K-tiles, so m/l/O state
+// stays in registers); only the bf16 P tile touches LDS (for the PV
+// A-fragment layout), padded conflict-free.
+//
+//   block = 256 threads (4 waves); BM = 64 query rows (wave w owns rows
+//   w*16); K-tiles of BN = 64 staged once per block (K direct-read
+//   fragments, V transposed into LDS for contiguous PV B-fragments).
 //
 // q [B,S,Hq,D], k,v [B,S,Hkv,D] bf16 row-major; out [B,S,Hq,D].
 
@@ -19,187 +24,173 @@ namespace {
 constexpr int BLOCK = 256;   // 4 waves
 constexpr int BM = 64;       // query rows per block
 constexpr int BN = 64;       // key cols per tile
-constexpr int PPAD = 8;      // LDS padding for P tile
+constexpr int PPAD = 8;      // P row pad: stride 72 elems = 144 B
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
-using bf16x4_t = __attribute__((ext_vector_type(4))) __bf16;
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 
-// mfma_f32_16x16x32_bf16 fragment maps (A 16x32, B 32x16, C 16x16):
-//   A: lane l holds A[l&15][(l>>4)*8 + j], j=0..7   (8 consecutive k)
-//   B: lane l holds B[(l>>4)*8 + j][l&15]
-//   C: lane l reg r holds C[(l>>4)*4 + r][l&15]
 template <int D>
 __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out, int B, int S, int Hq,
     int Hkv, float scale, int causal) {
   static_assert(D == 64 || D == 128);
-  constexpr int NSUB_PV = D / 16;  // MFMA col tiles for PV (output cols)
+  constexpr int KCH = D / 32;      // QK^T k-chunks
+  constexpr int NSUB = BN / 16;    // score col tiles = 4
+  constexpr int NS_PV = D / 16;    // PV output col tiles
 
-  __shared__ bf16 q_lds[BM][D];
-  __shared__ bf16 k_lds[BN][D];
-  __shared__ bf16 vt_lds[D][BN + 8];   // V transposed [d][n]; +8 keeps
-                                       // 16-B alignment of b128 reads
-  __shared__ float s_lds[BM][BN];
-  __shared__ bf16 p_lds[BM][BN + PPAD];
-  __shared__ float m_lds[BM], l_lds[BM], alpha_lds[BM];
+  // k_lds rows padded +8 elems: stride 272 B (17 slots) -> fragment reads
+  // of 16 different rows land on distinct banks (an unpadded 256-B stride
+  // puts EVERY row on bank 0: 16-way conflict).
+  __shared__ bf16 k_lds[BN][D + 8];
+  __shared__ bf16 vt_lds[D][BN + 8];
+  __shared__ bf16 p_lds[4][16][BN + PPAD];   // per wave: 16 rows x 64 keys
 
-  const int qt = blockIdx.x;           // query tile index
-  const int h = blockIdx.y;            // query head
+  const int qt = blockIdx.x;
+  const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
-  const int q0 = qt * BM;              // first query row (global)
+  const int q0 = qt * BM;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int r0 = wave * 16;            // wave's first row within tile
+  const int col = lane & 15;
+  const int quad = lane >> 4;
+  const int r0 = wave * 16;                  // wave's first row in tile
 
-  // ---- load Q tile [BM][D] ----
+  // ---- Q fragments in registers: rows r0+col, all KCH chunks ----
+  bf16x8_t qfrag[KCH];
   {
-    const int64_t qbase = ((int64_t)b * S) * Hq * D + (int64_t)h * D;
-    for (int i = threadIdx.x * 8; i < BM * D; i += BLOCK * 8) {
-      const int row = i / D, col = i % D;
-      bf16x8 val;
-      if (q0 + row < S)
-        val = *reinterpret_cast<const bf16x8*>(
-            q + qbase + (int64_t)(q0 + row) * Hq * D + col);
-      else
+    const int qrow = q0 + r0 + col;
+    const int safe = min(qrow, S - 1);
+    const bf16* qp = q + (((int64_t)b * S + safe) * Hq + h) * D;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) val.v[j] = f2bf(0.f);
-      *reinterpret_cast<bf16x8*>(&q_lds[row][col]) = val;
-    }
+    for (int c = 0; c < KCH; ++c)
+      qfrag[c] = *reinterpret_cast<const bf16x8_t*>(qp + c * 32 + quad * 8);
   }
-  if (threadIdx.x < BM) {
-    m_lds[threadIdx.x] = -1e30f;
-    l_lds[threadIdx.x] = 0.f;
-  }
-  __syncthreads();
 
-  // O accumulators: wave's 16 rows x D cols -> NSUB_PV x 4 regs per lane
-  f32x4 oacc[NSUB_PV];
+  float m_run[4], l_run[4];
+  f32x4 oacc[NS_PV];
 #pragma unroll
-  for (int i = 0; i < NSUB_PV; ++i) oacc[i] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+#pragma unroll
+  for (int n = 0; n < NS_PV; ++n) oacc[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? min(S, q0 + BM) : S;
   const int64_t kbase = ((int64_t)b * S) * Hkv * D + (int64_t)hkv * D;
 
   for (int kt0 = 0; kt0 < kv_end; kt0 += BN) {
-    // ---- load K tile [BN][D]; V transposed into vt_lds[D][BN] ----
+    // ---- stage K tile [BN][D]; V transposed [D][BN] ----
     for (int i = threadIdx.x * 8; i < BN * D; i += BLOCK * 8) {
-      const int row = i / D, col = i % D;
-      bf16x8 kval, vval;
-      if (kt0 + row < S) {
-        const int64_t off = kbase + (int64_t)(kt0 + row) * Hkv * D + col;
-        kval = *reinterpret_cast<const bf16x8*>(k + off);
-        vval = *reinterpret_cast<const bf16x8*>(v + off);
-      } else {
+      const int row = i / D, c = i % D;
+      const int key = kt0 + row;
+      const int64_t off = kbase + (int64_t)min(key, S - 1) * Hkv * D + c;
+      bf16x8_t kval = *reinterpret_cast<const bf16x8_t*>(k + off);
+      bf16x8_t vval = *reinterpret_cast<const bf16x8_t*>(v + off);
+      if (key >= S) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) { kval.v[j] = f2bf(0.f); vval.v[j] = f2bf(0.f); }
+        for (int j = 0; j < 8; ++j) { kval[j] = (__bf16)0.f; vval[j] = (__bf16)0.f; }
       }
-      *reinterpret_cast<bf16x8*>(&k_lds[row][col]) = kval;
+      *reinterpret_cast<bf16x8_t*>(&k_lds[row][c]) = kval;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vt_lds[col + j][row] = vval.v[j];
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<__bf16*>(&vt_lds[c + j][row]) = vval[j];
     }
     __syncthreads();
 
-    // ---- S = Q K^T for wave's 16 rows x BN cols ----
+    // ---- QK^T for all 4 col-subtiles (scores stay in registers) ----
+    f32x4 sacc[NSUB];
 #pragma unroll
-    for (int ns = 0; ns < BN / 16; ++ns) {
-      f32x4 acc = (f32x4){0.f, 0.f, 0.f, 0.f};
+    for (int ns = 0; ns < NSUB; ++ns) {
+      sacc[ns] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int ks = 0; ks < D / 32; ++ks) {
-        bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-            &q_lds[r0 + (lane & 15)][ks * 32 + (lane >> 4) * 8]);
+      for (int c = 0; c < KCH; ++c) {
         bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-            &k_lds[ns * 16 + (lane & 15)][ks * 32 + (lane >> 4) * 8]);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0, 0, 0);
+            &k_lds[ns * 16 + col][c * 32 + quad * 8]);
+        sacc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfrag,
+                                                           sacc[ns], 0, 0, 0);
       }
+    }
+    // ---- mask + scale + ONE tile-wide row max (P for every subtile is
+    // written against the same m, so no stale-P rescale hazard) ----
+    float row_max[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) row_max[r] = -1e30f;
+#pragma unroll
+    for (int ns = 0; ns < NSUB; ++ns) {
+      const int kcol = kt0 + ns * 16 + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + r0 + quad * 4 + r;
+        const bool masked = (kcol >= S) || (causal && kcol > qrow);
+        sacc[ns][r] = masked ? -1e30f : sacc[ns][r] * scale;
+        row_max[r] = fmaxf(row_max[r], sacc[ns][r]);
+      }
+    }
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        s_lds[r0 + (lane >> 4) * 4 + r][ns * 16 + (lane & 15)] = acc[r];
-    }
-    __syncthreads();
-
-    // ---- softmax phase: 4 threads per row, 16 cols each ----
-    {
-      const int row = threadIdx.x >> 2;          // [0, 64)
-      const int c0 = (threadIdx.x & 3) * 16;
-      const int qrow = q0 + row;
-      float mx = -1e30f;
-      float sc[16];
+        row_max[r] = fmaxf(row_max[r],
+                           __shfl_xor(row_max[r], off, WAVE_SIZE));
+    float alpha[4], psum[4];
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int kcol = kt0 + c0 + j;
-        float val = s_lds[row][c0 + j] * scale;
-        const bool masked = (kcol >= S) || (causal && kcol > qrow);
-        sc[j] = masked ? -1e30f : val;
-        mx = fmaxf(mx, sc[j]);
-      }
-      // row reduce across the 4 sibling threads (lanes differ in bits 0..1)
-      mx = fmaxf(mx, __shfl_xor(mx, 1, WAVE_SIZE));
-      mx = fmaxf(mx, __shfl_xor(mx, 2, WAVE_SIZE));
-      const float m_old = m_lds[row];
-      const float m_new = fmaxf(m_old, mx);
-      float lsum = 0.f;
-#pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const float p = (sc[j] <= -1e29f) ? 0.f : __expf(sc[j] - m_new);
-        p_lds[row][c0 + j] = f2bf(p);
-        lsum += p;
-      }
-      lsum += __shfl_xor(lsum, 1, WAVE_SIZE);
-      lsum += __shfl_xor(lsum, 2, WAVE_SIZE);
-      if ((threadIdx.x & 3) == 0) {
-        const float alpha = __expf(m_old - m_new);
-        alpha_lds[row] = alpha;
-        l_lds[row] = l_lds[row] * alpha + lsum;
-        m_lds[row] = m_new;
-      }
+    for (int r = 0; r < 4; ++r) {
+      const float m_new = fmaxf(m_run[r], row_max[r]);
+      alpha[r] = __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+      psum[r] = 0.f;
     }
-    __syncthreads();
+#pragma unroll
+    for (int ns = 0; ns < NSUB; ++ns)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float p = (sacc[ns][r] <= -1e29f)
+                            ? 0.f : __expf(sacc[ns][r] - m_run[r]);
+        p_lds[wave][quad * 4 + r][ns * 16 + col] = f2bf(p);
+        psum[r] += p;
+      }
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        psum[r] += __shfl_xor(psum[r], off, WAVE_SIZE);
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      l_run[r] = l_run[r] * alpha[r] + psum[r];
 
     // ---- O = O*alpha + P V ----
-    {
-      float al[4];
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        al[r] = alpha_lds[r0 + (lane >> 4) * 4 + r];
+    for (int n = 0; n < NS_PV; ++n)
 #pragma unroll
-      for (int ns = 0; ns < NSUB_PV; ++ns) {
+      for (int r = 0; r < 4; ++r) oacc[n][r] *= alpha[r];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) oacc[ns][r] *= al[r];
+    for (int ks = 0; ks < BN / 32; ++ks) {
+      bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(
+          &p_lds[wave][col][ks * 32 + quad * 8]);
 #pragma unroll
-        for (int ks = 0; ks < BN / 32; ++ks) {
-          bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-              &p_lds[r0 + (lane & 15)][ks * 32 + (lane >> 4) * 8]);
-          bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-              &vt_lds[ns * 16 + (lane & 15)][ks * 32 + (lane >> 4) * 8]);
-          oacc[ns] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, oacc[ns], 0, 0, 0);
-        }
+      for (int n = 0; n < NS_PV; ++n) {
+        bf16x8_t vb = *reinterpret_cast<const bf16x8_t*>(
+            &vt_lds[n * 16 + col][ks * 32 + quad * 8]);
+        oacc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, oacc[n],
+                                                          0, 0, 0);
       }
     }
-    __syncthreads();  // before next tile overwrites K/V/P
+    __syncthreads();  // before next tile overwrites K/VT
   }
 
   // ---- epilogue: divide by l, store ----
   {
-    float linv[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const float lv = l_lds[r0 + (lane >> 4) * 4 + r];
-      linv[r] = lv > 0.f ? 1.f / lv : 0.f;
-    }
     const int64_t obase = ((int64_t)b * S) * Hq * D + (int64_t)h * D;
 #pragma unroll
-    for (int ns = 0; ns < NSUB_PV; ++ns) {
+    for (int r = 0; r < 4; ++r) {
+      const int row = r0 + quad * 4 + r;
+      const float linv = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+      if (q0 + row < S) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = r0 + (lane >> 4) * 4 + r;
-        if (q0 + row < S)
-          out[obase + (int64_t)(q0 + row) * Hq * D + ns * 16 + (lane & 15)] =
-              f2bf(oacc[ns][r] * linv[r]);
+        for (int n = 0; n < NS_PV; ++n)
+          out[obase + (int64_t)(q0 + row) * Hq * D + n * 16 + col] =
+              f2bf(oacc[n][r] * linv);
       }
     }
   }
