@@ -73,6 +73,46 @@ __host__ __device__ inline int64_t cdiv64(int64_t a, int64_t b) {
   return (a + b - 1) / b;
 }
 
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 (gfx950 LDS transpose read) — no clang builtin; inline
+// asm. Measured semantics (scripts/tr16_probe.hip): within each 16-lane
+// group, the 16 lanes' 8-byte reads form a 64-element vector V (in lane
+// order, 4 elems each); lane c receives V[c], V[c+16], V[c+32], V[c+48].
+//
+// For an MFMA bf16 B-fragment (lane l needs B[k=(l>>4)*8+j][n=l&15] from a
+// row-major [32][16] panel), lane a=l&15 in group q=l>>4 supplies
+//   addr_elems = panel + (q*8 + (a>>2))*16 + 4*(a&3)
+// and the pair of reads (second at +128 B = +4 k-rows) yields the 8
+// fragment elements in order. The waitcnt lives INSIDE the asm so the
+// consuming MFMA (ordered by its register dependency) is safe.
+// ---------------------------------------------------------------------------
+typedef __bf16 tr_bf16x8 __attribute__((ext_vector_type(8)));
+
+DEVINLINE tr_bf16x8 ds_read_tr16_frag(unsigned addr_bytes) {
+  unsigned long long lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(lo), "=v"(hi)
+      : "v"(addr_bytes));
+  union {
+    struct { unsigned long long a, b; } u;
+    tr_bf16x8 v;
+  } cvt;
+  cvt.u.a = lo;
+  cvt.u.b = hi;
+  return cvt.v;
+}
+
+// Per-lane address (in BYTES) for the fragment above: panel_base_bytes must
+// be the LDS byte offset of a row-major [32][16] bf16 panel.
+DEVINLINE unsigned tr16_frag_addr(unsigned panel_base_bytes, int lane) {
+  const int a = lane & 15, q = lane >> 4;
+  return panel_base_bytes +
+         (unsigned)(((q * 8 + (a >> 2)) * 16 + 4 * (a & 3)) * 2);
+}
+
 #define HIP_CHECK(expr)                                                     \
   do {                                                                      \
     hipError_t _e = (expr);                                                 \

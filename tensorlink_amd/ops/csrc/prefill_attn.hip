@@ -39,11 +39,17 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
   constexpr int NSUB = BN / 16;    // score col tiles = 4
   constexpr int NS_PV = D / 16;    // PV output col tiles
 
+  constexpr int NPAN = (BN / 32) * (D / 16);  // V panels [32 keys][16 dv]
+  constexpr int PSTRIDE = 32 * 16 + 8;        // panel stride (elems), padded
   // k_lds rows padded +8 elems: stride 272 B (17 slots) -> fragment reads
   // of 16 different rows land on distinct banks (an unpadded 256-B stride
   // puts EVERY row on bank 0: 16-way conflict).
   __shared__ bf16 k_lds[BN][D + 8];
-  __shared__ bf16 vt_lds[D][BN + 8];
+  // V stored as row-major [32][16] panels, written with single b128 stores
+  // and consumed via ds_read_b64_tr_b16 (hardware transpose) — replaces the
+  // v1/v2 element-wise LDS scatter transpose that made the kernel
+  // LDS-bound (PMC: 43% WAIT_INST_LDS, 26% bank-conflict cycles).
+  __shared__ bf16 v_pan[NPAN * PSTRIDE];
   __shared__ bf16 p_lds[4][16][BN + PPAD];   // per wave: 16 rows x 64 keys
 
   const int qt = blockIdx.x;
@@ -80,7 +86,7 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
   const int64_t kbase = ((int64_t)b * S) * Hkv * D + (int64_t)hkv * D;
 
   for (int kt0 = 0; kt0 < kv_end; kt0 += BN) {
-    // ---- stage K tile [BN][D]; V transposed [D][BN] ----
+    // ---- stage K tile [BN][D]; V into [32][16] tr panels ----
     for (int i = threadIdx.x * 8; i < BN * D; i += BLOCK * 8) {
       const int row = i / D, c = i % D;
       const int key = kt0 + row;
@@ -92,9 +98,10 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
         for (int j = 0; j < 8; ++j) { kval[j] = (__bf16)0.f; vval[j] = (__bf16)0.f; }
       }
       *reinterpret_cast<bf16x8_t*>(&k_lds[row][c]) = kval;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *reinterpret_cast<__bf16*>(&vt_lds[c + j][row]) = vval[j];
+      // panel (ks = row/32, n = c/16), row kk = row%32, col c0 = c%16
+      const int pan = (row >> 5) * (D / 16) + (c >> 4);
+      *reinterpret_cast<bf16x8_t*>(
+          &v_pan[pan * PSTRIDE + (row & 31) * 16 + (c & 15)]) = vval;
     }
     __syncthreads();
 
@@ -170,8 +177,10 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
           &p_lds[wave][col][ks * 32 + quad * 8]);
 #pragma unroll
       for (int n = 0; n < NS_PV; ++n) {
-        bf16x8_t vb = *reinterpret_cast<const bf16x8_t*>(
-            &vt_lds[n * 16 + col][ks * 32 + quad * 8]);
+        const unsigned pan_base = (unsigned)(uintptr_t)(
+            &v_pan[(ks * (D / 16) + n) * PSTRIDE]);
+        bf16x8_t vb = (bf16x8_t)ds_read_tr16_frag(
+            tr16_frag_addr(pan_base, lane));
         oacc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, oacc[n],
                                                           0, 0, 0);
       }
