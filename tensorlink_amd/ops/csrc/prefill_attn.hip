@@ -85,47 +85,25 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
   const int kv_end = causal ? min(S, q0 + BM) : S;
   const int64_t kbase = ((int64_t)b * S) * Hkv * D + (int64_t)hkv * D;
 
-  // T14 split staging: tile t+1's global loads are ISSUED right after the
-  // barrier and land in registers while tile t's MFMA/softmax runs; the
-  // register tile is written to LDS at the top of the next iteration.
-  constexpr int NCHUNK = BN * D / (BLOCK * 8);
-  bf16x8_t kreg[NCHUNK], vreg[NCHUNK];
-
-  auto load_tile = [&](int kt0, bf16x8_t* kr, bf16x8_t* vr) {
-#pragma unroll
-    for (int ci = 0; ci < NCHUNK; ++ci) {
-      const int i = (ci * BLOCK + threadIdx.x) * 8;
+  for (int kt0 = 0; kt0 < kv_end; kt0 += BN) {
+    // ---- stage K tile [BN][D]; V into [32][16] tr panels ----
+    for (int i = threadIdx.x * 8; i < BN * D; i += BLOCK * 8) {
       const int row = i / D, c = i % D;
       const int key = kt0 + row;
       const int64_t off = kbase + (int64_t)min(key, S - 1) * Hkv * D + c;
-      kr[ci] = *reinterpret_cast<const bf16x8_t*>(k + off);
-      vr[ci] = *reinterpret_cast<const bf16x8_t*>(v + off);
+      bf16x8_t kval = *reinterpret_cast<const bf16x8_t*>(k + off);
+      bf16x8_t vval = *reinterpret_cast<const bf16x8_t*>(v + off);
       if (key >= S) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          kr[ci][j] = (__bf16)0.f;
-          vr[ci][j] = (__bf16)0.f;
-        }
+        for (int j = 0; j < 8; ++j) { kval[j] = (__bf16)0.f; vval[j] = (__bf16)0.f; }
       }
-    }
-  };
-
-  load_tile(0, kreg, vreg);
-
-  for (int kt0 = 0; kt0 < kv_end; kt0 += BN) {
-    // ---- write the prefetched tile: K row-major, V into tr panels ----
-#pragma unroll
-    for (int ci = 0; ci < NCHUNK; ++ci) {
-      const int i = (ci * BLOCK + threadIdx.x) * 8;
-      const int row = i / D, c = i % D;
-      *reinterpret_cast<bf16x8_t*>(&k_lds[row][c]) = kreg[ci];
+      *reinterpret_cast<bf16x8_t*>(&k_lds[row][c]) = kval;
+      // panel (ks = row/32, n = c/16), row kk = row%32, col c0 = c%16
       const int pan = (row >> 5) * (D / 16) + (c >> 4);
       *reinterpret_cast<bf16x8_t*>(
-          &v_pan[pan * PSTRIDE + (row & 31) * 16 + (c & 15)]) = vreg[ci];
+          &v_pan[pan * PSTRIDE + (row & 31) * 16 + (c & 15)]) = vval;
     }
     __syncthreads();
-    if (kt0 + BN < kv_end)
-      load_tile(kt0 + BN, kreg, vreg);   // in flight under the compute
 
     // ---- QK^T for all 4 col-subtiles (scores stay in registers) ----
     f32x4 sacc[NSUB];
