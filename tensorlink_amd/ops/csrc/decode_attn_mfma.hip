@@ -61,10 +61,26 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
   const int col = lane & 15;         // key-in-tile / head-col
   const int quad = lane >> 4;        // 0..3
 
-  __shared__ bf16 vt_lds[NWAVE][D][KT + VPAD];
-  __shared__ bf16 p_lds[NWAVE][16][KT];
+  // V as row-major [32 keys][16 dv] panels per (wave, dv-16-chunk): b128
+  // staging writes + ds_read_b64_tr_b16 fragment reads (no scatter
+  // transpose). Rows 16..31 stay zero — they feed the K=32 PV MFMA's
+  // upper half, so the quad<2 zero-branch disappears. Same for the upper
+  // half of the P tile.
+  constexpr int PSTRIDE = 32 * 16 + 8;
+  __shared__ bf16 v_pan[NWAVE * (D / 16) * PSTRIDE];
+  __shared__ bf16 p_lds[NWAVE][16][32 + 8];
   __shared__ float red_m[NWAVE][16], red_l[NWAVE][16];
   __shared__ float red_o[NWAVE][16][D];
+
+  // one-time zero of the never-written upper halves
+  for (int i = threadIdx.x; i < NWAVE * (D / 16); i += BLOCK) {
+    for (int e = 0; e < 16 * 16; ++e)
+      v_pan[i * PSTRIDE + 16 * 16 + e] = f2bf(0.f);
+  }
+  for (int i = threadIdx.x; i < NWAVE * 16; i += BLOCK) {
+    for (int e = 16; e < 32; ++e) p_lds[i / 16][i % 16][e] = f2bf(0.f);
+  }
+  __syncthreads();
 
   const bf16* kbase = k_cache + ((int64_t)b * Hkv + hkv) * Smax * D;
   const bf16* vbase = v_cache + ((int64_t)b * Hkv + hkv) * Smax * D;
@@ -109,13 +125,13 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
             krow + c * 32 + quad * 8);
         sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[c], kf, sacc,
                                                        0, 0, 0);
-        // V: same addressing; scatter-transpose into vt[d][key]
+        // V: same addressing; b128 write into the (wave, dv-chunk) panel
         bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
             vrow + c * 32 + quad * 8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<__bf16*>(
-              &vt_lds[wave][c * 32 + quad * 8 + j][col]) = vf[j];
+        const int d0 = c * 32 + quad * 8;
+        const int pan = wave * (D / 16) + (d0 >> 4);
+        *reinterpret_cast<bf16x8_t*>(
+            &v_pan[pan * PSTRIDE + col * 16 + (d0 & 15)]) = vf;
       }
     }
 
@@ -164,29 +180,15 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) oacc[n][r] *= alpha[r];
 
-    // ---- PV: K=32 MFMA with 16 valid keys + 16 zero keys ----
-    // A: lane holds P[row=col][key=quad*8+j]; keys 16..31 read zeros
-    bf16x8_t pa;
-    {
-      // quad 0,1 -> keys 0..15 from p_lds; quad 2,3 -> zero
-      if (quad < 2) {
-        pa = *reinterpret_cast<const bf16x8_t*>(&p_lds[wave][col][quad * 8]);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) pa[j] = (__bf16)0.f;
-      }
-    }
+    // ---- PV: K=32 MFMA; keys 16..31 are the pre-zeroed halves ----
+    bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(
+        &p_lds[wave][col][quad * 8]);
 #pragma unroll
     for (int n = 0; n < NS; ++n) {
-      // B: lane holds V[key=quad*8+j][dv=n*16+col] = vt[dv][key]
-      bf16x8_t vb;
-      if (quad < 2) {
-        vb = *reinterpret_cast<const bf16x8_t*>(
-            &vt_lds[wave][n * 16 + col][quad * 8]);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vb[j] = (__bf16)0.f;
-      }
+      const unsigned pan_base = (unsigned)(uintptr_t)(
+          &v_pan[(wave * (D / 16) + n) * PSTRIDE]);
+      bf16x8_t vb = (bf16x8_t)ds_read_tr16_frag(
+          tr16_frag_addr(pan_base, lane));
       oacc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, oacc[n],
                                                         0, 0, 0);
     }
