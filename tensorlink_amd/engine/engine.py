@@ -137,10 +137,11 @@ class InferenceEngine:
     def _build_runner(self, name: str, init: str, ckpt_dir: Optional[str],
                       **kwargs) -> ModelJob:
         config = get_config(name)
+        quantize = kwargs.pop("quantize", None)
         plan = plan_for_world(config, self.world, **kwargs)
         runner = PipelineRunner(plan, self.rank, self.world,
                                 device=self.device, init=init,
-                                ckpt_dir=ckpt_dir)
+                                ckpt_dir=ckpt_dir, quantize=quantize)
         tok = load_tokenizer(ckpt_dir or name)
         job = ModelJob(name=name, runner=runner, tokenizer=tok)
         self.jobs[name] = job

@@ -54,7 +54,7 @@ class PipelineRunner:
 
     def __init__(self, plan: StagePlan, rank: int, world: int,
                  device=None, init: str = "random", ckpt_dir: Optional[str] = None,
-                 dtype=None, seed: int = 0):
+                 dtype=None, seed: int = 0, quantize: Optional[str] = None):
         assert plan.num_stages == world, "plan stages must equal world size"
         self.plan = plan
         self.config: ModelConfig = plan.config
@@ -73,6 +73,9 @@ class PipelineRunner:
                                        device=self.device, dtype=self.dtype)
         elif init == "empty":
             self.stage.to(device=self.device, dtype=self.dtype)
+        if quantize == "fp8":
+            from tensorlink_amd.models.quant import quantize_experts_fp8
+            quantize_experts_fp8(self.stage)
         self.stage.eval()
         self.p2p = P2P(rank, world) if world > 1 else None
         self.is_first = rank == 0

@@ -1,0 +1,155 @@
+"""Aux subsystem tests: fp8 quant, proofs, watchdog, state keeper, config,
+CLI planner, formatter."""
+
+import json
+import os
+
+import torch
+
+from tensorlink_amd.engine.formatter import (ResponseFormatter,
+                                             extract_reasoning_and_answer,
+                                             format_chat_prompt,
+                                             normalize_generate_args)
+
+
+def test_normalize_generate_args_clamps():
+    out = normalize_generate_args({"max_new_tokens": 99999,
+                                   "temperature": 9.0, "top_p": 2.0})
+    assert out["max_new_tokens"] == 2048
+    assert out["temperature"] == 2.0
+    assert out["top_p"] == 1.0
+    out = normalize_generate_args({"temperature": -5, "max_new_tokens": "x"})
+    assert out["temperature"] == 0.01
+    assert out["max_new_tokens"] == 256
+    # beam + sample conflict resolves to sampling
+    out = normalize_generate_args({"num_beams": 4, "do_sample": True})
+    assert out["num_beams"] == 1
+    out = normalize_generate_args({"do_sample": False})
+    assert out["temperature"] == 0.0
+
+
+def test_chat_prompt_templates():
+    msgs = [{"role": "system", "content": "s"},
+            {"role": "user", "content": "hi"}]
+    qwen = format_chat_prompt(msgs, None, "Qwen/Qwen2.5-7B")
+    assert "<|im_start|>system" in qwen and qwen.endswith(
+        "<|im_start|>assistant\n")
+    llama = format_chat_prompt(msgs, None, "meta-llama/Llama-3-8B")
+    assert "<|start_header_id|>user<|end_header_id|>" in llama
+
+
+def test_reasoning_extraction():
+    think, ans = extract_reasoning_and_answer(
+        "<think>step by step</think>The answer is 4.")
+    assert think == "step by step"
+    assert ans == "The answer is 4."
+    think, ans = extract_reasoning_and_answer("plain")
+    assert think == "" and ans == "plain"
+
+
+def test_sse_chunk_shapes():
+    f = ResponseFormatter("m", "openai")
+    chunk = f.format_stream_chunk("hi", first=True)
+    assert chunk.startswith("data: ")
+    payload = json.loads(chunk[6:])
+    assert payload["choices"][0]["delta"]["role"] == "assistant"
+    final = f.format_final_chunk()
+    assert final.endswith("data: [DONE]\n\n")
+
+
+def test_fp8_quant_roundtrip():
+    from tensorlink_amd.models.quant import (Fp8Linear,
+                                             quantize_fp8_per_channel)
+    torch.manual_seed(0)
+    w = torch.randn(64, 32) * 3
+    w8, scale = quantize_fp8_per_channel(w)
+    assert w8.dtype == torch.float8_e4m3fn
+    deq = w8.float() * scale[:, None]
+    rel = (deq - w).abs().max() / w.abs().max()
+    assert rel < 0.05
+    lin = torch.nn.Linear(32, 64)
+    f8 = Fp8Linear.from_linear(lin)
+    x = torch.randn(4, 32)
+    torch.testing.assert_close(f8(x), lin(x), atol=0.1, rtol=0.1)
+
+
+def test_fp8_expert_conversion_and_forward():
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.quant import Fp8Linear, quantize_experts_fp8
+    cfg = get_config("tiny-moe")
+    m = build_full_model(cfg)
+    init_random_stage(m, dtype=torch.float32)
+    n = quantize_experts_fp8(m)
+    assert n == cfg.num_local_experts * 3 * cfg.num_hidden_layers
+    assert isinstance(m.layers[0].mlp.experts[0].gate_proj, Fp8Linear)
+    ids = torch.randint(0, cfg.vocab_size, (2, 8))
+    pos = torch.arange(8).unsqueeze(0).expand(2, -1).contiguous()
+    out = m(ids, pos)
+    assert torch.isfinite(out.float()).all()
+
+
+def test_gradient_hash_and_proofs():
+    from tensorlink_amd.utils.proofs import (gradient_hash,
+                                             verify_loss_trajectory)
+    g1 = [torch.ones(4), torch.zeros(3)]
+    g2 = [torch.ones(4), torch.zeros(3)]
+    g3 = [torch.ones(4) * 2, torch.zeros(3)]
+    assert gradient_hash(g1) == gradient_hash(g2)
+    assert gradient_hash(g1) != gradient_hash(g3)
+    assert verify_loss_trajectory(list(range(30, 0, -1)))
+    assert not verify_loss_trajectory([1.0] * 10 + [5.0] * 20)
+
+
+def test_watchdog_restarts_failed_job():
+    from tensorlink_amd.engine.engine import InferenceEngine, ModelJob
+    from tensorlink_amd.utils.watchdog import Watchdog
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    # simulate a failure
+    eng.jobs["tiny"] = ModelJob(name="tiny", runner=None, tokenizer=None,
+                                state="failed")
+    wd = Watchdog(eng, interval_s=999)
+    status = wd.check_once()
+    assert wd.restarts == 1
+    assert eng.jobs["tiny"].state == "ready"
+    assert status["status"] in ("ok", "degraded")
+
+
+def test_state_keeper_roundtrip(tmp_path):
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.utils.state import StateKeeper
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.demand["tiny"] = 5
+    path = str(tmp_path / "state.json")
+    keeper = StateKeeper(eng, path=path)
+    keeper.write_state()
+    eng2 = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    keeper2 = StateKeeper(eng2, path=path)
+    snap = keeper2.load_previous_state()
+    assert snap is not None
+    assert eng2.demand["tiny"] == 5
+
+
+def test_engine_config_load(tmp_path):
+    from tensorlink_amd.config import EngineConfig
+    p = tmp_path / "config.json"
+    p.write_text(json.dumps({"node": {"endpoint_port": 9123},
+                             "ml": {"default_models": ["tiny"]}}))
+    cfg = EngineConfig.load(str(p))
+    assert cfg.node.endpoint_port == 9123
+    assert cfg.ml.default_models == ["tiny"]
+
+
+def test_cli_plan(capsys):
+    import tlctl
+    import sys
+    argv = sys.argv
+    sys.argv = ["tlctl", "plan", "--model", "Qwen/Qwen3-8B", "--gpus", "4",
+                "--pp", "--training"]
+    try:
+        tlctl.main()
+    finally:
+        sys.argv = argv
+    out = capsys.readouterr().out
+    assert "PP=4" in out and "training=True" in out

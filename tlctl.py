@@ -1,0 +1,109 @@
+#!/usr/bin/env python3
+"""tlctl — operator CLI (reference bin/run_node.py analog).
+
+  tlctl serve  [--model NAME ...] [--port P] [--gpus N]   start the REST
+      serving engine (SPMD under torchrun for N>1)
+  tlctl plan   --model NAME [--gpus N] [--training]       print stage plan
+  tlctl bench  [bench.py args...]                          run the benchmark
+  tlctl health                                             device health
+"""
+
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def cmd_serve(args):
+    from tensorlink_amd.config import EngineConfig
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.utils.state import StateKeeper
+    from tensorlink_amd.utils.watchdog import Watchdog
+
+    cfg = EngineConfig.load(args.config)
+    rank, world = init_distributed()
+    engine = InferenceEngine(rank=rank, world=world,
+                             default_init=cfg.ml.init)
+    if rank != 0:
+        engine.worker_loop()
+        return
+
+    models = args.model or cfg.ml.default_models
+    for name in models:
+        print(f"loading {name} ...")
+        engine.load_model(name)
+    keeper = StateKeeper(engine)
+    keeper.load_previous_state()
+    keeper.start()
+    Watchdog(engine).start()
+
+    from tensorlink_amd.api.server import TensorlinkAPI
+    api = TensorlinkAPI(engine, host=cfg.node.endpoint_host,
+                        port=args.port or cfg.node.endpoint_port)
+    print(f"serving on http://{api.host}:{api.port}  "
+          f"(models: {', '.join(models)})")
+    try:
+        api.start(background=False)
+    finally:
+        keeper.stop()
+        engine.shutdown()
+
+
+def cmd_plan(args):
+    from tensorlink_amd.parallel.planner import ModelParser
+    parser = ModelParser(n_workers=args.gpus)
+    plan = parser.create_distributed_config(
+        args.model, batch_size=args.batch, seq_len=args.seq_len,
+        training=args.training,
+        num_stages=args.gpus if args.pp else None)
+    print(plan.describe())
+
+
+def cmd_bench(args, extra):
+    os.execv(sys.executable, [sys.executable,
+                              os.path.join(os.path.dirname(__file__),
+                                           "bench.py")] + extra)
+
+
+def cmd_health(args):
+    import json
+    from tensorlink_amd.utils.watchdog import check_gpu_health
+    print(json.dumps(check_gpu_health(), indent=2))
+
+
+def main():
+    p = argparse.ArgumentParser(prog="tlctl")
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    s = sub.add_parser("serve")
+    s.add_argument("--model", action="append")
+    s.add_argument("--port", type=int)
+    s.add_argument("--config")
+
+    pl = sub.add_parser("plan")
+    pl.add_argument("--model", required=True)
+    pl.add_argument("--gpus", type=int, default=8)
+    pl.add_argument("--batch", type=int, default=8)
+    pl.add_argument("--seq-len", type=int, default=4096)
+    pl.add_argument("--training", action="store_true")
+    pl.add_argument("--pp", action="store_true",
+                    help="force PP = --gpus instead of smallest fit")
+
+    sub.add_parser("bench")
+    sub.add_parser("health")
+
+    args, extra = p.parse_known_args()
+    if args.cmd == "serve":
+        cmd_serve(args)
+    elif args.cmd == "plan":
+        cmd_plan(args)
+    elif args.cmd == "bench":
+        cmd_bench(args, extra)
+    elif args.cmd == "health":
+        cmd_health(args)
+
+
+if __name__ == "__main__":
+    main()
