@@ -114,3 +114,30 @@ def test_graph_decode_matches_eager():
     out2_eager = r2.generate(ids2, SamplingParams(
         temperature=1e-6, top_k=1, max_new_tokens=12)).cpu()
     assert torch.equal(out2, out2_eager)
+
+
+def test_fp8_expert_gemm_gpu():
+    """fp8 e4m3 expert weights on GPU: _scaled_mm (hipBLASLt fp8) or
+    documented dequant fallback; numerics vs bf16 reference."""
+    from tensorlink_amd.models.quant import Fp8Linear
+    torch.manual_seed(2)
+    lin = torch.nn.Linear(256, 512, bias=False).to(DEV, torch.bfloat16)
+    f8 = Fp8Linear.from_linear(lin)
+    x = torch.randn(64, 256, device=DEV, dtype=torch.bfloat16)
+    out = f8(x)
+    ref = lin(x)
+    # fp8 weights + dynamic act scaling: ~2-3% relative error expected
+    rel = (out.float() - ref.float()).abs().mean() / ref.float().abs().mean()
+    assert rel < 0.05, rel
+
+
+def test_moe_fp8_generate_gpu():
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    plan = plan_for_world("tiny-moe", 1)
+    r = PipelineRunner(plan, 0, 1, device=DEV, quantize="fp8")
+    from tensorlink_amd.models.quant import Fp8Linear
+    assert isinstance(r.stage.layers[0].mlp.experts[0].gate_proj, Fp8Linear)
+    ids = torch.randint(0, 1024, (2, 12))
+    out = r.generate(ids, SamplingParams(max_new_tokens=4))
+    assert out.shape == (2, 4)
