@@ -43,7 +43,58 @@ def parse_args():
     p.add_argument("--new-tokens", type=int, default=128)
     p.add_argument("--micro-batches", type=int, default=0,
                    help="decode micro-batches in flight (0 => world size)")
+    p.add_argument("--mode", choices=["serve", "train"], default="serve")
+    p.add_argument("--dp", type=int, default=1,
+                   help="data-parallel replicas (train mode; world=dp*pp)")
     return p.parse_args()
+
+
+def run_train(args, rank, world, device, use_gpu, config):
+    """Training benchmark (BASELINE config #3 shape): 1F1B (+DP) steps,
+    metric = training tokens/s."""
+    from tensorlink_amd.parallel.dp import HybridTrainer
+    B = args.batch_per_gpu * world
+    S = args.prompt_len
+    trainer = HybridTrainer(config.name if config.name != "custom" else
+                            config, rank, world, dp=args.dp, device=device,
+                            lr=1e-4)
+    torch.manual_seed(17)
+    ids = torch.randint(0, config.vocab_size, (B, S)) if rank == 0 else None
+    n_micro = max(1, trainer.pp)
+
+    def sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        trainer.train_step(ids, ids, n_micro=n_micro)
+    sync()
+    t0 = time.perf_counter()
+    losses = [trainer.train_step(ids, ids, n_micro=n_micro)
+              for _ in range(args.steps)]
+    sync()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        e = torch.tensor([elapsed], dtype=torch.float64)
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e.item())
+    if rank == 0:
+        print(json.dumps({
+            "metric": "training tokens/sec (1F1B pipeline + fused AdamW)",
+            "value": B * S * args.steps / elapsed,
+            "unit": "tokens/s", "n_gpus": world, "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32", "data": "synthetic",
+            "loss_first": losses[0], "loss_last": losses[-1],
+            "config": {"model": args.model, "global_batch": B,
+                       "seq_len": S,
+                       "parallelism": f"dp{args.dp}xpp{world // args.dp}"},
+        }))
 
 
 def main():
@@ -65,6 +116,10 @@ def main():
     setup_tunableop(tune=bool(os.environ.get("TL_TUNE")))
 
     config = get_config(args.model)
+    if args.mode == "train":
+        run_train(args, rank, world, device, use_gpu, config)
+        return
+
     B = args.batch_per_gpu * world          # weak scaling: global batch
     S, T = args.prompt_len, args.new_tokens
     plan = plan_for_world(config, world, batch_size=B, seq_len=S + T)

@@ -62,6 +62,8 @@ def _cluster_worker(rank: int, world: int, port: int, model: str,
             break
         elif op == "generate":
             runner.generate(None, None)
+        elif op == "fwd_eval":
+            runner.forward_logits(None)
         elif op == "forward":
             holder.spmd_forward(None)
         elif op == "backward":
@@ -166,16 +168,8 @@ class DistributedModel(nn.Module):
         .backward()-ed (PP>1 routes through the pipeline). If labels are
         given, the shifted CE loss is attached as ``output.loss``."""
         if not self.training_mode:
-            # plain eval forward (PP=1 only for direct logits)
-            if self.world_size == 1:
-                B, S = input_ids.shape
-                pos = torch.arange(S).unsqueeze(0).expand(B, -1).contiguous()
-                with torch.no_grad():
-                    return self._runner.stage(
-                        input_ids.to(self._runner.device),
-                        pos.to(self._runner.device))
-            raise RuntimeError("eval forward with world_size>1: "
-                               "use generate()")
+            self._bcast(("fwd_eval",))
+            return self._runner.forward_logits(input_ids)
         if self.world_size == 1:
             B, S = input_ids.shape
             pos = torch.arange(S).unsqueeze(0).expand(B, -1).contiguous()

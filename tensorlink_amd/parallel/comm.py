@@ -78,31 +78,38 @@ class P2P:
     Shapes/dtypes are agreed at plan time, so no metadata travels on the
     hot path (the reference sends a JSON + safetensors blob per transfer —
     ``torch_node.py:825-836``).
+
+    `group`/`rank_base` support hybrid DP×PP: ranks passed to send/recv
+    are PIPELINE-local; `rank_base` maps them to global ranks (torch's
+    send/recv take global ranks regardless of group).
     """
 
-    def __init__(self, rank: int, world: int):
+    def __init__(self, rank: int, world: int, group=None, rank_base: int = 0):
         self.rank = rank
         self.world = world
+        self.group = group
+        self.base = rank_base
 
     def send(self, t: torch.Tensor, dst: int):
-        dist.send(t.contiguous(), dst)
+        dist.send(t.contiguous(), self.base + dst, group=self.group)
 
     def recv(self, shape, dtype, src: int, device) -> torch.Tensor:
         buf = torch.empty(*shape, dtype=dtype, device=device)
-        dist.recv(buf, src)
+        dist.recv(buf, self.base + src, group=self.group)
         return buf
 
     def isend(self, t: torch.Tensor, dst: int):
-        return dist.isend(t.contiguous(), dst)
+        return dist.isend(t.contiguous(), self.base + dst, group=self.group)
 
     def irecv_into(self, buf: torch.Tensor, src: int):
-        return dist.irecv(buf, src)
+        return dist.irecv(buf, self.base + src, group=self.group)
 
     def broadcast(self, t: torch.Tensor, src: int) -> torch.Tensor:
-        dist.broadcast(t, src)
+        dist.broadcast(t, self.base + src, group=self.group)
         return t
 
     def broadcast_obj(self, obj, src: int = 0):
         holder = [obj]
-        dist.broadcast_object_list(holder, src=src)
+        dist.broadcast_object_list(holder, src=self.base + src,
+                                   group=self.group)
         return holder[0]

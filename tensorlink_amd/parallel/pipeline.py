@@ -54,8 +54,11 @@ class PipelineRunner:
 
     def __init__(self, plan: StagePlan, rank: int, world: int,
                  device=None, init: str = "random", ckpt_dir: Optional[str] = None,
-                 dtype=None, seed: int = 0, quantize: Optional[str] = None):
+                 dtype=None, seed: int = 0, quantize: Optional[str] = None,
+                 group=None, rank_base: int = 0):
         assert plan.num_stages == world, "plan stages must equal world size"
+        self.group = group
+        self.rank_base = rank_base
         self.plan = plan
         self.config: ModelConfig = plan.config
         self.rank = rank
@@ -77,7 +80,7 @@ class PipelineRunner:
             from tensorlink_amd.models.quant import quantize_experts_fp8
             quantize_experts_fp8(self.stage)
         self.stage.eval()
-        self.p2p = P2P(rank, world) if world > 1 else None
+        self.p2p = P2P(rank, world, group, rank_base) if world > 1 else None
         self.is_first = rank == 0
         self.is_last = rank == world - 1
         self.next_rank = rank + 1
@@ -350,10 +353,15 @@ class PipelineTrainer:
 
     def __init__(self, plan: StagePlan, rank: int, world: int, device=None,
                  init: str = "random", ckpt_dir: Optional[str] = None,
-                 dtype=None, seed: int = 0, lr: float = 1e-4, **opt_kwargs):
+                 dtype=None, seed: int = 0, lr: float = 1e-4, group=None,
+                 rank_base: int = 0, **opt_kwargs):
         self.runner = PipelineRunner(plan, rank, world, device=device,
                                      init=init, ckpt_dir=ckpt_dir,
-                                     dtype=dtype, seed=seed)
+                                     dtype=dtype, seed=seed, group=group,
+                                     rank_base=rank_base)
+        # called with (self) right before optimizer.step(): the DP grad
+        # all-reduce hook attaches here (parallel/dp.py)
+        self.grad_hook = None
         self.stage = self.runner.stage
         self.stage.train()
         self.rank, self.world = rank, world
@@ -457,6 +465,8 @@ class PipelineTrainer:
         for w, _ in inflight:
             w.wait()
 
+        if self.grad_hook is not None:
+            self.grad_hook(self)
         self.optimizer.step()
         self.optimizer.zero_grad()
 
@@ -527,3 +537,37 @@ class PipelineTrainer:
             g = self.p2p.recv((B, S, self.H), self.dtype, r + 1, self.device)
             out.backward(g)
         self._fb_stash = None
+
+
+def _runner_forward_logits(self, input_ids=None):
+    """SPMD eval forward: full logits [B,S,V] on rank 0 (no grad, no KV
+    cache). The reference's remote-inference forward path
+    (tests/test_distributed_model.py:27-38)."""
+    with torch.no_grad():
+        if self.world == 1:
+            B, S = input_ids.shape
+            pos = torch.arange(S, device=self.device, dtype=torch.int32) \
+                .unsqueeze(0).expand(B, -1).contiguous()
+            return self.stage(input_ids.to(self.device), pos)
+        meta = tuple(input_ids.shape) if self.is_first else None
+        B, S = self.p2p.broadcast_obj(meta, src=0)
+        pos = torch.arange(S, device=self.device, dtype=torch.int32) \
+            .unsqueeze(0).expand(B, -1).contiguous()
+        if self.is_first:
+            h = self.stage(input_ids.to(self.device), pos,
+                           return_logits=False)
+            self.p2p.send(h.to(self.dtype), self.next_rank)
+            V = self.config.vocab_size
+            return self.p2p.recv((B, S, V), self.dtype, self.world - 1,
+                                 self.device)
+        h = self.p2p.recv((B, S, self.H), self.dtype, self.prev_rank,
+                          self.device)
+        out = self.stage(h, pos, return_logits=self.is_last)
+        if self.is_last:
+            self.p2p.send(out.to(self.dtype), 0)
+        else:
+            self.p2p.send(out, self.next_rank)
+        return None
+
+
+PipelineRunner.forward_logits = _runner_forward_logits

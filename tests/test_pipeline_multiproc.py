@@ -102,3 +102,48 @@ def test_pp2_1f1b_training_reduces_loss():
     losses = _run(_train_worker, 2, 29622)
     assert len(losses) == 4
     assert losses[-1] < losses[0]
+
+
+def _dp_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.dp import HybridTrainer
+    init_distributed(backend="gloo")
+    tr = HybridTrainer("tiny", rank, world, dp=2, device=torch.device("cpu"),
+                       lr=1e-3)
+    torch.manual_seed(5)
+    ids = torch.randint(0, 1024, (8, 16)) if rank == 0 else None
+    losses = [tr.train_step(ids, ids, n_micro=2) for _ in range(4)]
+    # replicas must remain bit-identical after synced steps
+    h = float(tr.optimizer.flat_param.double().sum())
+    q.put((rank, losses, h))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_training_replicas_stay_synced():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_dp_worker, args=(r, 2, 29633, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = {}
+    try:
+        for _ in range(2):
+            rank, losses, h = q.get(timeout=240)
+            results[rank] = (losses, h)
+    finally:
+        for p in ps:
+            p.join(30)
+            if p.is_alive():
+                p.terminate()
+    l0, h0 = results[0]
+    l1, h1 = results[1]
+    assert l0 == l1, "losses must agree across replicas"
+    assert l0[-1] < l0[0]
+    assert abs(h0 - h1) < 1e-6, "replica weights diverged"
