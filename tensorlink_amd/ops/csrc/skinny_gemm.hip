@@ -42,13 +42,20 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
   const int oc = blockIdx.x * 64 + wave * 16 + col;   // output column
 
   const int m_tiles = (M + 15) / 16;
-  const int rows = m_tiles * 16;
   const int split = SPLIT ? blockIdx.y : 0;
-  const int k_per = SPLIT ? ((K / 32 + n_split - 1) / n_split) * 32 : K;
+  const int k_per = SPLIT ? ((K / 64 + n_split - 1) / n_split) * 64 : K;
   const int k_begin = split * k_per;
   const int k_end = min(K, k_begin + k_per);
 
-  __shared__ bf16 x_lds[MT * 16][64 + XPAD];
+  // Double-buffered x slab staged by global_load_lds (HBM->LDS DMA: no
+  // staging registers — register staging at MT=16 spilled 272 B/lane).
+  // glds writes lane-linear (base + lane*16), so the LDS image is
+  // unpadded; the bank swizzle moves to the SOURCE address (guide rule
+  // 21): physical 16-B slot p of row r holds logical k-slot p^(r&7), and
+  // the fragment read applies the same XOR — 8-way read conflicts drop to
+  // 2-way with FETCH-neutral sources (the XOR permutes within one row's
+  // 128 B).
+  __shared__ bf16 x_lds[2][MT * 16][64];
 
   f32x4 acc[MT];
 #pragma unroll
@@ -56,22 +63,32 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
 
   const bf16* wrow = w + (int64_t)oc * K;   // this lane's W row
 
+  constexpr int GPW = MT > 1 ? MT / 2 : 1;  // glds chunks per wave
+  const int l8 = lane >> 3, c8 = lane & 7;
+
+#define SG_GLDS_TILE(bufi, k0)                                              \
+  if ((k0) < K) {                                                           \
+    _Pragma("unroll") for (int g = 0; g < GPW; ++g) {                       \
+      const int chunk = wave * GPW + g;      /* 8 rows per 1-KB chunk */    \
+      if (chunk >= MT * 2) break; /* MT=1: only 2 chunks (wave-uniform) */  \
+      const int row = chunk * 8 + l8;                                       \
+      const int xr = min(row, M - 1);                                       \
+      const int kc = min((k0) + (c8 ^ (row & 7)) * 8, K - 8);               \
+      auto gsrc = (const __attribute__((address_space(1))) void*)(          \
+          x + (int64_t)xr * K + kc);                                        \
+      auto ldst = (__attribute__((address_space(3))) void*)(                \
+          &x_lds[bufi][chunk * 8][0]);                                      \
+      __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);               \
+    }                                                                       \
+  }
+
+  SG_GLDS_TILE(0, k_begin);
+  int buf = 0;
   for (int k0 = k_begin; k0 < k_end; k0 += 64) {
-    const int tile_k = min(64, k_end - k0);            // 64 or 32
-    // ---- stage x[0:rows][k0:k0+tile_k] into LDS, full-line loads ----
-    {
-      const int vecs = rows * (tile_k / 8);            // 16-B units
-      for (int i = threadIdx.x; i < vecs; i += BLOCK) {
-        const int r = i / (tile_k / 8);
-        const int c8 = i - r * (tile_k / 8);
-        const int xr = min(r, M - 1);
-        bf16x8_t val = *reinterpret_cast<const bf16x8_t*>(
-            x + (int64_t)xr * K + k0 + c8 * 8);
-        *reinterpret_cast<bf16x8_t*>(&x_lds[r][c8 * 8]) = val;
-      }
-    }
-    __syncthreads();
-    // ---- compute: W streams from HBM, A-fragments from LDS ----
+    __syncthreads();                          // drains this tile's glds
+    if (k0 + 64 < k_end)
+      SG_GLDS_TILE(buf ^ 1, k0 + 64);         // in flight under compute
+    const int tile_k = min(64, k_end - k0);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       if (kk * 32 < tile_k) {
@@ -80,16 +97,19 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
 #pragma unroll
         for (int m = 0; m < MT; ++m) {
           if (m < m_tiles) {
+            const int row = m * 16 + col;
+            const int slot = (kk * 4 + quad) ^ (row & 7);
             bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-                &x_lds[m * 16 + col][kk * 32 + quad * 8]);
+                &x_lds[buf][row][slot * 8]);
             acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afrag, bfrag, acc[m], 0, 0, 0);
           }
         }
       }
     }
-    __syncthreads();
+    buf ^= 1;
   }
+#undef SG_GLDS_TILE
 
   // epilogue: lane holds C[row=quad*4+r][col] per m-tile
   const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
