@@ -28,7 +28,7 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 
 template <int MT, int HAS_BIAS, int SPLIT>
-__global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
+__global__ __launch_bounds__(BLOCK, 2) void skinny_gemm_kernel(
     const bf16* __restrict__ x,     // [M, K]
     const bf16* __restrict__ w,     // [N, K]
     const bf16* __restrict__ bias,  // [N] or null
@@ -96,14 +96,15 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
             wrow + k0 + kk * 32 + quad * 8);
 #pragma unroll
         for (int m = 0; m < MT; ++m) {
-          if (m < m_tiles) {
-            const int row = m * 16 + col;
-            const int slot = (kk * 4 + quad) ^ (row & 7);
-            bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-                &x_lds[buf][row][slot * 8]);
-            acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag, bfrag, acc[m], 0, 0, 0);
-          }
+          // unconditional across MT (dispatch picks smallest MT >=
+          // m_tiles; guarded acc updates made the compiler home the
+          // accumulator array in scratch)
+          const int row = m * 16 + col;
+          const int slot = (kk * 4 + quad) ^ (row & 7);
+          bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
+              &x_lds[buf][row][slot * 8]);
+          acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, bfrag, acc[m], 0, 0, 0);
         }
       }
     }
@@ -115,7 +116,6 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
   const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
 #pragma unroll
   for (int m = 0; m < MT; ++m) {
-    if (m >= m_tiles) break;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int orow = m * 16 + quad * 4 + r;
