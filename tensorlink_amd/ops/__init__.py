@@ -231,15 +231,22 @@ def attention_decode(q, k_cache, v_cache, seq_lens,
 # ---------------------------------------------------------------------------
 # Linear (skinny-M decode GEMM)
 # ---------------------------------------------------------------------------
-# Measured on MI355X: the hand-written streaming GEMM beats hipBLASLt up
-# to M≈64 (2.2 vs 1.3 TB/s at M=16); at M=256 hipBLASLt's tuned picks win
-# (profiles/skinny_gemm_ab.md), so large decode batches route to the lib.
-_SKINNY_MAX_M = 64
+def _use_skinny(M: int, N: int, K: int) -> bool:
+    """Measured dispatch policy (profiles/skinny_gemm_ab.md, MI355X):
+    the hand-written glds streaming GEMM beats hipBLASLt at small M
+    (M=16: 2.50 vs 1.33 TB/s) and on deep-K narrow-N shapes (down-proj
+    256x3584x18944: 72.7 vs 102 µs); hipBLASLt's tuned stream-K wins the
+    wide-N shapes (gate/up at 2.95 TB/s)."""
+    if K % 32 != 0 or N % 64 != 0 or M > 256:
+        return False
+    if M <= 64:
+        return True
+    return K >= 8192 and N <= 4096
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor,
            bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """x @ W^T + bias. Routes skinny decode shapes (M <= 512) to the
+    """x @ W^T + bias. Routes decode shapes where it wins to the
     hand-written MFMA streaming kernel; everything else (prefill M,
     training) to torch/hipBLASLt."""
     import os as _os
@@ -249,7 +256,7 @@ def linear(x: torch.Tensor, weight: torch.Tensor,
         K = x.shape[-1]
         M = x.numel() // K
         N = weight.shape[0]
-        if M <= _SKINNY_MAX_M and K % 32 == 0 and N % 64 == 0:
+        if _use_skinny(M, N, K):
             return _require_ext().skinny_gemm(
                 x.contiguous(), weight,
                 bias.to(torch.bfloat16) if bias is not None else None)
