@@ -241,7 +241,10 @@ def _use_skinny(M: int, N: int, K: int) -> bool:
         return False
     if M <= 64:
         return True
-    return K >= 8192 and N <= 4096
+    # deep-K won the isolated microbench (down-proj 72.7 vs 102 µs) but
+    # measured ~4% SLOWER decode in the real pipeline (L2 state differs);
+    # keep it off at large M until in-pipeline numbers say otherwise.
+    return False
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor,

@@ -76,9 +76,12 @@ class PipelineRunner:
                                        device=self.device, dtype=self.dtype)
         elif init == "empty":
             self.stage.to(device=self.device, dtype=self.dtype)
+        self._no_graph = False
         if quantize == "fp8":
             from tensorlink_amd.models.quant import quantize_experts_fp8
             quantize_experts_fp8(self.stage)
+            # torch._scaled_mm is not hipGraph-capture-safe on ROCm 7.2
+            self._no_graph = True
         self.stage.eval()
         self.p2p = P2P(rank, world, group, rank_base) if world > 1 else None
         self.is_first = rank == 0
@@ -195,14 +198,23 @@ class PipelineRunner:
             finished = (cur == eos) if eos is not None else None
             n_out = T
             import os as _os
-            if (self.device.type == "cuda" and sp.temperature <= 0
-                    and T > 4 and not _os.environ.get("TL_NO_GRAPH")
-                    and on_token is None and eos is None):
+            use_graph = (self.device.type == "cuda" and sp.temperature <= 0
+                         and T > 4 and not _os.environ.get("TL_NO_GRAPH")
+                         and not self._no_graph
+                         and on_token is None and eos is None)
+            if use_graph:
                 # hipGraph-captured decode: the whole per-token step (all
                 # layers + head + argmax) replays as one graph launch,
                 # eliminating ~10 kernel-launch gaps per layer.
-                self._graph_decode(cur, positions, out_tokens, T)
-            else:
+                try:
+                    self._graph_decode(cur, positions, out_tokens, T)
+                except RuntimeError:
+                    # capture-unsupported op somewhere in the stage: fall
+                    # back to the eager loop permanently for this runner
+                    self._no_graph = True
+                    self._decode_graph = None
+                    use_graph = False
+            if not use_graph:
                 for t in range(1, T):
                     pos = positions.unsqueeze(1)
                     logits = self.stage(cur.unsqueeze(1), pos.int(),
