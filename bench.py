@@ -37,7 +37,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", default="Qwen/Qwen2.5-7B-Instruct")
-    p.add_argument("--batch-per-gpu", type=int, default=32,
+    p.add_argument("--batch-per-gpu", type=int, default=256,
                    help="sequences per GPU (weak scaling)")
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--new-tokens", type=int, default=128)

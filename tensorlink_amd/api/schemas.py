@@ -27,6 +27,7 @@ class GenerationRequest(BaseModel):
     num_beams: int = 1
     stream: bool = False
     reasoning: bool = False
+    stop: Optional[List[str]] = None
     history: Optional[List[Dict[str, str]]] = None
     input_format: str = "chat"       # chat | raw
     output_format: str = "simple"    # simple | openai | raw

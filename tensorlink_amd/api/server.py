@@ -97,6 +97,7 @@ def create_app(engine: InferenceEngine,
             "top_p": req.top_p,
             "do_sample": req.temperature > 0,
             "stream": req.stream,
+            "stop": req.stop,
             "input_format": "chat",
             "output_format": "openai",
         }

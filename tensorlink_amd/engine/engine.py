@@ -209,7 +209,19 @@ class InferenceEngine:
             top_p=args["top_p"], top_k=int(args.get("top_k", 0) or 0),
             max_new_tokens=args["max_new_tokens"],
             eos_token_id=args.get("eos_token_id"))
-        return job, prompt, ids, sp
+        stop = request.get("stop") or []
+        if isinstance(stop, str):
+            stop = [stop]
+        return job, prompt, ids, sp, list(stop)
+
+    @staticmethod
+    def _apply_stop(text: str, stop) -> str:
+        """Truncate at the earliest stop string (OpenAI `stop` param)."""
+        for s in stop:
+            idx = text.find(s)
+            if idx >= 0:
+                text = text[:idx]
+        return text
 
     def generate(self, request: Dict) -> Dict:
         """Non-streaming completion; returns a formatted response dict."""
@@ -218,12 +230,13 @@ class InferenceEngine:
                                 request.get("model", ""),
                                 request.get("output_format", "openai"))
         try:
-            job, prompt, ids, sp = self._prepare(request)
+            job, prompt, ids, sp, stop = self._prepare(request)
             with self._serve_lock:
                 self._bcast(("generate", job.name))
                 out, stats = job.runner.generate(ids, sp, return_stats=True)
             text = job.tokenizer.decode(out[0].tolist(),
                                         skip_special_tokens=True)
+            text = self._apply_stop(text, stop)
             n_new = out.shape[1]
             job.requests_served += 1
             job.tokens_generated += n_new
@@ -244,7 +257,7 @@ class InferenceEngine:
                                 request.get("model", ""),
                                 request.get("output_format", "openai"))
         try:
-            job, prompt, ids, sp = self._prepare(request)
+            job, prompt, ids, sp, stop = self._prepare(request)
         except Exception as e:
             yield f"data: {__import__('json').dumps(fmt.format_error(str(e)))}\n\n"
             yield "data: [DONE]\n\n"
