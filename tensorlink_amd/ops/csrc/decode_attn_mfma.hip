@@ -25,7 +25,6 @@ namespace {
 constexpr int BLOCK = 256;
 constexpr int NWAVE = 4;
 constexpr int KT = 16;            // keys per MFMA tile
-constexpr int VPAD = 8;           // vt row pad (stride 40 elems = 80 B)
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));

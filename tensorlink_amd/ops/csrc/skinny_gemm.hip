@@ -21,8 +21,6 @@
 namespace {
 
 constexpr int BLOCK = 256;
-constexpr int XPAD = 8;              // x_lds row stride 72 elems (144 B):
-                                     // 16-B aligned, conflict-free b128
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));

@@ -79,24 +79,28 @@ class HybridTrainer:
             assert B % self.dp == 0, "batch must divide dp"
             b = B // self.dp
             shard_ids = shard_labels = None
+            # NCCL send/recv needs device tensors
+            comm_dev = self.device if self.device.type == "cuda" else "cpu"
             if self.global_rank == 0:
                 input_ids = input_ids.to(torch.int64)
                 labels = labels.to(torch.int64)
                 for d in range(1, self.dp):
                     dist.send(torch.stack(
                         [input_ids[d * b:(d + 1) * b],
-                         labels[d * b:(d + 1) * b]]).contiguous(),
+                         labels[d * b:(d + 1) * b]]).contiguous().to(comm_dev),
                         d * self.pp)
                 shard_ids = input_ids[:b]
                 shard_labels = labels[:b]
             elif self.pp_rank == 0:
-                buf = torch.empty(2, b, S, dtype=torch.int64)
+                buf = torch.empty(2, b, S, dtype=torch.int64, device=comm_dev)
                 dist.recv(buf, 0)
                 shard_ids, shard_labels = buf[0], buf[1]
             loss = self.trainer.train_step(shard_ids, shard_labels,
                                            n_micro=n_micro)
             # average reported loss across replicas
-            t = torch.tensor([loss], dtype=torch.float64)
+            t = torch.tensor([loss], dtype=torch.float64,
+                             device=self.device if self.device.type == "cuda"
+                             else "cpu")
             dist.all_reduce(t, group=self.dp_group)
             return float(t.item() / self.dp)
         return self.trainer.train_step(input_ids, labels, n_micro=n_micro)
