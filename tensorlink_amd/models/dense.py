@@ -75,10 +75,11 @@ class Attention(nn.Module):
         self.config = config
         h, q, kv = config.hidden_size, config.q_size, config.kv_size
         bias = config.qkv_bias
-        self.q_proj = TLLinear(h, q, bias=bias)
-        self.k_proj = TLLinear(h, kv, bias=bias)
-        self.v_proj = TLLinear(h, kv, bias=bias)
+        # fused qkv: ONE [q+2kv, h] GEMM instead of three (checkpoint
+        # loading maps HF q/k/v rows into slices — models/loader.py)
+        self.qkv_proj = TLLinear(h, q + 2 * kv, bias=bias)
         self.o_proj = TLLinear(q, h, bias=False)
+        self.q_size, self.kv_size = q, kv
         self.n_heads = config.num_attention_heads
         self.n_kv = config.num_key_value_heads
         self.head_dim = config.head_dim
@@ -93,52 +94,70 @@ class Attention(nn.Module):
             / self.head_dim))
         self.register_buffer("inv_freq", inv_freq, persistent=False)
 
+    def _split_qkv(self, qkv, B, S):
+        q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size],
+                            dim=-1)
+        q = q.reshape(B, S, self.n_heads, self.head_dim)
+        k = k.reshape(B, S, self.n_kv, self.head_dim)
+        v = v.reshape(B, S, self.n_kv, self.head_dim)
+        return q, k, v
+
     def forward(self, x: torch.Tensor, positions: torch.Tensor,
                 kv_cache: Optional[KVCache] = None, layer_idx: int = 0,
                 training: bool = False) -> torch.Tensor:
         B, S, H = x.shape
-        q = self.q_proj(x).view(B, S, self.n_heads, self.head_dim)
-        k = self.k_proj(x).view(B, S, self.n_kv, self.head_dim)
-        v = self.v_proj(x).view(B, S, self.n_kv, self.head_dim)
-        if self.use_qk_norm:
-            q = ops.rmsnorm(q, self.q_norm.to(q.dtype),
-                            self.config.rms_norm_eps)
-            k = ops.rmsnorm(k, self.k_norm.to(k.dtype),
-                            self.config.rms_norm_eps)
-
+        eps = self.config.rms_norm_eps
+        qkv = self.qkv_proj(x)
         flat_pos = positions.reshape(-1)
-        if training:
-            q2, k2 = ops.apply_rope(q.reshape(B * S, self.n_heads, -1),
-                                    k.reshape(B * S, self.n_kv, -1),
-                                    flat_pos, self.inv_freq)
-            q = q2.view(B, S, self.n_heads, -1)
-            k = k2.view(B, S, self.n_kv, -1)
-            out = ops.attention_train(q, k, v, causal=True, scale=self.scale)
-        else:
-            q = q.contiguous()
-            k = k.contiguous()
-            if kv_cache is None:
+
+        if training or kv_cache is None:
+            q, k, v = self._split_qkv(qkv, B, S)
+            if self.use_qk_norm:
+                q = ops.rmsnorm(q.contiguous(), self.q_norm.to(q.dtype), eps)
+                k = ops.rmsnorm(k.contiguous(), self.k_norm.to(k.dtype), eps)
+            if training:
+                q2, k2 = ops.apply_rope(q.reshape(B * S, self.n_heads, -1),
+                                        k.reshape(B * S, self.n_kv, -1),
+                                        flat_pos, self.inv_freq)
+                q = q2.view(B, S, self.n_heads, -1)
+                k = k2.view(B, S, self.n_kv, -1)
+                out = ops.attention_train(q, k, v, causal=True,
+                                          scale=self.scale)
+            else:
+                q = q.contiguous()
+                k = k.contiguous()
                 ops.apply_rope_(q.view(B * S, self.n_heads, -1),
                                 k.view(B * S, self.n_kv, -1), flat_pos,
                                 self.inv_freq)
-                out = ops.attention_prefill(q, k, v, causal=True,
-                                            scale=self.scale)
-            else:
-                v = v.contiguous()
-                ops.rope_append_(q.view(B * S, self.n_heads, -1),
-                                 k.view(B * S, self.n_kv, -1),
-                                 v.view(B * S, self.n_kv, -1),
-                                 kv_cache.k[layer_idx],
-                                 kv_cache.v[layer_idx], flat_pos,
-                                 self.inv_freq, S)
-                if S == 1:
-                    out = ops.attention_decode(
-                        q, kv_cache.k[layer_idx], kv_cache.v[layer_idx],
-                        kv_cache.seq_lens + 1, scale=self.scale)
-                else:
-                    # prefill with cache write (fresh prompt: cache was empty)
-                    out = ops.attention_prefill(q, k, v, causal=True,
-                                                scale=self.scale)
+                out = ops.attention_prefill(q, k, v.contiguous(),
+                                            causal=True, scale=self.scale)
+            return self.o_proj(out.reshape(B, S, -1))
+
+        # inference with KV cache: fused rope+append straight off the
+        # fused-QKV rows (q comes back contiguous, k/v land in the cache)
+        if self.use_qk_norm:
+            q, k, v = self._split_qkv(qkv, B, S)
+            q = ops.rmsnorm(q.contiguous(), self.q_norm.to(q.dtype), eps)
+            k = ops.rmsnorm(k.contiguous(), self.k_norm.to(k.dtype), eps)
+            qkv = torch.cat([q.reshape(B, S, -1), k.reshape(B, S, -1),
+                             v.reshape(B, S, -1)], dim=-1)
+        q = ops.rope_append_(qkv.reshape(B * S, -1), kv_cache.k[layer_idx],
+                             kv_cache.v[layer_idx], flat_pos, self.inv_freq,
+                             S, self.n_heads, self.n_kv)
+        q = q.view(B, S, self.n_heads, self.head_dim)
+        if S == 1:
+            out = ops.attention_decode(
+                q, kv_cache.k[layer_idx], kv_cache.v[layer_idx],
+                kv_cache.seq_lens + 1, scale=self.scale)
+        else:
+            # prefill with cache write (fresh prompt at positions 0..S-1):
+            # rotated k / raw v are read back from the cache slabs
+            k_attn = kv_cache.k[layer_idx][:, :, :S].permute(
+                0, 2, 1, 3).contiguous()
+            v_attn = kv_cache.v[layer_idx][:, :, :S].permute(
+                0, 2, 1, 3).contiguous()
+            out = ops.attention_prefill(q, k_attn, v_attn, causal=True,
+                                        scale=self.scale)
         return self.o_proj(out.reshape(B, S, -1))
 
 
@@ -146,12 +165,18 @@ class MLP(nn.Module):
     def __init__(self, config: ModelConfig):
         super().__init__()
         h, i = config.hidden_size, config.intermediate_size
-        self.gate_proj = TLLinear(h, i, bias=False)
-        self.up_proj = TLLinear(h, i, bias=False)
+        self.inter = i
+        # fused gate+up: ONE [2i, h] GEMM (HF gate/up rows map to slices)
+        self.gate_up_proj = TLLinear(h, 2 * i, bias=False)
         self.down_proj = TLLinear(i, h, bias=False)
 
     def forward(self, x):
-        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+        gu = self.gate_up_proj(x)
+        if torch.is_grad_enabled() and gu.requires_grad:
+            gate, up = gu.split([self.inter, self.inter], dim=-1)
+            return self.down_proj(ops.swiglu(gate.contiguous(),
+                                             up.contiguous()))
+        return self.down_proj(ops.swiglu_fused(gu))
 
 
 class MoEMLP(nn.Module):

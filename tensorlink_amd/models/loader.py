@@ -23,42 +23,50 @@ from tensorlink_amd.models.configs import ModelConfig
 from tensorlink_amd.models.dense import StageModel, build_stage
 
 
-# HF parameter name -> native name, relative to one decoder layer
+# HF parameter name -> (native name, row offset key), relative to one
+# decoder layer. Offsets are symbolic ("q"/"kv"/"i") because the fused
+# qkv/gate_up parameters pack several HF tensors as row slices.
 _HF_LAYER_MAP = {
-    "input_layernorm.weight": "input_layernorm",
-    "post_attention_layernorm.weight": "post_attention_layernorm",
-    "self_attn.q_proj.weight": "self_attn.q_proj.weight",
-    "self_attn.q_proj.bias": "self_attn.q_proj.bias",
-    "self_attn.k_proj.weight": "self_attn.k_proj.weight",
-    "self_attn.k_proj.bias": "self_attn.k_proj.bias",
-    "self_attn.v_proj.weight": "self_attn.v_proj.weight",
-    "self_attn.v_proj.bias": "self_attn.v_proj.bias",
-    "self_attn.o_proj.weight": "self_attn.o_proj.weight",
-    "self_attn.q_norm.weight": "self_attn.q_norm",
-    "self_attn.k_norm.weight": "self_attn.k_norm",
-    "mlp.gate_proj.weight": "mlp.gate_proj.weight",
-    "mlp.up_proj.weight": "mlp.up_proj.weight",
-    "mlp.down_proj.weight": "mlp.down_proj.weight",
-    # Mixtral MoE
-    "block_sparse_moe.gate.weight": "mlp.gate.weight",
+    "input_layernorm.weight": ("input_layernorm", None),
+    "post_attention_layernorm.weight": ("post_attention_layernorm", None),
+    "self_attn.q_proj.weight": ("self_attn.qkv_proj.weight", 0),
+    "self_attn.q_proj.bias": ("self_attn.qkv_proj.bias", 0),
+    "self_attn.k_proj.weight": ("self_attn.qkv_proj.weight", "q"),
+    "self_attn.k_proj.bias": ("self_attn.qkv_proj.bias", "q"),
+    "self_attn.v_proj.weight": ("self_attn.qkv_proj.weight", "q+kv"),
+    "self_attn.v_proj.bias": ("self_attn.qkv_proj.bias", "q+kv"),
+    "self_attn.o_proj.weight": ("self_attn.o_proj.weight", None),
+    "self_attn.q_norm.weight": ("self_attn.q_norm", None),
+    "self_attn.k_norm.weight": ("self_attn.k_norm", None),
+    "mlp.gate_proj.weight": ("mlp.gate_up_proj.weight", 0),
+    "mlp.up_proj.weight": ("mlp.gate_up_proj.weight", "i"),
+    "mlp.down_proj.weight": ("mlp.down_proj.weight", None),
+    # Mixtral MoE router
+    "block_sparse_moe.gate.weight": ("mlp.gate.weight", None),
 }
 
 
-def _map_hf_key(key: str, layer_start: int, layer_end: int,
-                stage: StageModel) -> Optional[str]:
-    """Map an HF checkpoint key to a native stage parameter name (or None if
-    the key belongs to another stage)."""
+def _resolve_offset(sym, config) -> int:
+    if sym in (None, 0):
+        return 0
+    return {"q": config.q_size, "q+kv": config.q_size + config.kv_size,
+            "i": config.intermediate_size}[sym]
+
+
+def _map_hf_key(key: str, layer_start: int, layer_end: int, stage):
+    """Map an HF checkpoint key to (native stage parameter name,
+    row offset) or None if the key belongs to another stage."""
     if key.startswith("model.embed_tokens.weight"):
         if stage.has_embedding:
-            return "embed_tokens.weight"
+            return "embed_tokens.weight", 0
         if stage.has_head and stage.config.tie_word_embeddings:
-            return "lm_head.weight"
+            return "lm_head.weight", 0
         return None
     if key == "model.norm.weight":
-        return "norm" if stage.has_head else None
+        return ("norm", 0) if stage.has_head else None
     if key == "lm_head.weight":
-        return "lm_head.weight" if (stage.has_head and
-                                    hasattr(stage, "lm_head")) else None
+        return ("lm_head.weight", 0) if (stage.has_head and
+                                         hasattr(stage, "lm_head")) else None
     if key.startswith("model.layers."):
         rest = key[len("model.layers."):]
         idx_str, _, sub = rest.partition(".")
@@ -67,14 +75,21 @@ def _map_hf_key(key: str, layer_start: int, layer_end: int,
             return None
         local = idx - layer_start
         if sub in _HF_LAYER_MAP:
-            return f"layers.{local}.{_HF_LAYER_MAP[sub]}"
+            native, off = _HF_LAYER_MAP[sub]
+            return (f"layers.{local}.{native}",
+                    _resolve_offset(off, stage.config))
         # Mixtral experts: block_sparse_moe.experts.E.w1/w3/w2
         if sub.startswith("block_sparse_moe.experts."):
             parts = sub.split(".")
             e = parts[2]
             wname = parts[3]
-            proj = {"w1": "gate_proj", "w3": "up_proj", "w2": "down_proj"}[wname]
-            return f"layers.{local}.mlp.experts.{e}.{proj}.weight"
+            native, off = {
+                "w1": ("gate_up_proj.weight", 0),
+                "w3": ("gate_up_proj.weight", "i"),
+                "w2": ("down_proj.weight", None),
+            }[wname]
+            return (f"layers.{local}.mlp.experts.{e}.{native}",
+                    _resolve_offset(off, stage.config))
     return None
 
 
@@ -104,13 +119,14 @@ def load_stage_from_checkpoint(stage: StageModel, ckpt_dir: str,
         with safe_open(path, framework="pt", device="cpu") as f:
             shard_keys = keys if keys is not None else f.keys()
             for key in shard_keys:
-                native = _map_hf_key(key, stage.layer_start, stage.layer_end,
+                mapped = _map_hf_key(key, stage.layer_start, stage.layer_end,
                                      stage)
-                if native is None or native not in params:
+                if mapped is None or mapped[0] not in params:
                     continue
+                native, off = mapped
                 t = f.get_tensor(key).to(dtype)
                 with torch.no_grad():
-                    params[native].copy_(t)
+                    params[native][off:off + t.shape[0]].copy_(t)
                 loaded += 1
     stage.to(device=device, dtype=dtype)
     return loaded

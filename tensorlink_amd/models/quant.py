@@ -81,7 +81,7 @@ def quantize_experts_fp8(stage: nn.Module) -> int:
     for mod in stage.modules():
         if isinstance(mod, MoEMLP):
             for expert in mod.experts:
-                for name in ("gate_proj", "up_proj", "down_proj"):
+                for name in ("gate_up_proj", "down_proj"):
                     lin = getattr(expert, name)
                     if isinstance(lin, nn.Linear):
                         setattr(expert, name, Fp8Linear.from_linear(lin))

@@ -105,26 +105,32 @@ def apply_rope_(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
     _require_ext().rope_(q, k, positions.int(), inv_freq.float(), sign)
 
 
-def rope_append_(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                 k_cache: torch.Tensor, v_cache: torch.Tensor,
-                 positions: torch.Tensor, inv_freq: torch.Tensor,
-                 S: int) -> None:
-    """Fused in-place RoPE on q/k + KV-cache append (inference hot path).
+def rope_append_(qkv: torch.Tensor, k_cache: torch.Tensor,
+                 v_cache: torch.Tensor, positions: torch.Tensor,
+                 inv_freq: torch.Tensor, S: int, Hq: int,
+                 Hkv: int) -> torch.Tensor:
+    """Fused RoPE + KV-cache append from the fused-QKV projection output.
 
-    q [T,Hq,D], k/v [T,Hkv,D] with T = B*S; caches [B,Hkv,Smax,D];
-    positions [T] gives the RoPE angle and the cache slot per token.
+    qkv [T, (Hq+2*Hkv)*D] with T = B*S (q | k | v per row); caches
+    [B,Hkv,Smax,D]; positions [T] gives the RoPE angle and cache slot.
+    Returns the rotated q as a contiguous [T, Hq, D] tensor.
     """
-    if _on_gpu(q):
-        _require_ext().rope_append_(q, k, v, k_cache, v_cache,
-                                    positions.int(), inv_freq.float(), S)
-        return
+    if _on_gpu(qkv):
+        return _require_ext().rope_append_(qkv.contiguous(), k_cache,
+                                           v_cache, positions.int(),
+                                           inv_freq.float(), S, Hq, Hkv)
+    D = k_cache.shape[-1]
+    T = qkv.numel() // qkv.shape[-1]
+    q = qkv[..., :Hq * D].reshape(T, Hq, D).contiguous()
+    k = qkv[..., Hq * D:(Hq + Hkv) * D].reshape(T, Hkv, D).contiguous()
+    v = qkv[..., (Hq + Hkv) * D:].reshape(T, Hkv, D).contiguous()
     apply_rope_(q, k, positions, inv_freq)
-    T, Hkv, D = k.shape
     B = T // S
     pos = positions.view(B, S).long()
     for b in range(B):
         k_cache[b, :, pos[b]] = k.view(B, S, Hkv, D)[b].transpose(0, 1)
         v_cache[b, :, pos[b]] = v.view(B, S, Hkv, D)[b].transpose(0, 1)
+    return q
 
 
 class _RopeFn(torch.autograd.Function):
@@ -174,6 +180,15 @@ def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     if torch.is_grad_enabled() and (gate.requires_grad or up.requires_grad):
         return _SwiGLUFn.apply(gate, up)
     return _require_ext().swiglu_fwd(gate.contiguous(), up.contiguous())
+
+
+def swiglu_fused(gate_up: torch.Tensor) -> torch.Tensor:
+    """silu(gu[..., :I]) * gu[..., I:] on the fused gate_up GEMM output
+    (inference path — avoids two .contiguous() splits)."""
+    if not _on_gpu(gate_up):
+        I = gate_up.shape[-1] // 2
+        return ref.swiglu(gate_up[..., :I], gate_up[..., I:])
+    return _require_ext().swiglu_fused(gate_up.contiguous())
 
 
 # ---------------------------------------------------------------------------

@@ -30,10 +30,12 @@ void tl_decode_attn(const void* q, const void* k_cache, const void* v_cache,
 void tl_prefill_attn(const void* q, const void* k, const void* v, void* out,
                      int B, int S, int Hq, int Hkv, int D, float scale,
                      int causal, hipStream_t stream);
-void tl_rope_append(void* q, void* k, const void* v, void* k_cache,
+void tl_rope_append(const void* qkv, void* q_out, void* k_cache,
                     void* v_cache, const void* positions,
-                    const void* inv_freq, int64_t T, int S, int Hq, int Hkv,
-                    int D, int Smax, hipStream_t stream);
+                    const void* inv_freq, int64_t T, int row_stride, int S,
+                    int Hq, int Hkv, int D, int Smax, hipStream_t stream);
+void tl_swiglu_fused(const void* gu, void* out, int64_t N, int I,
+                     hipStream_t stream);
 void tl_decode_attn_mfma(const void* q, const void* k_cache,
                          const void* v_cache, const void* seq_lens, void* out,
                          void* partial, void* partial_ml, int B, int Hq,
@@ -208,28 +210,44 @@ Tensor prefill_attn(Tensor q, Tensor k, Tensor v, double scale, bool causal) {
   return out;
 }
 
-void rope_append_(Tensor q, Tensor k, Tensor v, Tensor k_cache,
-                  Tensor v_cache, Tensor positions, Tensor inv_freq,
-                  int64_t S) {
-  CHECK_IN(q, torch::kBFloat16);
-  CHECK_IN(k, torch::kBFloat16);
-  CHECK_IN(v, torch::kBFloat16);
+// qkv: [T, row_stride] fused projection output (q | k | v per row).
+// Returns rotated q as a contiguous [T, Hq, D] tensor; k/v land in the
+// caches.
+Tensor rope_append_(Tensor qkv, Tensor k_cache, Tensor v_cache,
+                    Tensor positions, Tensor inv_freq, int64_t S,
+                    int64_t Hq, int64_t Hkv) {
+  CHECK_IN(qkv, torch::kBFloat16);
   CHECK_IN(k_cache, torch::kBFloat16);
   CHECK_IN(v_cache, torch::kBFloat16);
   CHECK_IN(positions, torch::kInt);
   CHECK_IN(inv_freq, torch::kFloat);
-  const int D = q.size(-1);
-  const int Hq = q.size(-2);
-  const int Hkv = k.size(-2);
-  const int64_t T = q.numel() / ((int64_t)Hq * D);
+  const int D = k_cache.size(3);
+  const int row_stride = qkv.size(-1);
+  TORCH_CHECK(row_stride == (Hq + 2 * Hkv) * D, "qkv width mismatch");
+  const int64_t T = qkv.numel() / row_stride;
   const int Smax = k_cache.size(2);
   TORCH_CHECK(T % S == 0, "T must be divisible by S");
   TORCH_CHECK(k_cache.size(0) * S == T, "cache batch mismatch");
   TORCH_CHECK(D % 16 == 0, "D must be divisible by 16");
-  tl_rope_append(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                 k_cache.data_ptr(), v_cache.data_ptr(),
-                 positions.data_ptr(), inv_freq.data_ptr(), T, (int)S, Hq,
-                 Hkv, D, Smax, cur_stream());
+  auto q_out = torch::empty({T, Hq, (int64_t)D}, qkv.options());
+  tl_rope_append(qkv.data_ptr(), q_out.data_ptr(), k_cache.data_ptr(),
+                 v_cache.data_ptr(), positions.data_ptr(),
+                 inv_freq.data_ptr(), T, row_stride, (int)S, (int)Hq,
+                 (int)Hkv, D, Smax, cur_stream());
+  return q_out;
+}
+
+Tensor swiglu_fused(Tensor gu) {
+  CHECK_IN(gu, torch::kBFloat16);
+  const int64_t I2 = gu.size(-1);
+  TORCH_CHECK(I2 % 16 == 0, "fused width must be divisible by 16");
+  const int64_t I = I2 / 2;
+  const int64_t N = gu.numel() / I2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = torch::empty(sizes, gu.options());
+  tl_swiglu_fused(gu.data_ptr(), out.data_ptr(), N, (int)I, cur_stream());
+  return out;
 }
 
 Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
@@ -270,7 +288,8 @@ Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("rope_append_", &rope_append_,
-          "fused in-place RoPE + KV-cache append");
+          "fused RoPE + KV-cache append from fused-QKV rows");
+  mod.def("swiglu_fused", &swiglu_fused, "silu-mul on fused [.,2I] rows");
   mod.def("skinny_gemm", &skinny_gemm, "decode-M GEMM: x @ W^T + bias");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm fwd (+residual)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm bwd");

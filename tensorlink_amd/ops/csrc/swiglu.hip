@@ -27,6 +27,23 @@ __global__ __launch_bounds__(BLOCK) void swiglu_fwd_kernel(
   reinterpret_cast<bf16x8*>(out)[i] = o;
 }
 
+// fused-layout variant: input gu [N, 2I] (gate | up per row) from the
+// fused gate_up GEMM; out [N, I]. Avoids two .contiguous() splits.
+__global__ __launch_bounds__(BLOCK) void swiglu_fused_kernel(
+    const bf16* __restrict__ gu, bf16* __restrict__ out, int64_t N, int I) {
+  const int64_t i = ((int64_t)blockIdx.x * BLOCK + threadIdx.x) * 8;
+  if (i >= N * (int64_t)I) return;
+  const int64_t r = i / I;
+  const int64_t c = i - r * I;
+  const bf16x8 g = *reinterpret_cast<const bf16x8*>(gu + r * 2 * I + c);
+  const bf16x8 u = *reinterpret_cast<const bf16x8*>(gu + r * 2 * I + I + c);
+  bf16x8 o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    o.v[j] = f2bf(silu(bf2f(g.v[j])) * bf2f(u.v[j]));
+  *reinterpret_cast<bf16x8*>(out + r * I + c) = o;
+}
+
 __global__ __launch_bounds__(BLOCK) void swiglu_bwd_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ gate,
     const bf16* __restrict__ up, bf16* __restrict__ dgate,
@@ -61,6 +78,14 @@ void tl_swiglu_fwd(const void* gate, const void* up, void* out, int64_t n,
   dim3 grid((uint32_t)cdiv(n8, BLOCK)), block(BLOCK);
   hipLaunchKernelGGL(swiglu_fwd_kernel, grid, block, 0, stream,
                      (const bf16*)gate, (const bf16*)up, (bf16*)out, n8);
+}
+
+void tl_swiglu_fused(const void* gu, void* out, int64_t N, int I,
+                     hipStream_t stream) {
+  const int64_t work = N * (int64_t)I / 8;
+  dim3 grid((uint32_t)((work + BLOCK - 1) / BLOCK)), block(BLOCK);
+  hipLaunchKernelGGL(swiglu_fused_kernel, grid, block, 0, stream,
+                     (const bf16*)gu, (bf16*)out, N, I);
 }
 
 void tl_swiglu_bwd(const void* dout, const void* gate, const void* up,

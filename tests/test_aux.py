@@ -81,8 +81,8 @@ def test_fp8_expert_conversion_and_forward():
     m = build_full_model(cfg)
     init_random_stage(m, dtype=torch.float32)
     n = quantize_experts_fp8(m)
-    assert n == cfg.num_local_experts * 3 * cfg.num_hidden_layers
-    assert isinstance(m.layers[0].mlp.experts[0].gate_proj, Fp8Linear)
+    assert n == cfg.num_local_experts * 2 * cfg.num_hidden_layers
+    assert isinstance(m.layers[0].mlp.experts[0].gate_up_proj, Fp8Linear)
     ids = torch.randint(0, cfg.vocab_size, (2, 8))
     pos = torch.arange(8).unsqueeze(0).expand(2, -1).contiguous()
     out = m(ids, pos)

@@ -88,14 +88,18 @@ def test_hf_key_mapping():
     plan = plan_for_world(cfg, 2)
     s1 = build_stage(cfg, plan.stage_for_rank(1))  # layers 2..4, head
     assert _map_hf_key("model.layers.2.self_attn.q_proj.weight", 2, 4, s1) \
-        == "layers.0.self_attn.q_proj.weight"
+        == ("layers.0.self_attn.qkv_proj.weight", 0)
+    assert _map_hf_key("model.layers.2.self_attn.k_proj.weight", 2, 4, s1) \
+        == ("layers.0.self_attn.qkv_proj.weight", cfg.q_size)
+    assert _map_hf_key("model.layers.2.mlp.up_proj.weight", 2, 4, s1) \
+        == ("layers.0.mlp.gate_up_proj.weight", cfg.intermediate_size)
     assert _map_hf_key("model.layers.1.self_attn.q_proj.weight", 2, 4, s1) \
         is None
-    assert _map_hf_key("model.norm.weight", 2, 4, s1) == "norm"
-    assert _map_hf_key("lm_head.weight", 2, 4, s1) == "lm_head.weight"
+    assert _map_hf_key("model.norm.weight", 2, 4, s1) == ("norm", 0)
+    assert _map_hf_key("lm_head.weight", 2, 4, s1) == ("lm_head.weight", 0)
     s0 = build_stage(cfg, plan.stage_for_rank(0))
     assert _map_hf_key("model.embed_tokens.weight", 0, 2, s0) \
-        == "embed_tokens.weight"
+        == ("embed_tokens.weight", 0)
     assert _map_hf_key("model.norm.weight", 0, 2, s0) is None
 
 

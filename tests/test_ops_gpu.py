@@ -187,26 +187,26 @@ def test_decode_attn_softmax_stability():
 
 
 def test_rope_append_fused():
+    """Fused-QKV rope+append vs reference rotate-then-place."""
     ops = _ext()
     from tensorlink_amd.ops import reference as ref
     torch.manual_seed(7)
     B, S, Hq, Hkv, D, Smax = 3, 5, 8, 2, 128, 64
     T = B * S
-    q = torch.randn(T, Hq, D, device=DEV, dtype=torch.bfloat16)
-    k = torch.randn(T, Hkv, D, device=DEV, dtype=torch.bfloat16)
-    v = torch.randn(T, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, device=DEV,
+                      dtype=torch.bfloat16)
+    q = qkv[:, :Hq * D].reshape(T, Hq, D).contiguous()
+    k = qkv[:, Hq * D:(Hq + Hkv) * D].reshape(T, Hkv, D).contiguous()
+    v = qkv[:, (Hq + Hkv) * D:].reshape(T, Hkv, D).contiguous()
     kc = torch.zeros(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
     vc = torch.zeros_like(kc)
     pos = torch.arange(S, device=DEV, dtype=torch.int32).repeat(B) + 3
     inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2, device=DEV).float() / D))
-    q2, k2 = q.clone(), k.clone()
-    ops.rope_append_(q2, k2, v, kc, vc, pos, inv, S)
-    # reference: rotate then place
+    q_out = ops.rope_append_(qkv, kc, vc, pos, inv, S, Hq, Hkv)
     cos, sin = ref.rope_cos_sin(D, pos, device=DEV)
     qr, kr = ref.apply_rope(q.float().unsqueeze(0), k.float().unsqueeze(0),
                             cos, sin)
-    torch.testing.assert_close(q2.float(), qr[0], atol=2e-2, rtol=2e-2)
-    torch.testing.assert_close(k2.float(), kr[0], atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(q_out.float(), qr[0], atol=2e-2, rtol=2e-2)
     krB = kr[0].view(B, S, Hkv, D)
     vB = v.view(B, S, Hkv, D).float()
     for b in range(B):
@@ -216,6 +216,16 @@ def test_rope_append_fused():
                                        atol=2e-2, rtol=2e-2)
             torch.testing.assert_close(vc[b, :, p].float(), vB[b, s],
                                        atol=1e-3, rtol=1e-3)
+
+
+def test_swiglu_fused_gpu():
+    ops = _ext()
+    torch.manual_seed(8)
+    N, I = 33, 512
+    gu = torch.randn(N, 2 * I, device=DEV, dtype=torch.bfloat16)
+    out = ops.swiglu_fused(gu)
+    ref32 = torch.nn.functional.silu(gu[:, :I].float()) * gu[:, I:].float()
+    torch.testing.assert_close(out.float(), ref32, atol=2e-2, rtol=2e-2)
 
 
 @pytest.mark.parametrize("B,Hq,Hkv,D,L,ns", [

@@ -137,7 +137,7 @@ def test_moe_fp8_generate_gpu():
     plan = plan_for_world("tiny-moe", 1)
     r = PipelineRunner(plan, 0, 1, device=DEV, quantize="fp8")
     from tensorlink_amd.models.quant import Fp8Linear
-    assert isinstance(r.stage.layers[0].mlp.experts[0].gate_proj, Fp8Linear)
+    assert isinstance(r.stage.layers[0].mlp.experts[0].gate_up_proj, Fp8Linear)
     ids = torch.randint(0, 1024, (2, 12))
     out = r.generate(ids, SamplingParams(max_new_tokens=4))
     assert out.shape == (2, 4)
