@@ -20,3 +20,15 @@ out3 = r3.generate(ids, SamplingParams(max_new_tokens=12)).cpu()
 print("eager-greedy:", out3.tolist())
 print("graph==eager-greedy:", torch.equal(out_graph, out3))
 print("eager-sample==eager-greedy:", torch.equal(out_eager, out3))
+
+del os.environ["TL_NO_GRAPH"]
+ids2 = torch.randint(0, 1024, (4, 16))
+out2 = r.generate(ids2, SamplingParams(max_new_tokens=12)).cpu()
+out2e = r2.generate(ids2, SamplingParams(temperature=1e-6, top_k=1, max_new_tokens=12)).cpu()
+os.environ["TL_NO_GRAPH"] = "1"
+out2g = r3.generate(ids2, SamplingParams(max_new_tokens=12)).cpu()
+print("2nd: graph==eager-sample:", torch.equal(out2, out2e))
+print("2nd: graph==eager-greedy:", torch.equal(out2, out2g))
+print("g:", out2[1].tolist())
+print("e:", out2e[1].tolist())
+print("n:", out2g[1].tolist())
