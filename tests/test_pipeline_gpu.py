@@ -150,3 +150,24 @@ def test_moe_fp8_generate_gpu():
     ids = torch.randint(0, 1024, (2, 12))
     out = r.generate(ids, SamplingParams(max_new_tokens=4))
     assert out.shape == (2, 4)
+
+
+def test_distributed_model_gpu():
+    """DistributedModel user API end-to-end on the HIP path."""
+    from tensorlink_amd import ops
+    from tensorlink_amd.module import DistributedModel
+    m = DistributedModel("tiny", training=True, lr=1e-3)
+    opt = m.create_optimizer(lr=1e-3)
+    torch.manual_seed(0)
+    ids = torch.randint(0, 1024, (4, 24))
+    losses = []
+    for _ in range(3):
+        logits = m(ids)
+        loss = ops.causal_lm_loss(logits, ids.to(logits.device))
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0]
+    out = m.generate(ids, max_new_tokens=6)
+    assert out.shape == (4, 30)

@@ -147,3 +147,51 @@ def test_dp2_training_replicas_stay_synced():
     assert l0 == l1, "losses must agree across replicas"
     assert l0[-1] < l0[0]
     assert abs(h0 - h1) < 1e-6, "replica weights diverged"
+
+
+def _engine_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.parallel.comm import init_distributed
+    init_distributed(backend="gloo")
+    eng = InferenceEngine(rank=rank, world=world,
+                          device=torch.device("cpu"))
+    if rank != 0:
+        eng.worker_loop()
+        return
+    eng.load_model("tiny")
+    resp = eng.generate({"hf_name": "tiny", "message": "hello",
+                         "max_new_tokens": 6, "do_sample": False,
+                         "output_format": "simple"})
+    chunks = list(eng.generate_stream({
+        "hf_name": "tiny", "message": "hi", "max_new_tokens": 4,
+        "do_sample": False, "output_format": "simple"}))
+    eng.shutdown()
+    q.put((resp, chunks))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_engine_serving_over_pp2():
+    """Full serving stack (engine + worker_loop) across 2 pipeline ranks —
+    the reference's validator+worker serving topology on gloo loopback."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_engine_worker, args=(r, 2, 29641, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    try:
+        resp, chunks = q.get(timeout=240)
+    finally:
+        for p in ps:
+            p.join(30)
+            if p.is_alive():
+                p.terminate()
+    assert "response" in resp and resp["model"] == "tiny"
+    assert chunks[-1] == "data: [DONE]\n\n"
+    assert len(chunks) >= 2
