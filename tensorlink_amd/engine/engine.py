@@ -29,9 +29,9 @@ from tensorlink_amd.engine.formatter import (ResponseFormatter,
                                              normalize_generate_args)
 from tensorlink_amd.engine.tokenizer import load_tokenizer
 from tensorlink_amd.models.configs import get_config
-from tensorlink_amd.parallel.comm import P2P, device_for_rank, is_distributed
+from tensorlink_amd.parallel.comm import P2P, device_for_rank
 from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
-from tensorlink_amd.parallel.planner import ModelParser, plan_for_world
+from tensorlink_amd.parallel.planner import plan_for_world
 
 
 @dataclass

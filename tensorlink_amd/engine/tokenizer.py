@@ -10,7 +10,7 @@ end-to-end on synthetic/random-weight models.
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+from typing import List
 
 
 class ByteTokenizer:

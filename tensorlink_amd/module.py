@@ -22,8 +22,7 @@ Execution backends:
 from __future__ import annotations
 
 import os
-import time
-from typing import Dict, Optional
+from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -9,7 +9,7 @@ implementations in :mod:`tensorlink_amd.ops.reference` run instead.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 
