@@ -173,6 +173,15 @@ def _tiny() -> ModelConfig:
         max_position_embeddings=512, rope_theta=10000.0, architecture="llama")
 
 
+def _tiny_qwen3() -> ModelConfig:
+    # exercises the qk-norm (Qwen3) attention path at kernel-supported D=64
+    return ModelConfig(
+        name="tiny-qwen3", vocab_size=1024, hidden_size=256,
+        intermediate_size=512, num_hidden_layers=4, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=64, max_position_embeddings=512,
+        rope_theta=10000.0, architecture="qwen3")
+
+
 def _tiny_moe() -> ModelConfig:
     return ModelConfig(
         name="tiny-moe", vocab_size=1024, hidden_size=256,
@@ -201,6 +210,7 @@ PRESETS = {
     "meta-llama/Llama-3-8B": _llama3_8b,
     "mistralai/Mixtral-8x7B-v0.1": _mixtral_8x7b,
     "tiny": _tiny,
+    "tiny-qwen3": _tiny_qwen3,
     "tiny-moe": _tiny_moe,
 }
 

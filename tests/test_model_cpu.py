@@ -114,3 +114,22 @@ def test_hf_config_mapping():
     cfg = ModelConfig.from_hf_config(hf, name="qwen")
     assert cfg.qkv_bias and cfg.architecture == "qwen2"
     assert cfg.head_dim == 128
+
+
+def test_qwen3_qk_norm_decode_matches_full():
+    """Qwen3-style per-head q/k RMSNorm: cached decode must equal the
+    uncached full forward (exercises the norm+re-fuse branch)."""
+    cfg, m = _make("tiny-qwen3")
+    torch.manual_seed(3)
+    B, S = 2, 10
+    ids = torch.randint(0, cfg.vocab_size, (B, S))
+    pos = torch.arange(S).unsqueeze(0).expand(B, -1).contiguous()
+    logits = m(ids, pos)
+    cache = m.make_kv_cache(B, 32, "cpu")
+    m(ids, pos, kv_cache=cache)
+    nxt = logits[:, -1].argmax(-1, keepdim=True)
+    lg = m(nxt, torch.full((B, 1), S, dtype=torch.long), kv_cache=cache)
+    full = m(torch.cat([ids, nxt], 1),
+             torch.arange(S + 1).unsqueeze(0).expand(B, -1).contiguous())
+    torch.testing.assert_close(lg[:, 0], full[:, -1], atol=1e-4, rtol=1e-3)
+    assert m.layers[0].self_attn.use_qk_norm

@@ -171,3 +171,13 @@ def test_distributed_model_gpu():
     assert losses[-1] < losses[0]
     out = m.generate(ids, max_new_tokens=6)
     assert out.shape == (4, 30)
+
+
+def test_qwen3_qk_norm_generate_gpu():
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    plan = plan_for_world("tiny-qwen3", 1)
+    r = PipelineRunner(plan, 0, 1, device=DEV, dtype=torch.bfloat16)
+    ids = torch.randint(0, 1024, (2, 12))
+    out = r.generate(ids, SamplingParams(max_new_tokens=6))
+    assert out.shape == (2, 6)
