@@ -134,7 +134,9 @@ class Attention(nn.Module):
             return self.o_proj(out.reshape(B, S, -1))
 
         # inference with KV cache: fused rope+append straight off the
-        # fused-QKV rows (q comes back contiguous, k/v land in the cache)
+        # fused-QKV rows (q comes back contiguous, k/v land in the cache;
+        # paged caches pass their block table through to the kernels)
+        table = getattr(kv_cache, "table", None)
         if self.use_qk_norm:
             q, k, v = self._split_qkv(qkv, B, S)
             q = ops.rmsnorm(q.contiguous(), self.q_norm.to(q.dtype), eps)
@@ -143,19 +145,22 @@ class Attention(nn.Module):
                              v.reshape(B, S, -1)], dim=-1)
         q = ops.rope_append_(qkv.reshape(B * S, -1), kv_cache.k[layer_idx],
                              kv_cache.v[layer_idx], flat_pos, self.inv_freq,
-                             S, self.n_heads, self.n_kv)
+                             S, self.n_heads, self.n_kv, block_table=table)
         q = q.view(B, S, self.n_heads, self.head_dim)
         if S == 1:
             out = ops.attention_decode(
                 q, kv_cache.k[layer_idx], kv_cache.v[layer_idx],
-                kv_cache.seq_lens + 1, scale=self.scale)
+                kv_cache.seq_lens + 1, scale=self.scale, block_table=table)
         else:
             # prefill with cache write (fresh prompt at positions 0..S-1):
-            # rotated k / raw v are read back from the cache slabs
-            k_attn = kv_cache.k[layer_idx][:, :, :S].permute(
-                0, 2, 1, 3).contiguous()
-            v_attn = kv_cache.v[layer_idx][:, :, :S].permute(
-                0, 2, 1, 3).contiguous()
+            # rotated k / raw v are read back from the cache
+            if table is None:
+                k_attn = kv_cache.k[layer_idx][:, :, :S].permute(
+                    0, 2, 1, 3).contiguous()
+                v_attn = kv_cache.v[layer_idx][:, :, :S].permute(
+                    0, 2, 1, 3).contiguous()
+            else:
+                k_attn, v_attn = kv_cache.gather_contiguous(layer_idx, S)
             out = ops.attention_prefill(q, k_attn, v_attn, causal=True,
                                         scale=self.scale)
         return self.o_proj(out.reshape(B, S, -1))
@@ -266,10 +271,14 @@ class StageModel(nn.Module):
     def num_layers(self):
         return len(self.layers)
 
-    def make_kv_cache(self, batch: int, max_seq: int, device,
-                      dtype=None) -> KVCache:
+    def make_kv_cache(self, batch: int, max_seq: int, device, dtype=None,
+                      kv_mode: str = "contiguous"):
         if dtype is None:
             dtype = next(self.parameters()).dtype
+        if kv_mode == "paged":
+            from tensorlink_amd.models.paged import PagedKVCache
+            return PagedKVCache(self.num_layers, batch, max_seq,
+                                self.config, device, dtype)
         return KVCache(self.num_layers, batch, max_seq, self.config, device,
                        dtype)
 

@@ -121,9 +121,11 @@ class Gpt2StageModel(nn.Module):
     def num_layers(self):
         return len(self.layers)
 
-    def make_kv_cache(self, batch, max_seq, device, dtype=None) -> KVCache:
+    def make_kv_cache(self, batch, max_seq, device, dtype=None,
+                      kv_mode: str = "contiguous") -> KVCache:
         if dtype is None:
             dtype = next(self.parameters()).dtype
+        # GPT-2 (plumbing family) always uses the contiguous cache
         return KVCache(self.num_layers, batch, max_seq, self.config, device,
                        dtype)
 

@@ -1,0 +1,78 @@
+"""Paged KV cache (vLLM-style block tables) for long-context serving.
+
+SURVEY.md §2.4/§5 names the paged-KV decode kernel as the long-context
+story: instead of one contiguous [B, Hkv, Smax, D] slab per sequence, K/V
+live in a shared pool of 128-position pages and a per-sequence block table
+maps position/128 -> page id. The decode and rope+append kernels take the
+table (decode_attn_mfma.hip / rope_append.hip, PAGED variants), so
+sequences only consume memory for pages they actually touch and a long
+context never needs a contiguous reservation.
+
+Enabled per-runner with kv_mode="paged" (contiguous remains the default
+serving path). Pages are allocated from a free list in shuffled order so
+tests genuinely exercise the indirection.
+"""
+
+from __future__ import annotations
+
+import random
+from typing import List, Optional
+
+import torch
+
+PAGE = 128
+
+
+class PagedKVCache:
+    """Same duck-type surface as KVCache (k/v per layer, seq_lens, reset,
+    advance) plus a block table; pool tensors are shared by reference when
+    the cache is sliced for micro-batches."""
+
+    def __init__(self, n_layers: int, batch: int, max_seq: int, config,
+                 device, dtype=torch.bfloat16, shuffle_pages: bool = True):
+        pages_per_seq = (max_seq + PAGE - 1) // PAGE
+        n_pages = batch * pages_per_seq
+        self.k = [torch.zeros(n_pages, config.num_key_value_heads, PAGE,
+                              config.head_dim, device=device, dtype=dtype)
+                  for _ in range(n_layers)]
+        self.v = [torch.zeros_like(self.k[0]) for _ in range(n_layers)]
+        self.seq_lens = torch.zeros(batch, device=device, dtype=torch.int32)
+        self.max_seq = pages_per_seq * PAGE
+        self.batch = batch
+        self.pages_per_seq = pages_per_seq
+        ids = list(range(n_pages))
+        if shuffle_pages:
+            random.Random(1234).shuffle(ids)
+        self.table = torch.tensor(ids, device=device,
+                                  dtype=torch.int32).view(batch,
+                                                          pages_per_seq)
+
+    def reset(self):
+        self.seq_lens.zero_()
+
+    def advance(self, n: int):
+        self.seq_lens += n
+
+    def gather_contiguous(self, layer: int, S: int):
+        """Materialize the first S positions as [B, S, Hkv, D] (prefill
+        attention readback; once per request)."""
+        np = (S + PAGE - 1) // PAGE
+        idx = self.table[:, :np].long()                        # [B, np]
+        k = self.k[layer][idx]                                 # [B,np,H,PAGE,D]
+        v = self.v[layer][idx]
+        B, _, H, _, D = k.shape
+        k = k.permute(0, 1, 3, 2, 4).reshape(B, np * PAGE, H, D)
+        v = v.permute(0, 1, 3, 2, 4).reshape(B, np * PAGE, H, D)
+        return k[:, :S].contiguous(), v[:, :S].contiguous()
+
+
+def kv_slice_paged(cache: PagedKVCache, s: int, e: int) -> PagedKVCache:
+    view = object.__new__(PagedKVCache)
+    view.k = cache.k                       # pools shared
+    view.v = cache.v
+    view.seq_lens = cache.seq_lens[s:e]
+    view.table = cache.table[s:e]
+    view.max_seq = cache.max_seq
+    view.batch = e - s
+    view.pages_per_seq = cache.pages_per_seq
+    return view

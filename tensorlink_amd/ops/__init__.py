@@ -107,18 +107,20 @@ def apply_rope_(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
 
 def rope_append_(qkv: torch.Tensor, k_cache: torch.Tensor,
                  v_cache: torch.Tensor, positions: torch.Tensor,
-                 inv_freq: torch.Tensor, S: int, Hq: int,
-                 Hkv: int) -> torch.Tensor:
+                 inv_freq: torch.Tensor, S: int, Hq: int, Hkv: int,
+                 block_table: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Fused RoPE + KV-cache append from the fused-QKV projection output.
 
     qkv [T, (Hq+2*Hkv)*D] with T = B*S (q | k | v per row); caches
-    [B,Hkv,Smax,D]; positions [T] gives the RoPE angle and cache slot.
+    [B,Hkv,Smax,D] (or paged pools [n_pages,Hkv,128,D] with block_table
+    [B, pages]); positions [T] gives the RoPE angle and cache slot.
     Returns the rotated q as a contiguous [T, Hq, D] tensor.
     """
     if _on_gpu(qkv):
-        return _require_ext().rope_append_(qkv.contiguous(), k_cache,
-                                           v_cache, positions.int(),
-                                           inv_freq.float(), S, Hq, Hkv)
+        return _require_ext().rope_append_(
+            qkv.contiguous(), k_cache, v_cache, positions.int(),
+            inv_freq.float(), S, Hq, Hkv,
+            block_table.int() if block_table is not None else None)
     D = k_cache.shape[-1]
     T = qkv.numel() // qkv.shape[-1]
     q = qkv[..., :Hq * D].reshape(T, Hq, D).contiguous()
@@ -127,9 +129,19 @@ def rope_append_(qkv: torch.Tensor, k_cache: torch.Tensor,
     apply_rope_(q, k, positions, inv_freq)
     B = T // S
     pos = positions.view(B, S).long()
-    for b in range(B):
-        k_cache[b, :, pos[b]] = k.view(B, S, Hkv, D)[b].transpose(0, 1)
-        v_cache[b, :, pos[b]] = v.view(B, S, Hkv, D)[b].transpose(0, 1)
+    kB = k.view(B, S, Hkv, D)
+    vB = v.view(B, S, Hkv, D)
+    if block_table is None:
+        for b in range(B):
+            k_cache[b, :, pos[b]] = kB[b].transpose(0, 1)
+            v_cache[b, :, pos[b]] = vB[b].transpose(0, 1)
+    else:
+        for b in range(B):
+            for s_i in range(S):
+                p = int(pos[b, s_i])
+                page = int(block_table[b, p // 128])
+                k_cache[page, :, p % 128] = kB[b, s_i]
+                v_cache[page, :, p % 128] = vB[b, s_i]
     return q
 
 
@@ -221,10 +233,11 @@ def attention_train(q, k, v, causal: bool = True,
 
 
 def attention_decode(q, k_cache, v_cache, seq_lens,
-                     scale: Optional[float] = None,
-                     n_split: int = 0) -> torch.Tensor:
-    """q [B,1,Hq,D] or [B,Hq,D]; caches [B,Hkv,Smax,D]; seq_lens [B].
-    n_split: flash-decode seq splits (0 = auto heuristic)."""
+                     scale: Optional[float] = None, n_split: int = 0,
+                     block_table: Optional[torch.Tensor] = None
+                     ) -> torch.Tensor:
+    """q [B,1,Hq,D] or [B,Hq,D]; caches [B,Hkv,Smax,D] (or paged pools +
+    block_table); seq_lens [B]. n_split: flash-decode splits (0 = auto)."""
     squeeze = q.dim() == 4
     if squeeze:
         q3 = q.squeeze(1)
@@ -232,14 +245,27 @@ def attention_decode(q, k_cache, v_cache, seq_lens,
         q3 = q
     scale = scale if scale is not None else 1.0 / math.sqrt(q3.shape[-1])
     if not _on_gpu(q3):
-        # reference wants [B,Smax,Hkv,D]
-        out = ref.attention_decode(q3.unsqueeze(1),
-                                   k_cache.permute(0, 2, 1, 3),
-                                   v_cache.permute(0, 2, 1, 3), seq_lens,
-                                   scale)
+        if block_table is not None:
+            # gather pages -> contiguous [B, L, Hkv, D] for the reference
+            B = q3.shape[0]
+            L = int(seq_lens.max())
+            np_ = (L + 127) // 128
+            idx = block_table[:, :np_].long()
+            kc = k_cache[idx].permute(0, 1, 3, 2, 4)
+            vc = v_cache[idx].permute(0, 1, 3, 2, 4)
+            kc = kc.reshape(B, np_ * 128, *kc.shape[3:])[:, :L]
+            vc = vc.reshape(B, np_ * 128, *vc.shape[3:])[:, :L]
+            out = ref.attention_decode(q3.unsqueeze(1), kc, vc, seq_lens,
+                                       scale)
+        else:
+            out = ref.attention_decode(q3.unsqueeze(1),
+                                       k_cache.permute(0, 2, 1, 3),
+                                       v_cache.permute(0, 2, 1, 3),
+                                       seq_lens, scale)
         return out if squeeze else out.squeeze(1)
-    out = _require_ext().decode_attn(q3.contiguous(), k_cache, v_cache,
-                                     seq_lens.int(), scale, n_split)
+    out = _require_ext().decode_attn(
+        q3.contiguous(), k_cache, v_cache, seq_lens.int(), scale, n_split,
+        block_table.int() if block_table is not None else None)
     return out.unsqueeze(1) if squeeze else out
 
 

@@ -32,15 +32,17 @@ void tl_prefill_attn(const void* q, const void* k, const void* v, void* out,
                      int causal, hipStream_t stream);
 void tl_rope_append(const void* qkv, void* q_out, void* k_cache,
                     void* v_cache, const void* positions,
-                    const void* inv_freq, int64_t T, int row_stride, int S,
-                    int Hq, int Hkv, int D, int Smax, hipStream_t stream);
+                    const void* inv_freq, const void* block_table,
+                    int64_t T, int row_stride, int S, int Hq, int Hkv,
+                    int D, int Smax, int bt_stride, hipStream_t stream);
 void tl_swiglu_fused(const void* gu, void* out, int64_t N, int I,
                      hipStream_t stream);
 void tl_decode_attn_mfma(const void* q, const void* k_cache,
-                         const void* v_cache, const void* seq_lens, void* out,
+                         const void* v_cache, const void* seq_lens,
+                         const void* block_table, void* out,
                          void* partial, void* partial_ml, int B, int Hq,
                          int Hkv, int Smax, int D, float scale, int n_split,
-                         hipStream_t stream);
+                         int bt_stride, hipStream_t stream);
 void tl_skinny_gemm(const void* x, const void* w, const void* bias,
                     void* out, void* partial, int M, int N, int K,
                     int n_split, hipStream_t stream);
@@ -155,7 +157,8 @@ void adamw_(Tensor param, Tensor grad, Tensor m, Tensor v, double lr,
 }
 
 Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
-                   double scale, int64_t n_split) {
+                   double scale, int64_t n_split,
+                   c10::optional<Tensor> block_table) {
   CHECK_IN(q, torch::kBFloat16);
   CHECK_IN(k_cache, torch::kBFloat16);
   CHECK_IN(v_cache, torch::kBFloat16);
@@ -163,10 +166,18 @@ Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
   const int B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int Hkv = k_cache.size(1), Smax = k_cache.size(2);
   const int G = Hq / Hkv;
+  const void* bt = nullptr;
+  int bt_stride = 0;
+  if (block_table.has_value()) {
+    CHECK_IN(block_table.value(), torch::kInt);
+    bt = block_table->data_ptr();
+    bt_stride = block_table->size(1);
+  }
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   auto out = torch::empty_like(q);
-  const bool legacy = getenv("TL_DECODE_LEGACY") != nullptr || G > 16;
+  const bool legacy = (getenv("TL_DECODE_LEGACY") != nullptr || G > 16) &&
+                      bt == nullptr;
   if (legacy) {
     tl_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
                    seq_lens.data_ptr(), out.data_ptr(), B, Hq, Hkv, Smax, D,
@@ -191,9 +202,9 @@ Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
     pml = partial_ml.data_ptr();
   }
   tl_decode_attn_mfma(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-                      seq_lens.data_ptr(), out.data_ptr(), pp, pml, B, Hq,
-                      Hkv, Smax, D, (float)scale, (int)n_split,
-                      cur_stream());
+                      seq_lens.data_ptr(), bt, out.data_ptr(), pp, pml, B,
+                      Hq, Hkv, Smax, D, (float)scale, (int)n_split,
+                      bt_stride, cur_stream());
   return out;
 }
 
@@ -215,7 +226,8 @@ Tensor prefill_attn(Tensor q, Tensor k, Tensor v, double scale, bool causal) {
 // caches.
 Tensor rope_append_(Tensor qkv, Tensor k_cache, Tensor v_cache,
                     Tensor positions, Tensor inv_freq, int64_t S,
-                    int64_t Hq, int64_t Hkv) {
+                    int64_t Hq, int64_t Hkv,
+                    c10::optional<Tensor> block_table) {
   CHECK_IN(qkv, torch::kBFloat16);
   CHECK_IN(k_cache, torch::kBFloat16);
   CHECK_IN(v_cache, torch::kBFloat16);
@@ -227,13 +239,23 @@ Tensor rope_append_(Tensor qkv, Tensor k_cache, Tensor v_cache,
   const int64_t T = qkv.numel() / row_stride;
   const int Smax = k_cache.size(2);
   TORCH_CHECK(T % S == 0, "T must be divisible by S");
-  TORCH_CHECK(k_cache.size(0) * S == T, "cache batch mismatch");
   TORCH_CHECK(D % 16 == 0, "D must be divisible by 16");
+  const void* bt = nullptr;
+  int bt_stride = 0;
+  if (block_table.has_value()) {
+    CHECK_IN(block_table.value(), torch::kInt);
+    TORCH_CHECK(k_cache.size(2) == 128, "paged pool PAGE must be 128");
+    bt = block_table->data_ptr();
+    bt_stride = block_table->size(1);
+    TORCH_CHECK(block_table->size(0) * S == T, "table batch mismatch");
+  } else {
+    TORCH_CHECK(k_cache.size(0) * S == T, "cache batch mismatch");
+  }
   auto q_out = torch::empty({T, Hq, (int64_t)D}, qkv.options());
   tl_rope_append(qkv.data_ptr(), q_out.data_ptr(), k_cache.data_ptr(),
                  v_cache.data_ptr(), positions.data_ptr(),
-                 inv_freq.data_ptr(), T, row_stride, (int)S, (int)Hq,
-                 (int)Hkv, D, Smax, cur_stream());
+                 inv_freq.data_ptr(), bt, T, row_stride, (int)S, (int)Hq,
+                 (int)Hkv, D, Smax, bt_stride, cur_stream());
   return q_out;
 }
 

@@ -31,16 +31,19 @@ typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 
 // out layout when SPLIT > 1: partial [B, Hq, SPLIT, D] float, ml [B, Hq,
 // SPLIT, 2] float (m, l). SPLIT == 1 writes bf16 out directly.
-template <int D, int SPLIT_MODE>
+// PAGED: k_cache/v_cache are page pools [n_pages, Hkv, PAGE, D] and
+// block_table [B, bt_stride] maps position/PAGE -> page id (PAGE = 128).
+template <int D, int SPLIT_MODE, int PAGED>
 __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
     const bf16* __restrict__ q,        // [B, Hq, D]
-    const bf16* __restrict__ k_cache,  // [B, Hkv, Smax, D]
+    const bf16* __restrict__ k_cache,  // [B, Hkv, Smax, D] (or pool)
     const bf16* __restrict__ v_cache,
     const int* __restrict__ seq_lens,  // [B]
+    const int* __restrict__ block_table,
     bf16* __restrict__ out,            // [B, Hq, D]
     float* __restrict__ partial,       // [B, Hq, SPLIT, D]
     float* __restrict__ partial_ml,    // [B, Hq, SPLIT, 2]
-    int Hq, int Hkv, int Smax, float scale, int n_split) {
+    int Hq, int Hkv, int Smax, float scale, int n_split, int bt_stride) {
   constexpr int KCH = D / 32;        // MFMA k-chunks for QK^T
   constexpr int NS = D / 16;         // PV output col tiles
   const int b = blockIdx.x;
@@ -116,8 +119,17 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
     // B fragments straight from global K + stage V transposed into LDS
     f32x4 sacc = (f32x4){0.f, 0.f, 0.f, 0.f};
     {
-      const bf16* krow = kbase + (int64_t)safe_key * D;
-      const bf16* vrow = vbase + (int64_t)safe_key * D;
+      const bf16 *krow, *vrow;
+      if (PAGED) {
+        const int page = block_table[b * bt_stride + (safe_key >> 7)];
+        const int64_t off =
+            (((int64_t)page * Hkv + hkv) * 128 + (safe_key & 127)) * D;
+        krow = k_cache + off;
+        vrow = v_cache + off;
+      } else {
+        krow = kbase + (int64_t)safe_key * D;
+        vrow = vbase + (int64_t)safe_key * D;
+      }
 #pragma unroll
       for (int c = 0; c < KCH; ++c) {
         bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
@@ -271,22 +283,30 @@ extern "C" {
 // scratch: partial floats [B*Hq*n_split*D] + ml [B*Hq*n_split*2]; pass
 // nullptr when n_split == 1.
 void tl_decode_attn_mfma(const void* q, const void* k_cache,
-                         const void* v_cache, const void* seq_lens, void* out,
+                         const void* v_cache, const void* seq_lens,
+                         const void* block_table, void* out,
                          void* partial, void* partial_ml, int B, int Hq,
                          int Hkv, int Smax, int D, float scale, int n_split,
-                         hipStream_t stream) {
+                         int bt_stride, hipStream_t stream) {
   dim3 grid(B, Hkv, n_split), block(BLOCK);
-#define LAUNCH(DD, SM)                                                       \
-  hipLaunchKernelGGL((decode_attn_mfma_kernel<DD, SM>), grid, block, 0,      \
+#define LAUNCH(DD, SM, PG)                                                   \
+  hipLaunchKernelGGL((decode_attn_mfma_kernel<DD, SM, PG>), grid, block, 0,  \
                      stream, (const bf16*)q, (const bf16*)k_cache,           \
-                     (const bf16*)v_cache, (const int*)seq_lens, (bf16*)out, \
+                     (const bf16*)v_cache, (const int*)seq_lens,             \
+                     (const int*)block_table, (bf16*)out,                    \
                      (float*)partial, (float*)partial_ml, Hq, Hkv, Smax,     \
-                     scale, n_split)
-  if (D == 128) {
-    if (n_split == 1) LAUNCH(128, 0); else LAUNCH(128, 1);
-  } else if (D == 64) {
-    if (n_split == 1) LAUNCH(64, 0); else LAUNCH(64, 1);
-  }
+                     scale, n_split, bt_stride)
+#define PICK(DD)                                                             \
+  do {                                                                       \
+    if (block_table) {                                                       \
+      if (n_split == 1) LAUNCH(DD, 0, 1); else LAUNCH(DD, 1, 1);             \
+    } else {                                                                 \
+      if (n_split == 1) LAUNCH(DD, 0, 0); else LAUNCH(DD, 1, 0);             \
+    }                                                                        \
+  } while (0)
+  if (D == 128) PICK(128);
+  else if (D == 64) PICK(64);
+#undef PICK
 #undef LAUNCH
   if (n_split > 1) {
     dim3 cgrid(B * Hq), cblock(D);

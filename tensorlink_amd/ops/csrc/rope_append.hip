@@ -18,14 +18,26 @@ namespace {
 
 constexpr int BLOCK = 256;
 
+// PAGED != 0: caches are page pools [n_pages, Hkv, 128, D] addressed via
+// block_table [B, bt_stride].
+template <int PAGED>
 __global__ __launch_bounds__(BLOCK) void rope_append_kernel(
     const bf16* __restrict__ qkv, bf16* __restrict__ q_out,
     bf16* __restrict__ k_cache, bf16* __restrict__ v_cache,
     const int* __restrict__ positions, const float* __restrict__ inv_freq,
-    int row_stride, int S, int Hq, int Hkv, int D, int Smax) {
+    const int* __restrict__ block_table, int row_stride, int S, int Hq,
+    int Hkv, int D, int Smax, int bt_stride) {
   const int64_t token = blockIdx.x;
   const int b = token / S;
   const int pos = positions[token];
+  int64_t slot;   // cache slot index (in units of D elems, per kv head 0)
+  if (PAGED) {
+    const int page = block_table[b * bt_stride + (pos >> 7)];
+    slot = ((int64_t)page * Hkv) * 128 + (pos & 127);
+  } else {
+    slot = ((int64_t)b * Hkv) * Smax + pos;
+  }
+  const int64_t head_stride = PAGED ? 128 : Smax;
   const int D2 = D / 2;
   const float fpos = (float)pos;
   const bf16* row = qkv + token * row_stride;
@@ -51,7 +63,7 @@ __global__ __launch_bounds__(BLOCK) void rope_append_kernel(
       const bf16* src = krow + hk * D;
       const float x1 = bf2f(src[d]);
       const float x2 = bf2f(src[d + D2]);
-      bf16* kc = k_cache + (((int64_t)b * Hkv + hk) * Smax + pos) * D;
+      bf16* kc = k_cache + (slot + hk * head_stride) * D;
       kc[d] = f2bf(x1 * c - x2 * s);
       kc[d + D2] = f2bf(x2 * c + x1 * s);
     }
@@ -64,7 +76,7 @@ __global__ __launch_bounds__(BLOCK) void rope_append_kernel(
     const bf16x8 val =
         *reinterpret_cast<const bf16x8*>(vrow + h * D + d8 * 8);
     reinterpret_cast<bf16x8*>(
-        v_cache + (((int64_t)b * Hkv + h) * Smax + pos) * D)[d8] = val;
+        v_cache + (slot + h * head_stride) * D)[d8] = val;
   }
 }
 
@@ -74,14 +86,22 @@ extern "C" {
 
 void tl_rope_append(const void* qkv, void* q_out, void* k_cache,
                     void* v_cache, const void* positions,
-                    const void* inv_freq, int64_t T, int row_stride, int S,
-                    int Hq, int Hkv, int D, int Smax, hipStream_t stream) {
+                    const void* inv_freq, const void* block_table,
+                    int64_t T, int row_stride, int S, int Hq, int Hkv,
+                    int D, int Smax, int bt_stride, hipStream_t stream) {
   dim3 grid((uint32_t)T), block(BLOCK);
-  hipLaunchKernelGGL(rope_append_kernel, grid, block, 0, stream,
-                     (const bf16*)qkv, (bf16*)q_out, (bf16*)k_cache,
-                     (bf16*)v_cache, (const int*)positions,
-                     (const float*)inv_freq, row_stride, S, Hq, Hkv, D,
-                     Smax);
+  if (block_table)
+    hipLaunchKernelGGL((rope_append_kernel<1>), grid, block, 0, stream,
+                       (const bf16*)qkv, (bf16*)q_out, (bf16*)k_cache,
+                       (bf16*)v_cache, (const int*)positions,
+                       (const float*)inv_freq, (const int*)block_table,
+                       row_stride, S, Hq, Hkv, D, Smax, bt_stride);
+  else
+    hipLaunchKernelGGL((rope_append_kernel<0>), grid, block, 0, stream,
+                       (const bf16*)qkv, (bf16*)q_out, (bf16*)k_cache,
+                       (bf16*)v_cache, (const int*)positions,
+                       (const float*)inv_freq, nullptr, row_stride, S, Hq,
+                       Hkv, D, Smax, bt_stride);
 }
 
 }  // extern "C"

@@ -38,7 +38,10 @@ class SamplingParams:
     seed: Optional[int] = None
 
 
-def _kv_slice(cache: KVCache, s: int, e: int) -> KVCache:
+def _kv_slice(cache, s: int, e: int):
+    from tensorlink_amd.models.paged import PagedKVCache, kv_slice_paged
+    if isinstance(cache, PagedKVCache):
+        return kv_slice_paged(cache, s, e)
     view = object.__new__(KVCache)
     view.k = [k[s:e] for k in cache.k]
     view.v = [v[s:e] for v in cache.v]
@@ -55,7 +58,8 @@ class PipelineRunner:
     def __init__(self, plan: StagePlan, rank: int, world: int,
                  device=None, init: str = "random", ckpt_dir: Optional[str] = None,
                  dtype=None, seed: int = 0, quantize: Optional[str] = None,
-                 group=None, rank_base: int = 0):
+                 group=None, rank_base: int = 0,
+                 kv_mode: Optional[str] = None):
         assert plan.num_stages == world, "plan stages must equal world size"
         self.group = group
         self.rank_base = rank_base
@@ -76,6 +80,9 @@ class PipelineRunner:
                                        device=self.device, dtype=self.dtype)
         elif init == "empty":
             self.stage.to(device=self.device, dtype=self.dtype)
+        import os as _os
+        self.kv_mode = kv_mode or _os.environ.get("TL_KV_MODE",
+                                                  "contiguous")
         self._no_graph = False
         if quantize == "fp8":
             from tensorlink_amd.models.quant import quantize_experts_fp8
@@ -111,7 +118,8 @@ class PipelineRunner:
             c.reset()
             return c
         self.kv_cache = self.stage.make_kv_cache(batch, max_seq, self.device,
-                                                 self.dtype)
+                                                 self.dtype,
+                                                 kv_mode=self.kv_mode)
         self._decode_graph = None
         return self.kv_cache
 
