@@ -223,12 +223,18 @@ def attention_train(q, k, v, causal: bool = True,
     are a later milestone; forward inference never uses this."""
     B, S, Hq, D = q.shape
     Hkv = k.shape[2]
-    rep = Hq // Hkv
     qt = q.transpose(1, 2)
-    kt = k.transpose(1, 2).repeat_interleave(rep, dim=1)
-    vt = v.transpose(1, 2).repeat_interleave(rep, dim=1)
-    out = torch.nn.functional.scaled_dot_product_attention(
-        qt, kt, vt, is_causal=causal, scale=scale)
+    kt = k.transpose(1, 2)
+    vt = v.transpose(1, 2)
+    try:
+        out = torch.nn.functional.scaled_dot_product_attention(
+            qt, kt, vt, is_causal=causal, scale=scale, enable_gqa=True)
+    except (TypeError, RuntimeError):
+        # older torch / unsupported backend: materialize repeated heads
+        rep = Hq // Hkv
+        out = torch.nn.functional.scaled_dot_product_attention(
+            qt, kt.repeat_interleave(rep, dim=1),
+            vt.repeat_interleave(rep, dim=1), is_causal=causal, scale=scale)
     return out.transpose(1, 2)
 
 

@@ -202,3 +202,25 @@ def test_paged_kv_decode_gpu_matches_contiguous():
     finally:
         del os.environ["TL_NO_GRAPH"]
     assert torch.equal(o1, o2)
+
+
+def test_planner_estimate_vs_actual_memory():
+    """The planner's stage estimate must be a sane upper-ish bound on what
+    the runner actually allocates (memory model validation on hardware)."""
+    from tensorlink_amd.models.configs import get_config
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    cfg = get_config("Qwen/Qwen2.5-7B-Instruct")
+    B, S, T = 8, 256, 16
+    plan = plan_for_world(cfg, 1, batch_size=B, seq_len=S + T)
+    torch.cuda.empty_cache()
+    torch.cuda.reset_peak_memory_stats()
+    r = PipelineRunner(plan, 0, 1, device=DEV)
+    ids = torch.randint(0, cfg.vocab_size, (B, S))
+    r.generate(ids, SamplingParams(max_new_tokens=T))
+    actual = torch.cuda.max_memory_allocated()
+    est = plan.stages[0].est_bytes
+    # estimate covers actual (activations modeled generously) but is not
+    # wildly off (< 4x)
+    assert actual < est * 1.1, (actual, est)
+    assert est < actual * 4, (actual, est)
