@@ -44,6 +44,7 @@ class ModelJob:
     requests_served: int = 0
     tokens_generated: int = 0
     last_request: float = 0.0
+    batcher: object = None          # ContinuousBatcher when enabled
 
 
 class EngineMetrics:
@@ -138,12 +139,19 @@ class InferenceEngine:
                       **kwargs) -> ModelJob:
         config = get_config(name)
         quantize = kwargs.pop("quantize", None)
+        continuous = kwargs.pop("continuous", False)
+        max_slots = kwargs.pop("max_slots", 16)
+        max_ctx = kwargs.pop("max_ctx", 2048)
         plan = plan_for_world(config, self.world, **kwargs)
         runner = PipelineRunner(plan, self.rank, self.world,
                                 device=self.device, init=init,
                                 ckpt_dir=ckpt_dir, quantize=quantize)
         tok = load_tokenizer(ckpt_dir or name)
         job = ModelJob(name=name, runner=runner, tokenizer=tok)
+        if continuous and self.world == 1 and self.rank == 0:
+            from tensorlink_amd.engine.batcher import ContinuousBatcher
+            job.batcher = ContinuousBatcher(runner, max_slots=max_slots,
+                                            max_ctx=max_ctx).start()
         self.jobs[name] = job
         return job
 
@@ -166,7 +174,9 @@ class InferenceEngine:
     def unload_model(self, name: str):
         with self._lock:
             self._bcast(("unload", name))
-            self.jobs.pop(name, None)
+            job = self.jobs.pop(name, None)
+            if job is not None and job.batcher is not None:
+                job.batcher.stop()
 
     def model_status(self, name: str) -> Dict:
         job = self.jobs.get(name)
@@ -231,6 +241,24 @@ class InferenceEngine:
                                 request.get("output_format", "openai"))
         try:
             job, prompt, ids, sp, stop = self._prepare(request)
+            if job.batcher is not None:
+                req = job.batcher.submit(
+                    ids[0], max_new_tokens=sp.max_new_tokens,
+                    temperature=sp.temperature, top_p=sp.top_p,
+                    top_k=sp.top_k, eos_token_id=sp.eos_token_id)
+                out_tokens = req.result()
+                n_new = len(out_tokens)
+                text = job.tokenizer.decode(out_tokens,
+                                            skip_special_tokens=True)
+                text = self._apply_stop(text, stop)
+                job.requests_served += 1
+                job.tokens_generated += n_new
+                self.metrics.record(n_new, time.time() - t0)
+                return fmt.format_response(
+                    text, prompt_tokens=ids.shape[1],
+                    completion_tokens=n_new,
+                    reasoning=bool(request.get("reasoning")),
+                    processing_time=time.time() - t0)
             with self._serve_lock:
                 self._bcast(("generate", job.name))
                 out, stats = job.runner.generate(ids, sp, return_stats=True)
@@ -271,11 +299,20 @@ class InferenceEngine:
 
         def run():
             try:
-                with self._serve_lock:
-                    self._bcast(("generate", job.name))
-                    out = job.runner.generate(ids, sp, on_token=on_token)
-                if out is not None:
-                    q.put(("final", out[0].tolist()))
+                if job.batcher is not None:
+                    req = job.batcher.submit(
+                        ids[0], max_new_tokens=sp.max_new_tokens,
+                        temperature=sp.temperature, top_p=sp.top_p,
+                        top_k=sp.top_k, eos_token_id=sp.eos_token_id)
+                    for tok in req.stream():
+                        q.put(tok)
+                else:
+                    with self._serve_lock:
+                        self._bcast(("generate", job.name))
+                        out = job.runner.generate(ids, sp,
+                                                  on_token=on_token)
+                    if out is not None:
+                        q.put(("final", out[0].tolist()))
             except Exception as e:  # pragma: no cover
                 q.put(("error", str(e)))
             q.put(None)
