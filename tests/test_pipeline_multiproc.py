@@ -195,3 +195,51 @@ def test_engine_serving_over_pp2():
     assert "response" in resp and resp["model"] == "tiny"
     assert chunks[-1] == "data: [DONE]\n\n"
     assert len(chunks) >= 2
+
+
+def _gen4_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    init_distributed(backend="gloo")
+    plan = plan_for_world("tiny", world)
+    r = PipelineRunner(plan, rank, world, device=torch.device("cpu"))
+    torch.manual_seed(21)
+    ids = torch.randint(0, 1024, (8, 12)) if rank == 0 else None
+    out = r.generate(ids, SamplingParams(max_new_tokens=5))
+    if rank == 0:
+        q.put(out)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(420)
+def test_pp4_ring_decode_matches_reference():
+    """4-stage ring-pipelined decode (the SCALE-bench shape) vs a
+    single-process stage-chained reference."""
+    out = _run(_gen4_worker, 4, 29651, timeout=360)
+    assert out.shape == (8, 5)
+    from tensorlink_amd.models.dense import build_stage
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.parallel.planner import plan_for_world
+    plan = plan_for_world("tiny", 4)
+    stages = []
+    for rk in range(4):
+        st = build_stage(plan.config, plan.stage_for_rank(rk))
+        init_random_stage(st, dtype=torch.float32, seed=rk)
+        stages.append(st)
+    torch.manual_seed(21)
+    ids = torch.randint(0, 1024, (8, 12))
+    cur = ids
+    for _ in range(5):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).expand(8, -1).contiguous()
+        h = cur
+        for st in stages[:-1]:
+            h = st(h, pos, return_logits=False)
+        logits = stages[-1](h, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 12:], out)
