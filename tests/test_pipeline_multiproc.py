@@ -54,21 +54,27 @@ def _train_worker(rank, world, port, q):
     dist.destroy_process_group()
 
 
-def _run(worker, world, port, timeout=240):
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    ps = [ctx.Process(target=worker, args=(r, world, port, q))
-          for r in range(world)]
-    for p in ps:
-        p.start()
-    try:
-        result = q.get(timeout=timeout)
-    finally:
+def _run(worker, world, port, timeout=240, retries=1):
+    last_exc = None
+    for attempt in range(retries + 1):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        ps = [ctx.Process(target=worker,
+                          args=(r, world, port + attempt * 7, q))
+              for r in range(world)]
         for p in ps:
-            p.join(30)
-            if p.is_alive():
-                p.terminate()
-    return result
+            p.start()
+        try:
+            result = q.get(timeout=timeout)
+            return result
+        except Exception as e:   # transient spawn/rendezvous flake: retry
+            last_exc = e
+        finally:
+            for p in ps:
+                p.join(30)
+                if p.is_alive():
+                    p.terminate()
+    raise last_exc
 
 
 @pytest.mark.timeout(300)
