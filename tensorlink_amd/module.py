@@ -109,7 +109,15 @@ class DistributedModel(nn.Module):
         self._p2p = None
         self.device = device
 
-        if world_size > 1 and mode in ("auto", "local", "spawn"):
+        launched_world = int(os.environ.get("WORLD_SIZE", "1"))
+        if world_size > 1 and (mode == "torchrun" or
+                               (mode == "auto" and
+                                launched_world == world_size)):
+            # already launched SPMD (torchrun): join the existing group
+            # instead of spawning local workers
+            self._join_torchrun(model, training, world_size, init, ckpt_dir,
+                                seed, lr)
+        elif world_size > 1 and mode in ("auto", "local", "spawn"):
             self._start_cluster(model, training, world_size, init, ckpt_dir,
                                 seed, lr)
         else:
@@ -127,6 +135,59 @@ class DistributedModel(nn.Module):
         self.config = self._runner.config
 
     # ------------------------------------------------------------------
+    def _join_torchrun(self, model, training, world, init, ckpt_dir, seed,
+                       lr):
+        """Under torchrun every rank constructs DistributedModel; rank 0
+        is the user-facing instance, other ranks enter the worker loop
+        via serve_worker() (call it after construction on rank != 0)."""
+        from tensorlink_amd.parallel.comm import init_distributed
+        rank, _ = init_distributed()
+        self._rank = rank
+        plan = plan_for_world(model, world, training=training)
+        if training:
+            self._trainer = PipelineTrainer(plan, rank, world,
+                                            device=self.device, init=init,
+                                            ckpt_dir=ckpt_dir, seed=seed,
+                                            lr=lr)
+            self._runner = self._trainer.runner
+        else:
+            self._trainer = None
+            self._runner = PipelineRunner(plan, rank, world,
+                                          device=self.device, init=init,
+                                          ckpt_dir=ckpt_dir, seed=seed)
+        self._p2p = self._runner.p2p
+
+    def serve_worker(self):
+        """Worker loop for torchrun ranks != 0 (mirrors the spawned-
+        cluster workers)."""
+        holder = self._trainer
+        runner = self._runner
+        p2p = runner.p2p
+        while True:
+            cmd = p2p.broadcast_obj(None, src=0)
+            op = cmd[0]
+            if op == "shutdown":
+                return
+            elif op == "generate":
+                runner.generate(None, None)
+            elif op == "fwd_eval":
+                runner.forward_logits(None)
+            elif op == "forward":
+                holder.spmd_forward(None)
+            elif op == "backward":
+                holder.spmd_backward(None)
+            elif op == "train_step":
+                holder.train_step(None, None, n_micro=cmd[1])
+            elif op == "opt_step":
+                holder.optimizer.step()
+            elif op == "opt_zero":
+                holder.optimizer.zero_grad()
+            elif op == "save":
+                from tensorlink_amd.models.loader import \
+                    save_stage_to_safetensors
+                save_stage_to_safetensors(runner.stage, cmd[1],
+                                          self._rank)
+
     def _start_cluster(self, model, training, world, init, ckpt_dir, seed,
                        lr):
         import torch.multiprocessing as mp
