@@ -153,3 +153,24 @@ def test_paged_kv_cache_matches_contiguous():
     # table is genuinely shuffled (exercises indirection)
     tab = r2.kv_cache.table.flatten().tolist()
     assert tab != sorted(tab)
+
+
+def test_sampled_generation_deterministic_with_seed():
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    plan = plan_for_world("tiny", 1)
+    r1 = PipelineRunner(plan, 0, 1, device=torch.device("cpu"))
+    r2 = PipelineRunner(plan, 0, 1, device=torch.device("cpu"))
+    torch.manual_seed(0)
+    ids = torch.randint(0, 1024, (2, 8))
+    sp = SamplingParams(temperature=0.8, top_p=0.9, top_k=50,
+                        max_new_tokens=8, seed=42)
+    o1 = r1.generate(ids, sp)
+    o2 = r2.generate(ids, sp)
+    assert torch.equal(o1, o2)
+
+
+def test_planner_rejects_more_stages_than_layers():
+    import pytest as _pytest
+    from tensorlink_amd.parallel.planner import AssignmentError
+    with _pytest.raises(AssignmentError):
+        plan_for_world("tiny", 8)   # tiny has 4 layers
