@@ -209,6 +209,7 @@ class PipelineRunner:
             use_graph = (self.device.type == "cuda" and sp.temperature <= 0
                          and T > 4 and not _os.environ.get("TL_NO_GRAPH")
                          and not self._no_graph
+                         and self.kv_mode == "contiguous"
                          and on_token is None and eos is None)
             if use_graph:
                 # hipGraph-captured decode: the whole per-token step (all
