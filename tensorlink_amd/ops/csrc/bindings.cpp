@@ -170,6 +170,8 @@ Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
   int bt_stride = 0;
   if (block_table.has_value()) {
     CHECK_IN(block_table.value(), torch::kInt);
+    TORCH_CHECK(G <= 16, "paged decode requires GQA group <= 16");
+    TORCH_CHECK(k_cache.size(2) == 128, "paged pool PAGE must be 128");
     bt = block_table->data_ptr();
     bt_stride = block_table->size(1);
   }
