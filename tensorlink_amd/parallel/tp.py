@@ -1,0 +1,151 @@
+"""Tensor parallelism (opt-in, beyond reference parity).
+
+The reference has no intra-layer sharding ("TP: NO").
+This module adds Megatron-style TP for inference: attention heads and MLP
+intermediate are sharded column-parallel across ranks; the o/down
+projections are row-parallel and their outputs are summed with ONE RCCL
+all-reduce each over xGMI — so a pure-TP decode step needs exactly two
+all-reduces per layer and no point-to-point traffic (every rank holds the
+full residual stream and computes identical logits after the reductions).
+
+Design choice: sharding happens by *construction* — a rank builds the
+model from a head-sharded config (the layer code in models/dense.py is
+untouched), the o/down linears are swapped for all-reducing variants, and
+weights are sliced from a full-model state so TP output is comparable to
+the single-rank reference. KV caches hold only the local heads, so cache
+memory also divides by TP.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from tensorlink_amd.models.configs import ModelConfig, get_config
+from tensorlink_amd.models.dense import TLLinear, build_full_model
+from tensorlink_amd.models.loader import init_random_stage
+from tensorlink_amd.parallel.comm import device_for_rank
+from tensorlink_amd.parallel.pipeline import SamplingParams
+
+
+class AllReduceLinear(TLLinear):
+    """Row-parallel output projection: local GEMM then sum across the TP
+    group (inference path)."""
+
+    def forward(self, x):
+        y = super().forward(x)
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            y = y.contiguous()
+            dist.all_reduce(y)
+        return y
+
+
+def local_config(config: ModelConfig, tp: int) -> ModelConfig:
+    assert config.num_attention_heads % tp == 0, "heads must divide tp"
+    assert config.num_key_value_heads % tp == 0, "kv heads must divide tp"
+    assert config.intermediate_size % tp == 0
+    return dataclasses.replace(
+        config,
+        num_attention_heads=config.num_attention_heads // tp,
+        num_key_value_heads=config.num_key_value_heads // tp,
+        intermediate_size=config.intermediate_size // tp,
+        head_dim=config.head_dim)
+
+
+def shard_state(full_state: dict, config: ModelConfig, tp_rank: int,
+                tp: int) -> dict:
+    """Slice a full-model state dict into this rank's TP shard."""
+    q, kv, i = config.q_size, config.kv_size, config.intermediate_size
+    ql, kvl, il = q // tp, kv // tp, i // tp
+    out = {}
+    for name, w in full_state.items():
+        if "qkv_proj" in name:
+            qs = w[tp_rank * ql:(tp_rank + 1) * ql]
+            ks = w[q + tp_rank * kvl:q + (tp_rank + 1) * kvl]
+            vs = w[q + kv + tp_rank * kvl:q + kv + (tp_rank + 1) * kvl]
+            out[name] = torch.cat([qs, ks, vs], dim=0)
+        elif "gate_up_proj" in name:
+            gs = w[tp_rank * il:(tp_rank + 1) * il]
+            us = w[i + tp_rank * il:i + (tp_rank + 1) * il]
+            out[name] = torch.cat([gs, us], dim=0)
+        elif "o_proj.weight" in name:
+            out[name] = w[:, tp_rank * ql:(tp_rank + 1) * ql]
+        elif "down_proj.weight" in name:
+            out[name] = w[:, tp_rank * il:(tp_rank + 1) * il]
+        else:
+            out[name] = w          # norms, embeddings, head: replicated
+    return {k: v.contiguous() for k, v in out.items()}
+
+
+def build_tp_model(config_or_name, tp_rank: int, tp: int, device=None,
+                   dtype=None, seed: int = 0):
+    """Full model (PP=1) sharded TP-ways. Weights come from slicing the
+    SAME seeded full-model init every rank, so a TP group reproduces the
+    single-rank reference."""
+    config = (config_or_name if isinstance(config_or_name, ModelConfig)
+              else get_config(config_or_name))
+    device = device if device is not None else device_for_rank()
+    dtype = dtype or (torch.bfloat16 if device.type == "cuda"
+                      else torch.float32)
+
+    full = build_full_model(config)
+    init_random_stage(full, device="cpu", dtype=dtype, seed=seed)
+    shards = shard_state(full.state_dict(), config, tp_rank, tp)
+    del full
+
+    stage = build_full_model(local_config(config, tp))
+    # swap row-parallel outputs for all-reducing variants
+    for layer in stage.layers:
+        for holder, name in ((layer.self_attn, "o_proj"),
+                             (layer.mlp, "down_proj")):
+            old = getattr(holder, name)
+            new = AllReduceLinear(old.in_features, old.out_features,
+                                  bias=old.bias is not None)
+            setattr(holder, name, new)
+    stage.load_state_dict(shards)
+    stage.to(device=device, dtype=dtype)
+    stage.eval()
+    return stage
+
+
+class TPRunner:
+    """Pure-TP generation (SPMD across the TP group): every rank runs the
+    same decode loop on its head shard; logits agree after the per-layer
+    all-reduces, so sampling is local and no tokens travel."""
+
+    def __init__(self, model, rank: int, tp: int, device=None, seed: int = 0):
+        self.rank, self.tp = rank, tp
+        self.device = device if device is not None else device_for_rank()
+        self.stage = build_tp_model(model, rank, tp, device=self.device,
+                                    seed=seed)
+        self.config = self.stage.config
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor,
+                 sampling: Optional[SamplingParams] = None) -> torch.Tensor:
+        sp = sampling or SamplingParams()
+        B, S = input_ids.shape
+        ids = input_ids.to(self.device)
+        cache = self.stage.make_kv_cache(B, S + sp.max_new_tokens,
+                                         self.device)
+        pos = torch.arange(S, device=self.device,
+                           dtype=torch.int32).unsqueeze(0).expand(B, -1)
+        hidden = self.stage(ids, pos.contiguous(), kv_cache=cache,
+                            return_logits=False)
+        logits = self.stage.head(hidden[:, -1:]).squeeze(1)
+        out = torch.empty(B, sp.max_new_tokens, device=self.device,
+                          dtype=torch.int64)
+        cur = logits.argmax(-1)
+        out[:, 0] = cur
+        positions = torch.full((B,), S, device=self.device,
+                               dtype=torch.int32)
+        for t in range(1, sp.max_new_tokens):
+            lg = self.stage(cur.unsqueeze(1), positions.unsqueeze(1),
+                            kv_cache=cache).squeeze(1)
+            cur = lg.argmax(-1)
+            out[:, t] = cur
+            positions += 1
+        return out

@@ -13,6 +13,12 @@ import torch.multiprocessing as mp
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _port(offset: int) -> int:
+    """Distinct, pid-varying rendezvous ports (avoids TIME_WAIT rebind
+    flakes when suites re-run quickly)."""
+    return 20000 + (os.getpid() * 13 + offset * 101) % 20000
+
+
 def _gen_worker(rank, world, port, q):
     sys.path.insert(0, REPO)
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
@@ -79,7 +85,7 @@ def _run(worker, world, port, timeout=240, retries=1):
 
 @pytest.mark.timeout(300)
 def test_pp2_generate_matches_reference():
-    out = _run(_gen_worker, 2, 29621)
+    out = _run(_gen_worker, 2, _port(1))
     assert out.shape == (4, 6)
 
     # single-process reference with identical per-stage seeds
@@ -105,7 +111,7 @@ def test_pp2_generate_matches_reference():
 
 @pytest.mark.timeout(300)
 def test_pp2_1f1b_training_reduces_loss():
-    losses = _run(_train_worker, 2, 29622)
+    losses = _run(_train_worker, 2, _port(2))
     assert len(losses) == 4
     assert losses[-1] < losses[0]
 
@@ -134,7 +140,7 @@ def _dp_worker(rank, world, port, q):
 def test_dp2_training_replicas_stay_synced():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=_dp_worker, args=(r, 2, 29633, q))
+    ps = [ctx.Process(target=_dp_worker, args=(r, 2, _port(3), q))
           for r in range(2)]
     for p in ps:
         p.start()
@@ -187,7 +193,7 @@ def test_engine_serving_over_pp2():
     the reference's validator+worker serving topology on gloo loopback."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=_engine_worker, args=(r, 2, 29641, q))
+    ps = [ctx.Process(target=_engine_worker, args=(r, 2, _port(4), q))
           for r in range(2)]
     for p in ps:
         p.start()
@@ -227,7 +233,7 @@ def _gen4_worker(rank, world, port, q):
 def test_pp4_ring_decode_matches_reference():
     """4-stage ring-pipelined decode (the SCALE-bench shape) vs a
     single-process stage-chained reference."""
-    out = _run(_gen4_worker, 4, 29651, timeout=360)
+    out = _run(_gen4_worker, 4, _port(5), timeout=360)
     assert out.shape == (8, 5)
     from tensorlink_amd.models.dense import build_stage
     from tensorlink_amd.models.loader import init_random_stage
@@ -249,3 +255,60 @@ def test_pp4_ring_decode_matches_reference():
         logits = stages[-1](h, pos)
         cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(cur[:, 12:], out)
+
+
+def _tp_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.pipeline import SamplingParams
+    from tensorlink_amd.parallel.tp import TPRunner
+    init_distributed(backend="gloo")
+    r = TPRunner("tiny", rank, world, device=torch.device("cpu"), seed=7)
+    torch.manual_seed(31)
+    ids = torch.randint(0, 1024, (3, 14))
+    out = r.generate(ids, SamplingParams(max_new_tokens=6))
+    q.put((rank, out))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp2_matches_single_rank_reference():
+    """Tensor parallelism: 2 head-sharded ranks with per-layer all-reduce
+    reproduce the single-rank greedy output."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_tp_worker, args=(r, 2, _port(6), q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    outs = {}
+    try:
+        for _ in range(2):
+            rank, out = q.get(timeout=240)
+            outs[rank] = out
+    finally:
+        for p in ps:
+            p.join(30)
+            if p.is_alive():
+                p.terminate()
+    # both ranks computed the same tokens (logits agree post all-reduce)
+    assert torch.equal(outs[0], outs[1])
+
+    # single-rank reference from the same seeded full init
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.configs import get_config
+    m = build_full_model(get_config("tiny"))
+    init_random_stage(m, device="cpu", dtype=torch.float32, seed=7)
+    torch.manual_seed(31)
+    ids = torch.randint(0, 1024, (3, 14))
+    cur = ids
+    for _ in range(6):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).expand(3, -1).contiguous()
+        logits = m(cur, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 14:], outs[0])
