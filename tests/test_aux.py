@@ -153,3 +153,36 @@ def test_cli_plan(capsys):
         sys.argv = argv
     out = capsys.readouterr().out
     assert "PP=4" in out and "training=True" in out
+
+
+def test_tp_shard_state_partitions_exactly():
+    """TP sharding: the two ranks' shards reassemble to the full weights."""
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.parallel.tp import local_config, shard_state
+    cfg = get_config("tiny")
+    m = build_full_model(cfg)
+    init_random_stage(m, dtype=torch.float32, seed=3)
+    full = m.state_dict()
+    s0 = shard_state(full, cfg, 0, 2)
+    s1 = shard_state(full, cfg, 1, 2)
+    lc = local_config(cfg, 2)
+    assert lc.num_attention_heads == 2 and lc.intermediate_size == 256
+    name = "layers.0.self_attn.qkv_proj.weight"
+    q, kv = cfg.q_size, cfg.kv_size
+    # reassemble q/k/v blocks
+    re_q = torch.cat([s0[name][:q // 2], s1[name][:q // 2]], 0)
+    torch.testing.assert_close(re_q, full[name][:q])
+    re_k = torch.cat([s0[name][q // 2:q // 2 + kv // 2],
+                      s1[name][q // 2:q // 2 + kv // 2]], 0)
+    torch.testing.assert_close(re_k, full[name][q:q + kv])
+    name = "layers.0.mlp.down_proj.weight"
+    re_d = torch.cat([s0[name], s1[name]], 1)
+    torch.testing.assert_close(re_d, full[name])
+    # replicated tensors untouched
+    torch.testing.assert_close(s0["norm"], full["norm"])
+
+
+def test_tunableop_noop_on_cpu():
+    from tensorlink_amd.utils.tunable import setup_tunableop
+    assert setup_tunableop() is False or torch.cuda.is_available()
