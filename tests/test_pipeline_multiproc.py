@@ -312,3 +312,31 @@ def test_tp2_matches_single_rank_reference():
         logits = m(cur, pos)
         cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(cur[:, 14:], outs[0])
+
+
+def _train4_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    init_distributed(backend="gloo")
+    plan = plan_for_world("tiny", world, training=True)
+    tr = PipelineTrainer(plan, rank, world, device=torch.device("cpu"),
+                         lr=1e-3)
+    torch.manual_seed(13)
+    ids = torch.randint(0, 1024, (8, 16)) if rank == 0 else None
+    losses = [tr.train_step(ids, ids, n_micro=8) for _ in range(3)]
+    if rank == 0:
+        q.put(losses)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(420)
+def test_pp4_1f1b_training(port_offset=7):
+    """BASELINE config #3 shape: 1F1B training across 4 pipeline stages."""
+    losses = _run(_train4_worker, 4, _port(7), timeout=360)
+    assert losses[-1] < losses[0], losses

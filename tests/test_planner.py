@@ -80,3 +80,26 @@ def test_param_count_close_to_nominal():
     assert 6.5e10 < n70 < 7.5e10
     nmoe = get_config("mistralai/Mixtral-8x7B-v0.1").param_count()
     assert 4.4e10 < nmoe < 5.0e10
+
+
+def test_all_baseline_configs_plan():
+    """Every BASELINE.json config has a constructible plan in this build."""
+    # 1: GPT-2-small, 2 CPU workers (plumbing)
+    p1 = plan_for_world("gpt2-small", 2)
+    assert p1.num_stages == 2
+    # 2: Qwen2.5-7B PP=2
+    p2 = plan_for_world("Qwen/Qwen2.5-7B-Instruct", 2, batch_size=8,
+                        seq_len=4096)
+    assert p2.num_stages == 2
+    # 3: Qwen3-8B training PP=4
+    p3 = plan_for_world("Qwen/Qwen3-8B", 4, training=True, batch_size=8,
+                        seq_len=2048)
+    assert p3.training and p3.num_stages == 4
+    # 4: Llama-3-70B PP=8
+    p4 = plan_for_world("meta-llama/Llama-3-70B", 8, batch_size=8,
+                        seq_len=8192)
+    assert p4.num_stages == 8
+    # 5: Mixtral 8x7B (experts grouped per stage)
+    p5 = plan_for_world("mistralai/Mixtral-8x7B-v0.1", 2, batch_size=8,
+                        seq_len=4096)
+    assert p5.config.is_moe and p5.num_stages == 2
