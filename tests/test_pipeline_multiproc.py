@@ -340,3 +340,58 @@ def test_pp4_1f1b_training(port_offset=7):
     """BASELINE config #3 shape: 1F1B training across 4 pipeline stages."""
     losses = _run(_train4_worker, 4, _port(7), timeout=360)
     assert losses[-1] < losses[0], losses
+
+
+def _ep_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.ep import EPRunner
+    from tensorlink_amd.parallel.pipeline import SamplingParams
+    init_distributed(backend="gloo")
+    r = EPRunner("tiny-moe", rank, world, device=torch.device("cpu"), seed=9)
+    torch.manual_seed(41)
+    ids = torch.randint(0, 1024, (2, 10))
+    out = r.generate(ids, SamplingParams(max_new_tokens=5))
+    q.put((rank, out))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ep2_matches_single_rank_reference():
+    """Expert parallelism: 2 ranks each holding half the experts (partial
+    sums all-reduced) reproduce the single-rank MoE output."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_ep_worker, args=(r, 2, _port(8), q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    outs = {}
+    try:
+        for _ in range(2):
+            rank, out = q.get(timeout=240)
+            outs[rank] = out
+    finally:
+        for p in ps:
+            p.join(30)
+            if p.is_alive():
+                p.terminate()
+    assert torch.equal(outs[0], outs[1])
+
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.configs import get_config
+    m = build_full_model(get_config("tiny-moe"))
+    init_random_stage(m, device="cpu", dtype=torch.float32, seed=9)
+    torch.manual_seed(41)
+    ids = torch.randint(0, 1024, (2, 10))
+    cur = ids
+    for _ in range(5):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).expand(2, -1).contiguous()
+        logits = m(cur, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 10:], outs[0])
