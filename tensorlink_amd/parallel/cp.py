@@ -1,0 +1,155 @@
+"""Context (sequence) parallelism for prefill — ring attention (opt-in,
+beyond reference parity; SURVEY.md §2.2 lists SP/CP as a stretch goal).
+
+The sequence is sharded across ranks (weights replicated); each rank
+computes attention for its query chunk while K/V chunks rotate around the
+ring over xGMI P2P. Causality prunes the ring: chunk j's K/V only visits
+ranks i >= j, and partial results merge with the standard online-softmax
+(m, l, o) combination, so the result equals single-rank attention.
+
+Scope: inference prefill (logits). Decode after a CP prefill and the
+training backward are roadmap items; the kernel-grade version would fuse
+the merge into prefill_attn.hip.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from tensorlink_amd import ops
+from tensorlink_amd.models.configs import ModelConfig, get_config
+from tensorlink_amd.models.dense import build_full_model
+from tensorlink_amd.models.loader import init_random_stage
+from tensorlink_amd.parallel.comm import device_for_rank
+
+
+def _partial_attn(q, k, v, scale, causal_diag):
+    """Unnormalized attention partial of q [B,Sq,H,D] against one K/V
+    chunk. Returns (o_unnorm [B,Sq,H,D], m [B,H,Sq], l [B,H,Sq]).
+    causal_diag=True applies the in-chunk causal mask (q and k are the
+    same chunk); False means the whole chunk precedes q (no mask)."""
+    B, Sq, Hq, D = q.shape
+    Hkv = k.shape[2]
+    rep = Hq // Hkv
+    qf = q.float().permute(0, 2, 1, 3)
+    kf = k.float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
+    vf = v.float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal_diag:
+        Sk = k.shape[1]
+        mask = torch.triu(torch.ones(Sq, Sk, dtype=torch.bool,
+                                     device=q.device), 1)
+        scores = scores.masked_fill(mask, float("-inf"))
+    m = scores.amax(-1)                                   # [B,H,Sq]
+    p = torch.exp(scores - m.unsqueeze(-1))
+    p = torch.nan_to_num(p, nan=0.0)                      # fully-masked rows
+    l = p.sum(-1)
+    o = torch.matmul(p, vf)                               # unnormalized
+    return o.permute(0, 2, 1, 3), m, l
+
+
+def _merge(acc, nxt):
+    """Online-softmax merge of two (o, m, l) partials."""
+    if acc is None:
+        return nxt
+    o1, m1, l1 = acc
+    o2, m2, l2 = nxt
+    m = torch.maximum(m1, m2)
+    a1 = torch.exp(m1 - m)
+    a2 = torch.exp(m2 - m)
+    o = o1 * a1.permute(0, 2, 1).unsqueeze(-1) + \
+        o2 * a2.permute(0, 2, 1).unsqueeze(-1)
+    l = l1 * a1 + l2 * a2
+    return o, m, l
+
+
+class CPRunner:
+    """Sequence-sharded prefill across a CP group (weights replicated,
+    same seeded init on every rank)."""
+
+    def __init__(self, model, rank: int, cp: int, device=None, seed: int = 0):
+        self.rank, self.cp = rank, cp
+        self.device = device if device is not None else device_for_rank()
+        dtype = (torch.bfloat16 if self.device.type == "cuda"
+                 else torch.float32)
+        config = (model if isinstance(model, ModelConfig)
+                  else get_config(model))
+        self.stage = build_full_model(config)
+        init_random_stage(self.stage, device=self.device, dtype=dtype,
+                          seed=seed)
+        self.stage.eval()
+        self.config = config
+        self.dtype = dtype
+
+    def _ring_attention(self, attn, x, positions):
+        """One attention layer with K/V rotating around the CP ring."""
+        B, Sq, H = x.shape
+        qkv = attn.qkv_proj(x)
+        q, k, v = attn._split_qkv(qkv, B, Sq)
+        q = q.contiguous()
+        k = k.contiguous()
+        v = v.contiguous()
+        flat_pos = positions.reshape(-1)
+        ops.apply_rope_(q.view(B * Sq, attn.n_heads, -1),
+                        k.view(B * Sq, attn.n_kv, -1), flat_pos,
+                        attn.inv_freq)
+        scale = attn.scale
+
+        acc = None
+        cur_k, cur_v = k, v
+        for r in range(self.cp):
+            src_chunk = (self.rank - r) % self.cp
+            if src_chunk == self.rank:
+                part = _partial_attn(q, cur_k, cur_v, scale,
+                                     causal_diag=True)
+                acc = _merge(acc, part)
+            elif src_chunk < self.rank:
+                part = _partial_attn(q, cur_k, cur_v, scale,
+                                     causal_diag=False)
+                acc = _merge(acc, part)
+            # rotate K/V to the next rank (skip after the last step)
+            if r < self.cp - 1 and self.cp > 1:
+                nk = torch.empty_like(cur_k)
+                nv = torch.empty_like(cur_v)
+                nxt = (self.rank + 1) % self.cp
+                prv = (self.rank - 1) % self.cp
+                reqs = [dist.isend(cur_k.contiguous(), nxt),
+                        dist.isend(cur_v.contiguous(), nxt)]
+                dist.recv(nk, prv)
+                dist.recv(nv, prv)
+                for w in reqs:
+                    w.wait()
+                cur_k, cur_v = nk, nv
+        o, m, l = acc
+        out = (o / l.permute(0, 2, 1).unsqueeze(-1).clamp(min=1e-30))
+        return attn.o_proj(out.to(x.dtype).reshape(B, Sq, -1))
+
+    @torch.no_grad()
+    def forward_logits(self, input_ids: torch.Tensor) -> torch.Tensor:
+        """Prefill logits for this rank's sequence chunk. input_ids is the
+        FULL [B, S] prompt on every rank (S divisible by cp)."""
+        B, S = input_ids.shape
+        assert S % self.cp == 0, "S must divide cp"
+        sc = S // self.cp
+        ids = input_ids.to(self.device)[:, self.rank * sc:(self.rank + 1) * sc]
+        positions = (torch.arange(sc, device=self.device,
+                                  dtype=torch.int32) + self.rank * sc) \
+            .unsqueeze(0).expand(B, -1).contiguous()
+
+        stage = self.stage
+        hidden = stage.embed(ids)
+        eps = self.config.rms_norm_eps
+        for layer in stage.layers:
+            h = ops.rmsnorm(hidden, layer.input_layernorm.to(hidden.dtype),
+                            eps)
+            hidden = hidden + self._ring_attention(layer.self_attn, h,
+                                                   positions)
+            h = ops.rmsnorm(hidden,
+                            layer.post_attention_layernorm.to(hidden.dtype),
+                            eps)
+            hidden = hidden + layer.mlp(h)
+        return stage.head(hidden)       # [B, sc, V] — this rank's chunk

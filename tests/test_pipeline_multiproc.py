@@ -395,3 +395,54 @@ def test_ep2_matches_single_rank_reference():
         logits = m(cur, pos)
         cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(cur[:, 10:], outs[0])
+
+
+def _cp_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.cp import CPRunner
+    init_distributed(backend="gloo")
+    r = CPRunner("tiny", rank, world, device=torch.device("cpu"), seed=17)
+    torch.manual_seed(51)
+    ids = torch.randint(0, 1024, (2, 24))
+    logits = r.forward_logits(ids)
+    q.put((rank, logits))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_cp2_ring_attention_matches_single_rank():
+    """Context parallelism: 2 sequence-sharded ranks with ring attention
+    reproduce single-rank prefill logits."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_cp_worker, args=(r, 2, _port(9), q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    outs = {}
+    try:
+        for _ in range(2):
+            rank, lg = q.get(timeout=240)
+            outs[rank] = lg
+    finally:
+        for p in ps:
+            p.join(30)
+            if p.is_alive():
+                p.terminate()
+    cp_logits = torch.cat([outs[0], outs[1]], dim=1)     # [B, S, V]
+
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.configs import get_config
+    m = build_full_model(get_config("tiny"))
+    init_random_stage(m, device="cpu", dtype=torch.float32, seed=17)
+    torch.manual_seed(51)
+    ids = torch.randint(0, 1024, (2, 24))
+    pos = torch.arange(24).unsqueeze(0).expand(2, -1).contiguous()
+    ref = m(ids, pos)
+    torch.testing.assert_close(cp_logits, ref, atol=1e-4, rtol=1e-4)
