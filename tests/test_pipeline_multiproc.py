@@ -83,6 +83,33 @@ def _run(worker, world, port, timeout=240, retries=1):
     raise last_exc
 
 
+def _run_collect(worker, world, port, n_results=None, timeout=240,
+                 retries=1):
+    """Like _run but collects one result per expected sender, with the
+    same transient-flake retry."""
+    n_results = world if n_results is None else n_results
+    last_exc = None
+    for attempt in range(retries + 1):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        ps = [ctx.Process(target=worker,
+                          args=(r, world, port + attempt * 7, q))
+              for r in range(world)]
+        for p in ps:
+            p.start()
+        try:
+            results = [q.get(timeout=timeout) for _ in range(n_results)]
+            return results
+        except Exception as e:
+            last_exc = e
+        finally:
+            for p in ps:
+                p.join(30)
+                if p.is_alive():
+                    p.terminate()
+    raise last_exc
+
+
 @pytest.mark.timeout(300)
 def test_pp2_generate_matches_reference():
     out = _run(_gen_worker, 2, _port(1))
@@ -138,22 +165,8 @@ def _dp_worker(rank, world, port, q):
 
 @pytest.mark.timeout(300)
 def test_dp2_training_replicas_stay_synced():
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    ps = [ctx.Process(target=_dp_worker, args=(r, 2, _port(3), q))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    results = {}
-    try:
-        for _ in range(2):
-            rank, losses, h = q.get(timeout=240)
-            results[rank] = (losses, h)
-    finally:
-        for p in ps:
-            p.join(30)
-            if p.is_alive():
-                p.terminate()
+    results = {rank: (losses, h) for rank, losses, h in
+               _run_collect(_dp_worker, 2, _port(3))}
     l0, h0 = results[0]
     l1, h1 = results[1]
     assert l0 == l1, "losses must agree across replicas"
@@ -191,19 +204,8 @@ def _engine_worker(rank, world, port, q):
 def test_engine_serving_over_pp2():
     """Full serving stack (engine + worker_loop) across 2 pipeline ranks —
     the reference's validator+worker serving topology on gloo loopback."""
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    ps = [ctx.Process(target=_engine_worker, args=(r, 2, _port(4), q))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    try:
-        resp, chunks = q.get(timeout=240)
-    finally:
-        for p in ps:
-            p.join(30)
-            if p.is_alive():
-                p.terminate()
+    resp, chunks = _run_collect(_engine_worker, 2, _port(4),
+                                n_results=1)[0]
     assert "response" in resp and resp["model"] == "tiny"
     assert chunks[-1] == "data: [DONE]\n\n"
     assert len(chunks) >= 2
@@ -279,22 +281,7 @@ def _tp_worker(rank, world, port, q):
 def test_tp2_matches_single_rank_reference():
     """Tensor parallelism: 2 head-sharded ranks with per-layer all-reduce
     reproduce the single-rank greedy output."""
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    ps = [ctx.Process(target=_tp_worker, args=(r, 2, _port(6), q))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    outs = {}
-    try:
-        for _ in range(2):
-            rank, out = q.get(timeout=240)
-            outs[rank] = out
-    finally:
-        for p in ps:
-            p.join(30)
-            if p.is_alive():
-                p.terminate()
+    outs = dict(_run_collect(_tp_worker, 2, _port(6)))
     # both ranks computed the same tokens (logits agree post all-reduce)
     assert torch.equal(outs[0], outs[1])
 
@@ -364,22 +351,7 @@ def _ep_worker(rank, world, port, q):
 def test_ep2_matches_single_rank_reference():
     """Expert parallelism: 2 ranks each holding half the experts (partial
     sums all-reduced) reproduce the single-rank MoE output."""
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    ps = [ctx.Process(target=_ep_worker, args=(r, 2, _port(8), q))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    outs = {}
-    try:
-        for _ in range(2):
-            rank, out = q.get(timeout=240)
-            outs[rank] = out
-    finally:
-        for p in ps:
-            p.join(30)
-            if p.is_alive():
-                p.terminate()
+    outs = dict(_run_collect(_ep_worker, 2, _port(8)))
     assert torch.equal(outs[0], outs[1])
 
     from tensorlink_amd.models.dense import build_full_model
@@ -418,22 +390,7 @@ def _cp_worker(rank, world, port, q):
 def test_cp2_ring_attention_matches_single_rank():
     """Context parallelism: 2 sequence-sharded ranks with ring attention
     reproduce single-rank prefill logits."""
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    ps = [ctx.Process(target=_cp_worker, args=(r, 2, _port(9), q))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    outs = {}
-    try:
-        for _ in range(2):
-            rank, lg = q.get(timeout=240)
-            outs[rank] = lg
-    finally:
-        for p in ps:
-            p.join(30)
-            if p.is_alive():
-                p.terminate()
+    outs = dict(_run_collect(_cp_worker, 2, _port(9)))
     cp_logits = torch.cat([outs[0], outs[1]], dim=1)     # [B, S, V]
 
     from tensorlink_amd.models.dense import build_full_model
