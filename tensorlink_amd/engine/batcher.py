@@ -24,7 +24,8 @@ from typing import List, Optional
 import torch
 
 from tensorlink_amd import ops
-from tensorlink_amd.models.paged import PAGE, PagedKVCache
+from tensorlink_amd.models.paged import (PAGE, DynamicPagedKVCache,
+                                         PagedKVCache)
 
 
 @dataclass
@@ -80,15 +81,21 @@ class _SlotView:
 
 
 class ContinuousBatcher:
-    def __init__(self, runner, max_slots: int = 16, max_ctx: int = 2048):
+    def __init__(self, runner, max_slots: int = 16, max_ctx: int = 2048,
+                 pool_pages: Optional[int] = None):
         assert runner.world == 1, "continuous batching is single-rank"
         self.runner = runner
         self.stage = runner.stage
         self.device = runner.device
         self.max_ctx = max_ctx
-        self.cache = self.stage.make_kv_cache(max_slots, max_ctx,
-                                              runner.device, runner.dtype,
-                                              kv_mode="paged")
+        # dynamic page pool: slots lease pages as sequences grow and
+        # release them on retirement; pool_pages < slots*max_ctx/PAGE
+        # oversubscribes memory (admission blocks when exhausted)
+        if pool_pages is None:
+            pool_pages = max_slots * ((max_ctx + PAGE - 1) // PAGE)
+        self.cache = DynamicPagedKVCache(
+            self.stage.num_layers, max_slots, pool_pages, max_ctx,
+            self.stage.config, runner.device, runner.dtype)
         self.max_slots = max_slots
         self.slots: List[Optional[Request]] = [None] * max_slots
         self.slot_len = [0] * max_slots       # tokens resident in cache
@@ -132,6 +139,7 @@ class ContinuousBatcher:
     def _prefill(self, slot: int, req: Request):
         ids = req.input_ids.to(self.device).unsqueeze(0)
         S = ids.shape[1]
+        self.cache.ensure(slot, S + 1)
         pos = torch.arange(S, device=self.device,
                            dtype=torch.int32).unsqueeze(0)
         view = _SlotView(self.cache, [slot],
@@ -164,6 +172,7 @@ class ContinuousBatcher:
             req.tokens.put(None)
             req.done.set()
             self.slots[slot] = None
+            self.cache.release_slot(slot)
 
     @torch.no_grad()
     def _decode_step(self):
@@ -171,6 +180,8 @@ class ContinuousBatcher:
         if not active:
             return False
         reqs = [self.slots[i] for i in active]
+        for i in active:
+            self.cache.ensure(i, self.slot_len[i] + 2)
         toks = torch.tensor([self.slot_last[i] for i in active],
                             device=self.device, dtype=torch.int64)
         lens = torch.tensor([self.slot_len[i] for i in active],
@@ -195,6 +206,10 @@ class ContinuousBatcher:
                 slot = self._free_slot()
                 if slot is None:
                     break
+                # admission control: need pages for prompt + first tokens
+                need = (self._queue.queue[0].input_ids.numel() + PAGE) // PAGE + 1
+                if self.cache.allocator.n_free < need:
+                    break
                 req = self._queue.get_nowait()
                 try:
                     self.slots[slot] = req
@@ -205,6 +220,7 @@ class ContinuousBatcher:
                     req.tokens.put(None)
                     req.done.set()
                     self.slots[slot] = None
+                    self.cache.release_slot(slot)
             busy = self._decode_step()
             if not busy and not admitted:
                 self._wake.wait(0.05)

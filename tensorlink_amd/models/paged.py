@@ -66,6 +66,69 @@ class PagedKVCache:
         return k[:, :S].contiguous(), v[:, :S].contiguous()
 
 
+class PageAllocator:
+    """Free-list allocator over a shared page pool: slots lease pages on
+    demand and release them on retirement, so long-lived short sequences
+    do not reserve worst-case context (the memory-elasticity half of
+    paged attention; per-slot static reservation is the fallback)."""
+
+    def __init__(self, n_pages: int):
+        self.free = list(range(n_pages - 1, -1, -1))
+        self.n_pages = n_pages
+
+    def alloc(self, n: int):
+        if n > len(self.free):
+            raise RuntimeError(
+                f"KV page pool exhausted (need {n}, free {len(self.free)})")
+        return [self.free.pop() for _ in range(n)]
+
+    def release(self, pages):
+        self.free.extend(int(p) for p in pages)
+
+    @property
+    def n_free(self):
+        return len(self.free)
+
+
+class DynamicPagedKVCache(PagedKVCache):
+    """Paged cache whose slot tables start EMPTY; `ensure(slot, length)`
+    leases pages as the sequence grows and `release_slot` returns them."""
+
+    def __init__(self, n_layers: int, slots: int, pool_pages: int,
+                 max_seq: int, config, device, dtype=torch.bfloat16):
+        self.k = [torch.zeros(pool_pages, config.num_key_value_heads, PAGE,
+                              config.head_dim, device=device, dtype=dtype)
+                  for _ in range(n_layers)]
+        self.v = [torch.zeros_like(self.k[0]) for _ in range(n_layers)]
+        self.seq_lens = torch.zeros(slots, device=device, dtype=torch.int32)
+        self.pages_per_seq = (max_seq + PAGE - 1) // PAGE
+        self.max_seq = self.pages_per_seq * PAGE
+        self.batch = slots
+        self.allocator = PageAllocator(pool_pages)
+        self.table = torch.zeros(slots, self.pages_per_seq, device=device,
+                                 dtype=torch.int32)
+        self._slot_pages = [[] for _ in range(slots)]
+
+    def ensure(self, slot: int, length: int):
+        """Lease enough pages for `length` positions in `slot`."""
+        need = (length + PAGE - 1) // PAGE
+        have = len(self._slot_pages[slot])
+        if need > have:
+            new = self.allocator.alloc(need - have)
+            self._slot_pages[slot].extend(new)
+            self.table[slot, have:need] = torch.tensor(
+                new, dtype=torch.int32, device=self.table.device)
+
+    def release_slot(self, slot: int):
+        self.allocator.release(self._slot_pages[slot])
+        self._slot_pages[slot] = []
+        self.seq_lens[slot] = 0
+
+    def reset(self):
+        for s in range(self.batch):
+            self.release_slot(s)
+
+
 def kv_slice_paged(cache: PagedKVCache, s: int, e: int) -> PagedKVCache:
     view = object.__new__(PagedKVCache)
     view.k = cache.k                       # pools shared
