@@ -7,11 +7,16 @@ on the paged KV cache: every slot owns a page table, the decode kernels
 take per-sequence lengths, so a step runs over whichever slots are live.
 
 Flow per scheduler iteration:
-  1. admit queued requests into free slots (prefill writes their pages);
-  2. one decode step over all running slots (ragged lengths are native to
+  1. admit queued requests into free slots;
+  2. run ONE prefill chunk for one admitted-but-unfilled slot (chunked
+     prefill: long prompts are processed ``prefill_chunk`` tokens at a
+     time so running decodes are not stalled behind a long prompt —
+     the attention kernel takes a ``q_off`` so chunk N attends to all
+     keys written by chunks 0..N);
+  3. one decode step over all running slots (ragged lengths are native to
      decode_attn_mfma via seq_lens);
-  3. emitted tokens go to per-request queues (SSE streams read them);
-  4. slots retire on EOS / max_new_tokens.
+  4. emitted tokens go to per-request queues (SSE streams read them);
+  5. slots retire on EOS / max_new_tokens (pages return to the pool).
 """
 
 from __future__ import annotations
@@ -82,7 +87,8 @@ class _SlotView:
 
 class ContinuousBatcher:
     def __init__(self, runner, max_slots: int = 16, max_ctx: int = 2048,
-                 pool_pages: Optional[int] = None):
+                 pool_pages: Optional[int] = None,
+                 prefill_chunk: Optional[int] = None):
         assert runner.world == 1, "continuous batching is single-rank"
         self.runner = runner
         self.stage = runner.stage
@@ -97,10 +103,13 @@ class ContinuousBatcher:
             self.stage.num_layers, max_slots, pool_pages, max_ctx,
             self.stage.config, runner.device, runner.dtype)
         self.max_slots = max_slots
+        self.prefill_chunk = prefill_chunk    # None = whole prompt at once
         self.slots: List[Optional[Request]] = [None] * max_slots
         self.slot_len = [0] * max_slots       # tokens resident in cache
         self.slot_emitted = [0] * max_slots
         self.slot_last = [0] * max_slots      # last sampled token
+        self.slot_prompt: List[Optional[torch.Tensor]] = [None] * max_slots
+        self.slot_filled = [0] * max_slots    # prompt tokens prefilled
         self._queue: "queue.Queue[Request]" = queue.Queue()
         self._wake = threading.Event()
         self._stop = threading.Event()
@@ -135,20 +144,33 @@ class ContinuousBatcher:
                 return i
         return None
 
+    def _prefilling(self, slot: int) -> bool:
+        p = self.slot_prompt[slot]
+        return p is not None and self.slot_filled[slot] < p.numel()
+
     @torch.no_grad()
-    def _prefill(self, slot: int, req: Request):
-        ids = req.input_ids.to(self.device).unsqueeze(0)
-        S = ids.shape[1]
-        self.cache.ensure(slot, S + 1)
-        pos = torch.arange(S, device=self.device,
+    def _prefill_chunk(self, slot: int):
+        """Process the next chunk of slot's prompt; on the last chunk,
+        sample the first token and enter decode."""
+        req = self.slots[slot]
+        prompt = self.slot_prompt[slot]
+        S_total = prompt.numel()
+        start = self.slot_filled[slot]
+        end = (S_total if self.prefill_chunk is None
+               else min(start + self.prefill_chunk, S_total))
+        ids = prompt[start:end].to(self.device).unsqueeze(0)
+        self.cache.ensure(slot, end + 1)
+        pos = torch.arange(start, end, device=self.device,
                            dtype=torch.int32).unsqueeze(0)
-        view = _SlotView(self.cache, [slot],
-                         torch.zeros(1, device=self.device,
-                                     dtype=torch.int32))
+        lens = torch.tensor([start], device=self.device, dtype=torch.int32)
+        view = _SlotView(self.cache, [slot], lens)
         hidden = self.stage(ids, pos, kv_cache=view, return_logits=False)
+        self.slot_filled[slot] = end
+        if end < S_total:
+            return
         logits = self.stage.head(hidden[:, -1:]).squeeze(1)
         tok = int(self._sample(logits, [req])[0])
-        self.slot_len[slot] = S
+        self.slot_len[slot] = S_total
         self.slot_last[slot] = tok
         self.slot_emitted[slot] = 1
         req.tokens.put(tok)
@@ -172,11 +194,14 @@ class ContinuousBatcher:
             req.tokens.put(None)
             req.done.set()
             self.slots[slot] = None
+            self.slot_prompt[slot] = None
             self.cache.release_slot(slot)
 
     @torch.no_grad()
     def _decode_step(self):
-        active = [i for i, s in enumerate(self.slots) if s is not None]
+        active = [i for i, s in enumerate(self.slots)
+                  if s is not None and not self._prefilling(i)
+                  and self.slot_filled[i] > 0]
         if not active:
             return False
         reqs = [self.slots[i] for i in active]
@@ -211,17 +236,30 @@ class ContinuousBatcher:
                 if self.cache.allocator.n_free < need:
                     break
                 req = self._queue.get_nowait()
-                try:
-                    self.slots[slot] = req
-                    self._prefill(slot, req)
-                    admitted = True
-                except Exception as e:   # pragma: no cover
-                    req.error = str(e)
-                    req.tokens.put(None)
-                    req.done.set()
-                    self.slots[slot] = None
-                    self.cache.release_slot(slot)
+                self.slots[slot] = req
+                self.slot_prompt[slot] = req.input_ids
+                self.slot_filled[slot] = 0
+                self.slot_len[slot] = 0
+                admitted = True
+            # one prefill chunk per iteration: bounds how long running
+            # decodes wait behind a new long prompt
+            prefilled = False
+            for slot in range(self.max_slots):
+                if self.slots[slot] is not None and (
+                        self._prefilling(slot) or self.slot_filled[slot] == 0):
+                    req = self.slots[slot]
+                    try:
+                        self._prefill_chunk(slot)
+                        prefilled = True
+                    except Exception as e:   # pragma: no cover
+                        req.error = str(e)
+                        req.tokens.put(None)
+                        req.done.set()
+                        self.slots[slot] = None
+                        self.slot_prompt[slot] = None
+                        self.cache.release_slot(slot)
+                    break
             busy = self._decode_step()
-            if not busy and not admitted:
+            if not busy and not admitted and not prefilled:
                 self._wake.wait(0.05)
                 self._wake.clear()

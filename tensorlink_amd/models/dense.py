@@ -152,17 +152,21 @@ class Attention(nn.Module):
                 q, kv_cache.k[layer_idx], kv_cache.v[layer_idx],
                 kv_cache.seq_lens + 1, scale=self.scale, block_table=table)
         else:
-            # prefill with cache write (fresh prompt at positions 0..S-1):
-            # rotated k / raw v are read back from the cache
+            # prefill with cache write; rotated k / raw v are read back from
+            # the cache. off > 0 = chunked prefill continuing an existing
+            # sequence (the chunk's queries sit at global rows off..off+S-1
+            # and attend to all off+S cached keys).
+            off = int(kv_cache.seq_lens.max())
+            kv_len = off + S
             if table is None:
-                k_attn = kv_cache.k[layer_idx][:, :, :S].permute(
+                k_attn = kv_cache.k[layer_idx][:, :, :kv_len].permute(
                     0, 2, 1, 3).contiguous()
-                v_attn = kv_cache.v[layer_idx][:, :, :S].permute(
+                v_attn = kv_cache.v[layer_idx][:, :, :kv_len].permute(
                     0, 2, 1, 3).contiguous()
             else:
-                k_attn, v_attn = kv_cache.gather_contiguous(layer_idx, S)
+                k_attn, v_attn = kv_cache.gather_contiguous(layer_idx, kv_len)
             out = ops.attention_prefill(q, k_attn, v_attn, causal=True,
-                                        scale=self.scale)
+                                        scale=self.scale, q_off=off)
         return self.o_proj(out.reshape(B, S, -1))
 
 

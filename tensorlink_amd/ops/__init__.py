@@ -207,13 +207,16 @@ def swiglu_fused(gate_up: torch.Tensor) -> torch.Tensor:
 # Attention
 # ---------------------------------------------------------------------------
 def attention_prefill(q, k, v, causal: bool = True,
-                      scale: Optional[float] = None) -> torch.Tensor:
-    """q [B,S,Hq,D], k/v [B,S,Hkv,D] -> [B,S,Hq,D]. Inference path."""
+                      scale: Optional[float] = None,
+                      q_off: int = 0) -> torch.Tensor:
+    """q [B,Sq,Hq,D], k/v [B,Skv,Hkv,D] -> [B,Sq,Hq,D]. Inference path.
+    q_off: query row i is at global position q_off+i (chunked prefill)."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if not _on_gpu(q):
-        return ref.attention_prefill(q, k, v, causal, scale)
+        return ref.attention_prefill(q, k, v, causal, scale, q_off)
     return _require_ext().prefill_attn(q.contiguous(), k.contiguous(),
-                                       v.contiguous(), scale, causal)
+                                       v.contiguous(), scale, causal,
+                                       q_off)
 
 
 def attention_train(q, k, v, causal: bool = True,

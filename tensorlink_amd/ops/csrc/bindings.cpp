@@ -28,8 +28,8 @@ void tl_decode_attn(const void* q, const void* k_cache, const void* v_cache,
                     const void* seq_lens, void* out, int B, int Hq, int Hkv,
                     int Smax, int D, float scale, hipStream_t stream);
 void tl_prefill_attn(const void* q, const void* k, const void* v, void* out,
-                     int B, int S, int Hq, int Hkv, int D, float scale,
-                     int causal, hipStream_t stream);
+                     int B, int Sq, int Skv, int q_off, int Hq, int Hkv,
+                     int D, float scale, int causal, hipStream_t stream);
 void tl_rope_append(const void* qkv, void* q_out, void* k_cache,
                     void* v_cache, const void* positions,
                     const void* inv_freq, const void* block_table,
@@ -210,16 +210,20 @@ Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
   return out;
 }
 
-Tensor prefill_attn(Tensor q, Tensor k, Tensor v, double scale, bool causal) {
+Tensor prefill_attn(Tensor q, Tensor k, Tensor v, double scale, bool causal,
+                    int64_t q_off) {
   CHECK_IN(q, torch::kBFloat16);
   CHECK_IN(k, torch::kBFloat16);
   CHECK_IN(v, torch::kBFloat16);
-  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1);
   const int Hkv = k.size(2);
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  TORCH_CHECK(q_off + Sq <= Skv || !causal, "q_off + Sq must fit Skv");
   auto out = torch::empty_like(q);
-  tl_prefill_attn(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(), B,
-                  S, Hq, Hkv, D, (float)scale, causal ? 1 : 0, cur_stream());
+  tl_prefill_attn(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+                  B, Sq, Skv, (int)q_off, Hq, Hkv, D, (float)scale,
+                  causal ? 1 : 0, cur_stream());
   return out;
 }
 

@@ -29,11 +29,14 @@ constexpr int PPAD = 8;      // P row pad: stride 72 elems = 144 B
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 
+// Sq = query length, Skv = key/value length, q_off = global position of
+// query row 0 (chunked prefill: Skv = q_off + Sq history+chunk keys; the
+// plain full prefill is Sq == Skv, q_off == 0).
 template <int D>
 __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
-    const bf16* __restrict__ v, bf16* __restrict__ out, int B, int S, int Hq,
-    int Hkv, float scale, int causal) {
+    const bf16* __restrict__ v, bf16* __restrict__ out, int B, int Sq,
+    int Skv, int q_off, int Hq, int Hkv, float scale, int causal) {
   static_assert(D == 64 || D == 128);
   constexpr int KCH = D / 32;      // QK^T k-chunks
   constexpr int NSUB = BN / 16;    // score col tiles = 4
@@ -68,8 +71,8 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
   bf16x8_t qfrag[KCH];
   {
     const int qrow = q0 + r0 + col;
-    const int safe = min(qrow, S - 1);
-    const bf16* qp = q + (((int64_t)b * S + safe) * Hq + h) * D;
+    const int safe = min(qrow, Sq - 1);
+    const bf16* qp = q + (((int64_t)b * Sq + safe) * Hq + h) * D;
 #pragma unroll
     for (int c = 0; c < KCH; ++c)
       qfrag[c] = *reinterpret_cast<const bf16x8_t*>(qp + c * 32 + quad * 8);
@@ -82,18 +85,18 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
 #pragma unroll
   for (int n = 0; n < NS_PV; ++n) oacc[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  const int kv_end = causal ? min(S, q0 + BM) : S;
-  const int64_t kbase = ((int64_t)b * S) * Hkv * D + (int64_t)hkv * D;
+  const int kv_end = causal ? min(Skv, q_off + q0 + BM) : Skv;
+  const int64_t kbase = ((int64_t)b * Skv) * Hkv * D + (int64_t)hkv * D;
 
   for (int kt0 = 0; kt0 < kv_end; kt0 += BN) {
     // ---- stage K tile [BN][D]; V into [32][16] tr panels ----
     for (int i = threadIdx.x * 8; i < BN * D; i += BLOCK * 8) {
       const int row = i / D, c = i % D;
       const int key = kt0 + row;
-      const int64_t off = kbase + (int64_t)min(key, S - 1) * Hkv * D + c;
+      const int64_t off = kbase + (int64_t)min(key, Skv - 1) * Hkv * D + c;
       bf16x8_t kval = *reinterpret_cast<const bf16x8_t*>(k + off);
       bf16x8_t vval = *reinterpret_cast<const bf16x8_t*>(v + off);
-      if (key >= S) {
+      if (key >= Skv) {
 #pragma unroll
         for (int j = 0; j < 8; ++j) { kval[j] = (__bf16)0.f; vval[j] = (__bf16)0.f; }
       }
@@ -128,8 +131,8 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
       const int kcol = kt0 + ns * 16 + col;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + r0 + quad * 4 + r;
-        const bool masked = (kcol >= S) || (causal && kcol > qrow);
+        const int qrow = q_off + q0 + r0 + quad * 4 + r;
+        const bool masked = (kcol >= Skv) || (causal && kcol > qrow);
         sacc[ns][r] = masked ? -1e30f : sacc[ns][r] * scale;
         row_max[r] = fmaxf(row_max[r], sacc[ns][r]);
       }
@@ -190,12 +193,12 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
 
   // ---- epilogue: divide by l, store ----
   {
-    const int64_t obase = ((int64_t)b * S) * Hq * D + (int64_t)h * D;
+    const int64_t obase = ((int64_t)b * Sq) * Hq * D + (int64_t)h * D;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = r0 + quad * 4 + r;
       const float linv = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
-      if (q0 + row < S) {
+      if (q0 + row < Sq) {
 #pragma unroll
         for (int n = 0; n < NS_PV; ++n)
           out[obase + (int64_t)(q0 + row) * Hq * D + n * 16 + col] =
@@ -210,17 +213,19 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
 extern "C" {
 
 void tl_prefill_attn(const void* q, const void* k, const void* v, void* out,
-                     int B, int S, int Hq, int Hkv, int D, float scale,
-                     int causal, hipStream_t stream) {
-  dim3 grid((S + BM - 1) / BM, Hq, B), block(BLOCK);
+                     int B, int Sq, int Skv, int q_off, int Hq, int Hkv,
+                     int D, float scale, int causal, hipStream_t stream) {
+  dim3 grid((Sq + BM - 1) / BM, Hq, B), block(BLOCK);
   if (D == 128)
     hipLaunchKernelGGL((prefill_attn_kernel<128>), grid, block, 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (bf16*)out, B, S, Hq, Hkv, scale, causal);
+                       (bf16*)out, B, Sq, Skv, q_off, Hq, Hkv, scale,
+                       causal);
   else if (D == 64)
     hipLaunchKernelGGL((prefill_attn_kernel<64>), grid, block, 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (bf16*)out, B, S, Hq, Hkv, scale, causal);
+                       (bf16*)out, B, Sq, Skv, q_off, Hq, Hkv, scale,
+                       causal);
 }
 
 }  // extern "C"

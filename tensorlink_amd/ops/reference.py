@@ -91,20 +91,24 @@ def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
 # Attention
 # --------------------------------------------------------------------------
 def attention_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                      causal: bool = True, scale: Optional[float] = None
-                      ) -> torch.Tensor:
-    """Causal GQA attention. q [B,S,Hq,D], k/v [B,S,Hkv,D] → [B,S,Hq,D]."""
-    B, S, Hq, D = q.shape
+                      causal: bool = True, scale: Optional[float] = None,
+                      q_off: int = 0) -> torch.Tensor:
+    """Causal GQA attention. q [B,Sq,Hq,D], k/v [B,Skv,Hkv,D] →
+    [B,Sq,Hq,D]. q_off: query row i sits at global position q_off+i
+    (chunked prefill against Skv = q_off + Sq history+chunk keys)."""
+    B, Sq, Hq, D = q.shape
+    Skv = k.shape[1]
     Hkv = k.shape[2]
     scale = scale or 1.0 / math.sqrt(D)
     rep = Hq // Hkv
-    qf = q.float().transpose(1, 2)                       # [B,Hq,S,D]
+    qf = q.float().transpose(1, 2)                       # [B,Hq,Sq,D]
     kf = k.float().transpose(1, 2).repeat_interleave(rep, dim=1)
     vf = v.float().transpose(1, 2).repeat_interleave(rep, dim=1)
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
     if causal:
-        mask = torch.triu(torch.ones(S, S, dtype=torch.bool, device=q.device), 1)
-        scores = scores.masked_fill(mask, float("-inf"))
+        rows = torch.arange(Sq, device=q.device).unsqueeze(1) + q_off
+        cols = torch.arange(Skv, device=q.device).unsqueeze(0)
+        scores = scores.masked_fill(cols > rows, float("-inf"))
     p = scores.softmax(-1)
     out = torch.matmul(p, vf)
     return out.transpose(1, 2).to(q.dtype)

@@ -145,6 +145,26 @@ def test_prefill_attn(B, S, Hq, Hkv, D):
     torch.testing.assert_close(out.float(), out_ref, atol=2e-2, rtol=2e-2)
 
 
+@pytest.mark.parametrize("B,Sq,Skv,q_off,Hq,Hkv,D", [
+    (1, 64, 192, 128, 8, 2, 128),   # aligned chunk
+    (2, 50, 127, 77, 4, 2, 128),    # ragged chunk + ragged history
+    (1, 33, 99, 66, 4, 4, 64),
+])
+def test_prefill_attn_q_off(B, Sq, Skv, q_off, Hq, Hkv, D):
+    """Chunked prefill: queries at global rows q_off..q_off+Sq-1 attend to
+    all Skv cached keys with the causal mask applied at global positions."""
+    ops = _ext()
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(15)
+    q = torch.randn(B, Sq, Hq, D, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(B, Skv, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(B, Skv, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    out = ops.attention_prefill(q, k, v, causal=True, q_off=q_off)
+    out_ref = ref.attention_prefill(q.float(), k.float(), v.float(),
+                                    causal=True, q_off=q_off)
+    torch.testing.assert_close(out.float(), out_ref, atol=2e-2, rtol=2e-2)
+
+
 def test_adamw_matches_torch():
     ops = _ext()
     torch.manual_seed(6)
