@@ -142,6 +142,8 @@ class InferenceEngine:
         continuous = kwargs.pop("continuous", False)
         max_slots = kwargs.pop("max_slots", 16)
         max_ctx = kwargs.pop("max_ctx", 2048)
+        pool_pages = kwargs.pop("pool_pages", None)
+        prefill_chunk = kwargs.pop("prefill_chunk", None)
         plan = plan_for_world(config, self.world, **kwargs)
         runner = PipelineRunner(plan, self.rank, self.world,
                                 device=self.device, init=init,
@@ -150,8 +152,10 @@ class InferenceEngine:
         job = ModelJob(name=name, runner=runner, tokenizer=tok)
         if continuous and self.world == 1 and self.rank == 0:
             from tensorlink_amd.engine.batcher import ContinuousBatcher
-            job.batcher = ContinuousBatcher(runner, max_slots=max_slots,
-                                            max_ctx=max_ctx).start()
+            job.batcher = ContinuousBatcher(
+                runner, max_slots=max_slots, max_ctx=max_ctx,
+                pool_pages=pool_pages,
+                prefill_chunk=prefill_chunk).start()
         self.jobs[name] = job
         return job
 
