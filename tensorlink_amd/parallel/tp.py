@@ -31,16 +31,66 @@ from tensorlink_amd.parallel.comm import device_for_rank
 from tensorlink_amd.parallel.pipeline import SamplingParams
 
 
+class _ReduceFromTP(torch.autograd.Function):
+    """Megatron's g: all-reduce forward, identity backward (row-parallel
+    outputs — each rank holds a partial sum, gradients are replicated)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        dist.all_reduce(x)
+        return x
+
+    @staticmethod
+    def backward(ctx, dy):
+        return dy
+
+
+class _CopyToTP(torch.autograd.Function):
+    """Megatron's f: identity forward, all-reduce backward (column-parallel
+    inputs — x is replicated, each rank produces a partial dx)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        return x
+
+    @staticmethod
+    def backward(ctx, dy):
+        dy = dy.contiguous()
+        dist.all_reduce(dy)
+        return dy
+
+
+def _tp_active() -> bool:
+    return dist.is_initialized() and dist.get_world_size() > 1
+
+
 class AllReduceLinear(TLLinear):
     """Row-parallel output projection: local GEMM then sum across the TP
-    group (inference path)."""
+    group. Differentiable — the training path routes the all-reduce
+    through :class:`_ReduceFromTP` so gradients flow (identity backward)."""
 
     def forward(self, x):
         y = super().forward(x)
-        if dist.is_initialized() and dist.get_world_size() > 1:
-            y = y.contiguous()
-            dist.all_reduce(y)
+        if _tp_active():
+            if torch.is_grad_enabled() and y.requires_grad:
+                y = _ReduceFromTP.apply(y)
+            else:
+                y = y.contiguous()
+                dist.all_reduce(y)
         return y
+
+
+class ColumnParallelLinear(TLLinear):
+    """Column-parallel input projection (qkv / gate_up): weight rows are
+    sharded, x is replicated. Forward is the plain local GEMM; in training
+    the input passes through :class:`_CopyToTP` so each rank's partial dx
+    is summed in backward (inference: pure identity, zero overhead)."""
+
+    def forward(self, x):
+        if _tp_active() and torch.is_grad_enabled() and x.requires_grad:
+            x = _CopyToTP.apply(x)
+        return super().forward(x)
 
 
 def local_config(config: ModelConfig, tp: int) -> ModelConfig:
@@ -97,13 +147,17 @@ def build_tp_model(config_or_name, tp_rank: int, tp: int, device=None,
     del full
 
     stage = build_full_model(local_config(config, tp))
-    # swap row-parallel outputs for all-reducing variants
+    # swap row-parallel outputs for all-reducing variants and
+    # column-parallel inputs for the f-op (backward all-reduce) variants
     for layer in stage.layers:
-        for holder, name in ((layer.self_attn, "o_proj"),
-                             (layer.mlp, "down_proj")):
+        for holder, name, cls in (
+                (layer.self_attn, "o_proj", AllReduceLinear),
+                (layer.mlp, "down_proj", AllReduceLinear),
+                (layer.self_attn, "qkv_proj", ColumnParallelLinear),
+                (layer.mlp, "gate_up_proj", ColumnParallelLinear)):
             old = getattr(holder, name)
-            new = AllReduceLinear(old.in_features, old.out_features,
-                                  bias=old.bias is not None)
+            new = cls(old.in_features, old.out_features,
+                      bias=old.bias is not None)
             setattr(holder, name, new)
     stage.load_state_dict(shards)
     stage.to(device=device, dtype=dtype)
@@ -149,3 +203,57 @@ class TPRunner:
             out[:, t] = cur
             positions += 1
         return out
+
+
+class TPTrainer:
+    """SPMD tensor-parallel training (PP=1): every rank holds a head/
+    intermediate shard, runs the full replicated batch, and computes the
+    identical loss after the per-layer reductions (row-parallel forward
+    all-reduce; column-parallel backward all-reduce via _CopyToTP).
+
+    Replicated parameters (norms, embeddings, lm head) receive full
+    gradients on every rank from identical replicated activations; their
+    grads are all-reduce-averaged each step only to pin down fp
+    drift across ranks (exact arithmetic would make it a no-op).
+    """
+
+    def __init__(self, model, rank: int, tp: int, device=None, seed: int = 0,
+                 lr: float = 1e-3, weight_decay: float = 0.01):
+        from tensorlink_amd.optim import FusedAdamW
+        self.rank, self.tp = rank, tp
+        self.device = device if device is not None else device_for_rank()
+        self.stage = build_tp_model(model, rank, tp, device=self.device,
+                                    seed=seed)
+        self.stage.train()
+        for p in self.stage.parameters():
+            p.requires_grad_(True)
+        self.config = self.stage.config
+        self._replicated = [
+            p for n, p in self.stage.named_parameters()
+            if not any(t in n for t in ("qkv_proj", "gate_up_proj",
+                                        "o_proj", "down_proj"))]
+        self.opt = FusedAdamW(self.stage.parameters(), lr=lr,
+                              weight_decay=weight_decay)
+
+    def train_step(self, input_ids: torch.Tensor,
+                   labels: Optional[torch.Tensor] = None) -> float:
+        from tensorlink_amd import ops as tl_ops
+        ids = input_ids.to(self.device)
+        labels = ids if labels is None else labels.to(self.device)
+        B, S = ids.shape
+        pos = torch.arange(S, device=self.device,
+                           dtype=torch.int32).unsqueeze(0).expand(B, -1)
+        self.opt.zero_grad()
+        hidden = self.stage(ids, pos.contiguous(), training=True,
+                            return_logits=False)
+        logits = self.stage.head(hidden)
+        loss = tl_ops.causal_lm_loss(logits, labels)
+        loss.backward()
+        if _tp_active():
+            for p in self._replicated:
+                if p.grad is not None:
+                    g = p.grad.contiguous()
+                    dist.all_reduce(g)
+                    p.grad = g.div_(self.tp)
+        self.opt.step()
+        return float(loss.detach())
