@@ -9,10 +9,20 @@ of per-rank partial outputs — token t's contribution comes only from the
 rank(s) owning its routed experts, so the sum reconstructs the dense
 result exactly.
 
-This "partial-sum" EP trades the classic all-to-all token exchange for
-one hidden-sized all-reduce per MoE layer — the right starting point on
-xGMI where all-reduce is per-link bound but simple and overlap-friendly;
-token-routing all-to-all is the roadmap follow-up.
+Two exchange modes per MoE layer (``mode=`` on EPMoEMLP/EPRunner):
+
+- ``"allreduce"`` (default): every rank runs the router over ALL tokens,
+  computes its local experts' contribution and the layer output is the
+  all-reduce sum of partials. One hidden-sized all-reduce (2·T·H words
+  on a ring); simple and overlap-friendly on xGMI.
+- ``"alltoall"``: tokens are sharded across the EP group; each rank
+  routes its T/ep tokens and dispatches them to their experts' owner
+  ranks with an uneven ``all_to_all_single`` (dispatch), gets expert
+  outputs back (combine), and the token shards are all-gathered to
+  rebuild the replicated stream. Traffic ≈ (2·top_k + 1)·T/ep·H words —
+  cheaper than the all-reduce when ep > 2·top_k + 1, i.e. for large EP
+  groups with few active experts per token (the Mixtral regime at ep=8,
+  top_k=2 is the break-even point on paper; measure on xGMI).
 """
 
 from __future__ import annotations
@@ -33,23 +43,87 @@ from tensorlink_amd.parallel.pipeline import SamplingParams
 
 
 class EPMoEMLP(nn.Module):
-    """MoE block holding a shard of the experts; output is this rank's
-    partial sum, all-reduced across the EP group."""
+    """MoE block holding a shard of the experts (see module docstring for
+    the two exchange modes)."""
 
-    def __init__(self, full: MoEMLP, ep_rank: int, ep: int):
+    def __init__(self, full: MoEMLP, ep_rank: int, ep: int,
+                 mode: str = "allreduce"):
         super().__init__()
         n = full.num_experts
         assert n % ep == 0, "experts must divide ep"
+        assert mode in ("allreduce", "alltoall")
         self.num_experts = n
         self.top_k = full.top_k
         self.ep_rank, self.ep = ep_rank, ep
+        self.mode = mode
         self.local_n = n // ep
         self.local_base = ep_rank * self.local_n
         self.gate = full.gate                       # replicated router
         self.experts = nn.ModuleList(
             full.experts[self.local_base:self.local_base + self.local_n])
 
+    def _run_local_experts(self, x_recv: torch.Tensor,
+                           eid_recv: torch.Tensor) -> torch.Tensor:
+        """Apply the owning local expert to each received row."""
+        y = torch.empty_like(x_recv)
+        for le, expert in enumerate(self.experts):
+            m = (eid_recv == self.local_base + le).nonzero(as_tuple=True)[0]
+            if m.numel():
+                y[m] = expert(x_recv[m])
+        return y
+
+    def _forward_alltoall(self, x):
+        B, S, H = x.shape
+        flat = x.reshape(-1, H)
+        T = flat.shape[0]
+        ep, r = self.ep, self.ep_rank
+        shard = (T + ep - 1) // ep                   # padded shard size
+        t0, t1 = r * shard, min((r + 1) * shard, T)
+        my = flat[t0:t1]                             # [t, H] this rank's
+        weights, idx = ops.moe_topk_router(self.gate(my), self.top_k)
+        weights = weights.to(x.dtype)
+
+        # (token, slot) pairs sorted by destination rank
+        t = my.shape[0]
+        pair_tok = torch.arange(t, device=x.device
+                                ).unsqueeze(1).expand(-1, self.top_k)
+        dest = idx // self.local_n                   # [t, k]
+        order = dest.reshape(-1).argsort(stable=True)
+        send_eid = idx.reshape(-1)[order].contiguous()
+        send_tok = pair_tok.reshape(-1)[order].contiguous()
+        send_x = my[send_tok].contiguous()
+        send_counts = torch.bincount(dest.reshape(-1),
+                                     minlength=ep).tolist()
+
+        # exchange pair counts, then dispatch tokens + expert ids
+        cnt_in = torch.tensor(send_counts, dtype=torch.int64)
+        cnt_out = torch.empty_like(cnt_in)
+        dist.all_to_all_single(cnt_out, cnt_in)
+        recv_counts = cnt_out.tolist()
+        n_recv = int(sum(recv_counts))
+        x_recv = torch.empty(n_recv, H, device=x.device, dtype=x.dtype)
+        eid_recv = torch.empty(n_recv, device=x.device, dtype=send_eid.dtype)
+        dist.all_to_all_single(x_recv, send_x, recv_counts, send_counts)
+        dist.all_to_all_single(eid_recv, send_eid, recv_counts, send_counts)
+
+        # expert compute on owned rows, then combine (reverse exchange)
+        y_recv = self._run_local_experts(x_recv, eid_recv)
+        y_back = torch.empty_like(send_x)
+        dist.all_to_all_single(y_back, y_recv, send_counts, recv_counts)
+
+        out_my = torch.zeros(shard, H, device=x.device, dtype=x.dtype)
+        w_pairs = weights.reshape(-1)[order]
+        out_my.index_add_(0, send_tok, y_back * w_pairs[:, None])
+
+        # rebuild the replicated stream from the token shards
+        gathered = [torch.empty_like(out_my) for _ in range(ep)]
+        dist.all_gather(gathered, out_my)
+        return torch.cat(gathered, 0)[:T].reshape(B, S, H)
+
     def forward(self, x):
+        if (self.mode == "alltoall" and dist.is_initialized()
+                and dist.get_world_size() > 1):
+            return self._forward_alltoall(x)
         B, S, H = x.shape
         flat = x.reshape(-1, H)
         weights, idx = ops.moe_topk_router(self.gate(flat), self.top_k)
@@ -69,7 +143,8 @@ class EPMoEMLP(nn.Module):
 
 
 def build_ep_model(config_or_name, ep_rank: int, ep: int, device=None,
-                   dtype=None, seed: int = 0, quantize: Optional[str] = None):
+                   dtype=None, seed: int = 0, quantize: Optional[str] = None,
+                   mode: str = "allreduce"):
     """Full MoE model (PP=1) with experts sharded EP-ways; weights come
     from the same seeded full init on every rank so the EP group
     reproduces the single-rank reference."""
@@ -82,7 +157,7 @@ def build_ep_model(config_or_name, ep_rank: int, ep: int, device=None,
     stage = build_full_model(config)
     init_random_stage(stage, device="cpu", dtype=dtype, seed=seed)
     for layer in stage.layers:
-        layer.mlp = EPMoEMLP(layer.mlp, ep_rank, ep)
+        layer.mlp = EPMoEMLP(layer.mlp, ep_rank, ep, mode=mode)
     if quantize == "fp8":
         from tensorlink_amd.models.quant import Fp8Linear
         for layer in stage.layers:
@@ -100,11 +175,11 @@ class EPRunner:
     the full attention/embedding weights and 1/ep of the experts)."""
 
     def __init__(self, model, rank: int, ep: int, device=None, seed: int = 0,
-                 quantize: Optional[str] = None):
+                 quantize: Optional[str] = None, mode: str = "allreduce"):
         self.rank, self.ep = rank, ep
         self.device = device if device is not None else device_for_rank()
         self.stage = build_ep_model(model, rank, ep, device=self.device,
-                                    seed=seed, quantize=quantize)
+                                    seed=seed, quantize=quantize, mode=mode)
         self.config = self.stage.config
 
     @torch.no_grad()

@@ -456,3 +456,46 @@ def test_tp2_training_matches_single_rank():
         opt.step()
         ref_losses.append(float(loss.detach()))
     assert outs[0] == pytest.approx(ref_losses, rel=2e-3)
+
+
+def _ep_a2a_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.ep import EPRunner
+    from tensorlink_amd.parallel.pipeline import SamplingParams
+    init_distributed(backend="gloo")
+    r = EPRunner("tiny-moe", rank, world, device=torch.device("cpu"),
+                 seed=9, mode="alltoall")
+    torch.manual_seed(41)
+    ids = torch.randint(0, 1024, (2, 10))
+    out = r.generate(ids, SamplingParams(max_new_tokens=5))
+    q.put((rank, out))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ep2_alltoall_matches_single_rank():
+    """Token-routing EP: tokens sharded, dispatched to expert-owner ranks
+    with uneven all_to_all, combined and all-gathered — reproduces the
+    single-rank MoE output exactly (incl. S=1 decode where some ranks
+    get an empty token shard)."""
+    outs = dict(_run_collect(_ep_a2a_worker, 2, _port(13)))
+    assert torch.equal(outs[0], outs[1])
+
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.configs import get_config
+    m = build_full_model(get_config("tiny-moe"))
+    init_random_stage(m, device="cpu", dtype=torch.float32, seed=9)
+    torch.manual_seed(41)
+    ids = torch.randint(0, 1024, (2, 10))
+    cur = ids
+    for _ in range(5):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).expand(2, -1).contiguous()
+        logits = m(cur, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 10:], outs[0])
