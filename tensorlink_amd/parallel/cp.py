@@ -7,9 +7,12 @@ ring over xGMI P2P. Causality prunes the ring: chunk j's K/V only visits
 ranks i >= j, and partial results merge with the standard online-softmax
 (m, l, o) combination, so the result equals single-rank attention.
 
-Scope: inference prefill (logits). Decode after a CP prefill and the
-training backward are roadmap items; the kernel-grade version would fuse
-the merge into prefill_attn.hip.
+Scope: inference prefill (logits) and decode after a CP prefill: each
+rank keeps the K/V of ITS sequence chunk (the cache is sequence-sharded,
+so cache memory divides by cp); decode tokens append to the last rank
+and every step merges the per-rank partial attentions with one
+all-gather of (o, m, l). The training backward is a roadmap item; the
+kernel-grade version would fuse the merge into prefill_attn.hip.
 """
 
 from __future__ import annotations
@@ -85,8 +88,10 @@ class CPRunner:
         self.config = config
         self.dtype = dtype
 
-    def _ring_attention(self, attn, x, positions):
-        """One attention layer with K/V rotating around the CP ring."""
+    def _ring_attention(self, attn, x, positions, layer_idx=None):
+        """One attention layer with K/V rotating around the CP ring.
+        layer_idx is not None => cache this rank's (rotated) k/v chunk for
+        subsequent decode."""
         B, Sq, H = x.shape
         qkv = attn.qkv_proj(x)
         q, k, v = attn._split_qkv(qkv, B, Sq)
@@ -98,6 +103,8 @@ class CPRunner:
                         k.view(B * Sq, attn.n_kv, -1), flat_pos,
                         attn.inv_freq)
         scale = attn.scale
+        if layer_idx is not None:
+            self._kv[layer_idx] = (k.clone(), v.clone())
 
         acc = None
         cur_k, cur_v = k, v
@@ -129,9 +136,11 @@ class CPRunner:
         return attn.o_proj(out.to(x.dtype).reshape(B, Sq, -1))
 
     @torch.no_grad()
-    def forward_logits(self, input_ids: torch.Tensor) -> torch.Tensor:
+    def forward_logits(self, input_ids: torch.Tensor,
+                       keep_kv: bool = False) -> torch.Tensor:
         """Prefill logits for this rank's sequence chunk. input_ids is the
-        FULL [B, S] prompt on every rank (S divisible by cp)."""
+        FULL [B, S] prompt on every rank (S divisible by cp). keep_kv
+        caches this rank's K/V chunk for subsequent decode."""
         B, S = input_ids.shape
         assert S % self.cp == 0, "S must divide cp"
         sc = S // self.cp
@@ -141,15 +150,92 @@ class CPRunner:
             .unsqueeze(0).expand(B, -1).contiguous()
 
         stage = self.stage
+        if keep_kv:
+            self._kv = [None] * len(stage.layers)
         hidden = stage.embed(ids)
         eps = self.config.rms_norm_eps
-        for layer in stage.layers:
+        for li, layer in enumerate(stage.layers):
             h = ops.rmsnorm(hidden, layer.input_layernorm.to(hidden.dtype),
                             eps)
-            hidden = hidden + self._ring_attention(layer.self_attn, h,
-                                                   positions)
+            hidden = hidden + self._ring_attention(
+                layer.self_attn, h, positions,
+                layer_idx=li if keep_kv else None)
             h = ops.rmsnorm(hidden,
                             layer.post_attention_layernorm.to(hidden.dtype),
                             eps)
             hidden = hidden + layer.mlp(h)
         return stage.head(hidden)       # [B, sc, V] — this rank's chunk
+
+    # ------------------------------------------------------------------
+    # decode over the sequence-sharded cache
+    # ------------------------------------------------------------------
+    def _decode_attention(self, attn, x, pos: int, layer_idx: int):
+        """One decode step of one attention layer: partial attention over
+        this rank's K/V shard, merged across the group with one all-gather
+        of (o, m, l). New tokens' K/V append to the LAST rank."""
+        B = x.shape[0]
+        qkv = attn.qkv_proj(x)
+        q, k, v = attn._split_qkv(qkv, B, 1)
+        q = q.contiguous()
+        k = k.contiguous()
+        positions = torch.full((B,), pos, device=self.device,
+                               dtype=torch.int32)
+        ops.apply_rope_(q.view(B, attn.n_heads, -1),
+                        k.view(B, attn.n_kv, -1), positions, attn.inv_freq)
+        if self.rank == self.cp - 1:        # owner of all decode tokens
+            k_loc = torch.cat([self._kv[layer_idx][0], k], dim=1)
+            v_loc = torch.cat([self._kv[layer_idx][1], v], dim=1)
+            self._kv[layer_idx] = (k_loc, v_loc)
+        k_loc, v_loc = self._kv[layer_idx]
+        o, m, l = _partial_attn(q, k_loc, v_loc, attn.scale,
+                                causal_diag=False)
+        if self.cp > 1:
+            gath = []
+            for t in (o, m, l):
+                t = t.contiguous()
+                bufs = [torch.empty_like(t) for _ in range(self.cp)]
+                dist.all_gather(bufs, t)
+                gath.append(bufs)
+            acc = None
+            for r in range(self.cp):
+                acc = _merge(acc, (gath[0][r], gath[1][r], gath[2][r]))
+            o, m, l = acc
+        out = o / l.permute(0, 2, 1).unsqueeze(-1).clamp(min=1e-30)
+        return attn.o_proj(out.to(x.dtype).reshape(B, 1, -1))
+
+    @torch.no_grad()
+    def _decode_forward(self, tok: torch.Tensor, pos: int) -> torch.Tensor:
+        """Replicated single-token forward over the sharded cache."""
+        stage = self.stage
+        eps = self.config.rms_norm_eps
+        hidden = stage.embed(tok.view(-1, 1))
+        for li, layer in enumerate(stage.layers):
+            h = ops.rmsnorm(hidden, layer.input_layernorm.to(hidden.dtype),
+                            eps)
+            hidden = hidden + self._decode_attention(layer.self_attn, h,
+                                                     pos, li)
+            h = ops.rmsnorm(hidden,
+                            layer.post_attention_layernorm.to(hidden.dtype),
+                            eps)
+            hidden = hidden + layer.mlp(h)
+        return stage.head(hidden).squeeze(1)        # [B, V]
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor,
+                 max_new_tokens: int = 16) -> torch.Tensor:
+        """Greedy generation: CP ring prefill, then sharded-cache decode.
+        Every rank returns the full [B, max_new_tokens] output (sampling
+        is replicated — the merged logits agree on all ranks; the prompt's
+        last-token logits live on the last rank and are broadcast)."""
+        B, S = input_ids.shape
+        logits = self.forward_logits(input_ids, keep_kv=True)
+        cur = logits[:, -1].argmax(-1)              # correct on last rank
+        dist.broadcast(cur, self.cp - 1)
+        out = torch.empty(B, max_new_tokens, device=self.device,
+                          dtype=torch.int64)
+        out[:, 0] = cur
+        for t in range(1, max_new_tokens):
+            lg = self._decode_forward(cur, S + t - 1)
+            cur = lg.argmax(-1)
+            out[:, t] = cur
+        return out

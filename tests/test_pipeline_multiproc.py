@@ -499,3 +499,43 @@ def test_ep2_alltoall_matches_single_rank():
         logits = m(cur, pos)
         cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(cur[:, 10:], outs[0])
+
+
+def _cp_gen_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.cp import CPRunner
+    init_distributed(backend="gloo")
+    r = CPRunner("tiny", rank, world, device=torch.device("cpu"), seed=17)
+    torch.manual_seed(53)
+    ids = torch.randint(0, 1024, (2, 24))
+    out = r.generate(ids, max_new_tokens=6)
+    q.put((rank, out))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_cp2_decode_matches_single_rank():
+    """CP decode: sequence-sharded KV cache (prompt chunks stay on their
+    prefill rank, decode tokens append to the last rank) with per-step
+    (o,m,l) all-gather merge reproduces single-rank greedy decode."""
+    outs = dict(_run_collect(_cp_gen_worker, 2, _port(15)))
+    assert torch.equal(outs[0], outs[1])
+
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.configs import get_config
+    m = build_full_model(get_config("tiny"))
+    init_random_stage(m, device="cpu", dtype=torch.float32, seed=17)
+    torch.manual_seed(53)
+    ids = torch.randint(0, 1024, (2, 24))
+    cur = ids
+    for _ in range(6):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).expand(2, -1).contiguous()
+        logits = m(cur, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 24:], outs[0])
