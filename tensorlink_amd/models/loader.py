@@ -174,6 +174,41 @@ def save_stage_to_safetensors(stage: StageModel, out_dir: str,
     return path
 
 
+@torch.no_grad()
+def load_stage_from_stage_ckpt(stage: StageModel, out_dir: str) -> int:
+    """Re-partition-aware checkpoint load: assemble THIS stage's weights
+    from a directory of per-stage files saved under a possibly DIFFERENT
+    pipeline partitioning (elastic recovery after a world-size change —
+    parallel/elastic.py). The sidecar ``stage_N.json`` gives each saved
+    shard's global layer range; local layer indices are remapped through
+    it. Returns the number of tensors loaded."""
+    import glob
+
+    from safetensors.torch import load_file
+    target = dict(stage.state_dict())
+    loaded = 0
+    for meta_path in sorted(glob.glob(os.path.join(out_dir,
+                                                   "stage_*.json"))):
+        r = int(os.path.basename(meta_path)[6:-5])
+        with open(meta_path) as f:
+            meta = json.load(f)
+        state = load_file(os.path.join(out_dir, f"stage_{r}.safetensors"))
+        ls = meta["layer_start"]
+        for k, v in state.items():
+            if k.startswith("layers."):
+                _, loc, rest = k.split(".", 2)
+                gl = ls + int(loc)
+                if not (stage.layer_start <= gl < stage.layer_end):
+                    continue
+                nk = f"layers.{gl - stage.layer_start}.{rest}"
+            else:
+                nk = k                  # embed / final norm / lm head
+            if nk in target:
+                target[nk].copy_(v.to(target[nk].dtype))
+                loaded += 1
+    return loaded
+
+
 def load_stage_from_safetensors(stage: StageModel, out_dir: str,
                                 rank: int) -> None:
     from safetensors.torch import load_file

@@ -539,3 +539,49 @@ def test_cp2_decode_matches_single_rank():
         logits = m(cur, pos)
         cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(cur[:, 24:], outs[0])
+
+
+def _elastic_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import time
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.elastic import ElasticRunner
+    from tensorlink_amd.parallel.pipeline import SamplingParams
+    init_distributed(backend="gloo")
+    ckpt = os.environ["TL_ELASTIC_CKPT"]
+    r = ElasticRunner("tiny", device=torch.device("cpu"), seed=23,
+                      rendezvous_port=port + 100, grace_s=1.0)
+    r.save_checkpoint(ckpt)              # both stage shards on disk
+    dist.barrier()
+    torch.manual_seed(61)
+    ids = torch.randint(0, 1024, (1, 12))
+    out1 = r.generate(ids, SamplingParams(max_new_tokens=4))
+    if rank == 1:
+        os._exit(0)                      # simulated crash: no teardown
+    time.sleep(2.0)                      # let rank 1 die first
+    out2 = r.generate(ids, SamplingParams(max_new_tokens=4))
+    # plain lists: the worker exits right after put, so shared-memory
+    # tensor handles would race the parent's fd dup
+    q.put((rank, (out1.tolist(), out2.tolist(), r.world, r.generation)))
+
+
+@pytest.mark.timeout(300)
+def test_elastic_recovery_after_rank_loss(tmp_path):
+    """Rank 1 dies mid-service: rank 0's next collective fails, survivors
+    re-rendezvous on a side store, the group re-forms as world=1, the
+    pipeline is re-planned and reloaded from the per-stage checkpoint
+    (re-partitioned pp2 -> pp1), and the retried request reproduces the
+    pre-failure output."""
+    os.environ["TL_ELASTIC_CKPT"] = str(tmp_path / "ckpt")
+    try:
+        outs = dict(_run_collect(_elastic_worker, 2, _port(17),
+                                 n_results=1))
+    finally:
+        del os.environ["TL_ELASTIC_CKPT"]
+    out1, out2, new_world, gen = outs[0]
+    assert new_world == 1 and gen == 1
+    assert out1 == out2                  # same weights, same tokens
