@@ -502,6 +502,33 @@ class PipelineTrainer:
     # loss.backward(), which routes here (reference CustomAutogradRouter,
     # ml/module.py:126-144).
     # ------------------------------------------------------------------
+    # ------------------------------------------------------------------
+    # training resume: weights + optimizer moments + step count
+    # (the reference checkpoints weights only — parameter retrieval,
+    #  ml/module.py:577-670; moment/step resume exceeds parity)
+    # ------------------------------------------------------------------
+    def save_checkpoint(self, out_dir: str) -> str:
+        import os
+
+        from tensorlink_amd.models.loader import save_stage_to_safetensors
+        save_stage_to_safetensors(self.stage, out_dir, self.rank)
+        torch.save(self.optimizer.state_dict(),
+                   os.path.join(out_dir, f"optim_{self.rank}.pt"))
+        return out_dir
+
+    def load_checkpoint(self, out_dir: str) -> None:
+        """Restore weights + optimizer state saved by save_checkpoint
+        under the SAME pipeline partitioning. load_state_dict copies
+        in-place, so the flat-buffer views FusedAdamW re-pointed the
+        parameters into stay valid."""
+        import os
+
+        from tensorlink_amd.models.loader import load_stage_from_safetensors
+        load_stage_from_safetensors(self.stage, out_dir, self.rank)
+        sd = torch.load(os.path.join(out_dir, f"optim_{self.rank}.pt"),
+                        weights_only=False)
+        self.optimizer.load_state_dict(sd)
+
     def spmd_forward(self, input_ids: Optional[torch.Tensor] = None):
         """Single-micro-batch pipeline forward with autograd stashes.
         Returns logits [B,S,V] on rank 0 (detached), None elsewhere."""

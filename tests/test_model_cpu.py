@@ -3,6 +3,7 @@ checkpoint round-trip, HF key mapping."""
 
 import os
 
+import pytest
 import torch
 
 from tensorlink_amd.models import build_full_model, get_config
@@ -174,3 +175,35 @@ def test_planner_rejects_more_stages_than_layers():
     from tensorlink_amd.parallel.planner import AssignmentError
     with _pytest.raises(AssignmentError):
         plan_for_world("tiny", 8)   # tiny has 4 layers
+
+
+def test_training_checkpoint_resume():
+    """Save-at-step-2 / resume reproduces the straight 4-step loss
+    trajectory exactly (weights + Adam moments + step count restored)."""
+    import torch
+
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    from tensorlink_amd.parallel.planner import plan_for_world
+
+    def batches():
+        g = torch.Generator().manual_seed(71)
+        return [torch.randint(0, 1024, (2, 16), generator=g)
+                for _ in range(4)]
+
+    plan = plan_for_world("tiny", 1)
+    t_ref = PipelineTrainer(plan, 0, 1, device=torch.device("cpu"),
+                            seed=5, lr=1e-3)
+    ref_losses = [t_ref.train_step(b, labels=b) for b in batches()]
+
+    t_a = PipelineTrainer(plan, 0, 1, device=torch.device("cpu"),
+                          seed=5, lr=1e-3)
+    bs = batches()
+    a_losses = [t_a.train_step(b, labels=b) for b in bs[:2]]
+    import tempfile
+    with tempfile.TemporaryDirectory() as d:
+        t_a.save_checkpoint(d)
+        t_b = PipelineTrainer(plan, 0, 1, device=torch.device("cpu"),
+                              seed=99, lr=1e-3)   # different init: must load
+        t_b.load_checkpoint(d)
+        b_losses = [t_b.train_step(b, labels=b) for b in bs[2:]]
+    assert a_losses + b_losses == pytest.approx(ref_losses, rel=1e-5)
