@@ -171,6 +171,10 @@ def main():
         from tensorlink_amd.utils.tunable import save_tunableop
         save_tunableop()
 
+    if os.environ.get("TL_TRACE"):          # per-rank Chrome trace export
+        from tensorlink_amd.utils.tracing import export_from_env
+        export_from_env(runner.tracer)
+
     if rank == 0:
         print(json.dumps({
             "metric": "output tokens/sec (Qwen2.5-7B serving, PP sharded)",

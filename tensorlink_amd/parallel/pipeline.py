@@ -99,6 +99,15 @@ class PipelineRunner:
         self.kv_cache: Optional[KVCache] = None
         self._gen = None
         self._decode_graph = None       # (graph, tok_buf, pos_buf)
+        from tensorlink_amd.utils.tracing import tracer_from_env
+        self.tracer = tracer_from_env(rank)   # TL_TRACE=<prefix> enables
+
+    def _trace(self, name: str, t0: float, t1: float, **args) -> None:
+        if self.tracer is not None:
+            self.tracer.events.append(
+                {"name": name, "ph": "X", "pid": self.rank, "tid": 0,
+                 "ts": (t0 - self.tracer._t0) * 1e6,
+                 "dur": (t1 - t0) * 1e6, "args": args})
 
     # ------------------------------------------------------------------
     def _sample(self, logits: torch.Tensor, sp: SamplingParams) -> torch.Tensor:
@@ -187,6 +196,10 @@ class PipelineRunner:
                     t_first = time.perf_counter()
                 first_tokens.append(tok)
 
+        self._trace("prefill", t_start, time.perf_counter(), batch=B,
+                    seq=S, micro_batches=n_mb)
+        t_dec0 = time.perf_counter()
+
         # ---------------- ring-pipelined decode -------------------------
         T = sp.max_new_tokens
         positions = torch.full((B,), S, device=self.device,
@@ -239,6 +252,8 @@ class PipelineRunner:
                             n_out = t + 1
                             break
             out = out_tokens[:, :n_out]
+            self._trace("decode", t_dec0, time.perf_counter(), batch=B,
+                        new_tokens=n_out)
             stats = self._finish_stats(B, S, n_out, t_start, t_first)
             if t_prefill_end is not None and stats is not None:
                 dec = stats["total_s"] - (t_prefill_end - t_start)
@@ -300,6 +315,8 @@ class PipelineRunner:
         if self.is_first:
             result = self.p2p.recv((B, T), torch.int64, self.world - 1,
                                    self.device)
+        self._trace("decode", t_dec0, time.perf_counter(), batch=B,
+                    new_tokens=T)
         stats = self._finish_stats(B, S, T, t_start, t_first)
         if return_stats:
             return result, stats

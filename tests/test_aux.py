@@ -186,3 +186,49 @@ def test_tp_shard_state_partitions_exactly():
 def test_tunableop_noop_on_cpu():
     from tensorlink_amd.utils.tunable import setup_tunableop
     assert setup_tunableop() is False or torch.cuda.is_available()
+
+
+def test_tracer_spans_and_export(tmp_path):
+    """Chrome-trace export: spans, instants, gpu-arg plumbing, summary."""
+    import json
+    import time
+
+    from tensorlink_amd.utils.tracing import Tracer
+    tr = Tracer(rank=3, use_gpu_events=False)
+    with tr.span("prefill", batch=4):
+        time.sleep(0.01)
+    with tr.span("decode", tokens=8):
+        pass
+    tr.instant("token_emitted", t=0)
+    path = tr.export(str(tmp_path / "t.json"))
+    data = json.load(open(path))
+    names = [e["name"] for e in data["traceEvents"]]
+    assert names == ["prefill", "decode", "token_emitted"]
+    assert all(e["pid"] == 3 for e in data["traceEvents"])
+    pre = data["traceEvents"][0]
+    assert pre["ph"] == "X" and pre["dur"] >= 9_000   # >= 9 ms in us
+    assert pre["args"]["batch"] == 4
+    s = tr.summary()
+    assert s["prefill"]["count"] == 1 and s["prefill"]["mean_ms"] >= 9
+
+
+def test_runner_trace_env(tmp_path, monkeypatch):
+    """TL_TRACE wires a tracer into PipelineRunner: generate() emits
+    prefill + decode spans and export writes the per-rank file."""
+    import json
+
+    import torch
+
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.utils.tracing import export_from_env
+    monkeypatch.setenv("TL_TRACE", str(tmp_path / "tr"))
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                       device=torch.device("cpu"))
+    torch.manual_seed(1)
+    r.generate(torch.randint(0, 1024, (2, 8)),
+               SamplingParams(max_new_tokens=4))
+    path = export_from_env(r.tracer)
+    data = json.load(open(path))
+    names = [e["name"] for e in data["traceEvents"]]
+    assert "prefill" in names and "decode" in names
