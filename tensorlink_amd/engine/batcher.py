@@ -86,10 +86,13 @@ class _SlotView:
 
 
 class ContinuousBatcher:
+    SUPPORTS_PP = False
+
     def __init__(self, runner, max_slots: int = 16, max_ctx: int = 2048,
                  pool_pages: Optional[int] = None,
                  prefill_chunk: Optional[int] = None):
-        assert runner.world == 1, "continuous batching is single-rank"
+        assert runner.world == 1 or self.SUPPORTS_PP, \
+            "use PPContinuousBatcher for world > 1"
         self.runner = runner
         self.stage = runner.stage
         self.device = runner.device
@@ -263,3 +266,211 @@ class ContinuousBatcher:
             if not busy and not admitted and not prefilled:
                 self._wake.wait(0.05)
                 self._wake.clear()
+
+
+class PPContinuousBatcher(ContinuousBatcher):
+    """Continuous batching across a pipeline-parallel group.
+
+    Rank 0 owns the scheduler (queue, slots, finish decisions) and
+    broadcasts ONE compact command per iteration; every rank executes its
+    stage for the scheduled work — a prefill chunk and/or a decode batch —
+    with activations flowing rank→rank+1 and the last rank sampling and
+    isend-ing tokens back to rank 0. Follower ranks keep NO request state:
+    their paged-KV allocators stay in lockstep because ensure/release are
+    driven purely by the command stream (releases are deferred one
+    iteration on rank 0 so every rank frees pages on the same step).
+
+    The reference serves strictly one request at a time per job
+    (``ml/validator.py:642``) and cannot batch across a pipeline at all;
+    this is the PP-aware variant of the single-rank slot scheduler above.
+    """
+
+    SUPPORTS_PP = True
+
+    def __init__(self, runner, **kw):
+        super().__init__(runner, **kw)
+        self.slot_samp = [None] * self.max_slots   # (temp, top_p, top_k)
+        self._pending_releases: List[int] = []
+        self._send_keep = []                       # isend keep-alive
+
+    # ------------------------- shared execution -----------------------
+    def _sample_rows(self, logits, samps):
+        return [int(ops.sample_token(logits[i:i + 1], temperature=t,
+                                     top_p=p, top_k=k)[0])
+                for i, (t, p, k) in enumerate(samps)]
+
+    def _isend0(self, t):
+        t = t.contiguous()
+        self._send_keep.append((self.runner.p2p.isend(t, 0), t))
+        if len(self._send_keep) > 8:
+            w, _ = self._send_keep.pop(0)
+            w.wait()
+
+    @torch.no_grad()
+    def _exec_step(self, chunk, decodes, releases):
+        """Execute one broadcast command on THIS rank. Returns
+        (first_token or None, decode_tokens or None) on rank 0."""
+        r = self.runner
+        dev = self.device
+        for slot in releases:
+            self.cache.release_slot(slot)
+        chunk_tok = None
+        if chunk is not None:
+            slot, start, end, s_total, samp, ids = chunk
+            self.cache.ensure(slot, end + 1)
+            S = end - start
+            pos = torch.arange(start, end, device=dev,
+                               dtype=torch.int32).unsqueeze(0)
+            lens = torch.tensor([start], device=dev, dtype=torch.int32)
+            view = _SlotView(self.cache, [slot], lens)
+            if r.is_first:
+                x = torch.tensor(ids, device=dev,
+                                 dtype=torch.int64).unsqueeze(0)
+                hidden = self.stage(x, pos, kv_cache=view,
+                                    return_logits=False)
+            else:
+                hidden = r.p2p.recv((1, S, r.H), r.dtype, r.prev_rank, dev)
+                hidden = self.stage(hidden, pos, kv_cache=view,
+                                    return_logits=False)
+            if not r.is_last:
+                r.p2p.send(hidden.contiguous(), r.next_rank)
+            elif end == s_total:
+                logits = self.stage.head(hidden[:, -1:]).squeeze(1)
+                tok = self._sample_rows(logits, [samp])
+                self._isend0(torch.tensor(tok, dtype=torch.int64,
+                                          device=dev))
+        if decodes:
+            B = len(decodes)
+            slots = [d[0] for d in decodes]
+            for d in decodes:
+                self.cache.ensure(d[0], d[1] + 2)
+            lens = torch.tensor([d[1] for d in decodes], device=dev,
+                                dtype=torch.int32)
+            view = _SlotView(self.cache, slots, lens)
+            pos = lens.unsqueeze(1)
+            if r.is_first:
+                toks = torch.tensor([d[2] for d in decodes], device=dev,
+                                    dtype=torch.int64).unsqueeze(1)
+                hidden = self.stage(toks, pos, kv_cache=view,
+                                    return_logits=False)
+            else:
+                hidden = r.p2p.recv((B, 1, r.H), r.dtype, r.prev_rank, dev)
+                hidden = self.stage(hidden, pos, kv_cache=view,
+                                    return_logits=False)
+            if not r.is_last:
+                r.p2p.send(hidden.contiguous(), r.next_rank)
+            else:
+                logits = self.stage.head(hidden).squeeze(1)
+                new = self._sample_rows(logits, [d[3] for d in decodes])
+                self._isend0(torch.tensor(new, dtype=torch.int64,
+                                          device=dev))
+        # rank 0 collects the sampled tokens (isend on the last rank
+        # breaks the send/recv cycle between the two pipeline passes)
+        new_tokens = None
+        if r.is_first:
+            if chunk is not None and chunk[2] == chunk[3]:   # final chunk
+                chunk_tok = int(r.p2p.recv((1,), torch.int64, r.world - 1,
+                                           dev)[0])
+            if decodes:
+                new_tokens = r.p2p.recv((len(decodes),), torch.int64,
+                                        r.world - 1, dev).tolist()
+        return chunk_tok, new_tokens
+
+    # ------------------------- follower (ranks 1..N-1) ------------------
+    def serve_follower(self):
+        """Blocking command loop for non-scheduler ranks; returns on
+        batch_stop."""
+        r = self.runner
+        while True:
+            cmd = r.p2p.broadcast_obj(None, src=0)
+            if cmd[0] == "batch_stop":
+                for w, _ in self._send_keep:
+                    w.wait()
+                return
+            _, chunk, decodes, releases = cmd
+            self._exec_step(chunk, decodes, releases)
+
+    # ------------------------- scheduler (rank 0) ----------------------
+    def _finish0(self, slot: int, tok: int):
+        req = self.slots[slot]
+        if ((req.eos_token_id is not None and tok == req.eos_token_id)
+                or self.slot_emitted[slot] >= req.max_new_tokens
+                or self.slot_len[slot] + 1 >= self.max_ctx):
+            req.tokens.put(None)
+            req.done.set()
+            self.slots[slot] = None
+            self.slot_prompt[slot] = None
+            # defer the page release so every rank frees on the same step
+            self._pending_releases.append(slot)
+
+    def _loop(self):
+        assert self.runner.rank == 0
+        r = self.runner
+        while not self._stop.is_set():
+            while not self._queue.empty():
+                slot = self._free_slot()
+                if slot is None:
+                    break
+                need = (self._queue.queue[0].input_ids.numel()
+                        + PAGE) // PAGE + 1
+                if self.cache.allocator.n_free < need:
+                    break
+                req = self._queue.get_nowait()
+                self.slots[slot] = req
+                self.slot_prompt[slot] = req.input_ids
+                self.slot_filled[slot] = 0
+                self.slot_len[slot] = 0
+                self.slot_samp[slot] = (req.temperature, req.top_p,
+                                        req.top_k)
+            # one prefill chunk per iteration
+            chunk = None
+            chunk_slot = None
+            for slot in range(self.max_slots):
+                if self.slots[slot] is not None and (
+                        self._prefilling(slot)
+                        or self.slot_filled[slot] == 0):
+                    prompt = self.slot_prompt[slot]
+                    s_total = prompt.numel()
+                    start = self.slot_filled[slot]
+                    end = (s_total if self.prefill_chunk is None
+                           else min(start + self.prefill_chunk, s_total))
+                    chunk = (slot, start, end, s_total,
+                             self.slot_samp[slot],
+                             prompt[start:end].tolist())
+                    chunk_slot = slot
+                    break
+            decodes = [(i, self.slot_len[i], self.slot_last[i],
+                        self.slot_samp[i])
+                       for i in range(self.max_slots)
+                       if self.slots[i] is not None
+                       and not self._prefilling(i)
+                       and self.slot_filled[i] > 0
+                       and i != chunk_slot]
+            if chunk is None and not decodes and not self._pending_releases:
+                self._wake.wait(0.05)
+                self._wake.clear()
+                continue
+            rel, self._pending_releases = self._pending_releases, []
+            r.p2p.broadcast_obj(("batch_step", chunk, decodes, rel), src=0)
+            chunk_tok, new_tokens = self._exec_step(chunk, decodes, rel)
+            if chunk is not None:
+                slot, start, end, s_total = chunk[:4]
+                self.slot_filled[slot] = end
+                if chunk_tok is not None:
+                    req = self.slots[slot]
+                    self.slot_len[slot] = s_total
+                    self.slot_last[slot] = chunk_tok
+                    self.slot_emitted[slot] = 1
+                    req.tokens.put(chunk_tok)
+                    self._finish0(slot, chunk_tok)
+            if new_tokens:
+                for (slot, _, _, _), tok in zip(decodes, new_tokens):
+                    self.slot_len[slot] += 1
+                    self.slot_emitted[slot] += 1
+                    self.slots[slot].tokens.put(tok)
+                    self.slot_last[slot] = tok
+                    self._finish0(slot, tok)
+                self.steps += 1
+        r.p2p.broadcast_obj(("batch_stop",), src=0)
+        for w, _ in self._send_keep:
+            w.wait()

@@ -156,6 +156,18 @@ class InferenceEngine:
                 runner, max_slots=max_slots, max_ctx=max_ctx,
                 pool_pages=pool_pages,
                 prefill_chunk=prefill_chunk).start()
+        elif continuous and self.world > 1:
+            from tensorlink_amd.engine.batcher import PPContinuousBatcher
+            b = PPContinuousBatcher(runner, max_slots=max_slots,
+                                    max_ctx=max_ctx, pool_pages=pool_pages,
+                                    prefill_chunk=prefill_chunk)
+            if self.rank == 0:
+                job.batcher = b.start()
+                self.jobs[name] = job
+            else:
+                self.jobs[name] = job
+                b.serve_follower()      # blocks until rank 0 stops it
+            return job
         self.jobs[name] = job
         return job
 
@@ -177,10 +189,12 @@ class InferenceEngine:
 
     def unload_model(self, name: str):
         with self._lock:
-            self._bcast(("unload", name))
             job = self.jobs.pop(name, None)
+            # stop the batcher FIRST: follower ranks sit in its command
+            # loop and must see batch_stop before the unload broadcast
             if job is not None and job.batcher is not None:
                 job.batcher.stop()
+            self._bcast(("unload", name))
 
     def model_status(self, name: str) -> Dict:
         job = self.jobs.get(name)

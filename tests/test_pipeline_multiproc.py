@@ -585,3 +585,74 @@ def test_elastic_recovery_after_rank_loss(tmp_path):
     out1, out2, new_world, gen = outs[0]
     assert new_world == 1 and gen == 1
     assert out1 == out2                  # same weights, same tokens
+
+
+def _pp_batch_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.parallel.comm import init_distributed
+    init_distributed(backend="gloo")
+    eng = InferenceEngine(rank=rank, world=world, device=torch.device("cpu"))
+    if rank != 0:
+        eng.worker_loop()
+        return
+    eng.load_model("tiny", continuous=True, max_slots=4, max_ctx=256,
+                   prefill_chunk=16)
+    b = eng.jobs["tiny"].batcher
+    torch.manual_seed(33)
+    prompts = [torch.randint(0, 1024, (n,)) for n in (30, 9, 21)]
+    reqs = [b.submit(p, max_new_tokens=8) for p in prompts]
+    outs = [rq.result(timeout=120) for rq in reqs]
+    # releases are deferred one scheduler iteration (rank sync) — wait
+    import time
+    for _ in range(200):
+        if b.cache.allocator.n_free == b.cache.allocator.n_pages:
+            break
+        time.sleep(0.02)
+    n_free = b.cache.allocator.n_free
+    n_pages = b.cache.allocator.n_pages
+    eng.unload_model("tiny")
+    eng.shutdown()
+    q.put((rank, ([p.tolist() for p in prompts], outs, n_free, n_pages,
+                  b.steps)))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_pp2_continuous_batching_matches_serial():
+    """PP-aware continuous batching: rank 0 schedules, followers execute
+    stage commands; concurrent chunked-prefill requests across a 2-stage
+    pipeline reproduce single-rank serial greedy outputs and all pages
+    return to the pool."""
+    outs = dict(_run_collect(_pp_batch_worker, 2, _port(19), n_results=1))
+    prompts, results, n_free, n_pages, steps = outs[0]
+    assert n_free == n_pages, "pages leaked"
+    assert steps < 3 * 8          # interleaved, not serial
+
+    import torch
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    # single-rank reference with the same per-stage seeds as the pp2 pair
+    from tensorlink_amd.models.dense import build_stage
+    from tensorlink_amd.models.loader import init_random_stage
+    plan = plan_for_world("tiny", 2)
+    stages = []
+    for rk in range(2):
+        st = build_stage(plan.config, plan.stage_for_rank(rk))
+        init_random_stage(st, dtype=torch.float32, seed=rk)
+        stages.append(st)
+    for p, o in zip(prompts, results):
+        cur = torch.tensor(p, dtype=torch.int64).unsqueeze(0)
+        toks = []
+        for _ in range(8):
+            pos = torch.arange(cur.shape[1]).unsqueeze(0)
+            h = stages[0](cur, pos.contiguous(), return_logits=False)
+            lg = stages[1](h, pos.contiguous())
+            t = int(lg[0, -1].argmax())
+            toks.append(t)
+            cur = torch.cat([cur, torch.tensor([[t]])], 1)
+        assert o == toks, (o, toks)
