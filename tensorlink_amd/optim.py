@@ -77,6 +77,27 @@ class FusedAdamW:
                 p.grad = self.flat_grad[off:off + n].view(p.shape)
             off += n
 
+    @torch.no_grad()
+    def grad_norm_sq(self) -> torch.Tensor:
+        """Squared L2 norm of the flat gradient (one reduction kernel;
+        callers all-reduce it across ranks for the global norm)."""
+        return self.flat_grad.float().pow(2).sum()
+
+    @torch.no_grad()
+    def scale_grads(self, scale: float) -> None:
+        self.flat_grad.mul_(scale)
+
+    def clip_grad_norm_(self, max_norm: float,
+                        norm_sq: "torch.Tensor | None" = None) -> float:
+        """Clip by global norm over the flat buffer (semantics of
+        torch.nn.utils.clip_grad_norm_). Pass an all-reduced norm_sq for
+        the cross-rank global norm in pipeline training."""
+        ns = self.grad_norm_sq() if norm_sq is None else norm_sq
+        total = float(ns.sqrt())
+        if total > max_norm:
+            self.scale_grads(max_norm / (total + 1e-6))
+        return total
+
     def state_dict(self):
         return {"step": self.step_count, "exp_avg": self.exp_avg,
                 "exp_avg_sq": self.exp_avg_sq, "lr": self.lr}
@@ -101,3 +122,37 @@ class DistributedOptimizer:
 
     def zero_grad(self, set_to_none: bool = False):
         self.inner.zero_grad(set_to_none)
+
+
+class WarmupCosineLR:
+    """Linear warmup then cosine decay to min_lr — the standard LLM
+    schedule (the reference exposes raw optimizer kwargs only; a
+    scheduler is this build's addition for the training path)."""
+
+    def __init__(self, optimizer, max_lr: float, warmup_steps: int,
+                 total_steps: int, min_lr: float = 0.0):
+        self.opt = optimizer
+        self.max_lr = max_lr
+        self.warmup = max(1, warmup_steps)
+        self.total = total_steps
+        self.min_lr = min_lr
+        self.t = 0
+
+    def step(self) -> float:
+        import math
+        self.t += 1
+        if self.t <= self.warmup:
+            lr = self.max_lr * self.t / self.warmup
+        else:
+            frac = min(1.0, (self.t - self.warmup)
+                       / max(1, self.total - self.warmup))
+            lr = self.min_lr + 0.5 * (self.max_lr - self.min_lr) * (
+                1 + math.cos(math.pi * frac))
+        self.opt.lr = lr
+        return lr
+
+    def state_dict(self):
+        return {"t": self.t}
+
+    def load_state_dict(self, sd):
+        self.t = sd["t"]

@@ -392,7 +392,8 @@ class PipelineTrainer:
     def __init__(self, plan: StagePlan, rank: int, world: int, device=None,
                  init: str = "random", ckpt_dir: Optional[str] = None,
                  dtype=None, seed: int = 0, lr: float = 1e-4, group=None,
-                 rank_base: int = 0, **opt_kwargs):
+                 rank_base: int = 0, max_grad_norm: Optional[float] = None,
+                 **opt_kwargs):
         self.runner = PipelineRunner(plan, rank, world, device=device,
                                      init=init, ckpt_dir=ckpt_dir,
                                      dtype=dtype, seed=seed, group=group,
@@ -410,6 +411,8 @@ class PipelineTrainer:
         from tensorlink_amd.optim import FusedAdamW
         self.optimizer = FusedAdamW(self.stage.parameters(), lr=lr,
                                     **opt_kwargs)
+        self.max_grad_norm = max_grad_norm
+        self.lr_scheduler = None        # optional WarmupCosineLR
 
     def train_step(self, input_ids: Optional[torch.Tensor] = None,
                    labels: Optional[torch.Tensor] = None,
@@ -505,6 +508,16 @@ class PipelineTrainer:
 
         if self.grad_hook is not None:
             self.grad_hook(self)
+        if self.max_grad_norm is not None:
+            # global norm across pipeline stages: each rank reduces its
+            # flat buffer, then one scalar all-reduce
+            ns = self.optimizer.grad_norm_sq()
+            if P > 1:
+                import torch.distributed as dist
+                dist.all_reduce(ns, group=self.runner.group)
+            self.optimizer.clip_grad_norm_(self.max_grad_norm, norm_sq=ns)
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
         self.optimizer.step()
         self.optimizer.zero_grad()
 

@@ -207,3 +207,49 @@ def test_training_checkpoint_resume():
         t_b.load_checkpoint(d)
         b_losses = [t_b.train_step(b, labels=b) for b in bs[2:]]
     assert a_losses + b_losses == pytest.approx(ref_losses, rel=1e-5)
+
+
+def test_grad_clip_matches_torch():
+    """Flat-buffer clip_grad_norm_ reproduces torch.nn.utils semantics."""
+    import torch
+
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.optim import FusedAdamW
+    m = build_full_model(get_config("tiny"))
+    init_random_stage(m, dtype=torch.float32, seed=3)
+    for p in m.parameters():
+        p.requires_grad_(True)
+    opt = FusedAdamW(m.parameters(), lr=1e-3)
+    ids = torch.randint(0, 1024, (2, 12))
+    pos = torch.arange(12).unsqueeze(0).expand(2, -1).contiguous()
+    logits = m.head(m(ids, pos, training=True, return_logits=False))
+    (logits.float().pow(2).mean() * 50).backward()
+
+    ref_norm = torch.sqrt(sum(p.grad.float().pow(2).sum()
+                              for p in m.parameters()))
+    grads_ref = [p.grad.clone() * min(1.0, 0.5 / (float(ref_norm) + 1e-6))
+                 for p in m.parameters()]
+    total = opt.clip_grad_norm_(0.5)
+    assert total == pytest.approx(float(ref_norm), rel=1e-5)
+    for p, g in zip(m.parameters(), grads_ref):
+        torch.testing.assert_close(p.grad, g, atol=1e-7, rtol=1e-5)
+
+
+def test_warmup_cosine_schedule():
+    import math
+
+    from tensorlink_amd.optim import WarmupCosineLR
+
+    class _O:
+        lr = 0.0
+    o = _O()
+    sch = WarmupCosineLR(o, max_lr=1.0, warmup_steps=10, total_steps=110,
+                         min_lr=0.1)
+    lrs = [sch.step() for _ in range(110)]
+    assert lrs[0] == pytest.approx(0.1)          # 1/10 of max
+    assert lrs[9] == pytest.approx(1.0)          # end of warmup
+    assert lrs[59] == pytest.approx(0.1 + 0.45 * (
+        1 + math.cos(math.pi * 50 / 100)), rel=1e-6)
+    assert lrs[-1] == pytest.approx(0.1, abs=1e-3)
+    assert o.lr == lrs[-1]
