@@ -36,14 +36,14 @@ class _ReduceFromTP(torch.autograd.Function):
     outputs — each rank holds a partial sum, gradients are replicated)."""
 
     @staticmethod
-    def forward(ctx, x):
+    def forward(ctx, x, group):
         x = x.contiguous()
-        dist.all_reduce(x)
+        dist.all_reduce(x, group=group)
         return x
 
     @staticmethod
     def backward(ctx, dy):
-        return dy
+        return dy, None
 
 
 class _CopyToTP(torch.autograd.Function):
@@ -51,33 +51,37 @@ class _CopyToTP(torch.autograd.Function):
     inputs — x is replicated, each rank produces a partial dx)."""
 
     @staticmethod
-    def forward(ctx, x):
+    def forward(ctx, x, group):
+        ctx.group = group
         return x
 
     @staticmethod
     def backward(ctx, dy):
         dy = dy.contiguous()
-        dist.all_reduce(dy)
-        return dy
+        dist.all_reduce(dy, group=ctx.group)
+        return dy, None
 
 
-def _tp_active() -> bool:
-    return dist.is_initialized() and dist.get_world_size() > 1
+def _tp_active(group=None) -> bool:
+    return dist.is_initialized() and dist.get_world_size(group) > 1
 
 
 class AllReduceLinear(TLLinear):
     """Row-parallel output projection: local GEMM then sum across the TP
-    group. Differentiable — the training path routes the all-reduce
-    through :class:`_ReduceFromTP` so gradients flow (identity backward)."""
+    group (``self.group``; None = world). Differentiable — the training
+    path routes the all-reduce through :class:`_ReduceFromTP` so
+    gradients flow (identity backward)."""
+
+    group = None            # TP subgroup (TP x PP grids set it per-stage)
 
     def forward(self, x):
         y = super().forward(x)
-        if _tp_active():
+        if _tp_active(self.group):
             if torch.is_grad_enabled() and y.requires_grad:
-                y = _ReduceFromTP.apply(y)
+                y = _ReduceFromTP.apply(y, self.group)
             else:
                 y = y.contiguous()
-                dist.all_reduce(y)
+                dist.all_reduce(y, group=self.group)
         return y
 
 
@@ -87,9 +91,12 @@ class ColumnParallelLinear(TLLinear):
     the input passes through :class:`_CopyToTP` so each rank's partial dx
     is summed in backward (inference: pure identity, zero overhead)."""
 
+    group = None
+
     def forward(self, x):
-        if _tp_active() and torch.is_grad_enabled() and x.requires_grad:
-            x = _CopyToTP.apply(x)
+        if _tp_active(self.group) and torch.is_grad_enabled() \
+                and x.requires_grad:
+            x = _CopyToTP.apply(x, self.group)
         return super().forward(x)
 
 
@@ -257,3 +264,92 @@ class TPTrainer:
                     p.grad = g.div_(self.tp)
         self.opt.step()
         return float(loss.detach())
+
+
+# ---------------------------------------------------------------------------
+# TP x PP grid
+# ---------------------------------------------------------------------------
+def _swap_tp_linears(stage, tp_group):
+    """Swap a StageModel's projections for TP variants bound to
+    tp_group."""
+    for layer in stage.layers:
+        for holder, name, cls in (
+                (layer.self_attn, "o_proj", AllReduceLinear),
+                (layer.mlp, "down_proj", AllReduceLinear),
+                (layer.self_attn, "qkv_proj", ColumnParallelLinear),
+                (layer.mlp, "gate_up_proj", ColumnParallelLinear)):
+            old = getattr(holder, name)
+            new = cls(old.in_features, old.out_features,
+                      bias=old.bias is not None)
+            new.group = tp_group
+            setattr(holder, name, new)
+
+
+class TPPPRunner:
+    """2-D TP x PP inference grid on a world of pp*tp ranks.
+
+    Grid: ``rank = tp_rank * pp + stage`` — each pipeline replica's ranks
+    are consecutive (P2P's rank_base offset maps pipeline-local ranks),
+    and the TP group for stage s is the stride-pp set {t*pp + s}. Stage
+    weights come from the SAME per-stage seeded init a pure-PP run uses,
+    sliced by tp_rank, so a tp x pp grid reproduces the pp-only output
+    exactly. The pipeline replicas are NOT independent: every o/down
+    projection all-reduces across the tp group, which keeps all replicas
+    in lockstep layer by layer. Sampling must therefore be deterministic
+    across replicas: greedy, or an identical SamplingParams.seed.
+
+    The planner runs on the TP-LOCAL config, so per-stage memory checks
+    see the sharded sizes — folding TP into the stage-assignment grid.
+    """
+
+    def __init__(self, model, rank: int, world: int, tp: int, device=None,
+                 dtype=None, seed: int = 0):
+        import dataclasses as _dc
+
+        from tensorlink_amd.models.dense import build_stage
+        from tensorlink_amd.parallel.pipeline import PipelineRunner
+        from tensorlink_amd.parallel.planner import plan_for_world
+        config = (model if isinstance(model, ModelConfig)
+                  else get_config(model))
+        assert world % tp == 0, "world must divide tp"
+        pp = world // tp
+        self.pp, self.tp = pp, tp
+        self.stage_idx, self.tp_rank = rank % pp, rank // pp
+        self.device = device if device is not None else device_for_rank()
+        dtype = dtype or (torch.bfloat16 if self.device.type == "cuda"
+                          else torch.float32)
+        # same group-creation order on every rank (dist.new_group is
+        # collective over the world)
+        pp_groups = [dist.new_group(list(range(t * pp, (t + 1) * pp)))
+                     for t in range(tp)] if pp > 1 and tp > 1 else \
+            [None] * tp
+        tp_groups = [dist.new_group([t * pp + s for t in range(tp)])
+                     for s in range(pp)] if tp > 1 and pp > 1 else \
+            [None] * pp
+        self.tp_group = tp_groups[self.stage_idx]
+
+        local_cfg = local_config(config, tp)
+        plan = plan_for_world(local_cfg, pp)
+        self.runner = PipelineRunner(
+            plan, self.stage_idx, pp, device=self.device, init="empty",
+            dtype=dtype, group=pp_groups[self.tp_rank],
+            rank_base=self.tp_rank * pp)
+
+        # seed the FULL stage exactly as a pure-PP rank would, then slice
+        spec = plan.stage_for_rank(self.stage_idx)
+        full = build_stage(config, spec)
+        init_random_stage(full, device="cpu", dtype=dtype,
+                          seed=seed + self.stage_idx)
+        shards = shard_state(full.state_dict(), config, self.tp_rank, tp)
+        del full
+        if tp > 1:
+            _swap_tp_linears(self.runner.stage, self.tp_group)
+        self.runner.stage.load_state_dict(shards)
+        self.runner.stage.to(device=self.device, dtype=dtype)
+        self.runner.stage.eval()
+        self.config = self.runner.config
+
+    def generate(self, input_ids, sampling=None, **kw):
+        """SPMD over the whole grid; input_ids significant on every
+        pipeline replica's first rank (stage 0); tokens return on them."""
+        return self.runner.generate(input_ids, sampling, **kw)
