@@ -41,6 +41,7 @@ class Request:
     top_p: float = 1.0
     top_k: int = 0
     eos_token_id: Optional[int] = None
+    priority: int = 0                    # higher admits first
     tokens: "queue.Queue" = field(default_factory=queue.Queue)
     done: threading.Event = field(default_factory=threading.Event)
     error: Optional[str] = None
@@ -113,7 +114,11 @@ class ContinuousBatcher:
         self.slot_last = [0] * max_slots      # last sampled token
         self.slot_prompt: List[Optional[torch.Tensor]] = [None] * max_slots
         self.slot_filled = [0] * max_slots    # prompt tokens prefilled
-        self._queue: "queue.Queue[Request]" = queue.Queue()
+        # admission queue ordered by (priority desc, arrival): a
+        # high-priority request jumps the queue but never preempts a
+        # running slot
+        self._queue: "queue.PriorityQueue" = queue.PriorityQueue()
+        self._seq = 0
         self._wake = threading.Event()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -136,7 +141,8 @@ class ContinuousBatcher:
         if req.input_ids.numel() + req.max_new_tokens > self.max_ctx:
             keep = self.max_ctx - req.max_new_tokens
             req.input_ids = req.input_ids[-keep:]
-        self._queue.put(req)
+        self._seq += 1
+        self._queue.put((-req.priority, self._seq, req))
         self._wake.set()
         return req
 
@@ -235,10 +241,12 @@ class ContinuousBatcher:
                 if slot is None:
                     break
                 # admission control: need pages for prompt + first tokens
-                need = (self._queue.queue[0].input_ids.numel() + PAGE) // PAGE + 1
+                head = self._queue.queue[0][2]
+                # pages for the prompt plus its first generated token
+                need = (head.input_ids.numel() + PAGE) // PAGE
                 if self.cache.allocator.n_free < need:
                     break
-                req = self._queue.get_nowait()
+                req = self._queue.get_nowait()[2]
                 self.slots[slot] = req
                 self.slot_prompt[slot] = req.input_ids
                 self.slot_filled[slot] = 0
@@ -411,11 +419,12 @@ class PPContinuousBatcher(ContinuousBatcher):
                 slot = self._free_slot()
                 if slot is None:
                     break
-                need = (self._queue.queue[0].input_ids.numel()
-                        + PAGE) // PAGE + 1
+                head = self._queue.queue[0][2]
+                # pages for the prompt plus its first generated token
+                need = (head.input_ids.numel() + PAGE) // PAGE
                 if self.cache.allocator.n_free < need:
                     break
-                req = self._queue.get_nowait()
+                req = self._queue.get_nowait()[2]
                 self.slots[slot] = req
                 self.slot_prompt[slot] = req.input_ids
                 self.slot_filled[slot] = 0
