@@ -307,10 +307,10 @@ class PPContinuousBatcher(ContinuousBatcher):
                                      top_p=p, top_k=k)[0])
                 for i, (t, p, k) in enumerate(samps)]
 
-    def _isend0(self, t):
+    def _isend(self, t, dst):
         t = t.contiguous()
-        self._send_keep.append((self.runner.p2p.isend(t, 0), t))
-        if len(self._send_keep) > 8:
+        self._send_keep.append((self.runner.p2p.isend(t, dst), t))
+        if len(self._send_keep) > 16:
             w, _ = self._send_keep.pop(0)
             w.wait()
 
@@ -341,23 +341,33 @@ class PPContinuousBatcher(ContinuousBatcher):
                 hidden = self.stage(hidden, pos, kv_cache=view,
                                     return_logits=False)
             if not r.is_last:
-                r.p2p.send(hidden.contiguous(), r.next_rank)
+                self._isend(hidden, r.next_rank)
             elif end == s_total:
                 logits = self.stage.head(hidden[:, -1:]).squeeze(1)
                 tok = self._sample_rows(logits, [samp])
-                self._isend0(torch.tensor(tok, dtype=torch.int64,
-                                          device=dev))
+                self._isend(torch.tensor(tok, dtype=torch.int64,
+                                         device=dev), 0)
+        # decode: split the batch into up to `world` groups so successive
+        # groups overlap across stages (rank s works on group g while
+        # rank s-1 already runs group g+1 — all inter-stage transfers are
+        # isends, only recv blocks)
+        groups = []
         if decodes:
-            B = len(decodes)
-            slots = [d[0] for d in decodes]
+            n_groups = min(r.world, len(decodes))
+            gsz = (len(decodes) + n_groups - 1) // n_groups
+            groups = [decodes[i:i + gsz]
+                      for i in range(0, len(decodes), gsz)]
             for d in decodes:
                 self.cache.ensure(d[0], d[1] + 2)
-            lens = torch.tensor([d[1] for d in decodes], device=dev,
+        for sub in groups:
+            B = len(sub)
+            slots = [d[0] for d in sub]
+            lens = torch.tensor([d[1] for d in sub], device=dev,
                                 dtype=torch.int32)
             view = _SlotView(self.cache, slots, lens)
             pos = lens.unsqueeze(1)
             if r.is_first:
-                toks = torch.tensor([d[2] for d in decodes], device=dev,
+                toks = torch.tensor([d[2] for d in sub], device=dev,
                                     dtype=torch.int64).unsqueeze(1)
                 hidden = self.stage(toks, pos, kv_cache=view,
                                     return_logits=False)
@@ -366,22 +376,25 @@ class PPContinuousBatcher(ContinuousBatcher):
                 hidden = self.stage(hidden, pos, kv_cache=view,
                                     return_logits=False)
             if not r.is_last:
-                r.p2p.send(hidden.contiguous(), r.next_rank)
+                self._isend(hidden, r.next_rank)
             else:
                 logits = self.stage.head(hidden).squeeze(1)
-                new = self._sample_rows(logits, [d[3] for d in decodes])
-                self._isend0(torch.tensor(new, dtype=torch.int64,
-                                          device=dev))
+                new = self._sample_rows(logits, [d[3] for d in sub])
+                self._isend(torch.tensor(new, dtype=torch.int64,
+                                         device=dev), 0)
         # rank 0 collects the sampled tokens (isend on the last rank
-        # breaks the send/recv cycle between the two pipeline passes)
+        # breaks the send/recv cycle between pipeline passes; same-pair
+        # P2P is FIFO so arrival order matches issue order)
         new_tokens = None
         if r.is_first:
             if chunk is not None and chunk[2] == chunk[3]:   # final chunk
                 chunk_tok = int(r.p2p.recv((1,), torch.int64, r.world - 1,
                                            dev)[0])
-            if decodes:
-                new_tokens = r.p2p.recv((len(decodes),), torch.int64,
-                                        r.world - 1, dev).tolist()
+            if groups:
+                new_tokens = []
+                for sub in groups:
+                    new_tokens += r.p2p.recv((len(sub),), torch.int64,
+                                             r.world - 1, dev).tolist()
         return chunk_tok, new_tokens
 
     # ------------------------- follower (ranks 1..N-1) ------------------

@@ -600,11 +600,12 @@ def _pp_batch_worker(rank, world, port, q):
     if rank != 0:
         eng.worker_loop()
         return
-    eng.load_model("tiny", continuous=True, max_slots=4, max_ctx=256,
+    eng.load_model("tiny", continuous=True, max_slots=5, max_ctx=256,
                    prefill_chunk=16)
     b = eng.jobs["tiny"].batcher
     torch.manual_seed(33)
-    prompts = [torch.randint(0, 1024, (n,)) for n in (30, 9, 21)]
+    # 5 concurrent slots > world: exercises multi-group pipelined decode
+    prompts = [torch.randint(0, 1024, (n,)) for n in (30, 9, 21, 40, 5)]
     reqs = [b.submit(p, max_new_tokens=8) for p in prompts]
     outs = [rq.result(timeout=120) for rq in reqs]
     # releases are deferred one scheduler iteration (rank sync) — wait
