@@ -99,12 +99,13 @@ class DistributedModel(nn.Module):
                  world_size: int = 1, mode: str = "auto",
                  device=None, init: str = "random",
                  ckpt_dir: Optional[str] = None, seed: int = 0,
-                 lr: float = 1e-4, n_pipelines: int = 1):
+                 lr: float = 1e-4, n_pipelines: int = 1, tp: int = 1):
         super().__init__()
         self.model_name = model
         self.training_mode = training
         self.world_size = world_size
         self.n_pipelines = n_pipelines
+        self.tp = tp
         self._procs = []
         self._p2p = None
         self.device = device
@@ -139,10 +140,23 @@ class DistributedModel(nn.Module):
                        lr):
         """Under torchrun every rank constructs DistributedModel; rank 0
         is the user-facing instance, other ranks enter the worker loop
-        via serve_worker() (call it after construction on rank != 0)."""
+        via serve_worker() (call it after construction on rank != 0).
+
+        ``tp > 1`` shards each pipeline stage tensor-parallel across the
+        tp x (world//tp) grid (inference; random init — the grid slices
+        the same seeded per-stage init a pure-PP run uses)."""
         from tensorlink_amd.parallel.comm import init_distributed
         rank, _ = init_distributed()
         self._rank = rank
+        if self.tp > 1:
+            assert not training, "tp>1 training: use parallel.tp.TPTrainer"
+            from tensorlink_amd.parallel.tp import TPPPRunner
+            grid = TPPPRunner(model, rank, world, self.tp,
+                              device=self.device, seed=seed)
+            self._trainer = None
+            self._runner = grid.runner
+            self._p2p = self._runner.p2p
+            return
         plan = plan_for_world(model, world, training=training)
         if training:
             self._trainer = PipelineTrainer(plan, rank, world,
@@ -159,9 +173,22 @@ class DistributedModel(nn.Module):
 
     def serve_worker(self):
         """Worker loop for torchrun ranks != 0 (mirrors the spawned-
-        cluster workers)."""
+        cluster workers). tp>1: commands carry the inputs and travel over
+        the WORLD group (every pipeline replica's first rank needs the
+        prompt — its own pp-group broadcast source is itself)."""
         holder = self._trainer
         runner = self._runner
+        if self.tp > 1:
+            import torch.distributed as dist
+            while True:
+                box = [None]
+                dist.broadcast_object_list(box, src=0)
+                cmd = box[0]
+                if cmd[0] == "shutdown":
+                    return
+                assert cmd[0] == "generate"
+                ids = torch.tensor(cmd[1], dtype=torch.int64)
+                runner.generate(ids if runner.is_first else None, cmd[2])
         p2p = runner.p2p
         while True:
             cmd = p2p.broadcast_obj(None, src=0)
@@ -279,9 +306,15 @@ class DistributedModel(nn.Module):
             temperature=temperature if do_sample else 0.0, top_p=top_p,
             top_k=top_k, max_new_tokens=max_new_tokens,
             eos_token_id=eos_token_id)
-        self._bcast(("generate",))
-        out = self._runner.generate(input_ids, sp,
-                                    micro_batches=self.n_pipelines)
+        if self.tp > 1 and self.world_size > 1:
+            import torch.distributed as dist
+            box = [("generate", input_ids.tolist(), sp)]
+            dist.broadcast_object_list(box, src=0)
+            out = self._runner.generate(input_ids, sp)
+        else:
+            self._bcast(("generate",))
+            out = self._runner.generate(input_ids, sp,
+                                        micro_batches=self.n_pipelines)
         return torch.cat([input_ids.to(out.device), out], dim=1)
 
     # ------------------------------------------------------------------
@@ -322,7 +355,17 @@ class DistributedModel(nn.Module):
         return out_dir
 
     # ------------------------------------------------------------------
+    def _shutdown_grid(self):
+        import torch.distributed as dist
+        box = [("shutdown",)]
+        dist.broadcast_object_list(box, src=0)
+
     def shutdown(self):
+        if self.tp > 1 and self.world_size > 1 and not self._procs \
+                and torch.distributed.is_initialized() \
+                and getattr(self, "_rank", 0) == 0:
+            self._shutdown_grid()
+            return
         if self._procs:
             self._bcast(("shutdown",))
             for p in self._procs:

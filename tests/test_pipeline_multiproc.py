@@ -709,3 +709,32 @@ def test_tp2_pp2_grid_matches_pp2():
         logits = stages[1](h, pos)
         cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert cur[:, 12:].tolist() == outs[0]
+
+
+def _dm_tp_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.module import DistributedModel
+    m = DistributedModel("tiny", world_size=world, mode="torchrun", tp=2,
+                         device=torch.device("cpu"), seed=0)
+    torch.manual_seed(88)
+    ids = torch.randint(0, 1024, (1, 10))
+    if rank == 0:
+        out = m.generate(ids, max_new_tokens=4)
+        m.shutdown()
+        q.put((rank, out.tolist()))
+    else:
+        m.serve_worker()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_model_tp_grid():
+    """DistributedModel(tp=2) on a 4-rank torchrun-style world: the user
+    API drives the TP x PP grid and returns prompt+tokens on rank 0."""
+    outs = dict(_run_collect(_dm_tp_worker, 4, _port(23), n_results=1))
+    out = outs[0]
+    assert len(out) == 1 and len(out[0]) == 14    # 10 prompt + 4 new
