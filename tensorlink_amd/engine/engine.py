@@ -59,6 +59,10 @@ class EngineMetrics:
         self.latencies: collections.deque = collections.deque(maxlen=512)
         self.history: collections.deque = collections.deque(maxlen=1440)
         self._lock = threading.Lock()
+        self._auto_lock = threading.Lock()
+        self._auto_loaded: set = set()
+        self._auto_demand_mark: Dict[str, int] = {}
+        self._auto_cold: Dict[str, int] = {}   # unloaded-at demand level
         self._last_roll = time.time()
 
     def record(self, tokens: int, latency_s: float, error: bool = False):
@@ -103,6 +107,10 @@ class InferenceEngine:
         self.demand: Dict[str, int] = collections.Counter()
         self.p2p = P2P(rank, world) if world > 1 else None
         self._lock = threading.Lock()
+        self._auto_lock = threading.Lock()
+        self._auto_loaded: set = set()
+        self._auto_demand_mark: Dict[str, int] = {}
+        self._auto_cold: Dict[str, int] = {}   # unloaded-at demand level
         # serializes SPMD runner use: one in-flight generation per engine
         # (continuous batching is a later milestone)
         self._serve_lock = threading.Lock()
@@ -384,6 +392,45 @@ class InferenceEngine:
             free, total = torch.cuda.mem_get_info(self.device)
             s["gpu_memory"] = {"free": free, "total": total}
         return s
+
+    def manage_auto_loaded_models(self, max_auto: int = 2,
+                                  min_demand: int = 1) -> None:
+        """Demand-driven auto load/unload of the configured default
+        models (reference ``ml/validator.py:278``:
+        ``_manage_auto_loaded_models`` keeps DEFAULT_MODELS hot and drops
+        cold ones). Called periodically — e.g. from a Watchdog tick.
+
+        - every configured default model not yet loaded is loaded (up to
+          ``max_auto`` auto-loaded models);
+        - an auto-loaded model whose demand since the last sweep is
+          below ``min_demand`` is unloaded (explicitly-loaded models are
+          never touched).
+        """
+        from tensorlink_amd.config import DEFAULT_MODELS
+        with self._auto_lock:
+            # unload cold auto-loaded models
+            for name in list(self._auto_loaded):
+                if self.demand.get(name, 0) -                         self._auto_demand_mark.get(name, 0) < min_demand                         and name in self.jobs:
+                    self.unload_model(name)
+                    self._auto_loaded.discard(name)
+                    # cooled: only reload once demand grows again
+                    self._auto_cold[name] = self.demand.get(name, 0)
+            # load configured defaults
+            for name in DEFAULT_MODELS:
+                if len(self._auto_loaded) >= max_auto:
+                    break
+                if name not in self.jobs:
+                    if name in self._auto_cold and \
+                            self.demand.get(name, 0) <= \
+                            self._auto_cold[name]:
+                        continue
+                    try:
+                        self.load_model(name)
+                        self._auto_loaded.add(name)
+                    except Exception:   # unknown config etc. — skip
+                        continue
+            for name in self._auto_loaded:
+                self._auto_demand_mark[name] = self.demand.get(name, 0)
 
     def model_demand(self) -> Dict[str, int]:
         return dict(self.demand)

@@ -232,3 +232,31 @@ def test_runner_trace_env(tmp_path, monkeypatch):
     data = json.load(open(path))
     names = [e["name"] for e in data["traceEvents"]]
     assert "prefill" in names and "decode" in names
+
+
+def test_auto_loaded_model_management(monkeypatch):
+    """Demand-driven auto load/unload (reference
+    _manage_auto_loaded_models, ml/validator.py:278): defaults load on
+    the sweep, cold auto-loads unload, hot ones and explicit loads
+    stay."""
+    import torch
+
+    import tensorlink_amd.config as cfg
+    from tensorlink_amd.engine.engine import InferenceEngine
+    monkeypatch.setattr(cfg, "DEFAULT_MODELS", ["tiny"])
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny-moe")                       # explicit
+    eng.manage_auto_loaded_models()
+    assert "tiny" in eng.jobs                        # auto-loaded
+
+    # demand arrives -> stays on next sweep
+    eng.generate({"hf_name": "tiny", "message": "x", "max_new_tokens": 2,
+                  "do_sample": False, "output_format": "simple"})
+    eng.manage_auto_loaded_models()
+    assert "tiny" in eng.jobs
+
+    # no demand since the mark -> unloaded; explicit model untouched
+    eng.manage_auto_loaded_models()
+    assert "tiny" not in eng.jobs
+    assert "tiny-moe" in eng.jobs
+    eng.unload_model("tiny-moe")
