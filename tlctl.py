@@ -33,7 +33,9 @@ def cmd_serve(args):
     models = args.model or cfg.ml.default_models
     for name in models:
         print(f"loading {name} ...")
-        engine.load_model(name)
+        engine.load_model(name, continuous=args.continuous,
+                          max_slots=args.max_slots, max_ctx=args.max_ctx,
+                          prefill_chunk=args.prefill_chunk)
     keeper = StateKeeper(engine)
     keeper.load_previous_state()
     keeper.start()
@@ -81,6 +83,12 @@ def main():
     s.add_argument("--model", action="append")
     s.add_argument("--port", type=int)
     s.add_argument("--config")
+    s.add_argument("--continuous", action="store_true",
+                   help="continuous batching (slot scheduler; PP-aware "
+                        "when world > 1)")
+    s.add_argument("--max-slots", type=int, default=16)
+    s.add_argument("--max-ctx", type=int, default=4096)
+    s.add_argument("--prefill-chunk", type=int, default=None)
 
     pl = sub.add_parser("plan")
     pl.add_argument("--model", required=True)
