@@ -186,6 +186,14 @@ class InferenceEngine:
         with self._lock:
             if name in self.jobs:
                 return self.jobs[name]
+            if (kwargs.get("continuous") and self.world > 1
+                    and any(j.batcher is not None
+                            for j in self.jobs.values())):
+                # follower ranks serve ONE PP batcher command stream at a
+                # time (single collective channel per group)
+                raise ValueError(
+                    "one continuous PP job per engine; unload the "
+                    "existing one first")
             init = init or self.default_init
             self._bcast(("load", name, init, ckpt_dir, kwargs))
             try:
