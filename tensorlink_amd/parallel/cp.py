@@ -7,12 +7,17 @@ ring over xGMI P2P. Causality prunes the ring: chunk j's K/V only visits
 ranks i >= j, and partial results merge with the standard online-softmax
 (m, l, o) combination, so the result equals single-rank attention.
 
-Scope: inference prefill (logits) and decode after a CP prefill: each
-rank keeps the K/V of ITS sequence chunk (the cache is sequence-sharded,
-so cache memory divides by cp); decode tokens append to the last rank
-and every step merges the per-rank partial attentions with one
-all-gather of (o, m, l). The training backward is a roadmap item; the
-kernel-grade version would fuse the merge into prefill_attn.hip.
+Scope: inference prefill (logits), decode after a CP prefill (each rank
+keeps the K/V of ITS sequence chunk — the cache is sequence-sharded, so
+cache memory divides by cp; decode tokens append to the last rank and
+every step merges the per-rank partial attentions with one all-gather of
+(o, m, l)), and TRAINING: the ring rotation is an autograd Function
+(:class:`_RingPass` — backward rotates gradients the opposite way), the
+online-softmax partials/merge are plain differentiable torch math with
+the row max detached (softmax is shift-invariant, so detaching is
+exact), and per-chunk losses + an all-reduce grad sum reproduce the
+full-sequence gradient. The kernel-grade version would fuse the merge
+into prefill_attn.hip.
 """
 
 from __future__ import annotations
@@ -239,3 +244,169 @@ class CPRunner:
             cur = lg.argmax(-1)
             out[:, t] = cur
         return out
+
+
+class _RingPass(torch.autograd.Function):
+    """Differentiable ring rotation: forward sends x to rank+1 and
+    receives from rank-1; backward routes the received tensor's gradient
+    back to its origin (send to rank-1, receive own grad from rank+1).
+    SPMD-symmetric: every rank executes the same pass sequence, so the
+    blocking recv always has a matching isend."""
+
+    @staticmethod
+    def forward(ctx, x, rank, world):
+        ctx.ring = (rank, world)
+        y = torch.empty_like(x)
+        w = dist.isend(x.contiguous(), (rank + 1) % world)
+        dist.recv(y, (rank - 1) % world)
+        w.wait()
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        rank, world = ctx.ring
+        dy = dy.contiguous()
+        dx = torch.empty_like(dy)
+        w = dist.isend(dy, (rank - 1) % world)
+        dist.recv(dx, (rank + 1) % world)
+        w.wait()
+        return dx, None, None
+
+
+def _partial_attn_train(q, k, v, scale, causal_diag):
+    """Differentiable version of :func:`_partial_attn` (row max
+    detached — exact, since softmax is shift-invariant)."""
+    B, Sq, Hq, D = q.shape
+    Hkv = k.shape[2]
+    rep = Hq // Hkv
+    qf = q.float().permute(0, 2, 1, 3)
+    kf = k.float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
+    vf = v.float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal_diag:
+        Sk = k.shape[1]
+        mask = torch.triu(torch.ones(Sq, Sk, dtype=torch.bool,
+                                     device=q.device), 1)
+        scores = scores.masked_fill(mask, float("-inf"))
+    m = scores.amax(-1).detach()
+    p = torch.exp(scores - m.unsqueeze(-1))
+    p = torch.nan_to_num(p, nan=0.0)
+    l = p.sum(-1)
+    o = torch.matmul(p, vf)
+    return o.permute(0, 2, 1, 3), m, l
+
+
+class CPTrainer:
+    """Context-parallel training (weights replicated, sequence sharded).
+
+    Every rank holds the full model and trains on its sequence chunk;
+    attention runs the differentiable K/V ring. Losses are per-chunk
+    sums normalized by the GLOBAL token count, so after backward each
+    rank holds the partial dW that flows through its chunk's compute —
+    one all-reduce SUM over the flat gradient completes dL/dW exactly
+    (weights replicated => gradients must be summed, not averaged).
+    """
+
+    def __init__(self, model, rank: int, cp: int, device=None, seed: int = 0,
+                 lr: float = 1e-3, weight_decay: float = 0.01):
+        from tensorlink_amd.optim import FusedAdamW
+        self.rank, self.cp = rank, cp
+        self.device = device if device is not None else device_for_rank()
+        self.dtype = (torch.bfloat16 if self.device.type == "cuda"
+                      else torch.float32)
+        config = (model if isinstance(model, ModelConfig)
+                  else get_config(model))
+        self.stage = build_full_model(config)
+        init_random_stage(self.stage, device=self.device, dtype=self.dtype,
+                          seed=seed)
+        self.stage.train()
+        for p in self.stage.parameters():
+            p.requires_grad_(True)
+        self.config = config
+        self.opt = FusedAdamW(self.stage.parameters(), lr=lr,
+                              weight_decay=weight_decay)
+
+    def _ring_attention_train(self, attn, x, positions):
+        B, sc, H = x.shape
+        qkv = attn.qkv_proj(x)
+        q, k, v = attn._split_qkv(qkv, B, sc)
+        flat_pos = positions.reshape(-1)
+        q2, k2 = ops.apply_rope(q.reshape(B * sc, attn.n_heads, -1),
+                                k.reshape(B * sc, attn.n_kv, -1),
+                                flat_pos, attn.inv_freq)
+        q = q2.view(B, sc, attn.n_heads, -1)
+        k = k2.view(B, sc, attn.n_kv, -1)
+        acc = None
+        cur_k, cur_v = k, v
+        for r in range(self.cp):
+            src = (self.rank - r) % self.cp
+            if src == self.rank:
+                acc = _merge(acc, _partial_attn_train(q, cur_k, cur_v,
+                                                      attn.scale, True))
+            elif src < self.rank:
+                acc = _merge(acc, _partial_attn_train(q, cur_k, cur_v,
+                                                      attn.scale, False))
+            if r < self.cp - 1 and self.cp > 1:
+                cur_k = _RingPass.apply(cur_k, self.rank, self.cp)
+                cur_v = _RingPass.apply(cur_v, self.rank, self.cp)
+        o, m, l = acc
+        out = o / l.permute(0, 2, 1).unsqueeze(-1).clamp(min=1e-30)
+        out = out.to(x.dtype).reshape(B, sc, -1)
+        # anchor the final (possibly unused) ring buffers so every rank
+        # runs the SAME backward pass sequence (a pruned chunk's
+        # _RingPass would otherwise never fire its backward and deadlock
+        # the symmetric grad rotation); the added grads are exactly zero
+        if self.cp > 1:
+            out = out + 0.0 * (cur_k.sum() + cur_v.sum()).to(out.dtype)
+        return attn.o_proj(out)
+
+    def _forward_chunk(self, ids_chunk, positions):
+        stage = self.stage
+        eps = self.config.rms_norm_eps
+        hidden = stage.embed(ids_chunk)
+        for layer in stage.layers:
+            h = ops.rmsnorm(hidden, layer.input_layernorm.to(hidden.dtype),
+                            eps)
+            hidden = hidden + self._ring_attention_train(layer.self_attn,
+                                                         h, positions)
+            h = ops.rmsnorm(hidden,
+                            layer.post_attention_layernorm.to(hidden.dtype),
+                            eps)
+            hidden = hidden + layer.mlp(h)
+        return stage.head(hidden)
+
+    def train_step(self, input_ids: torch.Tensor) -> float:
+        """One optimizer step; input_ids [B, S] (full sequence, identical
+        on every rank; S divisible by cp). Returns the global mean loss."""
+        import torch.nn.functional as F
+        B, S = input_ids.shape
+        assert S % self.cp == 0
+        sc = S // self.cp
+        ids = input_ids.to(self.device)
+        t0 = self.rank * sc
+        chunk = ids[:, t0:t0 + sc]
+        positions = (torch.arange(sc, device=self.device,
+                                  dtype=torch.int32) + t0) \
+            .unsqueeze(0).expand(B, -1).contiguous()
+        self.opt.zero_grad()
+        logits = self._forward_chunk(chunk, positions)
+        # shifted CE with one-token lookahead across the chunk boundary;
+        # the global last position has no label
+        if self.rank == self.cp - 1:
+            lg = logits[:, :-1]
+            lb = ids[:, t0 + 1:t0 + sc]
+        else:
+            lg = logits
+            lb = ids[:, t0 + 1:t0 + sc + 1]
+        n_total = B * (S - 1)
+        loss = F.cross_entropy(lg.reshape(-1, lg.shape[-1]).float(),
+                               lb.reshape(-1), reduction="sum") / n_total
+        loss.backward()
+        if self.cp > 1:
+            dist.all_reduce(self.opt.flat_grad)     # SUM of partial dW
+            total = loss.detach().clone()
+            dist.all_reduce(total)
+        else:
+            total = loss.detach()
+        self.opt.step()
+        return float(total)
