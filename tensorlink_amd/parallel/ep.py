@@ -126,6 +126,14 @@ class EPMoEMLP(nn.Module):
             return self._forward_alltoall(x)
         B, S, H = x.shape
         flat = x.reshape(-1, H)
+        if (dist.is_initialized() and dist.get_world_size() > 1
+                and torch.is_grad_enabled() and flat.requires_grad):
+            # f-collective on the block input (training): each rank's
+            # graph holds only its local experts' (and its own gate
+            # copy's) dL/dx terms, so dx must be SUMMED across the EP
+            # group in backward
+            from tensorlink_amd.parallel.tp import _CopyToTP
+            flat = _CopyToTP.apply(flat, None)
         weights, idx = ops.moe_topk_router(self.gate(flat), self.top_k)
         weights = weights.to(x.dtype)
         out = torch.zeros_like(flat)
@@ -135,10 +143,16 @@ class EPMoEMLP(nn.Module):
             tok, slot = mask.nonzero(as_tuple=True)
             if tok.numel() == 0:
                 continue
-            out.index_add_(0, tok, expert(flat[tok]) * weights[tok, slot,
-                                                              None])
+            out = out.index_add(0, tok, expert(flat[tok]) * weights[tok,
+                                                                    slot,
+                                                                    None])
         if dist.is_initialized() and dist.get_world_size() > 1:
-            dist.all_reduce(out)
+            if torch.is_grad_enabled() and out.requires_grad:
+                from tensorlink_amd.parallel.tp import _ReduceFromTP
+                out = _ReduceFromTP.apply(out, None)
+            else:
+                out = out.contiguous()
+                dist.all_reduce(out)
         return out.reshape(B, S, H)
 
 
@@ -207,3 +221,58 @@ class EPRunner:
             out[:, t] = cur
             positions += 1
         return out
+
+
+class EPTrainer:
+    """Expert-parallel training (partial-sum mode).
+
+    Every rank runs the full replicated batch; the MoE partial sums pass
+    through the differentiable all-reduce (identity backward), so after
+    ``loss.backward()``:
+
+    - expert grads are complete locally (only this rank computed those
+      experts' contributions);
+    - non-MoE grads (attention, norms, embeddings, head) are complete
+      AND identical on every rank (the computation is replicated);
+    - router/gate grads are PARTIAL (each rank's graph only saw its own
+      experts' weight usages) — one all-reduce SUM per gate completes
+      them.
+    """
+
+    def __init__(self, model, rank: int, ep: int, device=None, seed: int = 0,
+                 lr: float = 1e-3, weight_decay: float = 0.01):
+        from tensorlink_amd.optim import FusedAdamW
+        self.rank, self.ep = rank, ep
+        self.device = device if device is not None else device_for_rank()
+        self.stage = build_ep_model(model, rank, ep, device=self.device,
+                                    seed=seed)
+        self.stage.train()
+        for p in self.stage.parameters():
+            p.requires_grad_(True)
+        self.config = self.stage.config
+        self.gate_params = [layer.mlp.gate.weight
+                            for layer in self.stage.layers
+                            if isinstance(layer.mlp, EPMoEMLP)]
+        self.opt = FusedAdamW(self.stage.parameters(), lr=lr,
+                              weight_decay=weight_decay)
+
+    def train_step(self, input_ids: torch.Tensor) -> float:
+        from tensorlink_amd import ops as tl_ops
+        ids = input_ids.to(self.device)
+        B, S = ids.shape
+        pos = torch.arange(S, device=self.device,
+                           dtype=torch.int32).unsqueeze(0).expand(B, -1)
+        self.opt.zero_grad()
+        hidden = self.stage(ids, pos.contiguous(), training=True,
+                            return_logits=False)
+        logits = self.stage.head(hidden)
+        loss = tl_ops.causal_lm_loss(logits, ids)
+        loss.backward()
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            for g in self.gate_params:
+                if g.grad is not None:
+                    # in-place: .grad is a view into the optimizer's
+                    # flat gradient buffer
+                    dist.all_reduce(g.grad)  # SUM completes router grads
+        self.opt.step()
+        return float(loss.detach())

@@ -792,3 +792,56 @@ def test_cp2_training_matches_single_rank():
         opt.step()
         ref_losses.append(float(loss.detach()))
     assert outs[0] == pytest.approx(ref_losses, rel=2e-3)
+
+
+def _ep_train_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.ep import EPTrainer
+    init_distributed(backend="gloo")
+    tr = EPTrainer("tiny-moe", rank, world, device=torch.device("cpu"),
+                   seed=15, lr=1e-3)
+    torch.manual_seed(95)
+    batches = [torch.randint(0, 1024, (2, 16)) for _ in range(3)]
+    losses = [tr.train_step(b) for b in batches]
+    q.put((rank, losses))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ep2_training_matches_single_rank():
+    """EP training: differentiable partial-sum all-reduce + router grad
+    sum reproduce the single-rank MoE loss trajectory."""
+    outs = dict(_run_collect(_ep_train_worker, 2, _port(27)))
+    assert outs[0] == pytest.approx(outs[1], rel=1e-5)
+
+    import torch
+    from tensorlink_amd import ops as tl_ops
+    from tensorlink_amd.models.configs import get_config
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.optim import FusedAdamW
+    m = build_full_model(get_config("tiny-moe"))
+    init_random_stage(m, device="cpu", dtype=torch.float32, seed=15)
+    m.train()
+    for p in m.parameters():
+        p.requires_grad_(True)
+    opt = FusedAdamW(m.parameters(), lr=1e-3, weight_decay=0.01)
+    torch.manual_seed(95)
+    batches = [torch.randint(0, 1024, (2, 16)) for _ in range(3)]
+    ref_losses = []
+    for ids in batches:
+        B, S = ids.shape
+        pos = torch.arange(S, dtype=torch.int32
+                           ).unsqueeze(0).expand(B, -1).contiguous()
+        opt.zero_grad()
+        logits = m.head(m(ids, pos, training=True, return_logits=False))
+        loss = tl_ops.causal_lm_loss(logits, ids)
+        loss.backward()
+        opt.step()
+        ref_losses.append(float(loss.detach()))
+    assert outs[0] == pytest.approx(ref_losses, rel=2e-3)
