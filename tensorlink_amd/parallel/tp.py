@@ -369,7 +369,14 @@ class TPPPTrainer:
     """
 
     def __init__(self, model, rank: int, world: int, tp: int, device=None,
-                 dtype=None, seed: int = 0, lr: float = 1e-4, **opt_kwargs):
+                 dtype=None, seed: int = 0, lr: float = 1e-4,
+                 grid_base: int = 0, pp_groups=None, tp_groups=None,
+                 **opt_kwargs):
+        """rank/world are GRID-local (0..tp*pp). grid_base offsets them
+        into the global rank space, and pp_groups/tp_groups may be
+        pre-built by an outer 3-D (DP) wrapper — dist.new_group is
+        collective over the WORLD, so nested grids must hoist group
+        creation."""
         from tensorlink_amd.models.dense import build_stage
         from tensorlink_amd.optim import FusedAdamW
         from tensorlink_amd.parallel.pipeline import PipelineTrainer
@@ -383,12 +390,14 @@ class TPPPTrainer:
         self.device = device if device is not None else device_for_rank()
         dtype = dtype or (torch.bfloat16 if self.device.type == "cuda"
                           else torch.float32)
-        pp_groups = [dist.new_group(list(range(t * pp, (t + 1) * pp)))
-                     for t in range(tp)] if pp > 1 and tp > 1 else \
-            [None] * tp
-        tp_groups = [dist.new_group([t * pp + s for t in range(tp)])
-                     for s in range(pp)] if tp > 1 and pp > 1 else \
-            [None] * pp
+        if pp_groups is None:
+            pp_groups = [dist.new_group(
+                [grid_base + t * pp + s for s in range(pp)])
+                for t in range(tp)] if pp > 1 and tp > 1 else [None] * tp
+        if tp_groups is None:
+            tp_groups = [dist.new_group(
+                [grid_base + t * pp + s for t in range(tp)])
+                for s in range(pp)] if tp > 1 and pp > 1 else [None] * pp
         self.tp_group = tp_groups[self.stage_idx]
 
         local_cfg = local_config(config, tp)
@@ -396,7 +405,7 @@ class TPPPTrainer:
         self.trainer = PipelineTrainer(
             plan, self.stage_idx, pp, device=self.device, init="empty",
             dtype=dtype, seed=seed, lr=lr, group=pp_groups[self.tp_rank],
-            rank_base=self.tp_rank * pp, **opt_kwargs)
+            rank_base=grid_base + self.tp_rank * pp, **opt_kwargs)
         stage = self.trainer.stage
 
         spec = plan.stage_for_rank(self.stage_idx)
@@ -435,3 +444,58 @@ class TPPPTrainer:
         """SPMD over the whole grid; input_ids/labels significant on
         every replica's first rank (pass the same batch to each)."""
         return self.trainer.train_step(input_ids, labels, n_micro=n_micro)
+
+
+class Hybrid3DTrainer:
+    """DP x TP x PP 3-D training grid on world = dp * tp * pp ranks.
+
+    Rank layout: ``rank = dp_rank * (tp*pp) + tp_rank * pp + stage``.
+    Each DP replica is a TP x PP grid training its shard of the global
+    batch; before the optimizer step the gradients are (1) tp-synced
+    for replicated params (inside TPPPTrainer's hook), then (2)
+    all-reduce-AVERAGED across the dp peers holding the same
+    (tp_rank, stage) shard. Averaging matches mean-reduction losses, so
+    the trajectory equals a single replica training on the concatenated
+    batch. All dist.new_group calls are hoisted here in one
+    deterministic order (the call is collective over the world).
+    """
+
+    def __init__(self, model, rank: int, world: int, dp: int, tp: int,
+                 device=None, dtype=None, seed: int = 0, lr: float = 1e-4,
+                 **opt_kwargs):
+        rep = world // dp
+        pp = rep // tp
+        assert dp * tp * pp == world
+        self.dp, self.tp, self.pp = dp, tp, pp
+        self.dp_rank = rank // rep
+        lrank = rank % rep
+        g = lambda ranks: dist.new_group(ranks) if len(ranks) > 1 else None
+        pp_groups_all = [[g([d * rep + t * pp + s for s in range(pp)])
+                          for t in range(tp)] for d in range(dp)]
+        tp_groups_all = [[g([d * rep + t * pp + s for t in range(tp)])
+                          for s in range(pp)] for d in range(dp)]
+        dp_groups = [[g([d * rep + t * pp + s for d in range(dp)])
+                      for s in range(pp)] for t in range(tp)]
+        self.inner = TPPPTrainer(
+            model, lrank, rep, tp, device=device, dtype=dtype, seed=seed,
+            lr=lr, grid_base=self.dp_rank * rep,
+            pp_groups=pp_groups_all[self.dp_rank],
+            tp_groups=tp_groups_all[self.dp_rank], **opt_kwargs)
+        self.dp_group = dp_groups[self.inner.tp_rank][self.inner.stage_idx]
+        self.config = self.inner.config
+
+        tp_hook = self.inner.trainer.grad_hook
+
+        def _hook(tr):
+            if tp_hook is not None:
+                tp_hook(tr)
+            if self.dp > 1:
+                dist.all_reduce(tr.optimizer.flat_grad,
+                                group=self.dp_group)
+                tr.optimizer.flat_grad.div_(self.dp)
+        self.inner.trainer.grad_hook = _hook
+
+    def train_step(self, input_ids=None, labels=None, n_micro=None):
+        """input_ids/labels = THIS dp replica's batch shard, significant
+        on the replica's first rank. Returns the replica-shard loss."""
+        return self.inner.train_step(input_ids, labels, n_micro=n_micro)

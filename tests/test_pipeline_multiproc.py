@@ -894,3 +894,57 @@ def test_tp2_pp2_training_matches_pp2():
     ref = dict(_run_collect(_pp2_train_ref_worker, 2, _port(31)))
     assert grid[0] == pytest.approx(grid[2], rel=1e-5)  # replicas agree
     assert grid[0] == pytest.approx(ref[0], rel=2e-3)
+
+
+def _h3d_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.tp import Hybrid3DTrainer
+    init_distributed(backend="gloo")
+    tr = Hybrid3DTrainer("tiny", rank, world, dp=2, tp=2,
+                         device=torch.device("cpu"), seed=0, lr=1e-3)
+    torch.manual_seed(103)
+    full = [torch.randint(0, 1024, (4, 16)) for _ in range(2)]
+    shards = [b[tr.dp_rank * 2:(tr.dp_rank + 1) * 2] for b in full]
+    losses = [tr.train_step(b, labels=b) for b in shards]
+    q.put((rank, (tr.dp_rank, losses)))
+    dist.destroy_process_group()
+
+
+def _pp2_full_ref_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    from tensorlink_amd.parallel.planner import plan_for_world
+    init_distributed(backend="gloo")
+    plan = plan_for_world("tiny", 2, training=True)
+    tr = PipelineTrainer(plan, rank, 2, device=torch.device("cpu"),
+                         seed=0, lr=1e-3)
+    torch.manual_seed(103)
+    full = [torch.randint(0, 1024, (4, 16)) for _ in range(2)]
+    losses = [tr.train_step(b, labels=b) for b in full]
+    q.put((rank, losses))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_dp2_tp2_pp2_matches_pp2_full_batch():
+    """3-D grid (8 ranks, dp2 x tp2 x pp2): per-shard losses averaged
+    across the DP replicas reproduce the pure-PP2 full-batch loss
+    trajectory (grad averaging == mean-reduction on the union batch)."""
+    grid = dict(_run_collect(_h3d_worker, 8, _port(33), timeout=400))
+    ref = dict(_run_collect(_pp2_full_ref_worker, 2, _port(35)))
+    by_dp = {}
+    for rank, (dp_rank, losses) in grid.items():
+        by_dp.setdefault(dp_rank, losses)
+        assert losses == pytest.approx(by_dp[dp_rank], rel=1e-5)
+    mean = [(a + b) / 2 for a, b in zip(by_dp[0], by_dp[1])]
+    assert mean == pytest.approx(ref[0], rel=2e-3)
