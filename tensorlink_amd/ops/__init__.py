@@ -344,3 +344,6 @@ def sample_token(logits, *, temperature=1.0, top_p=1.0, top_k=0,
 moe_topk_router = ref.moe_topk_router
 causal_lm_loss = ref.causal_lm_loss
 rope_cos_sin = ref.rope_cos_sin
+
+
+chunked_causal_lm_loss = ref.chunked_causal_lm_loss

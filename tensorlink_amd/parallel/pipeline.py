@@ -413,6 +413,10 @@ class PipelineTrainer:
                                     **opt_kwargs)
         self.max_grad_norm = max_grad_norm
         self.lr_scheduler = None        # optional WarmupCosineLR
+        # chunked CE for big vocabularies (memory); plain CE for tiny
+        # test configs (keeps exact parity with existing expectations)
+        self._chunked_ce = (self.stage.has_head
+                            and plan.config.vocab_size >= 32000)
 
     def train_step(self, input_ids: Optional[torch.Tensor] = None,
                    labels: Optional[torch.Tensor] = None,
@@ -466,17 +470,18 @@ class PipelineTrainer:
 
         def fwd(i):
             s, e = i * b, (i + 1) * b
+            want_logits = is_last and not self._chunked_ce
             if is_first:
                 inp = None
                 out = self.stage(input_ids[s:e], pos, training=True,
-                                 return_logits=is_last)
+                                 return_logits=want_logits)
             else:
                 inp = self.p2p.recv((b, S, self.H), self.dtype, r - 1,
                                     self.device).requires_grad_(True)
                 out = self.stage(inp, pos, training=True,
-                                 return_logits=is_last)
+                                 return_logits=want_logits)
             if is_last:
-                loss = tl_ops.causal_lm_loss(out, labels_last[s:e]) / M
+                loss = self._loss(out, labels_last[s:e]) / M
                 losses.append(loss.detach())
                 stash[i] = (inp, loss)
             else:
@@ -532,6 +537,24 @@ class PipelineTrainer:
     # loss.backward(), which routes here (reference CustomAutogradRouter,
     # ml/module.py:126-144).
     # ------------------------------------------------------------------
+    def _loss(self, out, labels):
+        """CE over one micro-batch. Large vocabularies use the chunked
+        head-GEMM+CE (no [B,S,V] logits materialization — saves
+        ~0.6 MB/token at Qwen vocab); small ones take the plain path.
+        Both are exactly F.cross_entropy(head(hidden), labels)."""
+        from tensorlink_amd import ops as tl_ops
+        if self._chunked_ce:
+            # apply the final norm here (stage.head fuses norm+GEMM;
+            # the chunked CE takes the raw head weight)
+            h = tl_ops.rmsnorm(out, self.stage.norm.to(out.dtype),
+                               self.stage.config.rms_norm_eps)
+            w = (self.stage.embed_tokens.weight
+                 if self.stage.config.tie_word_embeddings
+                 and self.stage.has_embedding
+                 else self.stage.lm_head.weight)
+            return tl_ops.chunked_causal_lm_loss(h, w, labels)
+        return tl_ops.causal_lm_loss(out, labels)
+
     # ------------------------------------------------------------------
     # training resume: weights + optimizer moments + step count
     # (the reference checkpoints weights only — parameter retrieval,

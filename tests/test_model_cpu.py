@@ -253,3 +253,27 @@ def test_warmup_cosine_schedule():
         1 + math.cos(math.pi * 50 / 100)), rel=1e-6)
     assert lrs[-1] == pytest.approx(0.1, abs=1e-3)
     assert o.lr == lrs[-1]
+
+
+def test_trainer_chunked_ce_matches_plain():
+    """The chunked head-GEMM+CE loss path produces the identical loss
+    trajectory to the full-logits path (same seeds, 2 steps)."""
+    import torch
+
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    from tensorlink_amd.parallel.planner import plan_for_world
+
+    def run(chunked):
+        plan = plan_for_world("tiny", 1)
+        t = PipelineTrainer(plan, 0, 1, device=torch.device("cpu"),
+                            seed=7, lr=1e-3)
+        t._chunked_ce = chunked
+        g = torch.Generator().manual_seed(55)
+        out = []
+        for _ in range(2):
+            b = torch.randint(0, 1024, (2, 16), generator=g)
+            out.append(t.train_step(b, labels=b))
+        return out
+
+    a, b = run(False), run(True)
+    assert a == pytest.approx(b, rel=1e-5)

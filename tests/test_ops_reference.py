@@ -138,3 +138,31 @@ def test_causal_lm_loss_shift():
     labels = torch.tensor([[9, 5, 7]])
     loss = ref.causal_lm_loss(logits, labels)
     assert loss.item() < 1e-4
+
+
+def test_chunked_ce_matches_full():
+    """Chunked CE (loss + d_hidden + d_head) == full-logits CE exactly,
+    incl. ignore_index padding and chunk boundaries not dividing N."""
+    import torch
+
+    from tensorlink_amd.ops import reference as ref
+    torch.manual_seed(4)
+    B, S, H, V = 2, 13, 32, 97
+    hidden = torch.randn(B, S, H, requires_grad=True)
+    w = torch.randn(V, H, requires_grad=True)
+    labels = torch.randint(0, V, (B, S))
+    labels[0, 5:8] = -100
+
+    loss_c = ref.chunked_causal_lm_loss(hidden, w, labels, chunk=7)
+    loss_c.backward()
+    gh_c, gw_c = hidden.grad.clone(), w.grad.clone()
+
+    hidden2 = hidden.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    logits = hidden2 @ w2.t()
+    loss_f = ref.causal_lm_loss(logits, labels)
+    loss_f.backward()
+
+    assert torch.allclose(loss_c, loss_f, atol=1e-6)
+    assert torch.allclose(gh_c, hidden2.grad, atol=1e-6)
+    assert torch.allclose(gw_c, w2.grad, atol=1e-5)
