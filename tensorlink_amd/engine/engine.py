@@ -256,6 +256,7 @@ class InferenceEngine:
         stop = request.get("stop") or []
         if isinstance(stop, str):
             stop = [stop]
+        self._num_beams = int(args.get("num_beams", 1) or 1)
         return job, prompt, ids, sp, list(stop)
 
     @staticmethod
@@ -293,9 +294,17 @@ class InferenceEngine:
                     completion_tokens=n_new,
                     reasoning=bool(request.get("reasoning")),
                     processing_time=time.time() - t0)
-            with self._serve_lock:
-                self._bcast(("generate", job.name))
-                out, stats = job.runner.generate(ids, sp, return_stats=True)
+            nb = getattr(self, "_num_beams", 1)
+            if nb > 1 and self.world == 1:
+                with self._serve_lock:
+                    out = job.runner.generate_beam(
+                        ids, max_new_tokens=sp.max_new_tokens,
+                        num_beams=nb, eos_token_id=sp.eos_token_id)
+            else:
+                with self._serve_lock:
+                    self._bcast(("generate", job.name))
+                    out, stats = job.runner.generate(ids, sp,
+                                                     return_stats=True)
             text = job.tokenizer.decode(out[0].tolist(),
                                         skip_special_tokens=True)
             text = self._apply_stop(text, stop)

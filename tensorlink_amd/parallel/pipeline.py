@@ -322,6 +322,78 @@ class PipelineRunner:
             return result, stats
         return result
 
+    @torch.no_grad()
+    def generate_beam(self, input_ids: torch.Tensor,
+                      max_new_tokens: int = 64, num_beams: int = 4,
+                      length_penalty: float = 1.0,
+                      eos_token_id: Optional[int] = None) -> torch.Tensor:
+        """Beam-search decode (world==1). The reference exposes
+        ``num_beams`` through HF ``generate`` for whole-model jobs
+        (``ml/formatter.py:94-99`` normalizes the beam/sample conflict,
+        ``ml/worker.py:359`` runs generate); this is the native
+        equivalent: beams share the batch dimension of one KV cache and
+        the cache is index-reordered to each beam's parent between
+        steps. Returns [B, <=max_new_tokens] (best beam per sequence).
+        """
+        assert self.world == 1, "beam search is single-rank"
+        nb = num_beams
+        B, S = input_ids.shape
+        T = max_new_tokens
+        dev = self.device
+        ids = input_ids.to(dev).repeat_interleave(nb, 0)     # [B*nb, S]
+        # fresh cache (beam reorder swaps slabs; don't disturb the
+        # graph-captured cache reuse)
+        cache = self.stage.make_kv_cache(B * nb, S + T, dev, self.dtype,
+                                         kv_mode="contiguous")
+        pos = torch.arange(S, device=dev, dtype=torch.int32) \
+            .unsqueeze(0).expand(B * nb, -1).contiguous()
+        hidden = self.stage(ids, pos, kv_cache=cache, return_logits=False)
+        logp = torch.log_softmax(
+            self.stage.head(hidden[:, -1:]).squeeze(1).float(), -1)
+        V = logp.shape[-1]
+        # beam 0 seeds each sequence (identical prefills => mask others)
+        scores = torch.full((B, nb), float("-inf"), device=dev)
+        scores[:, 0] = 0.0
+        tokens = torch.empty(B * nb, 0, device=dev, dtype=torch.int64)
+        alive = torch.ones(B * nb, dtype=torch.bool, device=dev)
+        positions = torch.full((B * nb,), S, device=dev, dtype=torch.int32)
+        for t in range(T):
+            cand = scores.view(-1, 1) + logp                 # [B*nb, V]
+            if eos_token_id is not None and t > 0:
+                # a finished beam only extends with eos at no cost
+                frozen = ~alive
+                cand[frozen] = float("-inf")
+                cand[frozen, eos_token_id] = scores.view(-1)[frozen]
+            top_val, top_idx = cand.view(B, nb * V).topk(nb, -1)
+            parent = top_idx // V                            # [B, nb]
+            tok = (top_idx % V).to(torch.int64)
+            gather = (torch.arange(B, device=dev).unsqueeze(1) * nb
+                      + parent).view(-1)                     # [B*nb]
+            # reorder beam state to each survivor's parent
+            tokens = torch.cat([tokens[gather], tok.view(-1, 1)], 1)
+            scores = top_val
+            alive = alive[gather]
+            if eos_token_id is not None:
+                alive &= tok.view(-1) != eos_token_id
+            if t == T - 1 or not bool(alive.any()):
+                break
+            for li in range(len(cache.k)):
+                cache.k[li] = cache.k[li][gather].contiguous()
+                cache.v[li] = cache.v[li][gather].contiguous()
+            lg = self.stage(tokens[:, -1:],
+                            positions.unsqueeze(1).int(), kv_cache=cache)
+            logp = torch.log_softmax(lg.squeeze(1).float(), -1)
+            positions += 1
+        # pick the best beam per sequence under the GNMT length penalty
+        lengths = tokens.shape[1] - (tokens == eos_token_id).sum(1).float() \
+            if eos_token_id is not None else \
+            torch.full((B * nb,), float(tokens.shape[1]), device=dev)
+        lengths = lengths.clamp(min=1.0).view(B, nb)
+        final = scores / lengths.pow(length_penalty)
+        best = final.argmax(-1)                              # [B]
+        idx = torch.arange(B, device=dev) * nb + best
+        return tokens[idx]
+
     def _graph_decode(self, cur: torch.Tensor, positions: torch.Tensor,
                       out_tokens: torch.Tensor, T: int) -> None:
         """Greedy decode via a captured hipGraph (world==1).

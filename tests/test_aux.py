@@ -260,3 +260,17 @@ def test_auto_loaded_model_management(monkeypatch):
     assert "tiny" not in eng.jobs
     assert "tiny-moe" in eng.jobs
     eng.unload_model("tiny-moe")
+
+
+def test_engine_num_beams_request():
+    """num_beams>1 routes through beam search end-to-end (serial path)."""
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    out = eng.generate({"hf_name": "tiny", "message": "hello world",
+                        "max_new_tokens": 5, "do_sample": False,
+                        "num_beams": 3, "output_format": "simple"})
+    assert "response" in out and "error" not in out
+    eng.unload_model("tiny")

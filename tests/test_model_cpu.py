@@ -277,3 +277,36 @@ def test_trainer_chunked_ce_matches_plain():
 
     a, b = run(False), run(True)
     assert a == pytest.approx(b, rel=1e-5)
+
+
+def test_beam_search_decode():
+    """Beam search: nb=1 equals greedy; nb=4's best beam scores at least
+    the greedy sequence under teacher-forced log-prob; eos stops."""
+    import torch
+    import torch.nn.functional as F
+
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                       device=torch.device("cpu"), seed=4)
+    torch.manual_seed(17)
+    ids = torch.randint(0, 1024, (2, 12))
+    greedy = r.generate(ids, SamplingParams(max_new_tokens=6))
+    b1 = r.generate_beam(ids, max_new_tokens=6, num_beams=1)
+    assert torch.equal(b1, greedy)
+
+    b4 = r.generate_beam(ids, max_new_tokens=6, num_beams=4)
+    assert b4.shape == (2, 6)
+
+    def seq_logprob(prompt, cont):
+        full = torch.cat([prompt.unsqueeze(0), cont.unsqueeze(0)], 1)
+        pos = torch.arange(full.shape[1]).unsqueeze(0).contiguous()
+        logits = r.stage(full, pos)
+        lp = F.log_softmax(logits[0, :-1].float(), -1)
+        tgt = full[0, 1:]
+        S0 = prompt.shape[0]
+        return float(lp[torch.arange(len(tgt)), tgt][S0 - 1:].sum())
+
+    for b in range(2):
+        assert seq_logprob(ids[b], b4[b]) >= seq_logprob(
+            ids[b], greedy[b]) - 1e-4
