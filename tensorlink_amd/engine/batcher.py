@@ -45,6 +45,13 @@ class Request:
     tokens: "queue.Queue" = field(default_factory=queue.Queue)
     done: threading.Event = field(default_factory=threading.Event)
     error: Optional[str] = None
+    cancelled: bool = False
+
+    def cancel(self):
+        """Abort the request: queued requests are dropped at admission,
+        running ones retire at the next scheduler step (pages return to
+        the pool). The client sees the stream end."""
+        self.cancelled = True
 
     def result(self, timeout: float = 300.0) -> List[int]:
         """Block until finished; returns the generated token ids."""
@@ -138,8 +145,9 @@ class ContinuousBatcher:
 
     def submit(self, input_ids: torch.Tensor, **kw) -> Request:
         req = Request(input_ids=input_ids.reshape(-1), **kw)
+        req.max_new_tokens = min(req.max_new_tokens, self.max_ctx - 1)
         if req.input_ids.numel() + req.max_new_tokens > self.max_ctx:
-            keep = self.max_ctx - req.max_new_tokens
+            keep = max(1, self.max_ctx - req.max_new_tokens)
             req.input_ids = req.input_ids[-keep:]
         self._seq += 1
         self._queue.put((-req.priority, self._seq, req))
@@ -197,7 +205,9 @@ class ContinuousBatcher:
         req = self.slots[slot]
         if req is None:
             return
-        if ((req.eos_token_id is not None and tok == req.eos_token_id)
+        if (req.cancelled
+                or (req.eos_token_id is not None
+                    and tok == req.eos_token_id)
                 or self.slot_emitted[slot] >= req.max_new_tokens
                 or self.slot_len[slot] + 1 >= self.max_ctx):
             req.tokens.put(None)
@@ -247,6 +257,10 @@ class ContinuousBatcher:
                 if self.cache.allocator.n_free < need:
                     break
                 req = self._queue.get_nowait()[2]
+                if req.cancelled:
+                    req.tokens.put(None)
+                    req.done.set()
+                    continue
                 self.slots[slot] = req
                 self.slot_prompt[slot] = req.input_ids
                 self.slot_filled[slot] = 0
@@ -414,7 +428,9 @@ class PPContinuousBatcher(ContinuousBatcher):
     # ------------------------- scheduler (rank 0) ----------------------
     def _finish0(self, slot: int, tok: int):
         req = self.slots[slot]
-        if ((req.eos_token_id is not None and tok == req.eos_token_id)
+        if (req.cancelled
+                or (req.eos_token_id is not None
+                    and tok == req.eos_token_id)
                 or self.slot_emitted[slot] >= req.max_new_tokens
                 or self.slot_len[slot] + 1 >= self.max_ctx):
             req.tokens.put(None)
@@ -438,6 +454,10 @@ class PPContinuousBatcher(ContinuousBatcher):
                 if self.cache.allocator.n_free < need:
                     break
                 req = self._queue.get_nowait()[2]
+                if req.cancelled:
+                    req.tokens.put(None)
+                    req.done.set()
+                    continue
                 self.slots[slot] = req
                 self.slot_prompt[slot] = req.input_ids
                 self.slot_filled[slot] = 0

@@ -170,3 +170,37 @@ def test_priority_admission_order():
     # impossible here (not started yet) -> high must be first
     assert order[0] is high
     assert order[1:] == lows
+
+
+def test_request_cancellation():
+    """cancel() drops a queued request and retires a running one at the
+    next step; the pool drains fully afterwards."""
+    import time
+
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=1, max_ctx=128).start()
+    try:
+        torch.manual_seed(9)
+        running = b.submit(torch.randint(0, 1024, (8,)),
+                           max_new_tokens=100)
+        queued = b.submit(torch.randint(0, 1024, (8,)), max_new_tokens=4)
+        # let the first one start decoding, then cancel both
+        for _ in range(100):
+            if b.slot_emitted[0] > 2:
+                break
+            time.sleep(0.01)
+        running.cancel()
+        out_r = running.result(timeout=60)
+        assert 0 < len(out_r) < 100          # stopped early
+        out_q = queued.result(timeout=60)    # ran normally after slot freed
+        assert len(out_q) == 4
+        q2 = b.submit(torch.randint(0, 1024, (8,)), max_new_tokens=4)
+        q2.cancel()
+        assert q2.result(timeout=60) == []   # dropped at admission
+        for _ in range(200):
+            if b.cache.allocator.n_free == b.cache.allocator.n_pages:
+                break
+            time.sleep(0.01)
+    finally:
+        b.stop()
+    assert b.cache.allocator.n_free == b.cache.allocator.n_pages
