@@ -113,12 +113,18 @@ def local_config(config: ModelConfig, tp: int) -> ModelConfig:
 
 
 def shard_state(full_state: dict, config: ModelConfig, tp_rank: int,
-                tp: int) -> dict:
-    """Slice a full-model state dict into this rank's TP shard."""
+                tp: int, vocab_parallel: bool = False) -> dict:
+    """Slice a full-model state dict into this rank's TP shard.
+    vocab_parallel additionally shards the LM head over the vocab dim
+    (Megatron-style — training path; inference keeps it replicated)."""
     q, kv, i = config.q_size, config.kv_size, config.intermediate_size
     ql, kvl, il = q // tp, kv // tp, i // tp
     out = {}
     for name, w in full_state.items():
+        if vocab_parallel and name == "lm_head.weight":
+            vl = w.shape[0] // tp
+            out[name] = w[tp_rank * vl:(tp_rank + 1) * vl]
+            continue
         if "qkv_proj" in name:
             qs = w[tp_rank * ql:(tp_rank + 1) * ql]
             ks = w[q + tp_rank * kvl:q + (tp_rank + 1) * kvl]
@@ -138,7 +144,7 @@ def shard_state(full_state: dict, config: ModelConfig, tp_rank: int,
 
 
 def build_tp_model(config_or_name, tp_rank: int, tp: int, device=None,
-                   dtype=None, seed: int = 0):
+                   dtype=None, seed: int = 0, vocab_parallel: bool = False):
     """Full model (PP=1) sharded TP-ways. Weights come from slicing the
     SAME seeded full-model init every rank, so a TP group reproduces the
     single-rank reference."""
@@ -147,13 +153,21 @@ def build_tp_model(config_or_name, tp_rank: int, tp: int, device=None,
     device = device if device is not None else device_for_rank()
     dtype = dtype or (torch.bfloat16 if device.type == "cuda"
                       else torch.float32)
+    if vocab_parallel:
+        assert not config.tie_word_embeddings, \
+            "vocab-parallel head requires untied embeddings"
+        assert config.vocab_size % tp == 0
 
     full = build_full_model(config)
     init_random_stage(full, device="cpu", dtype=dtype, seed=seed)
-    shards = shard_state(full.state_dict(), config, tp_rank, tp)
+    shards = shard_state(full.state_dict(), config, tp_rank, tp,
+                         vocab_parallel=vocab_parallel)
     del full
 
     stage = build_full_model(local_config(config, tp))
+    if vocab_parallel:
+        stage.lm_head = TLLinear(config.hidden_size,
+                                 config.vocab_size // tp, bias=False)
     # swap row-parallel outputs for all-reducing variants and
     # column-parallel inputs for the f-op (backward all-reduce) variants
     for layer in stage.layers:
@@ -212,6 +226,49 @@ class TPRunner:
         return out
 
 
+class _VocabParallelCE(torch.autograd.Function):
+    """Cross-entropy over vocab-sharded logits without gathering them
+    (Megatron algorithm): all-reduce MAX for the row max, SUM for the
+    exp-denominator and the target logit; backward is purely local
+    (softmax_local minus the local one-hot)."""
+
+    @staticmethod
+    def forward(ctx, logits, labels, v0, group):
+        # logits [N, Vl] fp32; labels [N] global ids
+        lmax = logits.max(-1).values.contiguous()
+        dist.all_reduce(lmax, op=dist.ReduceOp.MAX, group=group)
+        z = (logits - lmax.unsqueeze(1)).exp()
+        denom = z.sum(-1).contiguous()
+        dist.all_reduce(denom, group=group)
+        Vl = logits.shape[1]
+        in_shard = (labels >= v0) & (labels < v0 + Vl)
+        idx = (labels - v0).clamp(0, Vl - 1)
+        tgt = (logits.gather(1, idx.unsqueeze(1)).squeeze(1)
+               - lmax) * in_shard
+        tgt = tgt.contiguous()
+        dist.all_reduce(tgt, group=group)
+        loss = denom.log() - tgt
+        ctx.save_for_backward(z, denom, idx, in_shard)
+        ctx.group = group
+        return loss.mean()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        z, denom, idx, in_shard = ctx.saved_tensors
+        N = z.shape[0]
+        dlogits = z / denom.unsqueeze(1)
+        dlogits.scatter_add_(
+            1, idx.unsqueeze(1),
+            -in_shard.to(dlogits.dtype).unsqueeze(1))
+        dlogits.mul_(dloss / N)
+        return dlogits, None, None, None
+
+
+def vocab_parallel_ce(logits_local, labels, vocab_start, group=None):
+    return _VocabParallelCE.apply(logits_local.float(), labels,
+                                  vocab_start, group)
+
+
 class TPTrainer:
     """SPMD tensor-parallel training (PP=1): every rank holds a head/
     intermediate shard, runs the full replicated batch, and computes the
@@ -225,20 +282,27 @@ class TPTrainer:
     """
 
     def __init__(self, model, rank: int, tp: int, device=None, seed: int = 0,
-                 lr: float = 1e-3, weight_decay: float = 0.01):
+                 lr: float = 1e-3, weight_decay: float = 0.01,
+                 vocab_parallel: bool = False):
         from tensorlink_amd.optim import FusedAdamW
         self.rank, self.tp = rank, tp
+        self.vocab_parallel = vocab_parallel
         self.device = device if device is not None else device_for_rank()
         self.stage = build_tp_model(model, rank, tp, device=self.device,
-                                    seed=seed)
+                                    seed=seed,
+                                    vocab_parallel=vocab_parallel)
         self.stage.train()
         for p in self.stage.parameters():
             p.requires_grad_(True)
         self.config = self.stage.config
+        sharded = ("qkv_proj", "gate_up_proj", "o_proj", "down_proj")
+        if vocab_parallel:
+            sharded = sharded + ("lm_head",)
         self._replicated = [
             p for n, p in self.stage.named_parameters()
-            if not any(t in n for t in ("qkv_proj", "gate_up_proj",
-                                        "o_proj", "down_proj"))]
+            if not any(t in n for t in sharded)]
+        self._v0 = rank * (self.config.vocab_size // tp
+                           if vocab_parallel else 0)
         self.opt = FusedAdamW(self.stage.parameters(), lr=lr,
                               weight_decay=weight_decay)
 
@@ -253,8 +317,13 @@ class TPTrainer:
         self.opt.zero_grad()
         hidden = self.stage(ids, pos.contiguous(), training=True,
                             return_logits=False)
-        logits = self.stage.head(hidden)
-        loss = tl_ops.causal_lm_loss(logits, labels)
+        logits = self.stage.head(hidden)     # [B,S,V] or [B,S,V/tp] (vp)
+        if self.vocab_parallel:
+            lg = logits[:, :-1].reshape(-1, logits.shape[-1])
+            lb = labels[:, 1:].reshape(-1)
+            loss = vocab_parallel_ce(lg, lb, self._v0)
+        else:
+            loss = tl_ops.causal_lm_loss(logits, labels)
         loss.backward()
         if _tp_active():
             for p in self._replicated:
