@@ -274,3 +274,38 @@ def test_engine_num_beams_request():
                         "num_beams": 3, "output_format": "simple"})
     assert "response" in out and "error" not in out
     eng.unload_model("tiny")
+
+
+def test_logging_and_dashboard(tmp_path, capsys):
+    """Rotating-file logger + colored console (reference debug_print,
+    smart_node.py:28-125) and the status dashboard (print_ui_status,
+    torch_node.py:963)."""
+    import os
+
+    import torch
+
+    from tensorlink_amd.utils.logging import (VERBOSE, enable_file_logging,
+                                              get_logger)
+    lg = get_logger("tensorlink_amd.test")
+    path = enable_file_logging(log_dir=str(tmp_path / "logs"),
+                               name="tensorlink_amd.test")
+    lg.info("hello %s", "world")
+    lg.verbose("fine detail")
+    lg.setLevel(VERBOSE)
+    lg.verbose("now visible")
+    for h in lg.handlers:
+        h.flush()
+    assert os.path.exists(path)
+    content = open(path).read()
+    assert "hello world" in content and "now visible" in content
+    assert "fine detail" not in content       # below level at emit time
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.utils.dashboard import render_status
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    eng.generate({"hf_name": "tiny", "message": "x", "max_new_tokens": 2,
+                  "do_sample": False, "output_format": "simple"})
+    txt = render_status(eng)
+    assert "tiny" in txt and "requests: 1" in txt and "[" in txt
+    eng.unload_model("tiny")

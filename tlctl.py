@@ -69,6 +69,23 @@ def cmd_bench(args, extra):
                                            "bench.py")] + extra)
 
 
+def cmd_status(args):
+    if args.url:
+        import json
+        import urllib.request
+        for ep in ("/stats", "/models", "/node-info"):
+            with urllib.request.urlopen(args.url.rstrip("/") + ep,
+                                        timeout=10) as r:
+                print(ep, json.dumps(json.load(r), indent=2))
+        return
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.utils.dashboard import render_status
+    eng = InferenceEngine(rank=0, world=1)
+    print(render_status(eng))
+
+
 def cmd_health(args):
     import json
     from tensorlink_amd.utils.watchdog import check_gpu_health
@@ -101,6 +118,10 @@ def main():
 
     sub.add_parser("bench")
     sub.add_parser("health")
+    st = sub.add_parser("status", help="terminal status dashboard "
+                        "(VRAM bars, jobs, throughput)")
+    st.add_argument("--url", default=None,
+                    help="query a running server instead of local state")
 
     args, extra = p.parse_known_args()
     if args.cmd == "serve":
@@ -111,6 +132,8 @@ def main():
         cmd_bench(args, extra)
     elif args.cmd == "health":
         cmd_health(args)
+    elif args.cmd == "status":
+        cmd_status(args)
 
 
 if __name__ == "__main__":
