@@ -19,10 +19,13 @@ def cmd_serve(args):
     from tensorlink_amd.config import EngineConfig
     from tensorlink_amd.engine.engine import InferenceEngine
     from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.utils.logging import enable_file_logging, get_logger
     from tensorlink_amd.utils.state import StateKeeper
     from tensorlink_amd.utils.watchdog import Watchdog
 
     cfg = EngineConfig.load(args.config)
+    log = get_logger("tensorlink_amd.serve")
+    enable_file_logging()                 # logs/runtime.log, daily x 7
     rank, world = init_distributed()
     engine = InferenceEngine(rank=rank, world=world,
                              default_init=cfg.ml.init)
@@ -32,7 +35,7 @@ def cmd_serve(args):
 
     models = args.model or cfg.ml.default_models
     for name in models:
-        print(f"loading {name} ...")
+        log.info("loading %s ...", name)
         engine.load_model(name, continuous=args.continuous,
                           max_slots=args.max_slots, max_ctx=args.max_ctx,
                           prefill_chunk=args.prefill_chunk)
