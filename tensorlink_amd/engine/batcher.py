@@ -30,7 +30,8 @@ import torch
 
 from tensorlink_amd import ops
 from tensorlink_amd.models.paged import (PAGE, DynamicPagedKVCache,
-                                         PagedKVCache)
+                                         PagedKVCache,
+                                         PrefixCachingKVCache)
 
 
 @dataclass
@@ -98,7 +99,8 @@ class ContinuousBatcher:
 
     def __init__(self, runner, max_slots: int = 16, max_ctx: int = 2048,
                  pool_pages: Optional[int] = None,
-                 prefill_chunk: Optional[int] = None):
+                 prefill_chunk: Optional[int] = None,
+                 prefix_caching: bool = False):
         assert runner.world == 1 or self.SUPPORTS_PP, \
             "use PPContinuousBatcher for world > 1"
         self.runner = runner
@@ -110,7 +112,10 @@ class ContinuousBatcher:
         # oversubscribes memory (admission blocks when exhausted)
         if pool_pages is None:
             pool_pages = max_slots * ((max_ctx + PAGE - 1) // PAGE)
-        self.cache = DynamicPagedKVCache(
+        cache_cls = (PrefixCachingKVCache if prefix_caching
+                     else DynamicPagedKVCache)
+        self.prefix_caching = prefix_caching
+        self.cache = cache_cls(
             self.stage.num_layers, max_slots, pool_pages, max_ctx,
             self.stage.config, runner.device, runner.dtype)
         self.max_slots = max_slots
@@ -185,6 +190,8 @@ class ContinuousBatcher:
         self.slot_filled[slot] = end
         if end < S_total:
             return
+        if self.prefix_caching:
+            self.cache.register_prefix(slot, prompt.tolist())
         logits = self.stage.head(hidden[:, -1:]).squeeze(1)
         tok = int(self._sample(logits, [req])[0])
         self.slot_len[slot] = S_total
@@ -252,9 +259,13 @@ class ContinuousBatcher:
                     break
                 # admission control: need pages for prompt + first tokens
                 head = self._queue.queue[0][2]
-                # pages for the prompt plus its first generated token
+                # pages for the prompt plus its first generated token,
+                # minus any published prefix it can adopt
                 need = (head.input_ids.numel() + PAGE) // PAGE
-                if self.cache.allocator.n_free < need:
+                if self.prefix_caching:
+                    need -= self.cache.match_prefix(
+                        head.input_ids.tolist()) // PAGE
+                if self.cache.available_pages() < need:
                     break
                 req = self._queue.get_nowait()[2]
                 if req.cancelled:
@@ -265,6 +276,9 @@ class ContinuousBatcher:
                 self.slot_prompt[slot] = req.input_ids
                 self.slot_filled[slot] = 0
                 self.slot_len[slot] = 0
+                if self.prefix_caching:
+                    self.slot_filled[slot] = self.cache.adopt_prefix(
+                        slot, req.input_ids.tolist())
                 admitted = True
             # one prefill chunk per iteration: bounds how long running
             # decodes wait behind a new long prompt

@@ -152,6 +152,7 @@ class InferenceEngine:
         max_ctx = kwargs.pop("max_ctx", 2048)
         pool_pages = kwargs.pop("pool_pages", None)
         prefill_chunk = kwargs.pop("prefill_chunk", None)
+        prefix_caching = kwargs.pop("prefix_caching", False)
         plan = plan_for_world(config, self.world, **kwargs)
         runner = PipelineRunner(plan, self.rank, self.world,
                                 device=self.device, init=init,
@@ -162,8 +163,8 @@ class InferenceEngine:
             from tensorlink_amd.engine.batcher import ContinuousBatcher
             job.batcher = ContinuousBatcher(
                 runner, max_slots=max_slots, max_ctx=max_ctx,
-                pool_pages=pool_pages,
-                prefill_chunk=prefill_chunk).start()
+                pool_pages=pool_pages, prefill_chunk=prefill_chunk,
+                prefix_caching=prefix_caching).start()
         elif continuous and self.world > 1:
             from tensorlink_amd.engine.batcher import PPContinuousBatcher
             b = PPContinuousBatcher(runner, max_slots=max_slots,

@@ -16,7 +16,8 @@ tests genuinely exercise the indirection.
 from __future__ import annotations
 
 import random
-from typing import List, Optional
+from collections import OrderedDict, defaultdict
+from typing import Dict, List, Optional, Tuple
 
 import torch
 
@@ -124,6 +125,9 @@ class DynamicPagedKVCache(PagedKVCache):
         self._slot_pages[slot] = []
         self.seq_lens[slot] = 0
 
+    def available_pages(self) -> int:
+        return self.allocator.n_free
+
     def reset(self):
         for s in range(self.batch):
             self.release_slot(s)
@@ -139,3 +143,133 @@ def kv_slice_paged(cache: PagedKVCache, s: int, e: int) -> PagedKVCache:
     view.batch = e - s
     view.pages_per_seq = cache.pages_per_seq
     return view
+
+
+class PrefixCachingKVCache(DynamicPagedKVCache):
+    """Dynamic paged cache with automatic prefix reuse (vLLM-style).
+
+    Every FULL page of a finished prompt is published under a chained
+    content hash (page i's key folds in the hash of pages 0..i-1, so a
+    page is only reused when the entire prefix matches). A new request
+    whose prompt shares a published prefix adopts those pages by
+    reference (refcount) and starts its chunked prefill at the first
+    uncached token — prompt compute and KV writes for the shared prefix
+    are skipped entirely. Published pages with no live references sit in
+    an LRU from which the allocator evicts when the free list runs dry,
+    so the prefix cache consumes only otherwise-idle pool memory.
+
+    At least the last prompt token is always re-prefilled (its logits
+    seed decoding), so ``match_prefix`` never covers the whole prompt.
+    """
+
+    def __init__(self, *a, **kw):
+        super().__init__(*a, **kw)
+        self.refcnt: Dict[int, int] = defaultdict(int)
+        self.hash_to_page: Dict[int, int] = {}
+        self.page_hash: Dict[int, int] = {}
+        self.lru: "OrderedDict[int, None]" = OrderedDict()
+        self.hits = 0                  # tokens of prefill skipped (stats)
+
+    # ---------------- hashing ----------------
+    @staticmethod
+    def _chain(tokens) -> List[int]:
+        hashes, h = [], 0
+        for i in range(len(tokens) // PAGE):
+            h = hash((h, tuple(int(t) for t in
+                               tokens[i * PAGE:(i + 1) * PAGE])))
+            hashes.append(h)
+        return hashes
+
+    def _usable(self, n_tokens: int) -> int:
+        """Cacheable prefix length: full pages, minus one page if that
+        would cover the whole prompt."""
+        n = (n_tokens // PAGE) * PAGE
+        if n >= n_tokens:
+            n -= PAGE
+        return max(0, n)
+
+    # ---------------- allocation with LRU eviction ----------------
+    def _take_pages(self, n: int) -> List[int]:
+        out = []
+        while len(out) < n:
+            if self.allocator.n_free:
+                out.extend(self.allocator.alloc(1))
+            elif self.lru:
+                page, _ = self.lru.popitem(last=False)
+                h = self.page_hash.pop(page)
+                self.hash_to_page.pop(h, None)
+                out.append(page)
+            else:
+                raise RuntimeError(
+                    "KV page pool exhausted (all pages referenced)")
+        return out
+
+    def available_pages(self) -> int:
+        return self.allocator.n_free + len(self.lru)
+
+    def ensure(self, slot: int, length: int):
+        need = (length + PAGE - 1) // PAGE
+        have = len(self._slot_pages[slot])
+        if need > have:
+            new = self._take_pages(need - have)
+            for p in new:
+                self.refcnt[p] += 1
+            self._slot_pages[slot].extend(new)
+            self.table[slot, have:need] = torch.tensor(
+                new, dtype=torch.int32, device=self.table.device)
+
+    def release_slot(self, slot: int):
+        for p in self._slot_pages[slot]:
+            self.refcnt[p] -= 1
+            if self.refcnt[p] == 0:
+                if p in self.page_hash:
+                    self.lru[p] = None          # evictable, still published
+                else:
+                    self.allocator.release([p])
+        self._slot_pages[slot] = []
+        self.seq_lens[slot] = 0
+
+    # ---------------- prefix API (used by the batcher) ----------------
+    def match_prefix(self, tokens) -> int:
+        """Longest published prefix (in tokens) WITHOUT adopting it."""
+        n = 0
+        for h in self._chain(tokens[:self._usable(len(tokens))]):
+            if h not in self.hash_to_page:
+                break
+            n += PAGE
+        return n
+
+    def adopt_prefix(self, slot: int, tokens) -> int:
+        """Point `slot`'s first pages at the published prefix; returns
+        the number of tokens covered (0 if none). Call on an empty
+        slot."""
+        assert not self._slot_pages[slot]
+        pages = []
+        for h in self._chain(tokens[:self._usable(len(tokens))]):
+            p = self.hash_to_page.get(h)
+            if p is None:
+                break
+            pages.append(p)
+        if not pages:
+            return 0
+        for p in pages:
+            self.refcnt[p] += 1
+            self.lru.pop(p, None)
+        self._slot_pages[slot].extend(pages)
+        self.table[slot, :len(pages)] = torch.tensor(
+            pages, dtype=torch.int32, device=self.table.device)
+        self.hits += len(pages) * PAGE
+        return len(pages) * PAGE
+
+    def register_prefix(self, slot: int, tokens):
+        """Publish `slot`'s full prompt pages for future reuse (call
+        after its prefill completes)."""
+        for i, h in enumerate(self._chain(
+                tokens[:self._usable(len(tokens))])):
+            if h in self.hash_to_page:
+                continue              # identical page already published
+            page = self._slot_pages[slot][i]
+            if page in self.page_hash:
+                continue              # page already carries another hash
+            self.hash_to_page[h] = page
+            self.page_hash[page] = h

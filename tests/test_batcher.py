@@ -204,3 +204,62 @@ def test_request_cancellation():
     finally:
         b.stop()
     assert b.cache.allocator.n_free == b.cache.allocator.n_pages
+
+
+def test_prefix_caching_reuses_pages_and_matches_serial():
+    """Prefix caching: a second request with the same long prompt adopts
+    the published pages (prefill skipped for the shared prefix) and
+    still reproduces serial greedy output exactly; distinct prompts do
+    not cross-match; pages fully drain after release."""
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=4, max_ctx=512, prefill_chunk=64,
+                          prefix_caching=True).start()
+    try:
+        torch.manual_seed(23)
+        prompt = torch.randint(0, 1024, (300,))     # 2 full pages usable
+        other = torch.randint(0, 1024, (300,))
+        r1 = b.submit(prompt.clone(), max_new_tokens=6)
+        out1 = r1.result(timeout=120)
+        assert b.cache.hits == 0                    # first sight: no reuse
+        r2 = b.submit(prompt.clone(), max_new_tokens=6)
+        out2 = r2.result(timeout=120)
+        assert b.cache.hits == 256                  # 2 pages adopted
+        assert out1 == out2
+        r3 = b.submit(other, max_new_tokens=6)
+        out3 = r3.result(timeout=120)
+        assert b.cache.hits == 256                  # no cross-match
+        ref = r.generate(prompt.unsqueeze(0), SamplingParams(max_new_tokens=6))
+        assert out1 == ref[0].tolist()
+        ref3 = r.generate(other.unsqueeze(0), SamplingParams(max_new_tokens=6))
+        assert out3 == ref3[0].tolist()
+    finally:
+        b.stop()
+    # all pages either free or parked in the evictable prefix LRU
+    assert (b.cache.allocator.n_free + len(b.cache.lru)
+            == b.cache.allocator.n_pages)
+
+
+def test_prefix_cache_eviction_under_pressure():
+    """A tiny pool forces LRU eviction of published prefixes; requests
+    still complete correctly afterwards."""
+    r = _runner()
+    # 6 pages total; each 300-token prompt needs 3
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=512, pool_pages=6,
+                          prefill_chunk=64, prefix_caching=True).start()
+    try:
+        torch.manual_seed(29)
+        prompts = [torch.randint(0, 1024, (300,)) for _ in range(3)]
+        outs = []
+        for p in prompts:
+            outs.append(b.submit(p.clone(), max_new_tokens=4
+                                 ).result(timeout=120))
+        # resubmit the FIRST prompt: its prefix may have been evicted,
+        # correctness must hold either way
+        again = b.submit(prompts[0].clone(), max_new_tokens=4
+                         ).result(timeout=120)
+        assert again == outs[0]
+        for p, o in zip(prompts, outs):
+            ref = r.generate(p.unsqueeze(0), SamplingParams(max_new_tokens=4))
+            assert o == ref[0].tolist()
+    finally:
+        b.stop()

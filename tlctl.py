@@ -38,7 +38,8 @@ def cmd_serve(args):
         log.info("loading %s ...", name)
         engine.load_model(name, continuous=args.continuous,
                           max_slots=args.max_slots, max_ctx=args.max_ctx,
-                          prefill_chunk=args.prefill_chunk)
+                          prefill_chunk=args.prefill_chunk,
+                          prefix_caching=args.prefix_caching)
     keeper = StateKeeper(engine)
     keeper.load_previous_state()
     keeper.start()
@@ -109,6 +110,9 @@ def main():
     s.add_argument("--max-slots", type=int, default=16)
     s.add_argument("--max-ctx", type=int, default=4096)
     s.add_argument("--prefill-chunk", type=int, default=None)
+    s.add_argument("--prefix-caching", action="store_true",
+                   help="reuse KV pages across requests sharing a "
+                        "prompt prefix")
 
     pl = sub.add_parser("plan")
     pl.add_argument("--model", required=True)
