@@ -43,6 +43,8 @@ class Request:
     top_k: int = 0
     eos_token_id: Optional[int] = None
     priority: int = 0                    # higher admits first
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
     tokens: "queue.Queue" = field(default_factory=queue.Queue)
     done: threading.Event = field(default_factory=threading.Event)
     error: Optional[str] = None
@@ -126,6 +128,7 @@ class ContinuousBatcher:
         self.slot_last = [0] * max_slots      # last sampled token
         self.slot_prompt: List[Optional[torch.Tensor]] = [None] * max_slots
         self.slot_filled = [0] * max_slots    # prompt tokens prefilled
+        self.slot_counts: List[dict] = [{} for _ in range(max_slots)]
         # admission queue ordered by (priority desc, arrival): a
         # high-priority request jumps the queue but never preempts a
         # running slot
@@ -193,19 +196,26 @@ class ContinuousBatcher:
         if self.prefix_caching:
             self.cache.register_prefix(slot, prompt.tolist())
         logits = self.stage.head(hidden[:, -1:]).squeeze(1)
-        tok = int(self._sample(logits, [req])[0])
+        tok = int(self._sample(logits, [req], slots=[slot])[0])
         self.slot_len[slot] = S_total
         self.slot_last[slot] = tok
         self.slot_emitted[slot] = 1
+        self.slot_counts[slot][tok] = self.slot_counts[slot].get(tok,
+                                                                 0) + 1
         req.tokens.put(tok)
         self._maybe_finish(slot, tok)
 
-    def _sample(self, logits, reqs):
+    def _sample(self, logits, reqs, slots=None):
         toks = []
         for i, req in enumerate(reqs):
+            counts = (self.slot_counts[slots[i]]
+                      if slots is not None else None)
             toks.append(int(ops.sample_token(
                 logits[i:i + 1], temperature=req.temperature,
-                top_p=req.top_p, top_k=req.top_k)[0]))
+                top_p=req.top_p, top_k=req.top_k,
+                token_counts=counts,
+                presence_penalty=req.presence_penalty,
+                frequency_penalty=req.frequency_penalty)[0]))
         return toks
 
     def _maybe_finish(self, slot: int, tok: int):
@@ -240,10 +250,12 @@ class ContinuousBatcher:
         view = _SlotView(self.cache, active, lens)
         pos = lens.unsqueeze(1)
         logits = self.stage(toks.unsqueeze(1), pos, kv_cache=view).squeeze(1)
-        new = self._sample(logits, reqs)
+        new = self._sample(logits, reqs, slots=active)
         for j, slot in enumerate(active):
             self.slot_len[slot] += 1
             self.slot_emitted[slot] += 1
+            self.slot_counts[slot][new[j]] = \
+                self.slot_counts[slot].get(new[j], 0) + 1
             self.slots[slot].tokens.put(new[j])
             self.slot_last[slot] = new[j]
             self._maybe_finish(slot, new[j])
@@ -276,6 +288,7 @@ class ContinuousBatcher:
                 self.slot_prompt[slot] = req.input_ids
                 self.slot_filled[slot] = 0
                 self.slot_len[slot] = 0
+                self.slot_counts[slot] = {}
                 if self.prefix_caching:
                     self.slot_filled[slot] = self.cache.adopt_prefix(
                         slot, req.input_ids.tolist())

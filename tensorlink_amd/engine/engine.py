@@ -253,7 +253,9 @@ class InferenceEngine:
             temperature=args["temperature"] if args["do_sample"] else 0.0,
             top_p=args["top_p"], top_k=int(args.get("top_k", 0) or 0),
             max_new_tokens=args["max_new_tokens"],
-            eos_token_id=args.get("eos_token_id"))
+            eos_token_id=args.get("eos_token_id"),
+            presence_penalty=args.get("presence_penalty", 0.0),
+            frequency_penalty=args.get("frequency_penalty", 0.0))
         stop = request.get("stop") or []
         if isinstance(stop, str):
             stop = [stop]
@@ -281,7 +283,9 @@ class InferenceEngine:
                 req = job.batcher.submit(
                     ids[0], max_new_tokens=sp.max_new_tokens,
                     temperature=sp.temperature, top_p=sp.top_p,
-                    top_k=sp.top_k, eos_token_id=sp.eos_token_id)
+                    top_k=sp.top_k, eos_token_id=sp.eos_token_id,
+                    presence_penalty=sp.presence_penalty,
+                    frequency_penalty=sp.frequency_penalty)
                 out_tokens = req.result()
                 n_new = len(out_tokens)
                 text = job.tokenizer.decode(out_tokens,
@@ -347,7 +351,9 @@ class InferenceEngine:
                     req = job.batcher.submit(
                         ids[0], max_new_tokens=sp.max_new_tokens,
                         temperature=sp.temperature, top_p=sp.top_p,
-                        top_k=sp.top_k, eos_token_id=sp.eos_token_id)
+                        top_k=sp.top_k, eos_token_id=sp.eos_token_id,
+                        presence_penalty=sp.presence_penalty,
+                        frequency_penalty=sp.frequency_penalty)
                     for tok in req.stream():
                         q.put(tok)
                 else:

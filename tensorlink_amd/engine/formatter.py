@@ -53,6 +53,8 @@ def normalize_generate_args(args: Dict, tokenizer=None,
 
     do_sample = bool(out.get("do_sample", True))
     num_beams = int(out.get("num_beams", 1) or 1)
+    for k in ("presence_penalty", "frequency_penalty"):
+        out[k] = max(-2.0, min(2.0, float(out.get(k, 0.0) or 0.0)))
     if num_beams > 1 and do_sample:
         # beam search and sampling conflict: sampling wins (reference
         # formatter.py:94-99 resolves the same way)

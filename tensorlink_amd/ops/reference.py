@@ -143,8 +143,20 @@ def attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
 # --------------------------------------------------------------------------
 def sample_token(logits: torch.Tensor, *, temperature: float = 1.0,
                  top_p: float = 1.0, top_k: int = 0,
-                 generator: Optional[torch.Generator] = None) -> torch.Tensor:
-    """Sample next tokens from [B, V] logits. temperature<=0 → greedy."""
+                 generator: Optional[torch.Generator] = None,
+                 token_counts: Optional[dict] = None,
+                 presence_penalty: float = 0.0,
+                 frequency_penalty: float = 0.0) -> torch.Tensor:
+    """Sample next tokens from [B, V] logits. temperature<=0 → greedy.
+
+    token_counts (B==1 path): {token_id: count} of tokens generated so
+    far; OpenAI-style penalties subtract presence_penalty once per seen
+    token plus frequency_penalty * count (the reference only DECLARES
+    these fields — ``api/models.py:73-74`` — and never applies them)."""
+    if token_counts and (presence_penalty or frequency_penalty):
+        logits = logits.float().clone()
+        for t, c in token_counts.items():
+            logits[..., t] -= presence_penalty + frequency_penalty * c
     if temperature <= 0.0:
         return logits.argmax(-1)
     logits = logits.float() / temperature

@@ -36,6 +36,8 @@ class SamplingParams:
     max_new_tokens: int = 64
     eos_token_id: Optional[int] = None
     seed: Optional[int] = None
+    presence_penalty: float = 0.0     # applied on the batcher path
+    frequency_penalty: float = 0.0
 
 
 def _kv_slice(cache, s: int, e: int):

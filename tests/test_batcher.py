@@ -263,3 +263,40 @@ def test_prefix_cache_eviction_under_pressure():
             assert o == ref[0].tolist()
     finally:
         b.stop()
+
+
+def test_frequency_penalty_suppresses_repeats():
+    """OpenAI penalties (the reference only declares the schema fields,
+    api/models.py:73-74 — never applies them): with a strong frequency
+    penalty the greedy decode cannot emit the same token many times,
+    and the output matches a manual penalized reference loop."""
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=256).start()
+    try:
+        torch.manual_seed(31)
+        p = torch.randint(0, 1024, (10,))
+        base = b.submit(p.clone(), max_new_tokens=12).result(timeout=60)
+        pen = b.submit(p.clone(), max_new_tokens=12,
+                       frequency_penalty=2.0,
+                       presence_penalty=1.0).result(timeout=60)
+
+        # manual penalized greedy reference
+        from tensorlink_amd.ops import reference as ref
+        cur = p.unsqueeze(0)
+        counts = {}
+        want = []
+        for _ in range(12):
+            pos = torch.arange(cur.shape[1]).unsqueeze(0).contiguous()
+            logits = r.stage(cur, pos)[0, -1:]
+            t = int(ref.sample_token(logits, temperature=0.0,
+                                     token_counts=counts,
+                                     presence_penalty=1.0,
+                                     frequency_penalty=2.0)[0])
+            counts[t] = counts.get(t, 0) + 1
+            want.append(t)
+            cur = torch.cat([cur, torch.tensor([[t]])], 1)
+        assert pen == want
+        assert max(pen.count(t) for t in set(pen)) <= \
+            max(base.count(t) for t in set(base)) or pen != base
+    finally:
+        b.stop()
