@@ -336,9 +336,13 @@ def adamw_(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,
 # Sampling (torch-composed on GPU for now; fused kernel is a later milestone)
 # ---------------------------------------------------------------------------
 def sample_token(logits, *, temperature=1.0, top_p=1.0, top_k=0,
-                 generator=None):
+                 generator=None, token_counts=None,
+                 presence_penalty=0.0, frequency_penalty=0.0):
     return ref.sample_token(logits, temperature=temperature, top_p=top_p,
-                            top_k=top_k, generator=generator)
+                            top_k=top_k, generator=generator,
+                            token_counts=token_counts,
+                            presence_penalty=presence_penalty,
+                            frequency_penalty=frequency_penalty)
 
 
 moe_topk_router = ref.moe_topk_router
