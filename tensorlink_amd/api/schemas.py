@@ -25,6 +25,7 @@ class GenerationRequest(BaseModel):
     top_k: int = 0
     do_sample: bool = True
     num_beams: int = 1
+    n: int = 1                        # OpenAI multiple completions
     presence_penalty: float = 0.0     # OpenAI range [-2, 2]
     frequency_penalty: float = 0.0
     stream: bool = False

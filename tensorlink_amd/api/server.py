@@ -98,6 +98,10 @@ def create_app(engine: InferenceEngine,
             "do_sample": req.temperature > 0,
             "stream": req.stream,
             "stop": req.stop,
+            "n": getattr(req, "n", 1) or 1,
+            "presence_penalty": getattr(req, "presence_penalty", 0.0) or 0.0,
+            "frequency_penalty": getattr(req, "frequency_penalty", 0.0)
+            or 0.0,
             "input_format": "chat",
             "output_format": "openai",
         }

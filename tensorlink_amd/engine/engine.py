@@ -279,18 +279,23 @@ class InferenceEngine:
                                 request.get("output_format", "openai"))
         try:
             job, prompt, ids, sp, stop = self._prepare(request)
+            n = max(1, int(request.get("n", 1) or 1))
             if job.batcher is not None:
-                req = job.batcher.submit(
-                    ids[0], max_new_tokens=sp.max_new_tokens,
-                    temperature=sp.temperature, top_p=sp.top_p,
-                    top_k=sp.top_k, eos_token_id=sp.eos_token_id,
-                    presence_penalty=sp.presence_penalty,
-                    frequency_penalty=sp.frequency_penalty)
-                out_tokens = req.result()
-                n_new = len(out_tokens)
-                text = job.tokenizer.decode(out_tokens,
-                                            skip_special_tokens=True)
-                text = self._apply_stop(text, stop)
+                kw = dict(max_new_tokens=sp.max_new_tokens,
+                          temperature=sp.temperature, top_p=sp.top_p,
+                          top_k=sp.top_k, eos_token_id=sp.eos_token_id,
+                          presence_penalty=sp.presence_penalty,
+                          frequency_penalty=sp.frequency_penalty)
+                reqs = [job.batcher.submit(ids[0], **kw)
+                        for _ in range(n)]
+                results = [rq.result() for rq in reqs]
+                out_tokens = results[0]
+                n_new = sum(len(t) for t in results)
+                texts = [self._apply_stop(
+                    job.tokenizer.decode(t, skip_special_tokens=True),
+                    stop) for t in results]
+                text = texts[0]
+                extra = texts[1:] or None
                 job.requests_served += 1
                 job.tokens_generated += n_new
                 self.metrics.record(n_new, time.time() - t0)
@@ -298,7 +303,8 @@ class InferenceEngine:
                     text, prompt_tokens=ids.shape[1],
                     completion_tokens=n_new,
                     reasoning=bool(request.get("reasoning")),
-                    processing_time=time.time() - t0)
+                    processing_time=time.time() - t0,
+                    extra_texts=extra)
             nb = getattr(self, "_num_beams", 1)
             if nb > 1 and self.world == 1:
                 with self._serve_lock:

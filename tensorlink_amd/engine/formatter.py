@@ -132,7 +132,11 @@ class ResponseFormatter:
     # ---- non-streaming ----
     def format_response(self, text: str, *, prompt_tokens: int = 0,
                         completion_tokens: int = 0, reasoning: bool = False,
-                        processing_time: Optional[float] = None) -> Dict:
+                        processing_time: Optional[float] = None,
+                        extra_texts: Optional[list] = None) -> Dict:
+        """extra_texts: additional sampled completions (OpenAI ``n>1``
+        — the reference declares ``n`` in its schema,
+        ``api/models.py:68``, but never produces extra choices)."""
         if reasoning:
             think, answer = extract_reasoning_and_answer(text)
         else:
@@ -149,13 +153,19 @@ class ResponseFormatter:
         message = {"role": "assistant", "content": answer}
         if think:
             message["reasoning_content"] = think
+        choices = [{"index": 0, "message": message,
+                    "finish_reason": "stop"}]
+        for i, extra in enumerate(extra_texts or [], start=1):
+            choices.append({"index": i,
+                            "message": {"role": "assistant",
+                                        "content": extra},
+                            "finish_reason": "stop"})
         return {
             "id": self.request_id,
             "object": "chat.completion",
             "created": self.created,
             "model": self.model_name,
-            "choices": [{"index": 0, "message": message,
-                         "finish_reason": "stop"}],
+            "choices": choices,
             "usage": {"prompt_tokens": prompt_tokens,
                       "completion_tokens": completion_tokens,
                       "total_tokens": prompt_tokens + completion_tokens},
