@@ -325,6 +325,76 @@ class PipelineRunner:
         return result
 
     @torch.no_grad()
+    def generate_speculative(self, input_ids: torch.Tensor,
+                             max_new_tokens: int = 64,
+                             lookup_n: int = 3, k: int = 8,
+                             eos_token_id: Optional[int] = None):
+        """Prompt-lookup speculative decoding (world==1, greedy, B==1):
+        propose the k tokens that followed the most recent occurrence of
+        the current lookup_n-gram earlier in the context, verify all of
+        them in ONE cached forward (the chunked-prefill q_off attention
+        path scores m+1 positions at once), and accept the longest
+        matching prefix plus the bonus token. Output is EXACTLY plain
+        greedy decoding — rejected positions simply roll seq_lens back
+        and are overwritten. No draft model needed; repetitive text
+        (code, extraction, chat history echoes) decodes several tokens
+        per forward. Returns (tokens [1, n], accepted_via_spec count).
+        """
+        assert self.world == 1, "speculative decode is single-rank"
+        B, S = input_ids.shape
+        assert B == 1, "prompt-lookup speculation is per-request (B=1)"
+        ids = input_ids.to(self.device)
+        cache = self.alloc_cache(1, S + max_new_tokens + k + 1)
+        pos = torch.arange(S, device=self.device,
+                           dtype=torch.int32).unsqueeze(0)
+        hidden = self.stage(ids, pos, kv_cache=cache, return_logits=False)
+        cur = int(self.stage.head(hidden[:, -1:]).squeeze(1).argmax(-1))
+        context = input_ids[0].tolist() + [cur]
+        out = [cur]
+        n_spec = 0
+        position = S                      # cache holds kv for 0..S-1
+        while len(out) < max_new_tokens and (eos_token_id is None
+                                             or cur != eos_token_id):
+            proposal = self._lookup_propose(context, lookup_n, k)
+            m = len(proposal)
+            feed = torch.tensor([[cur] + proposal, ], device=self.device,
+                                dtype=torch.int64)
+            p = (position + torch.arange(m + 1, device=self.device,
+                                         dtype=torch.int32)).unsqueeze(0)
+            logits = self.stage(feed, p, kv_cache=cache)     # [1, m+1, V]
+            targets = logits.argmax(-1)[0].tolist()
+            a = 0
+            while a < m and proposal[a] == targets[a]:
+                a += 1
+            accepted = proposal[:a] + [targets[a]]           # + bonus
+            n_spec += a
+            if eos_token_id is not None and eos_token_id in accepted:
+                accepted = accepted[:accepted.index(eos_token_id) + 1]
+            room = max_new_tokens - len(out)
+            accepted = accepted[:room]
+            out.extend(accepted)
+            context.extend(accepted)
+            cur = accepted[-1] if accepted else targets[0]
+            # roll back: kv is valid for cur + the a accepted proposals
+            position += 1 + a
+            cache.seq_lens.fill_(position)
+        return (torch.tensor([out], device=self.device,
+                             dtype=torch.int64), n_spec)
+
+    @staticmethod
+    def _lookup_propose(context, n, k):
+        """Most recent earlier occurrence of the trailing n-gram; the
+        tokens that followed it are the proposal."""
+        if len(context) <= n:
+            return []
+        tail = context[-n:]
+        # scan right-to-left, excluding the trailing occurrence itself
+        for i in range(len(context) - n - 1, -1, -1):
+            if context[i:i + n] == tail:
+                return context[i + n:i + n + k]
+        return []
+
+    @torch.no_grad()
     def generate_beam(self, input_ids: torch.Tensor,
                       max_new_tokens: int = 64, num_beams: int = 4,
                       length_penalty: float = 1.0,

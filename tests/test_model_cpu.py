@@ -310,3 +310,37 @@ def test_beam_search_decode():
     for b in range(2):
         assert seq_logprob(ids[b], b4[b]) >= seq_logprob(
             ids[b], greedy[b]) - 1e-4
+
+
+def test_speculative_decode_exact_greedy():
+    """Prompt-lookup speculative decoding reproduces plain greedy decode
+    EXACTLY; on a repetitive prompt the speculation actually accepts
+    proposals (n_spec > 0)."""
+    import torch
+
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    # seed 10's greedy decode collapses into a token loop, so the
+    # generated text re-matches its own n-grams and lookup fires
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                       device=torch.device("cpu"), seed=10)
+    torch.manual_seed(19)
+    rep = torch.randint(0, 1024, (1, 16))
+    ref = r.generate(rep, SamplingParams(max_new_tokens=40))
+    out, n_spec = r.generate_speculative(rep, max_new_tokens=40)
+    assert torch.equal(out, ref.to(out.device))
+    assert n_spec > 0, "no proposals accepted on a looping decode"
+
+    # random prompt: rarely matches, still exact
+    rnd = torch.randint(0, 1024, (1, 37))
+    ref2 = r.generate(rnd, SamplingParams(max_new_tokens=10))
+    out2, _ = r.generate_speculative(rnd, max_new_tokens=10)
+    assert torch.equal(out2, ref2.to(out2.device))
+
+    # eos early stop parity
+    eos = int(ref[0, 5])
+    ref3 = r.generate(rep, SamplingParams(max_new_tokens=24,
+                                          eos_token_id=eos))
+    out3, _ = r.generate_speculative(rep, max_new_tokens=24,
+                                     eos_token_id=eos)
+    assert torch.equal(out3, ref3.to(out3.device))
