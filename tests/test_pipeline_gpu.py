@@ -281,3 +281,22 @@ def test_chunked_prefill_gpu():
     finally:
         b.stop()
     assert b.cache.allocator.n_free == b.cache.allocator.n_pages
+
+
+def test_speculative_decode_gpu():
+    """Prompt-lookup speculation on the HIP path: exact greedy equality
+    (the verify step exercises the q_off cached-prefill kernel)."""
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    import os
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1, device=DEV,
+                       dtype=torch.bfloat16, seed=10)
+    torch.manual_seed(19)
+    ids = torch.randint(0, 1024, (1, 16))
+    os.environ["TL_NO_GRAPH"] = "1"      # compare true greedy eager
+    try:
+        ref = r.generate(ids, SamplingParams(max_new_tokens=40))
+        out, n_spec = r.generate_speculative(ids, max_new_tokens=40)
+    finally:
+        del os.environ["TL_NO_GRAPH"]
+    assert torch.equal(out.cpu(), ref.cpu())
