@@ -26,6 +26,7 @@ class GenerationRequest(BaseModel):
     do_sample: bool = True
     num_beams: int = 1
     n: int = 1                        # OpenAI multiple completions
+    speculative: bool = False         # prompt-lookup speculative decode
     presence_penalty: float = 0.0     # OpenAI range [-2, 2]
     frequency_penalty: float = 0.0
     stream: bool = False

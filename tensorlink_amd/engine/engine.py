@@ -311,6 +311,12 @@ class InferenceEngine:
                     out = job.runner.generate_beam(
                         ids, max_new_tokens=sp.max_new_tokens,
                         num_beams=nb, eos_token_id=sp.eos_token_id)
+            elif (request.get("speculative") and self.world == 1
+                    and sp.temperature <= 0):
+                with self._serve_lock:
+                    out, _ = job.runner.generate_speculative(
+                        ids, max_new_tokens=sp.max_new_tokens,
+                        eos_token_id=sp.eos_token_id)
             else:
                 with self._serve_lock:
                     self._bcast(("generate", job.name))

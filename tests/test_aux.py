@@ -309,3 +309,19 @@ def test_logging_and_dashboard(tmp_path, capsys):
     txt = render_status(eng)
     assert "tiny" in txt and "requests: 1" in txt and "[" in txt
     eng.unload_model("tiny")
+
+
+def test_engine_speculative_flag():
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    a = eng.generate({"hf_name": "tiny", "message": "hello",
+                      "max_new_tokens": 6, "do_sample": False,
+                      "output_format": "simple"})
+    b = eng.generate({"hf_name": "tiny", "message": "hello",
+                      "max_new_tokens": 6, "do_sample": False,
+                      "speculative": True, "output_format": "simple"})
+    assert a["response"] == b["response"]
+    eng.unload_model("tiny")
