@@ -175,6 +175,84 @@ def save_stage_to_safetensors(stage: StageModel, out_dir: str,
 
 
 @torch.no_grad()
+def save_hf_checkpoint(stage: StageModel, out_dir: str) -> str:
+    """Export a FULL model (single stage, PP=1) to the HuggingFace
+    safetensors layout — the exact inverse of :func:`_map_hf_key`: fused
+    qkv / gate_up rows are split back into q/k/v and gate/up, names get
+    the ``model.layers.N.`` prefix, and a ``model.safetensors.index.json``
+    is written so :func:`load_stage_from_checkpoint` (and HF tooling)
+    can read the result. Completes the round trip the reference only
+    half-has (parameter retrieval dumps raw per-module state dicts,
+    ``ml/worker.py:1395-1413`` — not a loadable HF layout)."""
+    from safetensors.torch import save_file
+    assert stage.has_embedding and stage.has_head, \
+        "export needs the full model (use the PP=1 stage or gather first)"
+    cfg = stage.config
+    q, kv, i = cfg.q_size, cfg.kv_size, cfg.intermediate_size
+    out = {}
+
+    def put(key, t):
+        out[key] = t.detach().cpu().contiguous()
+
+    sd = stage.state_dict()
+    put("model.embed_tokens.weight", sd["embed_tokens.weight"])
+    put("model.norm.weight", sd["norm"])
+    if not cfg.tie_word_embeddings and "lm_head.weight" in sd:
+        put("lm_head.weight", sd["lm_head.weight"])
+    for li in range(stage.layer_start, stage.layer_end):
+        loc = li - stage.layer_start
+        p = f"model.layers.{li}."
+        n = f"layers.{loc}."
+        put(p + "input_layernorm.weight", sd[n + "input_layernorm"])
+        put(p + "post_attention_layernorm.weight",
+            sd[n + "post_attention_layernorm"])
+        w = sd[n + "self_attn.qkv_proj.weight"]
+        put(p + "self_attn.q_proj.weight", w[:q])
+        put(p + "self_attn.k_proj.weight", w[q:q + kv])
+        put(p + "self_attn.v_proj.weight", w[q + kv:])
+        bkey = n + "self_attn.qkv_proj.bias"
+        if bkey in sd:
+            b = sd[bkey]
+            put(p + "self_attn.q_proj.bias", b[:q])
+            put(p + "self_attn.k_proj.bias", b[q:q + kv])
+            put(p + "self_attn.v_proj.bias", b[q + kv:])
+        put(p + "self_attn.o_proj.weight", sd[n + "self_attn.o_proj.weight"])
+        for extra in ("q_norm", "k_norm"):
+            k2 = n + f"self_attn.{extra}"
+            if k2 in sd:
+                put(p + f"self_attn.{extra}.weight", sd[k2])
+        if n + "mlp.gate_up_proj.weight" in sd:          # dense MLP
+            gu = sd[n + "mlp.gate_up_proj.weight"]
+            put(p + "mlp.gate_proj.weight", gu[:i])
+            put(p + "mlp.up_proj.weight", gu[i:])
+            put(p + "mlp.down_proj.weight", sd[n + "mlp.down_proj.weight"])
+        else:                                            # MoE block
+            put(p + "block_sparse_moe.gate.weight", sd[n + "mlp.gate.weight"])
+            e = 0
+            while n + f"mlp.experts.{e}.gate_up_proj.weight" in sd:
+                gu = sd[n + f"mlp.experts.{e}.gate_up_proj.weight"]
+                ep = p + f"block_sparse_moe.experts.{e}."
+                put(ep + "w1.weight", gu[:i])
+                put(ep + "w3.weight", gu[i:])
+                put(ep + "w2.weight",
+                    sd[n + f"mlp.experts.{e}.down_proj.weight"])
+                e += 1
+
+    os.makedirs(out_dir, exist_ok=True)
+    shard = "model.safetensors"
+    save_file(out, os.path.join(out_dir, shard))
+    index = {"metadata": {"total_size": sum(t.numel() * t.element_size()
+                                            for t in out.values())},
+             "weight_map": {k: shard for k in out}}
+    with open(os.path.join(out_dir, "model.safetensors.index.json"),
+              "w") as f:
+        json.dump(index, f)
+    with open(os.path.join(out_dir, "config.json"), "w") as f:
+        f.write(stage.config.to_json())      # JSON string
+    return out_dir
+
+
+@torch.no_grad()
 def load_stage_from_stage_ckpt(stage: StageModel, out_dir: str) -> int:
     """Re-partition-aware checkpoint load: assemble THIS stage's weights
     from a directory of per-stage files saved under a possibly DIFFERENT
