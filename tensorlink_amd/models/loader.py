@@ -187,6 +187,9 @@ def save_hf_checkpoint(stage: StageModel, out_dir: str) -> str:
     from safetensors.torch import save_file
     assert stage.has_embedding and stage.has_head, \
         "export needs the full model (use the PP=1 stage or gather first)"
+    assert type(stage).__name__ == "StageModel", \
+        "HF export covers the llama/qwen/mixtral families (GPT-2 uses "\
+        "the Conv1D layout; export for it is not implemented)"
     cfg = stage.config
     q, kv, i = cfg.q_size, cfg.kv_size, cfg.intermediate_size
     out = {}

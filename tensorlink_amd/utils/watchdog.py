@@ -52,10 +52,14 @@ class Watchdog:
     """Periodic health loop (reference: 30 s, job_monitor.py:99)."""
 
     def __init__(self, engine=None, interval_s: float = 30.0,
-                 on_failure: Optional[Callable] = None):
+                 on_failure: Optional[Callable] = None,
+                 manage_auto_models: bool = False):
         self.engine = engine
         self.interval_s = interval_s
         self.on_failure = on_failure
+        # demand-driven default-model load/unload each tick (reference
+        # _manage_auto_loaded_models runs inside the validator main loop)
+        self.manage_auto_models = manage_auto_models
         self.history = []
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -80,6 +84,11 @@ class Watchdog:
                 if not ok:
                     self._restart_job(name, job)
             status["jobs"] = jobs
+            if self.manage_auto_models:
+                try:
+                    self.engine.manage_auto_loaded_models()
+                except Exception:
+                    pass
         self.history.append(status)
         del self.history[:-120]
         if status["status"] == HealthStatus.FAILED and self.on_failure:
