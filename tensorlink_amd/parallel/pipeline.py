@@ -91,6 +91,10 @@ class PipelineRunner:
             quantize_experts_fp8(self.stage)
             # torch._scaled_mm is not hipGraph-capture-safe on ROCm 7.2
             self._no_graph = True
+        elif quantize == "fp8-dense":
+            from tensorlink_amd.models.quant import quantize_dense_fp8
+            quantize_dense_fp8(self.stage)
+            self._no_graph = True
         self.stage.eval()
         self.p2p = P2P(rank, world, group, rank_base) if world > 1 else None
         self.is_first = rank == 0
