@@ -53,13 +53,18 @@ class Watchdog:
 
     def __init__(self, engine=None, interval_s: float = 30.0,
                  on_failure: Optional[Callable] = None,
-                 manage_auto_models: bool = False):
+                 manage_auto_models: bool = False,
+                 job_ttl_s: Optional[float] = None):
         self.engine = engine
         self.interval_s = interval_s
         self.on_failure = on_failure
         # demand-driven default-model load/unload each tick (reference
         # _manage_auto_loaded_models runs inside the validator main loop)
         self.manage_auto_models = manage_auto_models
+        # idle-job eviction (reference FREE_JOB_MAX_TIME=3600 s caps a
+        # job's lifetime — validator_thread.py:19,518-520; here the cap
+        # is on IDLE time so active jobs never die mid-service)
+        self.job_ttl_s = job_ttl_s
         self.history = []
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -84,6 +89,16 @@ class Watchdog:
                 if not ok:
                     self._restart_job(name, job)
             status["jobs"] = jobs
+            if self.job_ttl_s is not None:
+                now = time.time()
+                for name, job in list(self.engine.jobs.items()):
+                    idle_since = max(job.last_request, job.created)
+                    if now - idle_since > self.job_ttl_s:
+                        try:
+                            self.engine.unload_model(name)
+                            jobs[name] = "evicted (idle ttl)"
+                        except Exception:
+                            pass
             if self.manage_auto_models:
                 try:
                     self.engine.manage_auto_loaded_models()

@@ -325,3 +325,27 @@ def test_engine_speculative_flag():
                       "speculative": True, "output_format": "simple"})
     assert a["response"] == b["response"]
     eng.unload_model("tiny")
+
+
+def test_watchdog_job_ttl_eviction():
+    """Idle jobs past the TTL are unloaded on a watchdog tick (reference
+    FREE_JOB_MAX_TIME job cap, validator_thread.py:19); active jobs
+    survive."""
+    import time
+
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.utils.watchdog import Watchdog
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    eng.load_model("tiny-moe")
+    time.sleep(0.15)
+    # touch tiny so it is recently active
+    eng.generate({"hf_name": "tiny", "message": "x", "max_new_tokens": 2,
+                  "do_sample": False, "output_format": "simple"})
+    wd = Watchdog(eng, job_ttl_s=0.1)
+    wd.check_once()
+    assert "tiny" in eng.jobs
+    assert "tiny-moe" not in eng.jobs
+    eng.unload_model("tiny")
