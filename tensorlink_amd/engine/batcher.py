@@ -341,6 +341,9 @@ class PPContinuousBatcher(ContinuousBatcher):
         self.slot_samp = [None] * self.max_slots   # (temp, top_p, top_k)
         self._pending_releases: List[int] = []
         self._send_keep = []                       # isend keep-alive
+        # follower-side prompt store (register_prefix needs the full
+        # prompt when the last chunk lands)
+        self._fol_prompt = [None] * self.max_slots
 
     # ------------------------- shared execution -----------------------
     def _sample_rows(self, logits, samps):
@@ -356,13 +359,20 @@ class PPContinuousBatcher(ContinuousBatcher):
             w.wait()
 
     @torch.no_grad()
-    def _exec_step(self, chunk, decodes, releases):
-        """Execute one broadcast command on THIS rank. Returns
-        (first_token or None, decode_tokens or None) on rank 0."""
+    def _exec_step(self, admissions, chunk, decodes, releases):
+        """Execute one broadcast command on THIS rank (identical order
+        everywhere: releases -> admissions/prefix-adopt -> prefill chunk
+        -> decode groups, so the per-rank page allocators stay in
+        lockstep). Returns (first_token or None, decode_tokens or None)
+        on rank 0."""
         r = self.runner
         dev = self.device
         for slot in releases:
             self.cache.release_slot(slot)
+        for slot, ids in admissions:
+            self._fol_prompt[slot] = ids
+            if self.prefix_caching:
+                self.cache.adopt_prefix(slot, ids)
         chunk_tok = None
         if chunk is not None:
             slot, start, end, s_total, samp, ids = chunk
@@ -381,6 +391,9 @@ class PPContinuousBatcher(ContinuousBatcher):
                 hidden = r.p2p.recv((1, S, r.H), r.dtype, r.prev_rank, dev)
                 hidden = self.stage(hidden, pos, kv_cache=view,
                                     return_logits=False)
+            if end == s_total and self.prefix_caching \
+                    and self._fol_prompt[slot] is not None:
+                self.cache.register_prefix(slot, self._fol_prompt[slot])
             if not r.is_last:
                 self._isend(hidden, r.next_rank)
             elif end == s_total:
@@ -449,8 +462,8 @@ class PPContinuousBatcher(ContinuousBatcher):
                 for w, _ in self._send_keep:
                     w.wait()
                 return
-            _, chunk, decodes, releases = cmd
-            self._exec_step(chunk, decodes, releases)
+            _, admissions, chunk, decodes, releases = cmd
+            self._exec_step(admissions, chunk, decodes, releases)
 
     # ------------------------- scheduler (rank 0) ----------------------
     def _finish0(self, slot: int, tok: int):
@@ -471,14 +484,19 @@ class PPContinuousBatcher(ContinuousBatcher):
         assert self.runner.rank == 0
         r = self.runner
         while not self._stop.is_set():
+            admissions = []
             while not self._queue.empty():
                 slot = self._free_slot()
                 if slot is None:
                     break
                 head = self._queue.queue[0][2]
-                # pages for the prompt plus its first generated token
+                ids_list = head.input_ids.tolist()
+                # pages for the prompt plus its first generated token,
+                # minus any published prefix it can adopt
                 need = (head.input_ids.numel() + PAGE) // PAGE
-                if self.cache.allocator.n_free < need:
+                matched = (self.cache.match_prefix(ids_list)
+                           if self.prefix_caching else 0)
+                if self.cache.available_pages() < need - matched // PAGE:
                     break
                 req = self._queue.get_nowait()[2]
                 if req.cancelled:
@@ -487,10 +505,11 @@ class PPContinuousBatcher(ContinuousBatcher):
                     continue
                 self.slots[slot] = req
                 self.slot_prompt[slot] = req.input_ids
-                self.slot_filled[slot] = 0
+                self.slot_filled[slot] = matched
                 self.slot_len[slot] = 0
                 self.slot_samp[slot] = (req.temperature, req.top_p,
                                         req.top_k)
+                admissions.append((slot, ids_list))
             # one prefill chunk per iteration
             chunk = None
             chunk_slot = None
@@ -515,13 +534,16 @@ class PPContinuousBatcher(ContinuousBatcher):
                        and not self._prefilling(i)
                        and self.slot_filled[i] > 0
                        and i != chunk_slot]
-            if chunk is None and not decodes and not self._pending_releases:
+            if chunk is None and not decodes and not admissions \
+                    and not self._pending_releases:
                 self._wake.wait(0.05)
                 self._wake.clear()
                 continue
             rel, self._pending_releases = self._pending_releases, []
-            r.p2p.broadcast_obj(("batch_step", chunk, decodes, rel), src=0)
-            chunk_tok, new_tokens = self._exec_step(chunk, decodes, rel)
+            r.p2p.broadcast_obj(("batch_step", admissions, chunk, decodes,
+                                 rel), src=0)
+            chunk_tok, new_tokens = self._exec_step(admissions, chunk,
+                                                    decodes, rel)
             if chunk is not None:
                 slot, start, end, s_total = chunk[:4]
                 self.slot_filled[slot] = end

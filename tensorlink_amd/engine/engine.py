@@ -169,7 +169,8 @@ class InferenceEngine:
             from tensorlink_amd.engine.batcher import PPContinuousBatcher
             b = PPContinuousBatcher(runner, max_slots=max_slots,
                                     max_ctx=max_ctx, pool_pages=pool_pages,
-                                    prefill_chunk=prefill_chunk)
+                                    prefill_chunk=prefill_chunk,
+                                    prefix_caching=prefix_caching)
             if self.rank == 0:
                 job.batcher = b.start()
                 self.jobs[name] = job

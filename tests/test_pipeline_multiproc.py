@@ -1002,3 +1002,50 @@ def test_tp2_vocab_parallel_training_matches_single_rank():
         opt.step()
         ref_losses.append(float(loss.detach()))
     assert outs[0] == pytest.approx(ref_losses, rel=2e-3)
+
+
+def _pp_prefix_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.parallel.comm import init_distributed
+    init_distributed(backend="gloo")
+    eng = InferenceEngine(rank=rank, world=world, device=torch.device("cpu"))
+    if rank != 0:
+        eng.worker_loop()
+        return
+    eng.load_model("tiny", continuous=True, max_slots=4, max_ctx=512,
+                   prefill_chunk=64, prefix_caching=True)
+    b = eng.jobs["tiny"].batcher
+    torch.manual_seed(47)
+    prompt = torch.randint(0, 1024, (300,))
+    out1 = b.submit(prompt.clone(), max_new_tokens=6).result(timeout=120)
+    out2 = b.submit(prompt.clone(), max_new_tokens=6).result(timeout=120)
+    hits = b.cache.hits
+    import time
+    for _ in range(200):
+        if (b.cache.allocator.n_free + len(b.cache.lru)
+                == b.cache.allocator.n_pages):
+            break
+        time.sleep(0.02)
+    drained = (b.cache.allocator.n_free + len(b.cache.lru)
+               == b.cache.allocator.n_pages)
+    eng.unload_model("tiny")
+    eng.shutdown()
+    q.put((rank, (out1, out2, hits, drained)))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_pp2_prefix_caching():
+    """Prefix caching across a 2-stage pipeline: admissions ride the
+    command stream, so follower allocators adopt/register the same pages
+    in lockstep; the repeat request reuses 2 pages and outputs match."""
+    outs = dict(_run_collect(_pp_prefix_worker, 2, _port(39), n_results=1))
+    out1, out2, hits, drained = outs[0]
+    assert out1 == out2
+    assert hits == 256
+    assert drained
