@@ -565,6 +565,9 @@ class PipelineTrainer:
         # test configs (keeps exact parity with existing expectations)
         self._chunked_ce = (self.stage.has_head
                             and plan.config.vocab_size >= 32000)
+        # (v0, group) when the LM head is vocab-sharded across a TP
+        # group (set by TPPPTrainer(vocab_parallel=True))
+        self._vp = None
 
     def train_step(self, input_ids: Optional[torch.Tensor] = None,
                    labels: Optional[torch.Tensor] = None,
@@ -618,7 +621,8 @@ class PipelineTrainer:
 
         def fwd(i):
             s, e = i * b, (i + 1) * b
-            want_logits = is_last and not self._chunked_ce
+            want_logits = is_last and not self._chunked_ce \
+                and self._vp is None
             if is_first:
                 inp = None
                 out = self.stage(input_ids[s:e], pos, training=True,
@@ -691,6 +695,15 @@ class PipelineTrainer:
         ~0.6 MB/token at Qwen vocab); small ones take the plain path.
         Both are exactly F.cross_entropy(head(hidden), labels)."""
         from tensorlink_amd import ops as tl_ops
+        if self._vp is not None:
+            from tensorlink_amd.parallel.tp import vocab_parallel_ce
+            v0, group = self._vp
+            h = tl_ops.rmsnorm(out, self.stage.norm.to(out.dtype),
+                               self.stage.config.rms_norm_eps)
+            lg = self.stage.lm_head(h)[:, :-1]
+            lb = labels[:, 1:].reshape(-1)
+            return vocab_parallel_ce(
+                lg.reshape(-1, lg.shape[-1]), lb, v0, group)
         if self._chunked_ce:
             # apply the final norm here (stage.head fuses norm+GEMM;
             # the chunked CE takes the raw head weight)

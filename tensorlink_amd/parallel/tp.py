@@ -166,8 +166,11 @@ def build_tp_model(config_or_name, tp_rank: int, tp: int, device=None,
 
     stage = build_full_model(local_config(config, tp))
     if vocab_parallel:
-        stage.lm_head = TLLinear(config.hidden_size,
-                                 config.vocab_size // tp, bias=False)
+        # column-parallel: output (vocab) dim sharded, so backward must
+        # all-reduce the input gradient across the group
+        stage.lm_head = ColumnParallelLinear(config.hidden_size,
+                                             config.vocab_size // tp,
+                                             bias=False)
     # swap row-parallel outputs for all-reducing variants and
     # column-parallel inputs for the f-op (backward all-reduce) variants
     for layer in stage.layers:
@@ -440,7 +443,7 @@ class TPPPTrainer:
     def __init__(self, model, rank: int, world: int, tp: int, device=None,
                  dtype=None, seed: int = 0, lr: float = 1e-4,
                  grid_base: int = 0, pp_groups=None, tp_groups=None,
-                 **opt_kwargs):
+                 vocab_parallel: bool = False, **opt_kwargs):
         """rank/world are GRID-local (0..tp*pp). grid_base offsets them
         into the global rank space, and pp_groups/tp_groups may be
         pre-built by an outer 3-D (DP) wrapper — dist.new_group is
@@ -481,10 +484,22 @@ class TPPPTrainer:
         full = build_stage(config, spec)
         init_random_stage(full, device="cpu", dtype=dtype,
                           seed=seed + self.stage_idx)
-        shards = shard_state(full.state_dict(), config, self.tp_rank, tp)
+        vp = vocab_parallel and stage.has_head and tp > 1
+        if vocab_parallel:
+            assert not config.tie_word_embeddings
+        shards = shard_state(full.state_dict(), config, self.tp_rank, tp,
+                             vocab_parallel=vp)
         del full
         if tp > 1:
             _swap_tp_linears(stage, self.tp_group)
+        if vp:
+            head = ColumnParallelLinear(config.hidden_size,
+                                        config.vocab_size // tp,
+                                        bias=False)
+            head.group = self.tp_group
+            stage.lm_head = head
+            self.trainer._vp = (self.tp_rank * (config.vocab_size // tp),
+                                self.tp_group)
         stage.load_state_dict(shards)
         stage.to(device=self.device, dtype=dtype)
         stage.train()
@@ -494,10 +509,12 @@ class TPPPTrainer:
         # optimizer — rebuild over the final parameter set
         self.trainer.optimizer = FusedAdamW(stage.parameters(), lr=lr,
                                             **opt_kwargs)
+        sharded = ("qkv_proj", "gate_up_proj", "o_proj", "down_proj")
+        if vp:
+            sharded = sharded + ("lm_head",)
         self._replicated = [
             p for n, p in stage.named_parameters()
-            if not any(t in n for t in ("qkv_proj", "gate_up_proj",
-                                        "o_proj", "down_proj"))]
+            if not any(t in n for t in sharded)]
 
         def _sync_replicated(tr):
             if self.tp <= 1:
