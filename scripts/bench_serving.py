@@ -24,7 +24,6 @@ import sys
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import json
 import random
-import statistics
 import threading
 import time
 

@@ -1,0 +1,97 @@
+"""Property-based fuzzing (hypothesis): planner robustness over random
+model/world shapes, paged-cache invariants under random lease/release
+traffic, and chunked-CE equality at arbitrary shapes."""
+
+import hypothesis.strategies as st
+import pytest
+import torch
+from hypothesis import given, settings
+
+from tensorlink_amd.models.configs import ModelConfig
+from tensorlink_amd.parallel.planner import AssignmentError, plan_for_world
+
+
+@settings(max_examples=40, deadline=None)
+@given(layers=st.integers(2, 48), world=st.integers(1, 8),
+       hidden=st.sampled_from([256, 512, 1024, 4096]),
+       heads=st.sampled_from([4, 8, 16, 32]))
+def test_planner_fuzz_valid_or_clean_error(layers, world, hidden, heads):
+    """Any config either plans cleanly (full cover, ordered stages,
+    embed-first/head-last) or raises AssignmentError — never crashes or
+    returns a malformed plan."""
+    cfg = ModelConfig(
+        name=f"fuzz-{layers}-{hidden}", vocab_size=1024,
+        hidden_size=hidden, intermediate_size=hidden * 3,
+        num_hidden_layers=layers, num_attention_heads=heads,
+        num_key_value_heads=max(1, heads // 2), head_dim=hidden // heads,
+        max_position_embeddings=4096)
+    try:
+        plan = plan_for_world(cfg, world)
+    except AssignmentError:
+        return
+    assert plan.num_stages == world
+    covered = []
+    for i in range(world):
+        spec = plan.stage_for_rank(i)
+        covered.extend(range(spec.layer_start, spec.layer_end))
+        assert spec.has_embedding == (i == 0)
+        assert spec.has_head == (i == world - 1)
+    assert covered == list(range(layers))
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.data())
+def test_paged_allocator_fuzz_invariants(data):
+    """Random ensure/release traffic never leaks or double-frees pages,
+    and table entries always point at leased pages."""
+    from tensorlink_amd.models.paged import PAGE, DynamicPagedKVCache
+    from tensorlink_amd.models.configs import get_config
+    slots, pool = 4, 12
+    cache = DynamicPagedKVCache(1, slots, pool, PAGE * 6,
+                                get_config("tiny"), torch.device("cpu"),
+                                torch.float32)
+    lens = [0] * slots
+    for _ in range(data.draw(st.integers(5, 40))):
+        slot = data.draw(st.integers(0, slots - 1))
+        if data.draw(st.booleans()):
+            want = data.draw(st.integers(1, PAGE * 6))
+            need = (want + PAGE - 1) // PAGE - len(cache._slot_pages[slot])
+            if need <= cache.allocator.n_free:
+                cache.ensure(slot, want)
+                lens[slot] = max(lens[slot], want)
+        else:
+            cache.release_slot(slot)
+            lens[slot] = 0
+        leased = sum(len(p) for p in cache._slot_pages)
+        assert leased + cache.allocator.n_free == pool
+        all_pages = [p for sp in cache._slot_pages for p in sp]
+        assert len(set(all_pages)) == len(all_pages)      # no double-lease
+    for s in range(slots):
+        cache.release_slot(s)
+    assert cache.allocator.n_free == pool
+
+
+@settings(max_examples=20, deadline=None)
+@given(B=st.integers(1, 3), S=st.integers(2, 20),
+       V=st.integers(16, 200), chunk=st.integers(1, 64),
+       pad=st.booleans())
+def test_chunked_ce_fuzz(B, S, V, chunk, pad):
+    """Chunked CE == full CE for arbitrary shapes/chunk sizes/padding."""
+    from tensorlink_amd.ops import reference as ref
+    H = 16
+    hidden = torch.randn(B, S, H, requires_grad=True)
+    w = torch.randn(V, H, requires_grad=True)
+    labels = torch.randint(0, V, (B, S))
+    if pad:
+        labels[:, -1] = -100
+    loss_c = ref.chunked_causal_lm_loss(hidden, w, labels, chunk=chunk)
+    h2 = hidden.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    loss_f = ref.causal_lm_loss(h2 @ w2.t(), labels)
+    if torch.isnan(loss_f):
+        return
+    loss_c.backward()
+    loss_f.backward()
+    assert torch.allclose(loss_c, loss_f, atol=1e-5)
+    assert torch.allclose(hidden.grad, h2.grad, atol=1e-5)
+    assert torch.allclose(w.grad, w2.grad, atol=1e-4)
