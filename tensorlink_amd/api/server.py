@@ -146,6 +146,14 @@ def create_app(engine: InferenceEngine,
     def network_history():
         return {"history": engine.metrics.snapshot()["history"]}
 
+    @app.get("/proposal-history")
+    def proposal_history():
+        # reference api/node.py:336 serves on-chain governance
+        # proposals; the chain layer is deleted by design on a single
+        # node, so the endpoint exists with an empty history
+        return {"proposals": [], "note": "no on-chain governance in "
+                "the single-node build"}
+
     @app.get("/node-info")
     def node_info_ep():
         info = {"version": "0.1.0", "role": "engine",
