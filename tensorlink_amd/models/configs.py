@@ -35,15 +35,20 @@ class ModelConfig:
     qkv_bias: bool = False
     hidden_act: str = "silu"
     dtype: str = "bfloat16"
-    # MoE (Mixtral-style); num_local_experts == 0 => dense MLP
+    # MoE; num_local_experts == 0 => dense MLP. expert_intermediate_size
+    # defaults to intermediate_size (Mixtral); Qwen3-MoE uses a smaller
+    # per-expert width (moe_intermediate_size)
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
+    expert_intermediate_size: Optional[int] = None
     # architecture tag for checkpoint key mapping
     architecture: str = "llama"
 
     def __post_init__(self):
         if self.head_dim is None:
             self.head_dim = self.hidden_size // self.num_attention_heads
+        if self.expert_intermediate_size is None:
+            self.expert_intermediate_size = self.intermediate_size
 
     @property
     def is_moe(self) -> bool:
@@ -65,7 +70,8 @@ class ModelConfig:
         if self.qkv_bias:
             attn += q + 2 * kv
         if self.is_moe:
-            mlp = self.num_local_experts * 3 * h * self.intermediate_size
+            mlp = (self.num_local_experts * 3 * h
+                   * self.expert_intermediate_size)
             mlp += h * self.num_local_experts  # router
         else:
             mlp = 3 * h * self.intermediate_size
@@ -96,6 +102,8 @@ class ModelConfig:
         arch = "llama"
         if any("Qwen" in a for a in archs):
             arch = "qwen2"
+        if any("Qwen3" in a for a in archs):
+            arch = "qwen3"
         if any("Mixtral" in a for a in archs):
             arch = "mixtral"
         get = cfg.get
@@ -114,8 +122,10 @@ class ModelConfig:
             rms_norm_eps=get("rms_norm_eps", 1e-6),
             tie_word_embeddings=get("tie_word_embeddings", False),
             qkv_bias=(arch == "qwen2" and "Qwen3" not in str(archs)),
-            num_local_experts=get("num_local_experts", 0) or 0,
+            num_local_experts=(get("num_local_experts", 0)
+                               or get("num_experts", 0) or 0),
             num_experts_per_tok=get("num_experts_per_tok", 2) or 2,
+            expert_intermediate_size=get("moe_intermediate_size", None),
             architecture=arch,
         )
 
@@ -197,6 +207,27 @@ def _gpt2(name):
     return make
 
 
+def _tiny_qwen3_moe() -> ModelConfig:
+    return ModelConfig(
+        name="tiny-qwen3-moe", vocab_size=1024, hidden_size=256,
+        intermediate_size=768, num_hidden_layers=4,
+        num_attention_heads=4, num_key_value_heads=2, head_dim=64,
+        max_position_embeddings=4096, num_local_experts=4,
+        num_experts_per_tok=2, expert_intermediate_size=192,
+        architecture="qwen3")
+
+
+def _qwen3_30b_a3b() -> ModelConfig:
+    # Qwen3-30B-A3B: 128 experts, 8 active, per-expert width 768
+    return ModelConfig(
+        name="Qwen/Qwen3-30B-A3B", vocab_size=151936, hidden_size=2048,
+        intermediate_size=6144, num_hidden_layers=48,
+        num_attention_heads=32, num_key_value_heads=4, head_dim=128,
+        max_position_embeddings=40960, rope_theta=1000000.0,
+        num_local_experts=128, num_experts_per_tok=8,
+        expert_intermediate_size=768, architecture="qwen3")
+
+
 PRESETS = {
     "gpt2-small": _gpt2("gpt2-small"),
     "gpt2": _gpt2("gpt2-small"),
@@ -212,6 +243,8 @@ PRESETS = {
     "tiny": _tiny,
     "tiny-qwen3": _tiny_qwen3,
     "tiny-moe": _tiny_moe,
+    "tiny-qwen3-moe": _tiny_qwen3_moe,
+    "Qwen/Qwen3-30B-A3B": _qwen3_30b_a3b,
 }
 
 

@@ -15,6 +15,8 @@ planner assigns to one GPU rank (cf. reference grouped entries
 
 from __future__ import annotations
 
+import dataclasses
+
 import math
 from typing import List, Optional
 
@@ -198,7 +200,12 @@ class MoEMLP(nn.Module):
         self.num_experts = config.num_local_experts
         self.top_k = config.num_experts_per_tok
         self.gate = nn.Linear(h, self.num_experts, bias=False)
-        self.experts = nn.ModuleList([MLP(config) for _ in range(self.num_experts)])
+        expert_cfg = (config if config.expert_intermediate_size == i
+                      else dataclasses.replace(
+                          config,
+                          intermediate_size=config.expert_intermediate_size))
+        self.experts = nn.ModuleList([MLP(expert_cfg)
+                                      for _ in range(self.num_experts)])
 
     def forward(self, x):
         B, S, H = x.shape

@@ -41,8 +41,9 @@ _HF_LAYER_MAP = {
     "mlp.gate_proj.weight": ("mlp.gate_up_proj.weight", 0),
     "mlp.up_proj.weight": ("mlp.gate_up_proj.weight", "i"),
     "mlp.down_proj.weight": ("mlp.down_proj.weight", None),
-    # Mixtral MoE router
+    # MoE routers: Mixtral / Qwen3-MoE
     "block_sparse_moe.gate.weight": ("mlp.gate.weight", None),
+    "mlp.gate.weight": ("mlp.gate.weight", None),
 }
 
 
@@ -50,7 +51,8 @@ def _resolve_offset(sym, config) -> int:
     if sym in (None, 0):
         return 0
     return {"q": config.q_size, "q+kv": config.q_size + config.kv_size,
-            "i": config.intermediate_size}[sym]
+            "i": config.intermediate_size,
+            "ei": config.expert_intermediate_size}[sym]
 
 
 def _map_hf_key(key: str, layer_start: int, layer_end: int, stage):
@@ -85,8 +87,20 @@ def _map_hf_key(key: str, layer_start: int, layer_end: int, stage):
             wname = parts[3]
             native, off = {
                 "w1": ("gate_up_proj.weight", 0),
-                "w3": ("gate_up_proj.weight", "i"),
+                "w3": ("gate_up_proj.weight", "ei"),
                 "w2": ("down_proj.weight", None),
+            }[wname]
+            return (f"layers.{local}.mlp.experts.{e}.{native}",
+                    _resolve_offset(off, stage.config))
+        # Qwen3-MoE experts: mlp.experts.E.gate_proj/up_proj/down_proj
+        if sub.startswith("mlp.experts."):
+            parts = sub.split(".")
+            e = parts[2]
+            wname = parts[3]
+            native, off = {
+                "gate_proj": ("gate_up_proj.weight", 0),
+                "up_proj": ("gate_up_proj.weight", "ei"),
+                "down_proj": ("down_proj.weight", None),
             }[wname]
             return (f"layers.{local}.mlp.experts.{e}.{native}",
                     _resolve_offset(off, stage.config))
@@ -230,15 +244,24 @@ def save_hf_checkpoint(stage: StageModel, out_dir: str) -> str:
             put(p + "mlp.up_proj.weight", gu[i:])
             put(p + "mlp.down_proj.weight", sd[n + "mlp.down_proj.weight"])
         else:                                            # MoE block
-            put(p + "block_sparse_moe.gate.weight", sd[n + "mlp.gate.weight"])
+            ei = cfg.expert_intermediate_size
+            mixtral = cfg.architecture == "mixtral"
+            put(p + ("block_sparse_moe.gate.weight" if mixtral
+                     else "mlp.gate.weight"), sd[n + "mlp.gate.weight"])
             e = 0
             while n + f"mlp.experts.{e}.gate_up_proj.weight" in sd:
                 gu = sd[n + f"mlp.experts.{e}.gate_up_proj.weight"]
-                ep = p + f"block_sparse_moe.experts.{e}."
-                put(ep + "w1.weight", gu[:i])
-                put(ep + "w3.weight", gu[i:])
-                put(ep + "w2.weight",
-                    sd[n + f"mlp.experts.{e}.down_proj.weight"])
+                dn = sd[n + f"mlp.experts.{e}.down_proj.weight"]
+                if mixtral:
+                    ep = p + f"block_sparse_moe.experts.{e}."
+                    put(ep + "w1.weight", gu[:ei])
+                    put(ep + "w3.weight", gu[ei:])
+                    put(ep + "w2.weight", dn)
+                else:
+                    ep = p + f"mlp.experts.{e}."
+                    put(ep + "gate_proj.weight", gu[:ei])
+                    put(ep + "up_proj.weight", gu[ei:])
+                    put(ep + "down_proj.weight", dn)
                 e += 1
 
     os.makedirs(out_dir, exist_ok=True)

@@ -50,7 +50,8 @@ def layer_param_bytes(config, dtype_bytes: int) -> int:
     if config.qkv_bias:
         attn += config.q_size + 2 * config.kv_size
     if config.is_moe:
-        mlp = config.num_local_experts * 3 * h * config.intermediate_size
+        mlp = (config.num_local_experts * 3 * h
+               * config.expert_intermediate_size)
         mlp += h * config.num_local_experts
     else:
         mlp = 3 * h * config.intermediate_size

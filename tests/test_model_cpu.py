@@ -359,7 +359,7 @@ def test_hf_export_roundtrip(tmp_path):
                                               save_hf_checkpoint)
     from tensorlink_amd.parallel.planner import plan_for_world
 
-    for preset in ("tiny", "tiny-qwen3", "tiny-moe"):
+    for preset in ("tiny", "tiny-qwen3", "tiny-moe", "tiny-qwen3-moe"):
         src = build_full_model(get_config(preset))
         init_random_stage(src, dtype=torch.float32, seed=21)
         d = str(tmp_path / preset)
@@ -412,3 +412,39 @@ def test_fp8_dense_quantization_close_to_ref():
     assert rel < 0.2 and cos > 0.98, (float(rel), float(cos))
     out = r_q.generate(ids, SamplingParams(max_new_tokens=4))
     assert out.shape == (1, 4)
+
+
+def test_qwen3_moe_family():
+    """Qwen3-MoE: smaller per-expert width, qk-norm attention, decode ==
+    full forward, plans across 2 stages, and the 30B-A3B preset counts
+    ~30.5B params."""
+    import torch
+
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+
+    cfg = get_config("tiny-qwen3-moe")
+    assert cfg.expert_intermediate_size == 192 != cfg.intermediate_size
+    m = build_full_model(cfg)
+    init_random_stage(m, dtype=torch.float32, seed=3)
+    assert m.layers[0].self_attn.use_qk_norm
+    assert m.layers[0].mlp.experts[0].gate_up_proj.weight.shape == (384, 256)
+
+    torch.manual_seed(40)
+    ids = torch.randint(0, 1024, (2, 10))
+    r = PipelineRunner(plan_for_world(cfg, 1), 0, 1,
+                       device=torch.device("cpu"), seed=3)
+    out = r.generate(ids, SamplingParams(max_new_tokens=5))
+    cur = ids
+    for _ in range(5):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).expand(2, -1).contiguous()
+        logits = m(cur, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 10:], out)
+
+    plan = plan_for_world("Qwen/Qwen3-30B-A3B", 2)
+    assert plan.num_stages == 2
+    assert abs(get_config("Qwen/Qwen3-30B-A3B").param_count() / 1e9
+               - 30.5) < 1.0
