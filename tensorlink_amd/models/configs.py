@@ -41,6 +41,8 @@ class ModelConfig:
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
     expert_intermediate_size: Optional[int] = None
+    # HF rope_scaling dict (Llama-3.1 "llama3" frequency scaling)
+    rope_scaling: Optional[dict] = None
     # architecture tag for checkpoint key mapping
     architecture: str = "llama"
 
@@ -126,6 +128,7 @@ class ModelConfig:
                                or get("num_experts", 0) or 0),
             num_experts_per_tok=get("num_experts_per_tok", 2) or 2,
             expert_intermediate_size=get("moe_intermediate_size", None),
+            rope_scaling=get("rope_scaling", None),
             architecture=arch,
         )
 
@@ -228,6 +231,18 @@ def _qwen3_30b_a3b() -> ModelConfig:
         expert_intermediate_size=768, architecture="qwen3")
 
 
+def _llama31_8b() -> ModelConfig:
+    return ModelConfig(
+        name="meta-llama/Llama-3.1-8B", vocab_size=128256,
+        hidden_size=4096, intermediate_size=14336, num_hidden_layers=32,
+        num_attention_heads=32, num_key_value_heads=8, head_dim=128,
+        max_position_embeddings=131072, rope_theta=500000.0,
+        rms_norm_eps=1e-5, rope_scaling={
+            "rope_type": "llama3", "factor": 8.0,
+            "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+            "original_max_position_embeddings": 8192})
+
+
 PRESETS = {
     "gpt2-small": _gpt2("gpt2-small"),
     "gpt2": _gpt2("gpt2-small"),
@@ -244,6 +259,7 @@ PRESETS = {
     "tiny-qwen3": _tiny_qwen3,
     "tiny-moe": _tiny_moe,
     "tiny-qwen3-moe": _tiny_qwen3_moe,
+    "meta-llama/Llama-3.1-8B": _llama31_8b,
     "Qwen/Qwen3-30B-A3B": _qwen3_30b_a3b,
 }
 

@@ -71,6 +71,29 @@ class TLLinear(nn.Linear):
         return ops.linear(x, self.weight, self.bias)
 
 
+def compute_inv_freq(config: ModelConfig) -> torch.Tensor:
+    """RoPE inverse frequencies, honoring HF rope_scaling. Implements
+    the Llama-3.1 "llama3" spectrum scaling (long wavelengths divided by
+    `factor`, short kept, smooth ramp between — matches transformers'
+    _compute_llama3_parameters); the kernels consume the resulting
+    buffer directly, so scaled-rope models need no kernel change."""
+    D = config.head_dim
+    inv = 1.0 / (config.rope_theta ** (
+        torch.arange(0, D, 2, dtype=torch.float32) / D))
+    rs = config.rope_scaling
+    if rs and rs.get("rope_type", rs.get("type")) == "llama3":
+        factor = rs["factor"]
+        lo, hi = rs["low_freq_factor"], rs["high_freq_factor"]
+        orig = rs["original_max_position_embeddings"]
+        wavelen = 2 * math.pi / inv
+        smooth = (orig / wavelen - lo) / (hi - lo)
+        smoothed = (1 - smooth) / factor * inv + smooth * inv
+        inv = torch.where(wavelen > orig / lo, inv / factor, inv)
+        mid = (wavelen <= orig / lo) & (wavelen >= orig / hi)
+        inv = torch.where(mid, smoothed, inv)
+    return inv
+
+
 class Attention(nn.Module):
     def __init__(self, config: ModelConfig):
         super().__init__()
@@ -91,10 +114,8 @@ class Attention(nn.Module):
         if self.use_qk_norm:
             self.q_norm = nn.Parameter(torch.ones(self.head_dim))
             self.k_norm = nn.Parameter(torch.ones(self.head_dim))
-        inv_freq = 1.0 / (config.rope_theta ** (
-            torch.arange(0, self.head_dim, 2, dtype=torch.float32)
-            / self.head_dim))
-        self.register_buffer("inv_freq", inv_freq, persistent=False)
+        self.register_buffer("inv_freq", compute_inv_freq(config),
+                             persistent=False)
 
     def _split_qkv(self, qkv, B, S):
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size],

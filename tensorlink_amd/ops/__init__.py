@@ -94,8 +94,10 @@ def apply_rope_(q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
     Inference path (no autograd). sign=-1 applies the inverse rotation.
     """
     if not _on_gpu(q):
-        T, Hq, D = q.shape
-        cos, sin = ref.rope_cos_sin(D, positions, device=q.device)
+        # compute cos/sin from the PASSED inv_freq (NOT a default theta:
+        # Qwen uses rope_theta=1e6, Llama-3.1 a scaled spectrum)
+        freqs = positions.float()[:, None] * inv_freq.float()[None, :]
+        cos, sin = freqs.cos(), freqs.sin()
         # reference expects [B,S,H,D]
         qq, kk = ref.apply_rope(q.unsqueeze(0), k.unsqueeze(0), cos,
                                 sin * sign)

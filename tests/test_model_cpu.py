@@ -448,3 +448,29 @@ def test_qwen3_moe_family():
     assert plan.num_stages == 2
     assert abs(get_config("Qwen/Qwen3-30B-A3B").param_count() / 1e9
                - 30.5) < 1.0
+
+
+def test_llama31_rope_scaling_matches_transformers():
+    """compute_inv_freq reproduces transformers' llama3 rope scaling
+    exactly, and the preset plans/builds."""
+    import torch
+    from transformers.modeling_rope_utils import ROPE_INIT_FUNCTIONS
+
+    from tensorlink_amd.models import get_config
+    from tensorlink_amd.models.dense import compute_inv_freq
+    from tensorlink_amd.parallel.planner import plan_for_world
+
+    cfg = get_config("meta-llama/Llama-3.1-8B")
+    mine = compute_inv_freq(cfg)
+
+    from transformers import LlamaConfig
+    hf = LlamaConfig(
+        rope_theta=cfg.rope_theta, head_dim=cfg.head_dim,
+        hidden_size=cfg.hidden_size,
+        num_attention_heads=cfg.num_attention_heads,
+        max_position_embeddings=cfg.max_position_embeddings,
+        rope_scaling=dict(cfg.rope_scaling))
+    ref, _ = ROPE_INIT_FUNCTIONS["llama3"](hf, "cpu")
+    torch.testing.assert_close(mine, ref.float(), atol=1e-6, rtol=1e-6)
+
+    assert plan_for_world(cfg, 2).num_stages == 2
