@@ -100,6 +100,10 @@ class ModelConfig:
         if not isinstance(cfg, dict):
             cfg = {k: getattr(cfg, k) for k in dir(cfg) if not k.startswith("_")
                    if not callable(getattr(cfg, k, None))}
+        if "architecture" in cfg and "architectures" not in cfg:
+            # our own to_json round-trip (save_hf_checkpoint writes it)
+            known = {f.name for f in dataclasses.fields(cls)}
+            return cls(**{k: v for k, v in cfg.items() if k in known})
         archs = cfg.get("architectures") or []
         arch = "llama"
         if any("Qwen" in a for a in archs):

@@ -474,3 +474,37 @@ def test_llama31_rope_scaling_matches_transformers():
     torch.testing.assert_close(mine, ref.float(), atol=1e-6, rtol=1e-6)
 
     assert plan_for_world(cfg, 2).num_stages == 2
+
+
+def test_export_then_serve_from_directory(tmp_path):
+    """Closest offline analog of the HF end-to-end flow (ROADMAP item
+    12): export a model to the HF layout, then serve it purely from the
+    directory path — get_config reads config.json, the runner loads the
+    safetensors — and greedy output matches the source model."""
+    import torch
+
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import (init_random_stage,
+                                              save_hf_checkpoint)
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+
+    src = build_full_model(get_config("tiny-qwen3"))
+    init_random_stage(src, dtype=torch.float32, seed=31)
+    d = str(tmp_path / "export")
+    save_hf_checkpoint(src, d)
+
+    cfg = get_config(d)                     # from the dir's config.json
+    assert cfg.architecture == "qwen3"      # qk-norm survives round-trip
+    r = PipelineRunner(plan_for_world(cfg, 1), 0, 1,
+                       device=torch.device("cpu"), init="checkpoint",
+                       ckpt_dir=d, dtype=torch.float32)
+    torch.manual_seed(73)
+    ids = torch.randint(0, 1024, (1, 9))
+    out = r.generate(ids, SamplingParams(max_new_tokens=5))
+    cur = ids
+    for _ in range(5):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).contiguous()
+        logits = src(cur, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 9:], out)
