@@ -146,6 +146,16 @@ def create_app(engine: InferenceEngine,
     def network_history():
         return {"history": engine.metrics.snapshot()["history"]}
 
+    @app.get("/metrics")
+    def prometheus_metrics():
+        # Prometheus exposition (beyond the reference's JSON /stats —
+        # production serving needs a scrapeable endpoint)
+        from prometheus_client import (CONTENT_TYPE_LATEST,
+                                       generate_latest)
+        from starlette.responses import Response
+        _sync_prometheus(engine)
+        return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
+
     @app.get("/proposal-history")
     def proposal_history():
         # reference api/node.py:336 serves on-chain governance
@@ -164,6 +174,31 @@ def create_app(engine: InferenceEngine,
         return info
 
     return app
+
+
+_PROM = {}
+
+
+def _sync_prometheus(engine):
+    """Mirror EngineMetrics counters into prometheus_client gauges
+    (created lazily so importing the server never requires a scrape)."""
+    from prometheus_client import Gauge
+    if not _PROM:
+        _PROM["requests"] = Gauge("tl_requests_total",
+                                  "requests served")
+        _PROM["errors"] = Gauge("tl_errors_total", "failed requests")
+        _PROM["tokens"] = Gauge("tl_tokens_total", "tokens generated")
+        _PROM["uptime"] = Gauge("tl_uptime_seconds", "engine uptime")
+        _PROM["latency_p50"] = Gauge("tl_latency_p50_seconds",
+                                     "request latency p50")
+        _PROM["models"] = Gauge("tl_models_loaded", "loaded models")
+    m = engine.metrics.snapshot()
+    _PROM["requests"].set(m.get("requests_total", 0))
+    _PROM["errors"].set(m.get("errors_total", 0))
+    _PROM["tokens"].set(m.get("tokens_total", 0))
+    _PROM["uptime"].set(m.get("uptime_s", 0))
+    _PROM["latency_p50"].set(m.get("latency_p50_s") or 0)
+    _PROM["models"].set(len(engine.jobs))
 
 
 class TensorlinkAPI:

@@ -99,3 +99,22 @@ def test_stats_and_info(client):
     assert client.get("/models").json()["models"]
     assert isinstance(client.get("/model-demand").json(), dict)
     assert "history" in client.get("/network-history").json()
+
+
+def test_metrics_prometheus_endpoint():
+    """/metrics serves Prometheus exposition with engine counters."""
+    import torch
+    from fastapi.testclient import TestClient
+
+    from tensorlink_amd.api.server import create_app
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    eng.generate({"hf_name": "tiny", "message": "x", "max_new_tokens": 2,
+                  "do_sample": False, "output_format": "simple"})
+    client = TestClient(create_app(eng))
+    resp = client.get("/metrics")
+    assert resp.status_code == 200
+    assert "tl_requests_total 1.0" in resp.text
+    assert "tl_models_loaded 1.0" in resp.text
+    eng.unload_model("tiny")
