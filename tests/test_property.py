@@ -95,3 +95,37 @@ def test_chunked_ce_fuzz(B, S, V, chunk, pad):
     assert torch.allclose(loss_c, loss_f, atol=1e-5)
     assert torch.allclose(hidden.grad, h2.grad, atol=1e-5)
     assert torch.allclose(w.grad, w2.grad, atol=1e-4)
+
+
+@settings(max_examples=8, deadline=None)
+@given(lens=st.lists(st.integers(3, 200), min_size=1, max_size=4),
+       chunk=st.sampled_from([None, 16, 64]),
+       prefix=st.booleans(), new=st.integers(1, 8))
+def test_batcher_fuzz_matches_serial(lens, chunk, prefix, new):
+    """Random prompt mixes through the batcher (any chunking / prefix
+    caching) always reproduce serial greedy outputs and drain the
+    pool."""
+    from tensorlink_amd.engine.batcher import ContinuousBatcher
+    from tensorlink_amd.parallel.pipeline import (PipelineRunner,
+                                                  SamplingParams)
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                       device=torch.device("cpu"))
+    b = ContinuousBatcher(r, max_slots=4, max_ctx=512,
+                          prefill_chunk=chunk,
+                          prefix_caching=prefix).start()
+    try:
+        g = torch.Generator().manual_seed(sum(lens) + new)
+        prompts = [torch.randint(0, 1024, (n,), generator=g)
+                   for n in lens]
+        reqs = [b.submit(p.clone(), max_new_tokens=new) for p in prompts]
+        outs = [rq.result(timeout=120) for rq in reqs]
+        for p, o in zip(prompts, outs):
+            ref = r.generate(p.unsqueeze(0),
+                             SamplingParams(max_new_tokens=new))
+            assert o == ref[0].tolist()
+    finally:
+        b.stop()
+    free = b.cache.allocator.n_free
+    if prefix:
+        free += len(b.cache.lru)
+    assert free == b.cache.allocator.n_pages
