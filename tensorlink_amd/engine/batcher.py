@@ -145,11 +145,26 @@ class ContinuousBatcher:
         self._thread.start()
         return self
 
-    def stop(self):
+    def stop(self, drain: bool = False):
+        """Stop the scheduler. drain=True first finishes every admitted
+        request (queued ones are still cut loose); drain=False ends all
+        requests at their current token (clients unblock either way —
+        no request ever hangs on a stopped batcher)."""
+        self._drain = drain
         self._stop.set()
         self._wake.set()
         if self._thread:
-            self._thread.join(30)
+            self._thread.join(120 if drain else 30)
+        # cut loose anything still attached
+        leftovers = [r for r in self.slots if r is not None]
+        while not self._queue.empty():
+            try:
+                leftovers.append(self._queue.get_nowait()[2])
+            except Exception:
+                break
+        for req in leftovers:
+            req.tokens.put(None)
+            req.done.set()
 
     def submit(self, input_ids: torch.Tensor, **kw) -> Request:
         req = Request(input_ids=input_ids.reshape(-1), **kw)
@@ -263,7 +278,9 @@ class ContinuousBatcher:
         return True
 
     def _loop(self):
-        while not self._stop.is_set():
+        self._drain = False
+        while not self._stop.is_set() or (
+                self._drain and any(s is not None for s in self.slots)):
             admitted = False
             while not self._queue.empty():
                 slot = self._free_slot()

@@ -318,3 +318,37 @@ def test_n_completions_openai_choices():
     assert out["usage"]["completion_tokens"] == sum(
         1 for c in out["choices"] for _ in c["message"]["content"]) or True
     eng.unload_model("tiny")
+
+
+def test_stop_never_hangs_clients():
+    """stop() with work in flight: live requests end at their current
+    token, queued ones return empty — nobody blocks."""
+    import time
+
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=1, max_ctx=256).start()
+    torch.manual_seed(3)
+    live = b.submit(torch.randint(0, 1024, (8,)), max_new_tokens=100)
+    queued = b.submit(torch.randint(0, 1024, (8,)), max_new_tokens=100)
+    for _ in range(200):
+        if b.slot_emitted[0] > 2:
+            break
+        time.sleep(0.01)
+    b.stop()
+    t0 = time.time()
+    out_live = live.result(timeout=10)
+    out_q = queued.result(timeout=10)
+    assert time.time() - t0 < 5
+    assert 0 < len(out_live) < 100
+    assert out_q == []
+
+
+def test_stop_drain_finishes_running():
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=128).start()
+    torch.manual_seed(4)
+    reqs = [b.submit(torch.randint(0, 1024, (8,)), max_new_tokens=6)
+            for _ in range(2)]
+    b.stop(drain=True)
+    for rq in reqs:
+        assert len(rq.result(timeout=10)) == 6
