@@ -33,7 +33,12 @@ def smoke(name, quantize=None, batch=2, prompt=64, new=16, vocab=32000):
 
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "both"
-    if which in ("both", "mixtral"):
+    if which in ("both", "all", "mixtral"):
         smoke("mistralai/Mixtral-8x7B-v0.1", quantize="fp8", vocab=32000)
-    if which in ("both", "llama70b"):
+    if which in ("both", "all", "llama70b"):
         smoke("meta-llama/Llama-3-70B", batch=8, new=32, vocab=128256)
+    if which in ("all", "qwen3moe"):
+        smoke("Qwen/Qwen3-30B-A3B", vocab=151936)
+    if which in ("all", "qwen7b-fp8"):
+        smoke("Qwen/Qwen2.5-7B-Instruct", quantize="fp8-dense",
+              batch=8, new=32, vocab=152064)
