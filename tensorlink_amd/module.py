@@ -95,6 +95,18 @@ class _Router(torch.autograd.Function):
 
 
 class DistributedModel(nn.Module):
+    def __new__(cls, model=None, *args, **kwargs):
+        # reference trusted mode: DistributedModel(model=CustomModel(),
+        # trusted=True) distributes an arbitrary nn.Module
+        # (ml/module.py:259, docs/examples/EXAMPLES.md)
+        if isinstance(model, nn.Module):
+            from tensorlink_amd.custom import CustomDistributedModel
+            return CustomDistributedModel(
+                model, world_size=kwargs.get("world_size", 2),
+                trusted=kwargs.get("trusted", False),
+                lr=kwargs.get("lr", 1e-3))
+        return super().__new__(cls)
+
     def __init__(self, model: str, training: bool = False,
                  world_size: int = 1, mode: str = "auto",
                  device=None, init: str = "random",

@@ -85,3 +85,51 @@ def test_gpt2_two_local_cpu_workers():
         assert out.shape == (2, 12)
     finally:
         m.shutdown()
+
+
+def test_custom_module_trusted_mode():
+    """Reference trusted mode: an arbitrary nn.Sequential distributes
+    across 2 worker processes; forward/backward/optimizer reproduce the
+    single-process module exactly."""
+    import torch
+    import torch.nn as nn
+
+    from tensorlink_amd.module import DistributedModel
+
+    def build():
+        torch.manual_seed(5)
+        return nn.Sequential(nn.Linear(16, 32), nn.Tanh(),
+                             nn.Linear(32, 32), nn.ReLU(),
+                             nn.Linear(32, 8))
+
+    ref = build()
+    opt_ref = torch.optim.AdamW(ref.parameters(), lr=1e-2)
+    with __import__("pytest").raises(ValueError):
+        DistributedModel(model=build(), world_size=2)   # trusted missing
+
+    m = DistributedModel(model=build(), world_size=2, trusted=True)
+    try:
+        m.train()
+        torch.manual_seed(9)
+        for _ in range(2):
+            x = torch.randn(4, 16)
+            y = m(x)
+            y_ref = ref(x)
+            torch.testing.assert_close(y, y_ref, atol=1e-6, rtol=1e-5)
+            opt = getattr(m, "_test_opt", None)
+            if opt is None:
+                opt = m.create_optimizer(lr=1e-2)
+                m._test_opt = opt
+            loss = y.pow(2).mean()
+            loss.backward()
+            ref(x).pow(2).mean().backward() if False else None
+            y_ref.pow(2).mean().backward()
+            opt.step()
+            opt_ref.step()
+            opt.zero_grad()
+            opt_ref.zero_grad()
+        # after 2 steps the rank-0 slice matches the reference slice
+        torch.testing.assert_close(m.stage[0].weight, ref[0].weight,
+                                   atol=1e-6, rtol=1e-5)
+    finally:
+        m.shutdown()
