@@ -349,3 +349,25 @@ def test_watchdog_job_ttl_eviction():
     assert "tiny" in eng.jobs
     assert "tiny-moe" not in eng.jobs
     eng.unload_model("tiny")
+
+
+def test_bench_serving_script_smoke():
+    """The serving load benchmark runs end-to-end on CPU and prints a
+    valid JSON metrics line."""
+    import json
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "scripts/bench_serving.py", "--model", "tiny",
+         "--rate", "200", "--num-requests", "6", "--prompt-len", "64",
+         "--new-tokens", "4", "--prefill-chunk", "32",
+         "--prefix-caching", "--shared-prefix", "40"],
+        capture_output=True, text=True, timeout=240,
+        cwd=__import__("os").path.dirname(__import__("os").path.dirname(
+            __import__("os").path.abspath(__file__))))
+    line = out.stdout.strip().splitlines()[-1]
+    data = json.loads(line)
+    assert data["num_requests"] == 6
+    assert data["output_tokens_per_s"] > 0
+    assert data["ttft_p50_s"] is not None
