@@ -344,11 +344,19 @@ def test_stop_never_hangs_clients():
 
 
 def test_stop_drain_finishes_running():
+    import time
+
     r = _runner()
     b = ContinuousBatcher(r, max_slots=2, max_ctx=128).start()
     torch.manual_seed(4)
     reqs = [b.submit(torch.randint(0, 1024, (8,)), max_new_tokens=6)
             for _ in range(2)]
+    # drain finishes ADMITTED work; wait for admission before stopping
+    for _ in range(300):
+        if all(s is not None for s in b.slots[:2]) or \
+                all(rq.done.is_set() for rq in reqs):
+            break
+        time.sleep(0.01)
     b.stop(drain=True)
     for rq in reqs:
         assert len(rq.result(timeout=10)) == 6
