@@ -27,6 +27,7 @@ class GenerationRequest(BaseModel):
     num_beams: int = 1
     n: int = 1                        # OpenAI multiple completions
     speculative: bool = False         # prompt-lookup speculative decode
+    seed: Optional[int] = None        # deterministic sampling seed
     presence_penalty: float = 0.0     # OpenAI range [-2, 2]
     frequency_penalty: float = 0.0
     stream: bool = False

@@ -45,6 +45,7 @@ class Request:
     priority: int = 0                    # higher admits first
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    seed: Optional[int] = None           # per-request sampling seed
     tokens: "queue.Queue" = field(default_factory=queue.Queue)
     done: threading.Event = field(default_factory=threading.Event)
     error: Optional[str] = None
@@ -225,9 +226,15 @@ class ContinuousBatcher:
         for i, req in enumerate(reqs):
             counts = (self.slot_counts[slots[i]]
                       if slots is not None else None)
+            gen = None
+            if req.seed is not None:
+                if not hasattr(req, "_gen"):
+                    req._gen = torch.Generator(device=logits.device)
+                    req._gen.manual_seed(req.seed)
+                gen = req._gen
             toks.append(int(ops.sample_token(
                 logits[i:i + 1], temperature=req.temperature,
-                top_p=req.top_p, top_k=req.top_k,
+                top_p=req.top_p, top_k=req.top_k, generator=gen,
                 token_counts=counts,
                 presence_penalty=req.presence_penalty,
                 frequency_penalty=req.frequency_penalty)[0]))

@@ -256,7 +256,8 @@ class InferenceEngine:
             max_new_tokens=args["max_new_tokens"],
             eos_token_id=args.get("eos_token_id"),
             presence_penalty=args.get("presence_penalty", 0.0),
-            frequency_penalty=args.get("frequency_penalty", 0.0))
+            frequency_penalty=args.get("frequency_penalty", 0.0),
+            seed=request.get("seed"))
         stop = request.get("stop") or []
         if isinstance(stop, str):
             stop = [stop]
@@ -286,7 +287,8 @@ class InferenceEngine:
                           temperature=sp.temperature, top_p=sp.top_p,
                           top_k=sp.top_k, eos_token_id=sp.eos_token_id,
                           presence_penalty=sp.presence_penalty,
-                          frequency_penalty=sp.frequency_penalty)
+                          frequency_penalty=sp.frequency_penalty,
+                          seed=sp.seed)
                 reqs = [job.batcher.submit(ids[0], **kw)
                         for _ in range(n)]
                 results = [rq.result() for rq in reqs]
@@ -366,7 +368,8 @@ class InferenceEngine:
                         temperature=sp.temperature, top_p=sp.top_p,
                         top_k=sp.top_k, eos_token_id=sp.eos_token_id,
                         presence_penalty=sp.presence_penalty,
-                        frequency_penalty=sp.frequency_penalty)
+                        frequency_penalty=sp.frequency_penalty,
+                        seed=sp.seed)
                     for tok in req.stream():
                         q.put(tok)
                 else:

@@ -360,3 +360,22 @@ def test_stop_drain_finishes_running():
     b.stop(drain=True)
     for rq in reqs:
         assert len(rq.result(timeout=10)) == 6
+
+
+def test_seeded_sampling_deterministic():
+    """OpenAI `seed`: two sampled requests with the same seed produce
+    identical tokens; different seeds (almost surely) differ."""
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=4, max_ctx=128).start()
+    try:
+        torch.manual_seed(5)
+        p = torch.randint(0, 1024, (8,))
+        kw = dict(max_new_tokens=8, temperature=1.0, seed=123)
+        a = b.submit(p.clone(), **kw).result(timeout=60)
+        c = b.submit(p.clone(), **kw).result(timeout=60)
+        d = b.submit(p.clone(), max_new_tokens=8, temperature=1.0,
+                     seed=77).result(timeout=60)
+        assert a == c
+        assert a != d
+    finally:
+        b.stop()
