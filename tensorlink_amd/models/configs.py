@@ -235,6 +235,16 @@ def _qwen3_30b_a3b() -> ModelConfig:
         expert_intermediate_size=768, architecture="qwen3")
 
 
+def _tiny_bigvocab() -> ModelConfig:
+    # tiny body with a >=32k vocabulary: exercises the chunked-CE loss
+    # path the real models take (vocab gate in PipelineTrainer._loss)
+    return ModelConfig(
+        name="tiny-bigvocab", vocab_size=38400, hidden_size=256,
+        intermediate_size=512, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, head_dim=64,
+        max_position_embeddings=512)
+
+
 def _llama31_8b() -> ModelConfig:
     return ModelConfig(
         name="meta-llama/Llama-3.1-8B", vocab_size=128256,
@@ -263,6 +273,7 @@ PRESETS = {
     "tiny-qwen3": _tiny_qwen3,
     "tiny-moe": _tiny_moe,
     "tiny-qwen3-moe": _tiny_qwen3_moe,
+    "tiny-bigvocab": _tiny_bigvocab,
     "meta-llama/Llama-3.1-8B": _llama31_8b,
     "Qwen/Qwen3-30B-A3B": _qwen3_30b_a3b,
 }

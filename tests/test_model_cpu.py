@@ -508,3 +508,18 @@ def test_export_then_serve_from_directory(tmp_path):
         logits = src(cur, pos)
         cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(cur[:, 9:], out)
+
+
+def test_chunked_ce_gate_on_big_vocab():
+    """The vocab>=32k gate selects chunked CE and the step runs (CPU)."""
+    import torch
+
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    from tensorlink_amd.parallel.planner import plan_for_world
+    t = PipelineTrainer(plan_for_world("tiny-bigvocab", 1), 0, 1,
+                        device=torch.device("cpu"), seed=1, lr=1e-3)
+    assert t._chunked_ce
+    ids = torch.randint(0, 38400, (1, 24))
+    loss = t.train_step(ids, labels=ids)
+    import math
+    assert math.isfinite(loss) and loss > 0
