@@ -130,6 +130,9 @@ class ContinuousBatcher:
         self.slot_prompt: List[Optional[torch.Tensor]] = [None] * max_slots
         self.slot_filled = [0] * max_slots    # prompt tokens prefilled
         self.slot_counts: List[dict] = [{} for _ in range(max_slots)]
+        # preempted-and-swapped requests waiting to resume:
+        # (req, blob, slot_len, slot_last, slot_emitted, counts, prompt)
+        self._swapped: List[tuple] = []
         # admission queue ordered by (priority desc, arrival): a
         # high-priority request jumps the queue but never preempts a
         # running slot
@@ -179,6 +182,65 @@ class ContinuousBatcher:
         return req
 
     # ------------------------------------------------------------------
+    def _committed_pages(self) -> int:
+        """Pages that admitted-but-still-prefilling slots will lease as
+        their chunks land (chunked prefill leases lazily, so admission
+        must count these as spoken for)."""
+        tot = 0
+        for i, r in enumerate(self.slots):
+            if r is not None and self._prefilling(i):
+                p = self.slot_prompt[i].numel()
+                tot += max(0, (p + PAGE) // PAGE
+                           - len(self.cache._slot_pages[i]))
+        return tot
+
+    # ---------------- preemption (swap to host) ----------------------
+    def _preempt_one(self, exclude: Optional[int] = None) -> bool:
+        """Swap out the youngest running slot (fewest emitted tokens —
+        least work lost per page freed) to host memory; it re-enters via
+        the resume path when pages free up. vLLM-style preemption keeps
+        admission live when the page pool starves."""
+        from tensorlink_amd.models.paged import swap_out
+        cands = [i for i, r in enumerate(self.slots)
+                 if r is not None and not self._prefilling(i)
+                 and self.slot_filled[i] > 0 and i != exclude]
+        if not cands:
+            return False
+        victim = min(cands, key=lambda i: self.slot_emitted[i])
+        req = self.slots[victim]
+        blob = swap_out(self.cache, victim)
+        self._swapped.append((req, blob, self.slot_len[victim],
+                              self.slot_last[victim],
+                              self.slot_emitted[victim],
+                              self.slot_counts[victim],
+                              self.slot_prompt[victim]))
+        self.slots[victim] = None
+        self.slot_prompt[victim] = None
+        self.preemptions = getattr(self, "preemptions", 0) + 1
+        return True
+
+    def _resume_swapped(self) -> bool:
+        from tensorlink_amd.models.paged import swap_in
+        resumed = False
+        while self._swapped:
+            slot = self._free_slot()
+            blob = self._swapped[0][1]
+            if slot is None or \
+                    self.cache.available_pages() < blob.n_pages + 1:
+                break
+            req, blob, s_len, s_last, s_emit, counts, prompt = \
+                self._swapped.pop(0)
+            swap_in(self.cache, slot, blob, s_len)
+            self.slots[slot] = req
+            self.slot_prompt[slot] = prompt
+            self.slot_filled[slot] = prompt.numel()
+            self.slot_len[slot] = s_len
+            self.slot_last[slot] = s_last
+            self.slot_emitted[slot] = s_emit
+            self.slot_counts[slot] = counts
+            resumed = True
+        return resumed
+
     def _free_slot(self) -> Optional[int]:
         for i, s in enumerate(self.slots):
             if s is None:
@@ -200,7 +262,13 @@ class ContinuousBatcher:
         end = (S_total if self.prefill_chunk is None
                else min(start + self.prefill_chunk, S_total))
         ids = prompt[start:end].to(self.device).unsqueeze(0)
-        self.cache.ensure(slot, end + 1)
+        try:
+            self.cache.ensure(slot, end + 1)
+        except RuntimeError:
+            # pool exhausted mid-prefill: park a decoding slot and retry
+            if not self._preempt_one(exclude=slot):
+                raise
+            self.cache.ensure(slot, end + 1)
         pos = torch.arange(start, end, device=self.device,
                            dtype=torch.int32).unsqueeze(0)
         lens = torch.tensor([start], device=self.device, dtype=torch.int32)
@@ -264,7 +332,16 @@ class ContinuousBatcher:
             return False
         reqs = [self.slots[i] for i in active]
         for i in active:
-            self.cache.ensure(i, self.slot_len[i] + 2)
+            try:
+                self.cache.ensure(i, self.slot_len[i] + 2)
+            except RuntimeError:
+                # pool exhausted mid-decode: park another slot and retry
+                if not self._preempt_one(exclude=i):
+                    raise
+                self.cache.ensure(i, self.slot_len[i] + 2)
+                active = [j for j in active
+                          if self.slots[j] is not None]
+                reqs = [self.slots[j] for j in active]
         toks = torch.tensor([self.slot_last[i] for i in active],
                             device=self.device, dtype=torch.int64)
         lens = torch.tensor([self.slot_len[i] for i in active],
@@ -288,7 +365,8 @@ class ContinuousBatcher:
         self._drain = False
         while not self._stop.is_set() or (
                 self._drain and any(s is not None for s in self.slots)):
-            admitted = False
+            admitted = self._resume_swapped()
+            starved = 0
             while not self._queue.empty():
                 slot = self._free_slot()
                 if slot is None:
@@ -301,7 +379,18 @@ class ContinuousBatcher:
                 if self.prefix_caching:
                     need -= self.cache.match_prefix(
                         head.input_ids.tolist()) // PAGE
-                if self.cache.available_pages() < need:
+                if self.cache.available_pages() \
+                        < need + self._committed_pages():
+                    # page starvation: swap out a young running slot so
+                    # the queue keeps moving (at most one per iteration,
+                    # never below two live slots, and not while others
+                    # already wait to resume)
+                    n_live = sum(r is not None for r in self.slots)
+                    if (starved == 0 and n_live >= 2
+                            and not self._swapped
+                            and self._preempt_one()):
+                        starved = 1
+                        continue
                     break
                 req = self._queue.get_nowait()[2]
                 if req.cancelled:

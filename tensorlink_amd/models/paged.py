@@ -145,6 +145,39 @@ def kv_slice_paged(cache: PagedKVCache, s: int, e: int) -> PagedKVCache:
     return view
 
 
+class _SwapBlob:
+    """Host-side copy of one slot's KV pages (preemption)."""
+
+    def __init__(self, k, v, n_pages):
+        self.k, self.v, self.n_pages = k, v, n_pages
+
+
+def swap_out(cache: DynamicPagedKVCache, slot: int) -> _SwapBlob:
+    """Copy `slot`'s leased pages to host memory and release them back
+    to the pool (vLLM-style preemption: the scheduler parks a running
+    sequence when the pool starves and resumes it later). Returns the
+    blob `swap_in` restores from."""
+    pages = list(cache._slot_pages[slot])
+    idx = torch.tensor(pages, dtype=torch.long, device=cache.k[0].device)
+    k = [layer[idx].to("cpu", copy=True) for layer in cache.k]
+    v = [layer[idx].to("cpu", copy=True) for layer in cache.v]
+    blob = _SwapBlob(k, v, len(pages))
+    cache.release_slot(slot)
+    return blob
+
+
+def swap_in(cache: DynamicPagedKVCache, slot: int, blob: _SwapBlob,
+            length: int) -> None:
+    """Lease fresh pages for `slot` and restore the swapped KV."""
+    assert not cache._slot_pages[slot], "slot must be empty to swap in"
+    cache.ensure(slot, blob.n_pages * PAGE)
+    pages = cache._slot_pages[slot][:blob.n_pages]
+    idx = torch.tensor(pages, dtype=torch.long, device=cache.k[0].device)
+    for li in range(len(cache.k)):
+        cache.k[li][idx] = blob.k[li].to(cache.k[li].device)
+        cache.v[li][idx] = blob.v[li].to(cache.v[li].device)
+
+
 class PrefixCachingKVCache(DynamicPagedKVCache):
     """Dynamic paged cache with automatic prefix reuse (vLLM-style).
 

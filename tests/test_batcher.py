@@ -379,3 +379,31 @@ def test_seeded_sampling_deterministic():
         assert a != d
     finally:
         b.stop()
+
+
+def test_preemption_swap_and_resume_matches_serial():
+    """Pool starvation preempts a young slot to host memory and resumes
+    it when pages free — every request still reproduces serial greedy
+    output exactly and the pool drains."""
+    import time
+
+    r = _runner()
+    # 2 slots; both prompts fit at admission (2x3 pages of 7) but decode
+    # growth needs 2x4 > 7 -> mid-decode exhaustion -> swap
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=512, pool_pages=7,
+                          prefill_chunk=64).start()
+    try:
+        torch.manual_seed(41)
+        prompts = [torch.randint(0, 1024, (300,)) for _ in range(2)]
+        # long decodes force page growth past the pool
+        reqs = [b.submit(p.clone(), max_new_tokens=90) for p in prompts]
+        outs = [rq.result(timeout=240) for rq in reqs]
+        assert all(len(o) == 90 for o in outs)
+        assert getattr(b, "preemptions", 0) >= 1, "never preempted"
+        for p, o in zip(prompts, outs):
+            ref = r.generate(p.unsqueeze(0),
+                             SamplingParams(max_new_tokens=90))
+            assert o == ref[0].tolist()
+    finally:
+        b.stop()
+    assert b.cache.allocator.n_free == b.cache.allocator.n_pages
