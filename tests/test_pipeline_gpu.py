@@ -387,3 +387,32 @@ def test_chunked_ce_training_gpu():
     losses = [t.train_step(ids, labels=ids) for _ in range(3)]
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+def test_preemption_swap_gpu():
+    """Preemption on the HIP paged path: device->host swap, resume,
+    serial-equal outputs."""
+    from tensorlink_amd.engine.batcher import ContinuousBatcher
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    plan = plan_for_world("tiny", 1)
+    r = PipelineRunner(plan, 0, 1, device=DEV, dtype=torch.bfloat16)
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=512, pool_pages=7,
+                          prefill_chunk=64).start()
+    try:
+        torch.manual_seed(41)
+        prompts = [torch.randint(0, 1024, (300,)) for _ in range(2)]
+        reqs = [b.submit(p.clone(), max_new_tokens=90) for p in prompts]
+        outs = [rq.result(timeout=240) for rq in reqs]
+        assert all(len(o) == 90 for o in outs)
+        import os
+        os.environ["TL_NO_GRAPH"] = "1"
+        try:
+            for p, o in zip(prompts, outs):
+                ref = r.generate(p.unsqueeze(0),
+                                 SamplingParams(max_new_tokens=90))
+                assert o == ref[0].cpu().tolist()
+        finally:
+            del os.environ["TL_NO_GRAPH"]
+    finally:
+        b.stop()
