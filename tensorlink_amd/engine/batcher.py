@@ -159,8 +159,10 @@ class ContinuousBatcher:
         self._wake.set()
         if self._thread:
             self._thread.join(120 if drain else 30)
-        # cut loose anything still attached
+        # cut loose anything still attached (incl. swapped-out)
         leftovers = [r for r in self.slots if r is not None]
+        leftovers += [t[0] for t in self._swapped]
+        self._swapped = []
         while not self._queue.empty():
             try:
                 leftovers.append(self._queue.get_nowait()[2])
@@ -331,7 +333,9 @@ class ContinuousBatcher:
         if not active:
             return False
         reqs = [self.slots[i] for i in active]
-        for i in active:
+        for i in list(active):
+            if self.slots[i] is None:      # preempted below this loop
+                continue
             try:
                 self.cache.ensure(i, self.slot_len[i] + 2)
             except RuntimeError:
@@ -339,9 +343,10 @@ class ContinuousBatcher:
                 if not self._preempt_one(exclude=i):
                     raise
                 self.cache.ensure(i, self.slot_len[i] + 2)
-                active = [j for j in active
-                          if self.slots[j] is not None]
-                reqs = [self.slots[j] for j in active]
+        active = [j for j in active if self.slots[j] is not None]
+        if not active:
+            return False
+        reqs = [self.slots[j] for j in active]
         toks = torch.tensor([self.slot_last[i] for i in active],
                             device=self.device, dtype=torch.int64)
         lens = torch.tensor([self.slot_len[i] for i in active],
