@@ -614,7 +614,8 @@ class PPContinuousBatcher(ContinuousBatcher):
                 need = (head.input_ids.numel() + PAGE) // PAGE
                 matched = (self.cache.match_prefix(ids_list)
                            if self.prefix_caching else 0)
-                if self.cache.available_pages() < need - matched // PAGE:
+                if self.cache.available_pages() < (need - matched // PAGE
+                                                   + self._committed_pages()):
                     break
                 req = self._queue.get_nowait()[2]
                 if req.cancelled:
