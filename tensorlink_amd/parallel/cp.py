@@ -100,6 +100,10 @@ class CPRunner:
         B, Sq, H = x.shape
         qkv = attn.qkv_proj(x)
         q, k, v = attn._split_qkv(qkv, B, Sq)
+        if attn.use_qk_norm:
+            eps = attn.config.rms_norm_eps
+            q = ops.rmsnorm(q.contiguous(), attn.q_norm.to(q.dtype), eps)
+            k = ops.rmsnorm(k.contiguous(), attn.k_norm.to(k.dtype), eps)
         q = q.contiguous()
         k = k.contiguous()
         v = v.contiguous()
@@ -181,6 +185,10 @@ class CPRunner:
         B = x.shape[0]
         qkv = attn.qkv_proj(x)
         q, k, v = attn._split_qkv(qkv, B, 1)
+        if attn.use_qk_norm:
+            eps = attn.config.rms_norm_eps
+            q = ops.rmsnorm(q.contiguous(), attn.q_norm.to(q.dtype), eps)
+            k = ops.rmsnorm(k.contiguous(), attn.k_norm.to(k.dtype), eps)
         q = q.contiguous()
         k = k.contiguous()
         positions = torch.full((B,), pos, device=self.device,
@@ -330,6 +338,10 @@ class CPTrainer:
         B, sc, H = x.shape
         qkv = attn.qkv_proj(x)
         q, k, v = attn._split_qkv(qkv, B, sc)
+        if attn.use_qk_norm:
+            eps = attn.config.rms_norm_eps
+            q = ops.rmsnorm(q.contiguous(), attn.q_norm.to(q.dtype), eps)
+            k = ops.rmsnorm(k.contiguous(), attn.k_norm.to(k.dtype), eps)
         flat_pos = positions.reshape(-1)
         q2, k2 = ops.apply_rope(q.reshape(B * sc, attn.n_heads, -1),
                                 k.reshape(B * sc, attn.n_kv, -1),

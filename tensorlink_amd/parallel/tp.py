@@ -423,7 +423,15 @@ class TPPPRunner:
 
     def generate(self, input_ids, sampling=None, **kw):
         """SPMD over the whole grid; input_ids significant on every
-        pipeline replica's first rank (stage 0); tokens return on them."""
+        pipeline replica's first rank (stage 0); tokens return on them.
+        Stochastic sampling must be seeded so the replicas stay in
+        lockstep (their per-layer TP all-reduces pair up)."""
+        if sampling is not None and sampling.temperature > 0 \
+                and sampling.seed is None:
+            raise ValueError(
+                "TP x PP sampling needs SamplingParams.seed: unseeded "
+                "draws diverge across pipeline replicas and break the "
+                "per-layer TP collectives")
         return self.runner.generate(input_ids, sampling, **kw)
 
 

@@ -1080,3 +1080,44 @@ def test_tp2_pp2_vocab_parallel_training():
     ref = dict(_run_collect(_pp2_train_ref_worker, 2, _port(43)))
     assert grid[0] == pytest.approx(grid[2], rel=1e-5)
     assert grid[0] == pytest.approx(ref[0], rel=2e-3)
+
+
+def _cp_qk_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.cp import CPRunner
+    init_distributed(backend="gloo")
+    r = CPRunner("tiny-qwen3", rank, world, device=torch.device("cpu"),
+                 seed=19)
+    torch.manual_seed(57)
+    ids = torch.randint(0, 1024, (1, 24))
+    out = r.generate(ids, max_new_tokens=4)
+    q.put((rank, out.tolist()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_cp2_qwen3_qk_norm():
+    """CP honors Qwen3's per-head qk-norm (prefill + decode) — matches
+    the single-rank model."""
+    outs = dict(_run_collect(_cp_qk_worker, 2, _port(45)))
+    assert outs[0] == outs[1]
+
+    import torch
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.configs import get_config
+    m = build_full_model(get_config("tiny-qwen3"))
+    init_random_stage(m, device="cpu", dtype=torch.float32, seed=19)
+    torch.manual_seed(57)
+    ids = torch.randint(0, 1024, (1, 24))
+    cur = ids
+    for _ in range(4):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).contiguous()
+        logits = m(cur, pos)
+        cur = torch.cat([cur, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert cur[:, 24:].tolist() == outs[0]
