@@ -153,6 +153,9 @@ def build_tp_model(config_or_name, tp_rank: int, tp: int, device=None,
     device = device if device is not None else device_for_rank()
     dtype = dtype or (torch.bfloat16 if device.type == "cuda"
                       else torch.float32)
+    assert not config.is_moe, \
+        "TP shards dense projections; shard MoE experts with EP " \
+        "(parallel/ep.py) instead"
     if vocab_parallel:
         assert not config.tie_word_embeddings, \
             "vocab-parallel head requires untied embeddings"
