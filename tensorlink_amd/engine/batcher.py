@@ -418,6 +418,13 @@ class ContinuousBatcher:
                 if self.slots[slot] is not None and (
                         self._prefilling(slot) or self.slot_filled[slot] == 0):
                     req = self.slots[slot]
+                    if req.cancelled:          # cancelled mid-prefill
+                        req.tokens.put(None)
+                        req.done.set()
+                        self.slots[slot] = None
+                        self.slot_prompt[slot] = None
+                        self.cache.release_slot(slot)
+                        continue
                     try:
                         self._prefill_chunk(slot)
                         prefilled = True
@@ -600,8 +607,10 @@ class PPContinuousBatcher(ContinuousBatcher):
 
     def _loop(self):
         assert self.runner.rank == 0
+        self._drain = False
         r = self.runner
-        while not self._stop.is_set():
+        while not self._stop.is_set() or (
+                self._drain and any(s is not None for s in self.slots)):
             admissions = []
             while not self._queue.empty():
                 slot = self._free_slot()
@@ -636,6 +645,13 @@ class PPContinuousBatcher(ContinuousBatcher):
                 if self.slots[slot] is not None and (
                         self._prefilling(slot)
                         or self.slot_filled[slot] == 0):
+                    if self.slots[slot].cancelled:
+                        self.slots[slot].tokens.put(None)
+                        self.slots[slot].done.set()
+                        self.slots[slot] = None
+                        self.slot_prompt[slot] = None
+                        self._pending_releases.append(slot)
+                        continue
                     prompt = self.slot_prompt[slot]
                     s_total = prompt.numel()
                     start = self.slot_filled[slot]

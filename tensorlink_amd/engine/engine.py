@@ -218,7 +218,17 @@ class InferenceEngine:
         job = self.jobs.get(name)
         if job is None:
             return {"model": name, "status": "not_loaded"}
-        return {"model": name, "status": job.state,
+        b = job.batcher
+        extra = {}
+        if b is not None:
+            extra = {"batcher": {
+                "steps": b.steps,
+                "preemptions": getattr(b, "preemptions", 0),
+                "swapped_waiting": len(getattr(b, "_swapped", [])),
+                "prefix_cache_hit_tokens": getattr(b.cache, "hits", 0),
+                "pages_free": b.cache.available_pages(),
+                "pages_total": b.cache.allocator.n_pages}}
+        return {"model": name, "status": job.state, **extra,
                 "requests_served": job.requests_served,
                 "tokens_generated": job.tokens_generated}
 
