@@ -107,3 +107,89 @@ def quantize_dense_fp8(stage: nn.Module) -> int:
                 setattr(mod, name, Fp8Linear.from_linear(lin))
                 n += 1
     return n
+
+
+# ---------------------------------------------------------------------------
+# MXFP4 (OCP microscaling fp4): the gfx950-native 4-bit format
+# ---------------------------------------------------------------------------
+_E2M1_VALUES = torch.tensor([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0])
+
+
+def quantize_mxfp4(w: torch.Tensor, group: int = 32):
+    """Quantize [N, K] weights to MXFP4: groups of 32 along K share one
+    power-of-two e8m0 scale; elements are e2m1 (±{0,.5,1,1.5,2,3,4,6}),
+    packed two per byte. This is the exact operand format of gfx950's
+    scaled-MFMA instructions (CDNA4 guide §MX), so round-2's 4-bit GEMM
+    kernel consumes these tensors directly; until then Fp4Linear
+    dequantizes on the fly. Returns (packed [N, K//2] uint8,
+    exponents [N, K/group] int8)."""
+    N, K = w.shape
+    assert K % group == 0
+    wf = w.float().reshape(N, K // group, group)
+    amax = wf.abs().amax(-1).clamp(min=1e-12)
+    # shared exponent: amax maps near the top code (6 = 1.5 * 2^2)
+    e = torch.floor(torch.log2(amax)) - 2
+    scale = torch.pow(2.0, e)
+    x = (wf / scale.unsqueeze(-1)).clamp(-6, 6)
+    vals = _E2M1_VALUES.to(w.device)
+    idx = (x.abs().unsqueeze(-1) - vals).abs().argmin(-1).to(torch.uint8)
+    code = idx | ((x < 0).to(torch.uint8) << 3)          # sign bit 3
+    code = code.reshape(N, K)
+    packed = (code[:, 0::2] | (code[:, 1::2] << 4)).contiguous()
+    return packed, e.to(torch.int8)
+
+
+def dequantize_mxfp4(packed: torch.Tensor, exponents: torch.Tensor,
+                     group: int = 32, dtype=torch.float32):
+    N = packed.shape[0]
+    K = packed.shape[1] * 2
+    code = torch.empty(N, K, dtype=torch.uint8, device=packed.device)
+    code[:, 0::2] = packed & 0xF
+    code[:, 1::2] = packed >> 4
+    vals = _E2M1_VALUES.to(packed.device)
+    mag = vals[(code & 0x7).long()]
+    sign = torch.where((code & 0x8) != 0, -1.0, 1.0)
+    scale = torch.pow(2.0, exponents.float()).repeat_interleave(group, -1)
+    return (mag * sign * scale).to(dtype)
+
+
+class Fp4Linear(nn.Module):
+    """Weight-only MXFP4 linear: 4x smaller weights than bf16. Forward
+    dequantizes per call (correct everywhere; the scaled-MFMA GEMM that
+    consumes the packed form directly is round-2 kernel work —
+    ROADMAP.md §6)."""
+
+    def __init__(self, packed, exponents, bias, in_features, out_features):
+        super().__init__()
+        self.register_buffer("packed", packed)
+        self.register_buffer("exponents", exponents)
+        self.bias = bias
+        self.in_features = in_features
+        self.out_features = out_features
+
+    @classmethod
+    def from_linear(cls, lin: nn.Linear) -> "Fp4Linear":
+        packed, e = quantize_mxfp4(lin.weight.detach())
+        return cls(packed, e, lin.bias, lin.in_features, lin.out_features)
+
+    def forward(self, x):
+        w = dequantize_mxfp4(self.packed, self.exponents, dtype=torch.float32)
+        y = torch.nn.functional.linear(x.float(), w,
+                                       self.bias.float()
+                                       if self.bias is not None else None)
+        return y.to(x.dtype)
+
+
+def quantize_dense_fp4(stage: nn.Module) -> int:
+    """Convert every projection Linear to weight-only MXFP4 in place
+    (4-bit analog of quantize_dense_fp8)."""
+    n = 0
+    targets = ("qkv_proj", "o_proj", "gate_up_proj", "down_proj")
+    for mod in stage.modules():
+        for name in targets:
+            lin = getattr(mod, name, None)
+            if isinstance(lin, nn.Linear) and not isinstance(
+                    lin, (Fp8Linear,)):
+                setattr(mod, name, Fp4Linear.from_linear(lin))
+                n += 1
+    return n

@@ -95,6 +95,10 @@ class PipelineRunner:
             from tensorlink_amd.models.quant import quantize_dense_fp8
             quantize_dense_fp8(self.stage)
             self._no_graph = True
+        elif quantize == "fp4-dense":
+            from tensorlink_amd.models.quant import quantize_dense_fp4
+            quantize_dense_fp4(self.stage)
+            self._no_graph = True
         self.stage.eval()
         self.p2p = P2P(rank, world, group, rank_base) if world > 1 else None
         self.is_first = rank == 0

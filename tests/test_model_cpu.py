@@ -523,3 +523,41 @@ def test_chunked_ce_gate_on_big_vocab():
     loss = t.train_step(ids, labels=ids)
     import math
     assert math.isfinite(loss) and loss > 0
+
+
+def test_mxfp4_quantization():
+    """MXFP4 pack/unpack round trip: exact e2m1 grid values survive,
+    random weights stay within the 4-bit group-relative error bound,
+    and the quantized model still decodes."""
+    import torch
+
+    from tensorlink_amd.models.quant import (dequantize_mxfp4,
+                                             quantize_mxfp4)
+
+    # exact grid: values representable in e2m1 * 2^e round-trip exactly
+    g = torch.tensor([[0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0] * 4,
+                      [-6.0, -4.0, -3.0, -2.0, -1.5, -1.0, -0.5, 0.0] * 4])
+    p, e = quantize_mxfp4(g)
+    torch.testing.assert_close(dequantize_mxfp4(p, e), g)
+
+    torch.manual_seed(8)
+    w = torch.randn(64, 256)
+    p, e = quantize_mxfp4(w)
+    wq = dequantize_mxfp4(p, e)
+    assert p.numel() * 1 + e.numel() == w.numel() // 2 + w.numel() // 32
+    # e2m1 relative step is <= 25% of the group max
+    grp = w.reshape(64, -1, 32)
+    gq = wq.reshape(64, -1, 32)
+    err = (grp - gq).abs().amax(-1)
+    bound = grp.abs().amax(-1) * 0.26
+    assert bool((err <= bound + 1e-6).all())
+
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    r_q = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                         device=torch.device("cpu"), seed=2,
+                         quantize="fp4-dense")
+    torch.manual_seed(33)
+    out = r_q.generate(torch.randint(0, 1024, (1, 12)),
+                       SamplingParams(max_new_tokens=4))
+    assert out.shape == (1, 4)
