@@ -333,8 +333,17 @@ class StageModel(nn.Module):
             hidden = self.embed(hidden_or_ids)
         else:
             hidden = hidden_or_ids
+        ckpt = (training and getattr(self, "grad_checkpointing", False)
+                and torch.is_grad_enabled())
         for i, layer in enumerate(self.layers):
-            hidden = layer(hidden, positions, kv_cache, i, training)
+            if ckpt:
+                # recompute the layer in backward instead of stashing
+                # activations (no dropout anywhere -> exact recompute)
+                hidden = torch.utils.checkpoint.checkpoint(
+                    layer, hidden, positions, kv_cache, i, training,
+                    use_reentrant=False)
+            else:
+                hidden = layer(hidden, positions, kv_cache, i, training)
         if kv_cache is not None:
             kv_cache.advance(hidden.shape[1])
         if self.has_head and return_logits:

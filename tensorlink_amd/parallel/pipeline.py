@@ -545,7 +545,7 @@ class PipelineTrainer:
                  init: str = "random", ckpt_dir: Optional[str] = None,
                  dtype=None, seed: int = 0, lr: float = 1e-4, group=None,
                  rank_base: int = 0, max_grad_norm: Optional[float] = None,
-                 **opt_kwargs):
+                 grad_checkpointing: bool = False, **opt_kwargs):
         self.runner = PipelineRunner(plan, rank, world, device=device,
                                      init=init, ckpt_dir=ckpt_dir,
                                      dtype=dtype, seed=seed, group=group,
@@ -565,6 +565,8 @@ class PipelineTrainer:
                                     **opt_kwargs)
         self.max_grad_norm = max_grad_norm
         self.lr_scheduler = None        # optional WarmupCosineLR
+        # trade activation memory for a recompute pass per layer
+        self.stage.grad_checkpointing = grad_checkpointing
         # chunked CE for big vocabularies (memory); plain CE for tiny
         # test configs (keeps exact parity with existing expectations)
         self._chunked_ce = (self.stage.has_head

@@ -561,3 +561,26 @@ def test_mxfp4_quantization():
     out = r_q.generate(torch.randint(0, 1024, (1, 12)),
                        SamplingParams(max_new_tokens=4))
     assert out.shape == (1, 4)
+
+
+def test_activation_checkpointing_exact():
+    """grad_checkpointing recomputes layers in backward: identical loss
+    trajectory to the stored-activation path (no dropout => exact)."""
+    import torch
+
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    from tensorlink_amd.parallel.planner import plan_for_world
+
+    def run2(ck):
+        t = PipelineTrainer(plan_for_world("tiny", 1), 0, 1,
+                            device=torch.device("cpu"), seed=7, lr=1e-3,
+                            grad_checkpointing=ck)
+        g = torch.Generator().manual_seed(77)
+        out = []
+        for _ in range(3):
+            b = torch.randint(0, 1024, (2, 16), generator=g)
+            out.append(t.train_step(b, labels=b))
+        return out
+
+    a, c = run2(False), run2(True)
+    assert a == __import__("pytest").approx(c, rel=1e-6)
