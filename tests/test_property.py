@@ -97,21 +97,22 @@ def test_chunked_ce_fuzz(B, S, V, chunk, pad):
     assert torch.allclose(w.grad, w2.grad, atol=1e-4)
 
 
-@settings(max_examples=8, deadline=None)
+@settings(max_examples=10, deadline=None)
 @given(lens=st.lists(st.integers(3, 200), min_size=1, max_size=4),
        chunk=st.sampled_from([None, 16, 64]),
-       prefix=st.booleans(), new=st.integers(1, 8))
-def test_batcher_fuzz_matches_serial(lens, chunk, prefix, new):
+       prefix=st.booleans(), new=st.integers(1, 8),
+       pool=st.sampled_from([None, 8, 12]))
+def test_batcher_fuzz_matches_serial(lens, chunk, prefix, new, pool):
     """Random prompt mixes through the batcher (any chunking / prefix
-    caching) always reproduce serial greedy outputs and drain the
-    pool."""
+    caching / pool pressure incl. preemption+eviction) always reproduce
+    serial greedy outputs and drain the pool."""
     from tensorlink_amd.engine.batcher import ContinuousBatcher
     from tensorlink_amd.parallel.pipeline import (PipelineRunner,
                                                   SamplingParams)
     r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
                        device=torch.device("cpu"))
     b = ContinuousBatcher(r, max_slots=4, max_ctx=512,
-                          prefill_chunk=chunk,
+                          prefill_chunk=chunk, pool_pages=pool,
                           prefix_caching=prefix).start()
     try:
         g = torch.Generator().manual_seed(sum(lens) + new)
