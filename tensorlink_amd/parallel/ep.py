@@ -96,7 +96,10 @@ class EPMoEMLP(nn.Module):
                                      minlength=ep).tolist()
 
         # exchange pair counts, then dispatch tokens + expert ids
-        cnt_in = torch.tensor(send_counts, dtype=torch.int64)
+        # (tensors live on x.device — NCCL requires device-resident
+        # collectives on GPU; gloo accepts CPU)
+        cnt_in = torch.tensor(send_counts, dtype=torch.int64,
+                              device=x.device)
         cnt_out = torch.empty_like(cnt_in)
         dist.all_to_all_single(cnt_out, cnt_in)
         recv_counts = cnt_out.tolist()
