@@ -13,6 +13,13 @@ class JobRequest(BaseModel):
     model_type: Optional[str] = None
     training: bool = False
     payment: int = 0
+    # serving options (beyond the reference's schema)
+    continuous: bool = False          # continuous batching
+    max_slots: int = 16
+    max_ctx: int = 4096
+    prefill_chunk: Optional[int] = None
+    prefix_caching: bool = False
+    quantize: Optional[str] = None    # fp8 | fp8-dense | fp4-dense
 
 
 class GenerationRequest(BaseModel):

@@ -118,8 +118,14 @@ def create_app(engine: InferenceEngine,
     def request_model(req: JobRequest, request: Request):
         _check_rate(request)
         try:
-            job = engine.load_model(req.hf_name)
-            return {"model": req.hf_name, "status": job.state}
+            job = engine.load_model(
+                req.hf_name, continuous=req.continuous,
+                max_slots=req.max_slots, max_ctx=req.max_ctx,
+                prefill_chunk=req.prefill_chunk,
+                prefix_caching=req.prefix_caching,
+                quantize=req.quantize)
+            return {"model": req.hf_name, "status": job.state,
+                    "continuous": job.batcher is not None}
         except KeyError as e:
             raise HTTPException(404, str(e))
         except Exception as e:
@@ -145,6 +151,16 @@ def create_app(engine: InferenceEngine,
     @app.get("/network-history")
     def network_history():
         return {"history": engine.metrics.snapshot()["history"]}
+
+    @app.get("/health")
+    def health():
+        from tensorlink_amd.utils.watchdog import check_gpu_health
+        h = check_gpu_health()
+        code = 200 if h["status"] != "failed" else 503
+        return JSONResponse({"status": h["status"],
+                             "checks": h["checks"],
+                             "models": len(engine.jobs)},
+                            status_code=code)
 
     @app.get("/metrics")
     def prometheus_metrics():

@@ -162,3 +162,21 @@ def test_http_streaming_over_continuous_batcher():
     # byte-fallback tokenizer decodes printable text)
     assert body.count("data:") >= 2
     eng.unload_model("tiny")
+
+
+def test_request_model_serving_options_and_health():
+    import torch
+    from fastapi.testclient import TestClient
+
+    from tensorlink_amd.api.server import create_app
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    client = TestClient(create_app(eng))
+    resp = client.post("/request-model", json={
+        "hf_name": "tiny", "continuous": True, "max_slots": 4,
+        "max_ctx": 256, "prefix_caching": True})
+    assert resp.status_code == 200 and resp.json()["continuous"] is True
+    assert eng.jobs["tiny"].batcher.prefix_caching
+    h = client.get("/health")
+    assert h.status_code == 200 and h.json()["models"] == 1
+    eng.unload_model("tiny")
