@@ -319,6 +319,7 @@ class InferenceEngine:
                     processing_time=time.time() - t0,
                     extra_texts=extra)
             nb = getattr(self, "_num_beams", 1)
+            extra = None
             if nb > 1 and self.world == 1:
                 with self._serve_lock:
                     out = job.runner.generate_beam(
@@ -335,6 +336,17 @@ class InferenceEngine:
                     self._bcast(("generate", job.name))
                     out, stats = job.runner.generate(ids, sp,
                                                      return_stats=True)
+                if n > 1 and sp.temperature > 0:
+                    # serial path: additional sampled completions
+                    extra = []
+                    for _ in range(n - 1):
+                        with self._serve_lock:
+                            self._bcast(("generate", job.name))
+                            o2 = job.runner.generate(ids, sp)
+                        extra.append(self._apply_stop(
+                            job.tokenizer.decode(o2[0].tolist(),
+                                                 skip_special_tokens=True),
+                            stop))
             text = job.tokenizer.decode(out[0].tolist(),
                                         skip_special_tokens=True)
             text = self._apply_stop(text, stop)
@@ -346,7 +358,7 @@ class InferenceEngine:
             return fmt.format_response(
                 text, prompt_tokens=ids.shape[1], completion_tokens=n_new,
                 reasoning=bool(request.get("reasoning")),
-                processing_time=time.time() - t0)
+                processing_time=time.time() - t0, extra_texts=extra)
         except Exception as e:
             self.metrics.record(0, time.time() - t0, error=True)
             return fmt.format_error(str(e))

@@ -371,3 +371,18 @@ def test_bench_serving_script_smoke():
     assert data["num_requests"] == 6
     assert data["output_tokens_per_s"] > 0
     assert data["ttft_p50_s"] is not None
+
+
+def test_n_completions_serial_path():
+    """n>1 on the serial (non-batcher) path also yields n choices."""
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    out = eng.generate({"hf_name": "tiny", "message": "x",
+                        "max_new_tokens": 4, "do_sample": True,
+                        "temperature": 1.0, "n": 2,
+                        "output_format": "openai"})
+    assert len(out["choices"]) == 2
+    eng.unload_model("tiny")
