@@ -142,6 +142,8 @@ class ContinuousBatcher:
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self.steps = 0
+        from tensorlink_amd.utils.tracing import tracer_from_env
+        self.tracer = tracer_from_env(runner.rank)  # TL_TRACE=<prefix>
 
     # ------------------------------------------------------------------
     def start(self):
@@ -171,6 +173,9 @@ class ContinuousBatcher:
         for req in leftovers:
             req.tokens.put(None)
             req.done.set()
+        if self.tracer is not None:
+            from tensorlink_amd.utils.tracing import export_from_env
+            export_from_env(self.tracer)
 
     def submit(self, input_ids: torch.Tensor, **kw) -> Request:
         req = Request(input_ids=input_ids.reshape(-1), **kw)
@@ -210,6 +215,8 @@ class ContinuousBatcher:
             return False
         victim = min(cands, key=lambda i: self.slot_emitted[i])
         req = self.slots[victim]
+        if self.tracer is not None:
+            self.tracer.instant("preempt", slot=victim)
         blob = swap_out(self.cache, victim)
         self._swapped.append((req, blob, self.slot_len[victim],
                               self.slot_last[victim],
@@ -426,7 +433,12 @@ class ContinuousBatcher:
                         self.cache.release_slot(slot)
                         continue
                     try:
-                        self._prefill_chunk(slot)
+                        if self.tracer is not None:
+                            with self.tracer.span("prefill_chunk",
+                                                  slot=slot):
+                                self._prefill_chunk(slot)
+                        else:
+                            self._prefill_chunk(slot)
                         prefilled = True
                     except Exception as e:   # pragma: no cover
                         req.error = str(e)
@@ -436,7 +448,13 @@ class ContinuousBatcher:
                         self.slot_prompt[slot] = None
                         self.cache.release_slot(slot)
                     break
-            busy = self._decode_step()
+            if self.tracer is not None:
+                with self.tracer.span("decode_step",
+                                      slots=sum(x is not None
+                                                for x in self.slots)):
+                    busy = self._decode_step()
+            else:
+                busy = self._decode_step()
             if not busy and not admitted and not prefilled:
                 self._wake.wait(0.05)
                 self._wake.clear()

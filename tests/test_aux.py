@@ -386,3 +386,27 @@ def test_n_completions_serial_path():
                         "output_format": "openai"})
     assert len(out["choices"]) == 2
     eng.unload_model("tiny")
+
+
+def test_batcher_tracing(tmp_path, monkeypatch):
+    """TL_TRACE captures scheduler spans (prefill chunks, decode steps)
+    and exports on stop."""
+    import json
+
+    import torch
+
+    from tensorlink_amd.engine.batcher import ContinuousBatcher
+    from tensorlink_amd.parallel.pipeline import PipelineRunner
+    from tensorlink_amd.parallel.planner import plan_for_world
+    monkeypatch.setenv("TL_TRACE", str(tmp_path / "bt"))
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                       device=torch.device("cpu"))
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=256,
+                          prefill_chunk=16).start()
+    torch.manual_seed(2)
+    b.submit(torch.randint(0, 1024, (40,)), max_new_tokens=4
+             ).result(timeout=60)
+    b.stop()
+    data = json.load(open(str(tmp_path / "bt") + "_rank0.json"))
+    names = {e["name"] for e in data["traceEvents"]}
+    assert "prefill_chunk" in names and "decode_step" in names
