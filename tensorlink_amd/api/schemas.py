@@ -35,6 +35,7 @@ class GenerationRequest(BaseModel):
     n: int = 1                        # OpenAI multiple completions
     speculative: bool = False         # prompt-lookup speculative decode
     seed: Optional[int] = None        # deterministic sampling seed
+    logprobs: bool = False            # return chosen-token logprobs
     presence_penalty: float = 0.0     # OpenAI range [-2, 2]
     frequency_penalty: float = 0.0
     stream: bool = False

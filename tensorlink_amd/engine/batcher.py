@@ -46,6 +46,8 @@ class Request:
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
     seed: Optional[int] = None           # per-request sampling seed
+    logprobs: bool = False               # record chosen-token logprobs
+    logprob_values: List[float] = field(default_factory=list)
     tokens: "queue.Queue" = field(default_factory=queue.Queue)
     done: threading.Event = field(default_factory=threading.Event)
     error: Optional[str] = None
@@ -309,12 +311,16 @@ class ContinuousBatcher:
                     req._gen = torch.Generator(device=logits.device)
                     req._gen.manual_seed(req.seed)
                 gen = req._gen
-            toks.append(int(ops.sample_token(
+            tok = int(ops.sample_token(
                 logits[i:i + 1], temperature=req.temperature,
                 top_p=req.top_p, top_k=req.top_k, generator=gen,
                 token_counts=counts,
                 presence_penalty=req.presence_penalty,
-                frequency_penalty=req.frequency_penalty)[0]))
+                frequency_penalty=req.frequency_penalty)[0])
+            if req.logprobs:
+                lp = torch.log_softmax(logits[i].float(), -1)[tok]
+                req.logprob_values.append(float(lp))
+            toks.append(tok)
         return toks
 
     def _maybe_finish(self, slot: int, tok: int):

@@ -298,7 +298,8 @@ class InferenceEngine:
                           top_k=sp.top_k, eos_token_id=sp.eos_token_id,
                           presence_penalty=sp.presence_penalty,
                           frequency_penalty=sp.frequency_penalty,
-                          seed=sp.seed)
+                          seed=sp.seed,
+                          logprobs=bool(request.get("logprobs")))
                 reqs = [job.batcher.submit(ids[0], **kw)
                         for _ in range(n)]
                 results = [rq.result() for rq in reqs]
@@ -312,12 +313,20 @@ class InferenceEngine:
                 job.requests_served += 1
                 job.tokens_generated += n_new
                 self.metrics.record(n_new, time.time() - t0)
-                return fmt.format_response(
+                resp = fmt.format_response(
                     text, prompt_tokens=ids.shape[1],
                     completion_tokens=n_new,
                     reasoning=bool(request.get("reasoning")),
                     processing_time=time.time() - t0,
                     extra_texts=extra)
+                if request.get("logprobs"):
+                    lp = reqs[0].logprob_values
+                    if "choices" in resp:
+                        resp["choices"][0]["logprobs"] = {
+                            "token_logprobs": lp}
+                    else:
+                        resp["logprobs"] = lp
+                return resp
             nb = getattr(self, "_num_beams", 1)
             extra = None
             if nb > 1 and self.world == 1:

@@ -407,3 +407,27 @@ def test_preemption_swap_and_resume_matches_serial():
     finally:
         b.stop()
     assert b.cache.allocator.n_free == b.cache.allocator.n_pages
+
+
+def test_logprobs_match_recompute():
+    """Requested logprobs equal a teacher-forced recompute of the greedy
+    sequence."""
+    import torch.nn.functional as F
+
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=128).start()
+    try:
+        torch.manual_seed(7)
+        p = torch.randint(0, 1024, (10,))
+        req = b.submit(p.clone(), max_new_tokens=5, logprobs=True)
+        out = req.result(timeout=60)
+        assert len(req.logprob_values) == 5
+        full = torch.cat([p, torch.tensor(out)]).unsqueeze(0)
+        pos = torch.arange(full.shape[1]).unsqueeze(0).contiguous()
+        logits = r.stage(full, pos)
+        for j, tok in enumerate(out):
+            row = logits[0, 9 + j].float()
+            want = float(F.log_softmax(row, -1)[tok])
+            assert abs(req.logprob_values[j] - want) < 1e-4
+    finally:
+        b.stop()
