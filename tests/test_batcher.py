@@ -424,7 +424,8 @@ def test_logprobs_match_recompute():
         assert len(req.logprob_values) == 5
         full = torch.cat([p, torch.tensor(out)]).unsqueeze(0)
         pos = torch.arange(full.shape[1]).unsqueeze(0).contiguous()
-        logits = r.stage(full, pos)
+        with torch.no_grad():
+            logits = r.stage(full, pos)
         for j, tok in enumerate(out):
             row = logits[0, 9 + j].float()
             want = float(F.log_softmax(row, -1)[tok])
