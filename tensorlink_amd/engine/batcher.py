@@ -105,7 +105,9 @@ class ContinuousBatcher:
     def __init__(self, runner, max_slots: int = 16, max_ctx: int = 2048,
                  pool_pages: Optional[int] = None,
                  prefill_chunk: Optional[int] = None,
-                 prefix_caching: bool = False):
+                 prefix_caching: bool = False,
+                 speculative: bool = False, lookup_n: int = 3,
+                 spec_k: int = 8):
         assert runner.world == 1 or self.SUPPORTS_PP, \
             "use PPContinuousBatcher for world > 1"
         self.runner = runner
@@ -120,6 +122,14 @@ class ContinuousBatcher:
         cache_cls = (PrefixCachingKVCache if prefix_caching
                      else DynamicPagedKVCache)
         self.prefix_caching = prefix_caching
+        # prompt-lookup speculation when exactly ONE slot is decoding
+        # greedily (the ragged multi-token verify is per-slot; batching
+        # it needs per-row q_off — roadmap). Exact greedy either way.
+        self.speculative = speculative
+        self.lookup_n = lookup_n
+        self.spec_k = spec_k
+        self.slot_ctx: List[Optional[list]] = [None] * max_slots
+        self.spec_accepted = 0
         self.cache = cache_cls(
             self.stage.num_layers, max_slots, pool_pages, max_ctx,
             self.stage.config, runner.device, runner.dtype)
@@ -243,6 +253,7 @@ class ContinuousBatcher:
                 self._swapped.pop(0)
             swap_in(self.cache, slot, blob, s_len)
             self.slots[slot] = req
+            self.slot_ctx[slot] = None   # resumed: plain decode
             self.slot_prompt[slot] = prompt
             self.slot_filled[slot] = prompt.numel()
             self.slot_len[slot] = s_len
@@ -295,6 +306,8 @@ class ContinuousBatcher:
         self.slot_len[slot] = S_total
         self.slot_last[slot] = tok
         self.slot_emitted[slot] = 1
+        if self.speculative:
+            self.slot_ctx[slot] = prompt.tolist() + [tok]
         self.slot_counts[slot][tok] = self.slot_counts[slot].get(tok,
                                                                  0) + 1
         req.tokens.put(tok)
@@ -375,7 +388,55 @@ class ContinuousBatcher:
                 self.slot_counts[slot].get(new[j], 0) + 1
             self.slots[slot].tokens.put(new[j])
             self.slot_last[slot] = new[j]
+            if self.slot_ctx[slot] is not None:
+                self.slot_ctx[slot].append(new[j])
             self._maybe_finish(slot, new[j])
+        self.steps += 1
+        return True
+
+    @torch.no_grad()
+    def _spec_step(self, slot: int) -> bool:
+        """One speculative round for the single decoding slot: propose
+        k tokens via prompt lookup, verify them all in ONE forward
+        (ragged q_off attention), accept the matching prefix + bonus.
+        Greedy-exact; emits 1..k+1 tokens per forward."""
+        from tensorlink_amd.parallel.pipeline import PipelineRunner
+        req = self.slots[slot]
+        ctx = self.slot_ctx[slot]
+        proposal = PipelineRunner._lookup_propose(ctx, self.lookup_n,
+                                                  self.spec_k)
+        # bound by remaining budget/context
+        room = min(req.max_new_tokens - self.slot_emitted[slot],
+                   self.max_ctx - 1 - self.slot_len[slot]) - 1
+        proposal = proposal[:max(0, room)]
+        m = len(proposal)
+        self.cache.ensure(slot, self.slot_len[slot] + m + 2)
+        feed = torch.tensor([[self.slot_last[slot]] + proposal],
+                            device=self.device, dtype=torch.int64)
+        pos = (self.slot_len[slot] + torch.arange(
+            m + 1, device=self.device, dtype=torch.int32)).unsqueeze(0)
+        lens = torch.tensor([self.slot_len[slot]], device=self.device,
+                            dtype=torch.int32)
+        view = _SlotView(self.cache, [slot], lens)
+        logits = self.stage(feed, pos, kv_cache=view)      # [1, m+1, V]
+        targets = logits.argmax(-1)[0].tolist()
+        a = 0
+        while a < m and proposal[a] == targets[a]:
+            a += 1
+        accepted = proposal[:a] + [targets[a]]
+        self.spec_accepted += a
+        # kv valid for the fed token + a accepted proposals
+        self.slot_len[slot] += 1 + a
+        for tok in accepted:
+            self.slot_emitted[slot] += 1
+            self.slot_counts[slot][tok] = \
+                self.slot_counts[slot].get(tok, 0) + 1
+            req.tokens.put(tok)
+            ctx.append(tok)
+            self.slot_last[slot] = tok
+            self._maybe_finish(slot, tok)
+            if self.slots[slot] is None:
+                break
         self.steps += 1
         return True
 
@@ -420,6 +481,7 @@ class ContinuousBatcher:
                 self.slot_filled[slot] = 0
                 self.slot_len[slot] = 0
                 self.slot_counts[slot] = {}
+                self.slot_ctx[slot] = None
                 if self.prefix_caching:
                     self.slot_filled[slot] = self.cache.adopt_prefix(
                         slot, req.input_ids.tolist())
@@ -454,7 +516,18 @@ class ContinuousBatcher:
                         self.slot_prompt[slot] = None
                         self.cache.release_slot(slot)
                     break
-            if self.tracer is not None:
+            spec_slot = None
+            if self.speculative:
+                decoding = [i for i, r in enumerate(self.slots)
+                            if r is not None and not self._prefilling(i)
+                            and self.slot_filled[i] > 0]
+                if (len(decoding) == 1
+                        and self.slots[decoding[0]].temperature <= 0
+                        and self.slot_ctx[decoding[0]] is not None):
+                    spec_slot = decoding[0]
+            if spec_slot is not None:
+                busy = self._spec_step(spec_slot)
+            elif self.tracer is not None:
                 with self.tracer.span("decode_step",
                                       slots=sum(x is not None
                                                 for x in self.slots)):

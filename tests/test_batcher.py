@@ -432,3 +432,38 @@ def test_logprobs_match_recompute():
             assert abs(req.logprob_values[j] - want) < 1e-4
     finally:
         b.stop()
+
+
+def test_batcher_speculative_single_slot():
+    """In-batcher speculation (single greedy slot): output equals the
+    non-speculative batcher and serial generation; on a looping model
+    proposals get accepted, cutting scheduler steps."""
+    from tensorlink_amd.parallel.planner import plan_for_world
+
+    def mk(spec):
+        # seed 10's decode loops (see test_speculative_decode_exact_greedy)
+        r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                           device=torch.device("cpu"), seed=10)
+        return r, ContinuousBatcher(r, max_slots=2, max_ctx=256,
+                                    speculative=spec).start()
+
+    torch.manual_seed(19)
+    p = torch.randint(0, 1024, (16,))
+
+    r1, b1 = mk(False)
+    try:
+        base = b1.submit(p.clone(), max_new_tokens=40).result(timeout=60)
+        base_steps = b1.steps
+    finally:
+        b1.stop()
+
+    r2, b2 = mk(True)
+    try:
+        spec = b2.submit(p.clone(), max_new_tokens=40).result(timeout=60)
+        assert spec == base
+        assert b2.spec_accepted > 0
+        assert b2.steps < base_steps
+        ref = r2.generate(p.unsqueeze(0), SamplingParams(max_new_tokens=40))
+        assert spec == ref[0].tolist()
+    finally:
+        b2.stop()
