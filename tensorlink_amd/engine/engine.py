@@ -153,6 +153,7 @@ class InferenceEngine:
         pool_pages = kwargs.pop("pool_pages", None)
         prefill_chunk = kwargs.pop("prefill_chunk", None)
         prefix_caching = kwargs.pop("prefix_caching", False)
+        speculative = kwargs.pop("speculative", False)
         plan = plan_for_world(config, self.world, **kwargs)
         runner = PipelineRunner(plan, self.rank, self.world,
                                 device=self.device, init=init,
@@ -164,7 +165,8 @@ class InferenceEngine:
             job.batcher = ContinuousBatcher(
                 runner, max_slots=max_slots, max_ctx=max_ctx,
                 pool_pages=pool_pages, prefill_chunk=prefill_chunk,
-                prefix_caching=prefix_caching).start()
+                prefix_caching=prefix_caching,
+                speculative=speculative).start()
         elif continuous and self.world > 1:
             from tensorlink_amd.engine.batcher import PPContinuousBatcher
             b = PPContinuousBatcher(runner, max_slots=max_slots,

@@ -39,7 +39,8 @@ def cmd_serve(args):
         engine.load_model(name, continuous=args.continuous,
                           max_slots=args.max_slots, max_ctx=args.max_ctx,
                           prefill_chunk=args.prefill_chunk,
-                          prefix_caching=args.prefix_caching)
+                          prefix_caching=args.prefix_caching,
+                          speculative=args.speculative)
     keeper = StateKeeper(engine)
     keeper.load_previous_state()
     keeper.start()
@@ -110,6 +111,9 @@ def main():
     s.add_argument("--max-slots", type=int, default=16)
     s.add_argument("--max-ctx", type=int, default=4096)
     s.add_argument("--prefill-chunk", type=int, default=None)
+    s.add_argument("--speculative", action="store_true",
+                   help="prompt-lookup speculation when a single "
+                        "request is decoding (exact greedy)")
     s.add_argument("--prefix-caching", action="store_true",
                    help="reuse KV pages across requests sharing a "
                         "prompt prefix")
