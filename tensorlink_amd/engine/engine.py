@@ -393,6 +393,8 @@ class InferenceEngine:
         def on_token(step, toks):
             q.put(int(toks[0]))
 
+        breq = {}                       # batcher Request (for cancel)
+
         def run():
             try:
                 if job.batcher is not None:
@@ -403,6 +405,7 @@ class InferenceEngine:
                         presence_penalty=sp.presence_penalty,
                         frequency_penalty=sp.frequency_penalty,
                         seed=sp.seed)
+                    breq["req"] = req
                     for tok in req.stream():
                         q.put(tok)
                 else:
@@ -420,6 +423,19 @@ class InferenceEngine:
         thread.start()
 
         sent: List[int] = []
+        first = True
+        prev_text = ""
+        try:
+            yield from self._stream_body(q, fmt, job, sp, sent, t0,
+                                         stop, request)
+        finally:
+            # client disconnected (GeneratorExit) or stream finished:
+            # stop the batcher request so its slot frees immediately
+            if "req" in breq and not breq["req"].done.is_set():
+                breq["req"].cancel()
+
+    def _stream_body(self, q, fmt, job, sp, sent, t0, stop, request):
+        emitted = 0
         first = True
         prev_text = ""
         while True:

@@ -180,3 +180,27 @@ def test_request_model_serving_options_and_health():
     h = client.get("/health")
     assert h.status_code == 200 and h.json()["models"] == 1
     eng.unload_model("tiny")
+
+
+def test_stream_disconnect_cancels_request():
+    """Closing the SSE stream mid-generation cancels the batcher
+    request so its slot frees (no zombie decode)."""
+    import time
+
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny", continuous=True, max_slots=2, max_ctx=256)
+    gen = eng.generate_stream({"hf_name": "tiny", "message": "hello",
+                               "max_new_tokens": 200, "do_sample": False,
+                               "output_format": "simple"})
+    next(gen)                      # first chunk arrives
+    gen.close()                    # client disconnects
+    b = eng.jobs["tiny"].batcher
+    for _ in range(300):
+        if all(s is None for s in b.slots):
+            break
+        time.sleep(0.01)
+    assert all(s is None for s in b.slots), "slot not freed after close"
+    eng.unload_model("tiny")
