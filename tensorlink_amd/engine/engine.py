@@ -499,7 +499,9 @@ class InferenceEngine:
         with self._auto_lock:
             # unload cold auto-loaded models
             for name in list(self._auto_loaded):
-                if self.demand.get(name, 0) -                         self._auto_demand_mark.get(name, 0) < min_demand                         and name in self.jobs:
+                idle = (self.demand.get(name, 0)
+                        - self._auto_demand_mark.get(name, 0))
+                if idle < min_demand and name in self.jobs:
                     self.unload_model(name)
                     self._auto_loaded.discard(name)
                     # cooled: only reload once demand grows again
