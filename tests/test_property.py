@@ -130,3 +130,23 @@ def test_batcher_fuzz_matches_serial(lens, chunk, prefix, new, pool):
     if prefix:
         free += len(b.cache.lru)
     assert free == b.cache.allocator.n_pages
+
+
+@settings(max_examples=30, deadline=None)
+@given(N=st.integers(1, 8), groups=st.integers(1, 6),
+       scale=st.floats(1e-3, 1e3))
+def test_mxfp4_fuzz_bounds(N, groups, scale):
+    """MXFP4 quantization error stays within the e2m1 grid bound for
+    arbitrary magnitudes, and packing is exactly invertible on its own
+    dequantized output."""
+    from tensorlink_amd.models.quant import (dequantize_mxfp4,
+                                             quantize_mxfp4)
+    K = groups * 32
+    w = torch.randn(N, K) * scale
+    p, e = quantize_mxfp4(w)
+    wq = dequantize_mxfp4(p, e)
+    grp = w.reshape(N, groups, 32)
+    err = (grp - wq.reshape(N, groups, 32)).abs().amax(-1)
+    assert bool((err <= grp.abs().amax(-1) * 0.26 + 1e-9).all())
+    p2, e2 = quantize_mxfp4(wq)
+    torch.testing.assert_close(dequantize_mxfp4(p2, e2), wq)
