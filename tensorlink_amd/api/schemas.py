@@ -25,6 +25,9 @@ class JobRequest(BaseModel):
 class GenerationRequest(BaseModel):
     hf_name: str
     message: str = ""
+    prompt: Optional[str] = None      # reference alias for message
+    model_type: Optional[str] = None  # architecture hint (auto-detected)
+    is_chat_completion: bool = False  # reference: openai response shape
     max_length: int = 2048
     max_new_tokens: int = 256
     temperature: float = 0.7

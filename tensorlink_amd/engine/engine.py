@@ -242,6 +242,14 @@ class InferenceEngine:
         name = request.get("hf_name") or request.get("model")
         if not name:
             raise ValueError("request needs 'hf_name' or 'model'")
+        # reference aliases (README.md request table): `prompt` is an
+        # alternative to `message`; `is_chat_completion` selects the
+        # OpenAI response shape
+        if not request.get("message") and request.get("prompt"):
+            request = dict(request, message=request["prompt"])
+        if request.get("is_chat_completion") and \
+                "output_format" not in request:
+            request = dict(request, output_format="openai")
         self.demand[name] += 1
         job = self.jobs.get(name)
         if job is None or job.state != "ready":

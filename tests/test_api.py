@@ -204,3 +204,18 @@ def test_stream_disconnect_cancels_request():
         time.sleep(0.01)
     assert all(s is None for s in b.slots), "slot not freed after close"
     eng.unload_model("tiny")
+
+
+def test_reference_request_aliases():
+    """Reference README request params: `prompt` aliases `message`;
+    `is_chat_completion` selects the OpenAI shape."""
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny")
+    out = eng.generate({"hf_name": "tiny", "prompt": "hello there",
+                        "max_new_tokens": 3, "do_sample": False,
+                        "is_chat_completion": True})
+    assert out.get("object") == "chat.completion"
+    eng.unload_model("tiny")
