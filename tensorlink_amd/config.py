@@ -29,6 +29,14 @@ class MLConfig:
     init: str = "random"                      # random | checkpoint
     checkpoint_root: Optional[str] = None
     dtype: str = "bfloat16"
+    # reference ml.trusted (bin/config.json): allow distributing
+    # user-supplied nn.Modules (module.py trusted mode)
+    trusted: bool = False
+    # serving defaults applied by `tlctl serve` when flags are absent
+    continuous: bool = False
+    prefix_caching: bool = False
+    prefill_chunk: Optional[int] = None
+    job_ttl_s: Optional[float] = None         # idle-job eviction
 
 
 @dataclass

@@ -36,15 +36,16 @@ def cmd_serve(args):
     models = args.model or cfg.ml.default_models
     for name in models:
         log.info("loading %s ...", name)
-        engine.load_model(name, continuous=args.continuous,
-                          max_slots=args.max_slots, max_ctx=args.max_ctx,
-                          prefill_chunk=args.prefill_chunk,
-                          prefix_caching=args.prefix_caching,
-                          speculative=args.speculative)
+        engine.load_model(
+            name, continuous=args.continuous or cfg.ml.continuous,
+            max_slots=args.max_slots, max_ctx=args.max_ctx,
+            prefill_chunk=args.prefill_chunk or cfg.ml.prefill_chunk,
+            prefix_caching=args.prefix_caching or cfg.ml.prefix_caching,
+            speculative=args.speculative)
     keeper = StateKeeper(engine)
     keeper.load_previous_state()
     keeper.start()
-    Watchdog(engine).start()
+    Watchdog(engine, job_ttl_s=cfg.ml.job_ttl_s).start()
 
     from tensorlink_amd.api.server import TensorlinkAPI
     api = TensorlinkAPI(engine, host=cfg.node.endpoint_host,
