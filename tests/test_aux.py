@@ -410,3 +410,20 @@ def test_batcher_tracing(tmp_path, monkeypatch):
     data = json.load(open(str(tmp_path / "bt") + "_rank0.json"))
     names = {e["name"] for e in data["traceEvents"]}
     assert "prefill_chunk" in names and "decode_step" in names
+
+
+def test_engine_config_serving_defaults(tmp_path):
+    import json
+
+    from tensorlink_amd.config import EngineConfig
+    p = tmp_path / "config.json"
+    p.write_text(json.dumps({
+        "node": {"endpoint_port": 9100},
+        "ml": {"continuous": True, "prefix_caching": True,
+               "prefill_chunk": 256, "job_ttl_s": 60.0,
+               "trusted": True}}))
+    cfg = EngineConfig.load(str(p))
+    assert cfg.node.endpoint_port == 9100
+    assert cfg.ml.continuous and cfg.ml.prefix_caching
+    assert cfg.ml.prefill_chunk == 256 and cfg.ml.job_ttl_s == 60.0
+    assert cfg.ml.trusted
