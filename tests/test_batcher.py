@@ -467,3 +467,24 @@ def test_batcher_speculative_single_slot():
         assert spec == ref[0].tolist()
     finally:
         b2.stop()
+
+
+def test_speculation_with_prefix_and_chunking():
+    """Speculation + prefix caching + chunked prefill compose: exact
+    serial equality on repeat submissions."""
+    from tensorlink_amd.parallel.planner import plan_for_world
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                       device=torch.device("cpu"), seed=10)
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=512, prefill_chunk=64,
+                          prefix_caching=True, speculative=True).start()
+    try:
+        torch.manual_seed(19)
+        p = torch.randint(0, 1024, (150,))
+        o1 = b.submit(p.clone(), max_new_tokens=30).result(timeout=120)
+        o2 = b.submit(p.clone(), max_new_tokens=30).result(timeout=120)
+        assert o1 == o2
+        assert b.cache.hits == 128
+        ref = r.generate(p.unsqueeze(0), SamplingParams(max_new_tokens=30))
+        assert o1 == ref[0].tolist()
+    finally:
+        b.stop()
