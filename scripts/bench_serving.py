@@ -53,6 +53,7 @@ def main():
     ap.add_argument("--max-ctx", type=int, default=2048)
     ap.add_argument("--prefill-chunk", type=int, default=None)
     ap.add_argument("--prefix-caching", action="store_true")
+    ap.add_argument("--speculative", action="store_true")
     ap.add_argument("--shared-prefix", type=int, default=0,
                     help="tokens of prompt shared by ALL requests "
                          "(exercises prefix caching)")
@@ -65,7 +66,8 @@ def main():
     batcher = ContinuousBatcher(
         runner, max_slots=args.max_slots, max_ctx=args.max_ctx,
         prefill_chunk=args.prefill_chunk,
-        prefix_caching=args.prefix_caching).start()
+        prefix_caching=args.prefix_caching,
+        speculative=args.speculative).start()
 
     rng = random.Random(args.seed)
     g = torch.Generator().manual_seed(args.seed)
@@ -118,6 +120,8 @@ def main():
         "tpot_p50_s": pct(tpots, 50), "tpot_p95_s": pct(tpots, 95),
         "latency_p50_s": pct([r["latency"] for r in results], 50),
         "prefix_cache_hit_tokens": getattr(batcher.cache, "hits", 0),
+        "spec_accepted_tokens": getattr(batcher, "spec_accepted", 0),
+        "preemptions": getattr(batcher, "preemptions", 0),
         "scheduler_steps": batcher.steps,
         "config": {"prompt_len": args.prompt_len,
                    "new_tokens": args.new_tokens,

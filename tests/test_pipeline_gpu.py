@@ -416,3 +416,31 @@ def test_preemption_swap_gpu():
             del os.environ["TL_NO_GRAPH"]
     finally:
         b.stop()
+
+
+def test_serving_feature_composition_gpu():
+    """Chunked prefill + prefix caching + speculation composed on the
+    HIP path: repeat prompt equals serial greedy."""
+    from tensorlink_amd.engine.batcher import ContinuousBatcher
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1, device=DEV,
+                       dtype=torch.bfloat16, seed=10)
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=512, prefill_chunk=64,
+                          prefix_caching=True, speculative=True).start()
+    try:
+        torch.manual_seed(19)
+        p = torch.randint(0, 1024, (150,))
+        o1 = b.submit(p.clone(), max_new_tokens=30).result(timeout=120)
+        o2 = b.submit(p.clone(), max_new_tokens=30).result(timeout=120)
+        assert o1 == o2 and b.cache.hits == 128
+        import os
+        os.environ["TL_NO_GRAPH"] = "1"
+        try:
+            ref = r.generate(p.unsqueeze(0),
+                             SamplingParams(max_new_tokens=30))
+        finally:
+            del os.environ["TL_NO_GRAPH"]
+        assert o1 == ref[0].cpu().tolist()
+    finally:
+        b.stop()
