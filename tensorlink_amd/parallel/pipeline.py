@@ -545,7 +545,9 @@ class PipelineTrainer:
                  init: str = "random", ckpt_dir: Optional[str] = None,
                  dtype=None, seed: int = 0, lr: float = 1e-4, group=None,
                  rank_base: int = 0, max_grad_norm: Optional[float] = None,
-                 grad_checkpointing: bool = False, **opt_kwargs):
+                 grad_checkpointing: bool = False,
+                 lora_r: Optional[int] = None, lora_alpha: float = 16.0,
+                 **opt_kwargs):
         self.runner = PipelineRunner(plan, rank, world, device=device,
                                      init=init, ckpt_dir=ckpt_dir,
                                      dtype=dtype, seed=seed, group=group,
@@ -561,8 +563,16 @@ class PipelineTrainer:
         self.p2p = self.runner.p2p
         self.H = self.runner.H
         from tensorlink_amd.optim import FusedAdamW
-        self.optimizer = FusedAdamW(self.stage.parameters(), lr=lr,
-                                    **opt_kwargs)
+        if lora_r:
+            # LoRA fine-tuning: freeze the stage, train only adapters
+            from tensorlink_amd.models.lora import (apply_lora,
+                                                    lora_parameters)
+            apply_lora(self.stage, r=lora_r, alpha=lora_alpha)
+            self.optimizer = FusedAdamW(lora_parameters(self.stage),
+                                        lr=lr, **opt_kwargs)
+        else:
+            self.optimizer = FusedAdamW(self.stage.parameters(), lr=lr,
+                                        **opt_kwargs)
         self.max_grad_norm = max_grad_norm
         self.lr_scheduler = None        # optional WarmupCosineLR
         # trade activation memory for a recompute pass per layer

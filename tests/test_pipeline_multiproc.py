@@ -1186,3 +1186,41 @@ def test_pp4_continuous_batching():
             toks.append(t)
             cur = torch.cat([cur, torch.tensor([[t]])], 1)
         assert o == toks, (o, toks)
+
+
+def _lora_pp_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    from tensorlink_amd.parallel.planner import plan_for_world
+    init_distributed(backend="gloo")
+    plan = plan_for_world("tiny", 2, training=True)
+    t = PipelineTrainer(plan, rank, 2, device=torch.device("cpu"),
+                        seed=0, lr=5e-3, lora_r=4)
+    base = {k: v.clone() for k, v in t.stage.state_dict().items()
+            if ".base.weight" in k}
+    torch.manual_seed(31)
+    losses = []
+    for _ in range(6):
+        b = torch.randint(0, 1024, (2, 16))
+        losses.append(t.train_step(b, labels=b))
+    frozen = all(torch.equal(v, t.stage.state_dict()[k])
+                 for k, v in base.items())
+    q.put((rank, (losses, frozen,
+                  t.optimizer.flat_param.numel())))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_lora_across_pipeline():
+    """LoRA over a 2-stage pipeline: adapters train (loss falls), base
+    weights stay frozen, optimizer state is tiny."""
+    outs = dict(_run_collect(_lora_pp_worker, 2, _port(49)))
+    for rank, (losses, frozen, n_opt) in outs.items():
+        assert losses[-1] < losses[0]
+        assert frozen
+        assert n_opt < 100_000          # adapters only (full stage ~1.5M)
