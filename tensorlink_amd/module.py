@@ -111,7 +111,8 @@ class DistributedModel(nn.Module):
                  world_size: int = 1, mode: str = "auto",
                  device=None, init: str = "random",
                  ckpt_dir: Optional[str] = None, seed: int = 0,
-                 lr: float = 1e-4, n_pipelines: int = 1, tp: int = 1):
+                 lr: float = 1e-4, n_pipelines: int = 1, tp: int = 1,
+                 **trainer_kw):
         super().__init__()
         self.model_name = model
         self.training_mode = training
@@ -138,7 +139,8 @@ class DistributedModel(nn.Module):
             if training:
                 self._trainer = PipelineTrainer(plan, 0, 1, device=device,
                                                 init=init, ckpt_dir=ckpt_dir,
-                                                seed=seed, lr=lr)
+                                                seed=seed, lr=lr,
+                                                **trainer_kw)
                 self._runner = self._trainer.runner
             else:
                 self._trainer = None

@@ -133,3 +133,21 @@ def test_custom_module_trusted_mode():
                                    atol=1e-6, rtol=1e-5)
     finally:
         m.shutdown()
+
+
+def test_distributed_model_lora():
+    """DistributedModel(training=True, lora_r=4): adapter-only training
+    through the user API."""
+    import torch
+
+    from tensorlink_amd.module import DistributedModel
+    m = DistributedModel("tiny", training=True, lora_r=4, lr=5e-3,
+                         device=torch.device("cpu"))
+    opt = m.create_optimizer()
+    torch.manual_seed(3)
+    ids = torch.randint(0, 1024, (2, 16))
+    l0 = m.train_step(ids, ids)
+    for _ in range(5):
+        l1 = m.train_step(ids, ids)
+    assert l1 < l0
+    assert m._trainer.optimizer.flat_param.numel() < 100_000
