@@ -154,6 +154,8 @@ class InferenceEngine:
         prefill_chunk = kwargs.pop("prefill_chunk", None)
         prefix_caching = kwargs.pop("prefix_caching", False)
         speculative = kwargs.pop("speculative", False)
+        lookup_n = kwargs.pop("lookup_n", 3)
+        spec_k = kwargs.pop("spec_k", 8)
         plan = plan_for_world(config, self.world, **kwargs)
         runner = PipelineRunner(plan, self.rank, self.world,
                                 device=self.device, init=init,
@@ -166,7 +168,8 @@ class InferenceEngine:
                 runner, max_slots=max_slots, max_ctx=max_ctx,
                 pool_pages=pool_pages, prefill_chunk=prefill_chunk,
                 prefix_caching=prefix_caching,
-                speculative=speculative).start()
+                speculative=speculative, lookup_n=lookup_n,
+                spec_k=spec_k).start()
         elif continuous and self.world > 1:
             from tensorlink_amd.engine.batcher import PPContinuousBatcher
             b = PPContinuousBatcher(runner, max_slots=max_slots,
