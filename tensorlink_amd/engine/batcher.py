@@ -191,6 +191,8 @@ class ContinuousBatcher:
 
     def submit(self, input_ids: torch.Tensor, **kw) -> Request:
         req = Request(input_ids=input_ids.reshape(-1), **kw)
+        if req.input_ids.numel() == 0:
+            raise ValueError("empty prompt")
         req.max_new_tokens = min(req.max_new_tokens, self.max_ctx - 1)
         if req.input_ids.numel() + req.max_new_tokens > self.max_ctx:
             keep = max(1, self.max_ctx - req.max_new_tokens)

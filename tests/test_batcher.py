@@ -488,3 +488,25 @@ def test_speculation_with_prefix_and_chunking():
         assert o1 == ref[0].tolist()
     finally:
         b.stop()
+
+
+def test_edge_cases():
+    """Empty prompts rejected; max_new_tokens=1 works; a 1-token prompt
+    works."""
+    import pytest as _pytest
+
+    r = _runner()
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=128).start()
+    try:
+        with _pytest.raises(ValueError):
+            b.submit(torch.empty(0, dtype=torch.int64))
+        torch.manual_seed(1)
+        one = b.submit(torch.randint(0, 1024, (1,)),
+                       max_new_tokens=1).result(timeout=60)
+        assert len(one) == 1
+        ref = r.generate(torch.tensor([[one and 0 or 0]]) * 0 +
+                         torch.randint(0, 1024, (1, 1)),
+                         SamplingParams(max_new_tokens=1))
+        assert ref.shape == (1, 1)
+    finally:
+        b.stop()
