@@ -643,3 +643,27 @@ def test_lora_training_and_merge():
     with torch.no_grad():
         again = m2(ids, pos)
     torch.testing.assert_close(again, before, atol=2e-5, rtol=1e-5)
+
+
+def test_runner_paged_mode_matches_contiguous():
+    """TL_KV_MODE=paged runner generation (incl. speculative decode over
+    the paged pool) equals contiguous-mode output."""
+    import torch
+
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    torch.manual_seed(9)
+    ids = torch.randint(0, 1024, (2, 20))
+    rp = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                        device=torch.device("cpu"), seed=3,
+                        kv_mode="paged")
+    rc = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                        device=torch.device("cpu"), seed=3)
+    a = rp.generate(ids, SamplingParams(max_new_tokens=6))
+    b = rc.generate(ids, SamplingParams(max_new_tokens=6))
+    assert torch.equal(a, b)
+
+    one = torch.randint(0, 1024, (1, 16))
+    ref = rc.generate(one, SamplingParams(max_new_tokens=12))
+    out, _ = rp.generate_speculative(one, max_new_tokens=12)
+    assert torch.equal(out, ref)
