@@ -98,6 +98,9 @@ class _SlotView:
     def gather_contiguous(self, layer, S):
         return PagedKVCache.gather_contiguous(self, layer, S)
 
+    def append(self, layer, k_new, v_new, positions):
+        return PagedKVCache.append(self, layer, k_new, v_new, positions)
+
 
 class ContinuousBatcher:
     SUPPORTS_PP = False

@@ -65,13 +65,27 @@ class Gpt2Attention(nn.Module):
                                         scale=self.scale)
         else:
             kv_cache.append(layer_idx, k, v, positions)
+            table = getattr(kv_cache, "table", None)
             if S == 1:
                 out = ops.attention_decode(
                     q, kv_cache.k[layer_idx], kv_cache.v[layer_idx],
-                    kv_cache.seq_lens + 1, scale=self.scale)
+                    kv_cache.seq_lens + 1, scale=self.scale,
+                    block_table=table)
             else:
-                out = ops.attention_prefill(q, k, v, causal=True,
-                                            scale=self.scale)
+                # prefill with cache; off > 0 = chunked continuation
+                # (same contract as models/dense.py Attention)
+                off = int(kv_cache.seq_lens.max())
+                kv_len = off + S
+                if table is None:
+                    k_attn = kv_cache.k[layer_idx][:, :, :kv_len].permute(
+                        0, 2, 1, 3).contiguous()
+                    v_attn = kv_cache.v[layer_idx][:, :, :kv_len].permute(
+                        0, 2, 1, 3).contiguous()
+                else:
+                    k_attn, v_attn = kv_cache.gather_contiguous(layer_idx,
+                                                                kv_len)
+                out = ops.attention_prefill(q, k_attn, v_attn, causal=True,
+                                            scale=self.scale, q_off=off)
         return self.c_proj(out.reshape(B, S, H))
 
 

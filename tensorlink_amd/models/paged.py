@@ -54,6 +54,21 @@ class PagedKVCache:
     def advance(self, n: int):
         self.seq_lens += n
 
+    def append(self, layer: int, k_new: torch.Tensor,
+               v_new: torch.Tensor, positions: torch.Tensor):
+        """Scatter new K/V into pages at `positions` (the contiguous
+        KVCache.append API — the GPT-2 family writes through it instead
+        of the fused rope_append kernel). k_new/v_new [B,S,Hkv,D],
+        positions [B,S]."""
+        B, S, H, D = k_new.shape
+        pos = positions.to(self.table.device, torch.long).reshape(B, S)
+        page = self.table.long().gather(1, pos // PAGE).reshape(-1)
+        off = (pos % PAGE).reshape(-1)
+        self.k[layer][page, :, off] = k_new.reshape(B * S, H, D).to(
+            self.k[layer].dtype)
+        self.v[layer][page, :, off] = v_new.reshape(B * S, H, D).to(
+            self.v[layer].dtype)
+
     def gather_contiguous(self, layer: int, S: int):
         """Materialize the first S positions as [B, S, Hkv, D] (prefill
         attention readback; once per request)."""

@@ -510,3 +510,21 @@ def test_edge_cases():
         assert ref.shape == (1, 1)
     finally:
         b.stop()
+
+
+def test_gpt2_through_batcher():
+    """The GPT-2 family (contiguous-API cache writes) serves through the
+    paged continuous batcher: chunked prefill + decode == serial."""
+    from tensorlink_amd.parallel.planner import plan_for_world
+    r = PipelineRunner(plan_for_world("gpt2-small", 1), 0, 1,
+                       device=torch.device("cpu"), seed=1)
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=256,
+                          prefill_chunk=16).start()
+    try:
+        torch.manual_seed(4)
+        p = torch.randint(0, 50257, (40,))
+        out = b.submit(p.clone(), max_new_tokens=6).result(timeout=120)
+        ref = r.generate(p.unsqueeze(0), SamplingParams(max_new_tokens=6))
+        assert out == ref[0].tolist()
+    finally:
+        b.stop()
