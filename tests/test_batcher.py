@@ -528,3 +528,26 @@ def test_gpt2_through_batcher():
         assert out == ref[0].tolist()
     finally:
         b.stop()
+
+
+@pytest.mark.parametrize("name", ["tiny-moe", "tiny-qwen3",
+                                  "tiny-qwen3-moe", "gpt2-small"])
+def test_all_families_through_batcher(name):
+    """Every model family serves through the batcher (chunked prefill +
+    prefix caching) and matches serial greedy; speculation is exact."""
+    from tensorlink_amd.parallel.planner import plan_for_world
+    r = PipelineRunner(plan_for_world(name, 1), 0, 1,
+                       device=torch.device("cpu"), seed=2)
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=256, prefill_chunk=16,
+                          prefix_caching=True).start()
+    try:
+        torch.manual_seed(5)
+        p = torch.randint(0, r.config.vocab_size, (40,))
+        out = b.submit(p.clone(), max_new_tokens=5).result(timeout=120)
+        ref = r.generate(p.unsqueeze(0), SamplingParams(max_new_tokens=5))
+        assert out == ref[0].tolist()
+    finally:
+        b.stop()
+    o, _ = r.generate_speculative(p.unsqueeze(0), max_new_tokens=8)
+    rr = r.generate(p.unsqueeze(0), SamplingParams(max_new_tokens=8))
+    assert torch.equal(o, rr)
