@@ -3,6 +3,7 @@ decode batch and reproduce serial-generation outputs exactly."""
 
 import threading
 
+import pytest
 import torch
 
 from tensorlink_amd.engine.batcher import ContinuousBatcher
