@@ -667,3 +667,39 @@ def test_runner_paged_mode_matches_contiguous():
     ref = rc.generate(one, SamplingParams(max_new_tokens=12))
     out, _ = rp.generate_speculative(one, max_new_tokens=12)
     assert torch.equal(out, ref)
+
+
+def test_quantized_and_beams_matrix():
+    """Quantized runners serve through the batcher; beam search works
+    for every family (nb=1 == greedy)."""
+    import torch
+
+    from tensorlink_amd.engine.batcher import ContinuousBatcher
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+
+    for q in ("fp8-dense", "fp4-dense"):
+        r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                           device=torch.device("cpu"), seed=2, quantize=q)
+        b = ContinuousBatcher(r, max_slots=2, max_ctx=256,
+                              prefill_chunk=16).start()
+        try:
+            torch.manual_seed(5)
+            p = torch.randint(0, 1024, (40,))
+            out = b.submit(p.clone(), max_new_tokens=5).result(timeout=60)
+            ref = r.generate(p.unsqueeze(0),
+                             SamplingParams(max_new_tokens=5))
+            assert out == ref[0].tolist()
+        finally:
+            b.stop()
+
+    for name in ("tiny-moe", "tiny-qwen3-moe", "gpt2-small"):
+        r = PipelineRunner(plan_for_world(name, 1), 0, 1,
+                           device=torch.device("cpu"), seed=2)
+        torch.manual_seed(7)
+        ids = torch.randint(0, r.config.vocab_size, (1, 12))
+        g = r.generate(ids, SamplingParams(max_new_tokens=4))
+        assert torch.equal(r.generate_beam(ids, max_new_tokens=4,
+                                           num_beams=1), g)
+        assert r.generate_beam(ids, max_new_tokens=4,
+                               num_beams=3).shape == (1, 4)
