@@ -703,3 +703,34 @@ def test_quantized_and_beams_matrix():
                                            num_beams=1), g)
         assert r.generate_beam(ids, max_new_tokens=4,
                                num_beams=3).shape == (1, 4)
+
+
+def test_moe_repartition_checkpoint(tmp_path):
+    """Re-partition-aware stage checkpoint reassembly covers MoE
+    (expert keys remap through the layer-range sidecars)."""
+    import torch
+
+    from tensorlink_amd.models.configs import get_config
+    from tensorlink_amd.models.dense import build_full_model, build_stage
+    from tensorlink_amd.models.loader import (init_random_stage,
+                                              load_stage_from_stage_ckpt,
+                                              save_stage_to_safetensors)
+    from tensorlink_amd.parallel.planner import plan_for_world
+    cfg = get_config("tiny-moe")
+    plan = plan_for_world(cfg, 2)
+    d = str(tmp_path)
+    stages = []
+    for r in range(2):
+        st = build_stage(plan.config, plan.stage_for_rank(r))
+        init_random_stage(st, dtype=torch.float32, seed=r)
+        save_stage_to_safetensors(st, d, r)
+        stages.append(st)
+    full = build_full_model(cfg)
+    init_random_stage(full, dtype=torch.float32, seed=99)
+    assert load_stage_from_stage_ckpt(full, d) > 0
+    torch.manual_seed(3)
+    ids = torch.randint(0, 1024, (1, 10))
+    pos = torch.arange(10).unsqueeze(0)
+    h = stages[0](ids, pos, return_logits=False)
+    torch.testing.assert_close(full(ids, pos), stages[1](h, pos),
+                               atol=1e-5, rtol=1e-5)
