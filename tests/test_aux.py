@@ -427,3 +427,25 @@ def test_engine_config_serving_defaults(tmp_path):
     assert cfg.ml.continuous and cfg.ml.prefix_caching
     assert cfg.ml.prefill_chunk == 256 and cfg.ml.job_ttl_s == 60.0
     assert cfg.ml.trusted
+
+
+def test_fp8_moe_through_engine():
+    """BASELINE config #5 shape on CPU: fp8-expert MoE served through
+    the engine (continuous batching on) and the Qwen3-MoE family too."""
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny-moe", quantize="fp8", continuous=True,
+                   max_slots=2, max_ctx=256)
+    out = eng.generate({"hf_name": "tiny-moe", "message": "hello",
+                        "max_new_tokens": 4, "do_sample": False,
+                        "output_format": "simple"})
+    assert "response" in out and "error" not in out
+    eng.unload_model("tiny-moe")
+    eng.load_model("tiny-qwen3-moe", quantize="fp8")
+    out = eng.generate({"hf_name": "tiny-qwen3-moe", "message": "hi",
+                        "max_new_tokens": 4, "do_sample": False,
+                        "output_format": "simple"})
+    assert "response" in out and "error" not in out
+    eng.unload_model("tiny-qwen3-moe")
