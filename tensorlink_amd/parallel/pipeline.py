@@ -580,7 +580,9 @@ class PipelineTrainer:
         # chunked CE for big vocabularies (memory); plain CE for tiny
         # test configs (keeps exact parity with existing expectations)
         self._chunked_ce = (self.stage.has_head
-                            and plan.config.vocab_size >= 32000)
+                            and plan.config.vocab_size >= 32000
+                            and hasattr(self.stage, "norm")
+                            and hasattr(self.stage, "lm_head"))
         # (v0, group) when the LM head is vocab-sharded across a TP
         # group (set by TPPPTrainer(vocab_parallel=True))
         self._vp = None
