@@ -22,7 +22,9 @@ from typing import Dict, Iterable, List, Optional
 import torch
 import torch.nn as nn
 
-DEFAULT_TARGETS = ("qkv_proj", "o_proj", "gate_up_proj", "down_proj")
+# llama/qwen/mixtral projections + the GPT-2 Conv1D-derived names
+DEFAULT_TARGETS = ("qkv_proj", "o_proj", "gate_up_proj", "down_proj",
+                   "c_attn", "c_proj", "c_fc")
 
 
 class LoRALinear(nn.Module):

@@ -752,3 +752,15 @@ def test_training_all_families():
         ids = torch.randint(0, V, (2, 16))
         losses = [t.train_step(ids, labels=ids) for _ in range(4)]
         assert losses[-1] < losses[0], (name, losses)
+
+
+def test_lora_gpt2_defaults():
+    """Default LoRA targets cover the GPT-2 layer names too."""
+    import torch
+
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.models.lora import apply_lora
+    m = build_full_model(get_config("gpt2-small"))
+    init_random_stage(m, dtype=torch.float32, seed=5)
+    assert apply_lora(m, r=4) == 4 * len(m.layers)
