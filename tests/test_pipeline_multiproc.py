@@ -1224,3 +1224,50 @@ def test_lora_across_pipeline():
         assert losses[-1] < losses[0]
         assert frozen
         assert n_opt < 100_000          # adapters only (full stage ~1.5M)
+
+
+def _family_par_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.ep import EPRunner
+    from tensorlink_amd.parallel.pipeline import SamplingParams
+    from tensorlink_amd.parallel.tp import TPRunner
+    init_distributed(backend="gloo")
+    r = EPRunner("tiny-qwen3-moe", rank, world,
+                 device=torch.device("cpu"), seed=4)
+    torch.manual_seed(6)
+    ids = torch.randint(0, 1024, (1, 10))
+    out = r.generate(ids, SamplingParams(max_new_tokens=4))
+    r2 = TPRunner("tiny-qwen3", rank, world, device=torch.device("cpu"),
+                  seed=4)
+    out2 = r2.generate(ids, SamplingParams(max_new_tokens=4))
+    q.put((rank, (out.tolist(), out2.tolist())))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ep_tp_on_new_families():
+    """EP handles Qwen3-MoE's per-expert width; TP shards qk-norm heads
+    (Qwen3) — both reproduce single-rank outputs."""
+    outs = dict(_run_collect(_family_par_worker, 2, _port(51)))
+    assert outs[0] == outs[1]
+
+    import torch
+    from tensorlink_amd.models.configs import get_config
+    from tensorlink_amd.models.dense import build_full_model
+    from tensorlink_amd.models.loader import init_random_stage
+    for preset, got in (("tiny-qwen3-moe", outs[0][0]),
+                        ("tiny-qwen3", outs[0][1])):
+        m = build_full_model(get_config(preset))
+        init_random_stage(m, device="cpu", dtype=torch.float32, seed=4)
+        torch.manual_seed(6)
+        cur = torch.randint(0, 1024, (1, 10))
+        for _ in range(4):
+            pos = torch.arange(cur.shape[1]).unsqueeze(0).contiguous()
+            lg = m(cur, pos)
+            cur = torch.cat([cur, lg[:, -1].argmax(-1, keepdim=True)], 1)
+        assert cur[:, 10:].tolist() == got
