@@ -75,6 +75,26 @@ def cmd_bench(args, extra):
                                            "bench.py")] + extra)
 
 
+def cmd_export(args):
+    import torch
+
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import (init_random_stage,
+                                              load_stage_from_checkpoint,
+                                              save_hf_checkpoint)
+    cfg = get_config(args.model)
+    m = build_full_model(cfg)
+    if args.ckpt_dir:
+        n = load_stage_from_checkpoint(m, args.ckpt_dir,
+                                       dtype=torch.bfloat16)
+        print(f"loaded {n} tensors from {args.ckpt_dir}")
+    else:
+        init_random_stage(m, dtype=torch.bfloat16, seed=args.seed)
+        print(f"random init (seed {args.seed})")
+    out = save_hf_checkpoint(m, args.out)
+    print(f"exported {cfg.name} -> {out} (HF safetensors layout)")
+
+
 def cmd_status(args):
     if args.url:
         import json
@@ -134,6 +154,15 @@ def main():
                         "(VRAM bars, jobs, throughput)")
     st.add_argument("--url", default=None,
                     help="query a running server instead of local state")
+    ex = sub.add_parser("export", help="export a model to the HF "
+                        "safetensors layout (loadable by HF tooling)")
+    ex.add_argument("--model", required=True,
+                    help="preset name or checkpoint dir")
+    ex.add_argument("--out", required=True)
+    ex.add_argument("--ckpt-dir", default=None,
+                    help="load weights from this HF-layout dir first "
+                         "(default: seeded random init)")
+    ex.add_argument("--seed", type=int, default=0)
 
     args, extra = p.parse_known_args()
     if args.cmd == "serve":
@@ -146,6 +175,8 @@ def main():
         cmd_health(args)
     elif args.cmd == "status":
         cmd_status(args)
+    elif args.cmd == "export":
+        cmd_export(args)
 
 
 if __name__ == "__main__":
