@@ -157,6 +157,15 @@ class InferenceEngine:
         lookup_n = kwargs.pop("lookup_n", 3)
         spec_k = kwargs.pop("spec_k", 8)
         plan = plan_for_world(config, self.world, **kwargs)
+        import os as _os
+        if init == "random" and ckpt_dir is None and _os.path.isdir(name) \
+                and (_os.path.exists(_os.path.join(
+                    name, "model.safetensors"))
+                     or _os.path.exists(_os.path.join(
+                         name, "model.safetensors.index.json"))):
+            # serving a checkpoint DIRECTORY: load its weights, don't
+            # random-init over them
+            init, ckpt_dir = "checkpoint", name
         runner = PipelineRunner(plan, self.rank, self.world,
                                 device=self.device, init=init,
                                 ckpt_dir=ckpt_dir, quantize=quantize)

@@ -449,3 +449,33 @@ def test_fp8_moe_through_engine():
                         "output_format": "simple"})
     assert "response" in out and "error" not in out
     eng.unload_model("tiny-qwen3-moe")
+
+
+def test_serve_checkpoint_directory(tmp_path):
+    """load_model on a checkpoint DIRECTORY serves its weights (not a
+    fresh random init): outputs match the exported source model."""
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import (init_random_stage,
+                                              save_hf_checkpoint)
+    from tensorlink_amd.parallel.pipeline import SamplingParams
+
+    src = build_full_model(get_config("tiny"))
+    init_random_stage(src, dtype=torch.float32, seed=77)
+    d = str(tmp_path / "m")
+    save_hf_checkpoint(src, d)
+
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    job = eng.load_model(d)
+    torch.manual_seed(5)
+    ids = torch.randint(0, 1024, (1, 8))
+    out = job.runner.generate(ids, SamplingParams(max_new_tokens=4))
+    cur = ids
+    for _ in range(4):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).contiguous()
+        lg = src(cur, pos)
+        cur = torch.cat([cur, lg[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur[:, 8:], out)
+    eng.unload_model(d)
