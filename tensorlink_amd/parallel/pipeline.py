@@ -74,6 +74,17 @@ class PipelineRunner:
                                else torch.float32)
         self.spec = plan.stage_for_rank(rank)
         self.stage: StageModel = build_stage(self.config, self.spec)
+        import os as _os
+        if init == "random" and ckpt_dir is None \
+                and isinstance(plan.model, str) \
+                and _os.path.isdir(plan.model) and (
+                    _os.path.exists(_os.path.join(
+                        plan.model, "model.safetensors"))
+                    or _os.path.exists(_os.path.join(
+                        plan.model, "model.safetensors.index.json"))):
+            # the model "name" IS a checkpoint directory: serve its
+            # weights instead of a fresh random init
+            init, ckpt_dir = "checkpoint", plan.model
         if init == "random":
             init_random_stage(self.stage, device=self.device,
                               dtype=self.dtype, seed=seed + rank)
