@@ -140,7 +140,12 @@ class ModelParser:
                 stages = self._plan_pp(config, pp, caps[:pp], batch_size,
                                        seq_len, training, dtype, micro_batches)
                 return StagePlan(
-                    model=config.name, config=config, stages=stages,
+                    # keep the CALLER's model string (it may be a
+                    # checkpoint directory the runner should load from;
+                    # config.name round-trips the original preset name)
+                    model=(model if isinstance(model, str)
+                           else config.name),
+                    config=config, stages=stages,
                     dtype=dtype, training=training, micro_batches=micro_batches,
                     tie_word_embeddings=config.tie_word_embeddings)
             except AssignmentError as e:

@@ -151,3 +151,27 @@ def test_distributed_model_lora():
         l1 = m.train_step(ids, ids)
     assert l1 < l0
     assert m._trainer.optimizer.flat_param.numel() < 100_000
+
+
+def test_distributed_model_from_checkpoint_dir(tmp_path):
+    """DistributedModel('<export dir>') serves the exported weights."""
+    import torch
+
+    from tensorlink_amd.models import build_full_model, get_config
+    from tensorlink_amd.models.loader import (init_random_stage,
+                                              save_hf_checkpoint)
+    from tensorlink_amd.module import DistributedModel
+    src = build_full_model(get_config("tiny"))
+    init_random_stage(src, dtype=torch.float32, seed=77)
+    d = str(tmp_path / "m")
+    save_hf_checkpoint(src, d)
+    m = DistributedModel(d, device=torch.device("cpu"))
+    torch.manual_seed(5)
+    ids = torch.randint(0, 1024, (1, 8))
+    out = m.generate(ids, max_new_tokens=4)
+    cur = ids
+    for _ in range(4):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0).contiguous()
+        lg = src(cur, pos)
+        cur = torch.cat([cur, lg[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(cur, out)
