@@ -125,9 +125,12 @@ class ContinuousBatcher:
         cache_cls = (PrefixCachingKVCache if prefix_caching
                      else DynamicPagedKVCache)
         self.prefix_caching = prefix_caching
-        # prompt-lookup speculation when exactly ONE slot is decoding
-        # greedily (the ragged multi-token verify is per-slot; batching
-        # it needs per-row q_off — roadmap). Exact greedy either way.
+        # prompt-lookup speculation: eligible greedy slots contribute
+        # 1+m ragged rows to the decode batch (rows share the slot's page
+        # table with seq_lens len..len+m), so the verify runs through the
+        # DECODE kernels — bitwise-identical to m+1 sequential decode
+        # steps by construction, and composable with sampled slots in the
+        # same batch. Exact greedy either way.
         self.speculative = speculative
         self.lookup_n = lookup_n
         self.spec_k = spec_k
@@ -356,92 +359,106 @@ class ContinuousBatcher:
             self.slot_prompt[slot] = None
             self.cache.release_slot(slot)
 
+    def _spec_eligible(self, slot: int) -> bool:
+        """Greedy-only slots with no sampling state speculate; penalties,
+        logprobs and sampled slots take the plain row (a greedy request
+        with penalties would otherwise silently lose them in verify)."""
+        req = self.slots[slot]
+        return (self.speculative and self.slot_ctx[slot] is not None
+                and req.temperature <= 0
+                and req.presence_penalty == 0.0
+                and req.frequency_penalty == 0.0
+                and not req.logprobs)
+
     @torch.no_grad()
     def _decode_step(self):
+        """One ragged decode batch over all running slots. A speculative
+        slot contributes 1+m rows (its last token plus m prompt-lookup
+        proposals) with per-row seq_lens len..len+m over the SAME page
+        table; rope_append writes each row's K/V before attention reads
+        them, so row j attends to keys 0..len+j exactly as m+1 serial
+        decode steps would — same kernels, same bits. Non-speculative
+        slots are single rows sampled with their penalties/seed."""
+        from tensorlink_amd.parallel.pipeline import PipelineRunner
         active = [i for i, s in enumerate(self.slots)
                   if s is not None and not self._prefilling(i)
                   and self.slot_filled[i] > 0]
         if not active:
             return False
-        reqs = [self.slots[i] for i in active]
+        # prompt-lookup proposals for eligible slots
+        props = {}
+        for i in active:
+            if self._spec_eligible(i):
+                req = self.slots[i]
+                room = min(req.max_new_tokens - self.slot_emitted[i],
+                           self.max_ctx - 1 - self.slot_len[i]) - 1
+                p = PipelineRunner._lookup_propose(self.slot_ctx[i],
+                                                   self.lookup_n,
+                                                   self.spec_k)
+                props[i] = p[:max(0, room)]
         for i in list(active):
             if self.slots[i] is None:      # preempted below this loop
                 continue
+            m = len(props.get(i, ()))
             try:
-                self.cache.ensure(i, self.slot_len[i] + 2)
+                self.cache.ensure(i, self.slot_len[i] + m + 2)
             except RuntimeError:
                 # pool exhausted mid-decode: park another slot and retry
                 if not self._preempt_one(exclude=i):
                     raise
-                self.cache.ensure(i, self.slot_len[i] + 2)
+                self.cache.ensure(i, self.slot_len[i] + m + 2)
         active = [j for j in active if self.slots[j] is not None]
         if not active:
             return False
-        reqs = [self.slots[j] for j in active]
-        toks = torch.tensor([self.slot_last[i] for i in active],
-                            device=self.device, dtype=torch.int64)
-        lens = torch.tensor([self.slot_len[i] for i in active],
-                            device=self.device, dtype=torch.int32)
-        view = _SlotView(self.cache, active, lens)
+        rows_tok, rows_len, rows_slot, seg = [], [], [], {}
+        for i in active:
+            prop = list(props.get(i, ()))
+            seg[i] = (len(rows_tok), len(prop))
+            rows_tok += [self.slot_last[i]] + prop
+            rows_len += [self.slot_len[i] + j
+                         for j in range(len(prop) + 1)]
+            rows_slot += [i] * (len(prop) + 1)
+        toks = torch.tensor(rows_tok, device=self.device, dtype=torch.int64)
+        lens = torch.tensor(rows_len, device=self.device, dtype=torch.int32)
+        view = _SlotView(self.cache, rows_slot, lens)
         pos = lens.unsqueeze(1)
         logits = self.stage(toks.unsqueeze(1), pos, kv_cache=view).squeeze(1)
-        new = self._sample(logits, reqs, slots=active)
-        for j, slot in enumerate(active):
-            self.slot_len[slot] += 1
-            self.slot_emitted[slot] += 1
-            self.slot_counts[slot][new[j]] = \
-                self.slot_counts[slot].get(new[j], 0) + 1
-            self.slots[slot].tokens.put(new[j])
-            self.slot_last[slot] = new[j]
-            if self.slot_ctx[slot] is not None:
-                self.slot_ctx[slot].append(new[j])
-            self._maybe_finish(slot, new[j])
-        self.steps += 1
-        return True
-
-    @torch.no_grad()
-    def _spec_step(self, slot: int) -> bool:
-        """One speculative round for the single decoding slot: propose
-        k tokens via prompt lookup, verify them all in ONE forward
-        (ragged q_off attention), accept the matching prefix + bonus.
-        Greedy-exact; emits 1..k+1 tokens per forward."""
-        from tensorlink_amd.parallel.pipeline import PipelineRunner
-        req = self.slots[slot]
-        ctx = self.slot_ctx[slot]
-        proposal = PipelineRunner._lookup_propose(ctx, self.lookup_n,
-                                                  self.spec_k)
-        # bound by remaining budget/context
-        room = min(req.max_new_tokens - self.slot_emitted[slot],
-                   self.max_ctx - 1 - self.slot_len[slot]) - 1
-        proposal = proposal[:max(0, room)]
-        m = len(proposal)
-        self.cache.ensure(slot, self.slot_len[slot] + m + 2)
-        feed = torch.tensor([[self.slot_last[slot]] + proposal],
-                            device=self.device, dtype=torch.int64)
-        pos = (self.slot_len[slot] + torch.arange(
-            m + 1, device=self.device, dtype=torch.int32)).unsqueeze(0)
-        lens = torch.tensor([self.slot_len[slot]], device=self.device,
-                            dtype=torch.int32)
-        view = _SlotView(self.cache, [slot], lens)
-        logits = self.stage(feed, pos, kv_cache=view)      # [1, m+1, V]
-        targets = logits.argmax(-1)[0].tolist()
-        a = 0
-        while a < m and proposal[a] == targets[a]:
-            a += 1
-        accepted = proposal[:a] + [targets[a]]
-        self.spec_accepted += a
-        # kv valid for the fed token + a accepted proposals
-        self.slot_len[slot] += 1 + a
-        for tok in accepted:
-            self.slot_emitted[slot] += 1
-            self.slot_counts[slot][tok] = \
-                self.slot_counts[slot].get(tok, 0) + 1
-            req.tokens.put(tok)
-            ctx.append(tok)
-            self.slot_last[slot] = tok
-            self._maybe_finish(slot, tok)
-            if self.slots[slot] is None:
-                break
+        for i in active:
+            off, m = seg[i]
+            req = self.slots[i]
+            if i in props:
+                # greedy accept: matching prefix + bonus token
+                targets = logits[off:off + m + 1].argmax(-1).tolist()
+                prop = props[i]
+                a = 0
+                while a < m and prop[a] == targets[a]:
+                    a += 1
+                accepted = prop[:a] + [targets[a]]
+                self.spec_accepted += a
+                # kv valid for the fed token + a accepted proposals
+                self.slot_len[i] += 1 + a
+                for tok in accepted:
+                    self.slot_emitted[i] += 1
+                    self.slot_counts[i][tok] = \
+                        self.slot_counts[i].get(tok, 0) + 1
+                    req.tokens.put(tok)
+                    self.slot_ctx[i].append(tok)
+                    self.slot_last[i] = tok
+                    self._maybe_finish(i, tok)
+                    if self.slots[i] is None:
+                        break
+            else:
+                tok = self._sample(logits[off:off + 1], [req],
+                                   slots=[i])[0]
+                self.slot_len[i] += 1
+                self.slot_emitted[i] += 1
+                self.slot_counts[i][tok] = \
+                    self.slot_counts[i].get(tok, 0) + 1
+                req.tokens.put(tok)
+                self.slot_last[i] = tok
+                if self.slot_ctx[i] is not None:
+                    self.slot_ctx[i].append(tok)
+                self._maybe_finish(i, tok)
         self.steps += 1
         return True
 
@@ -455,8 +472,16 @@ class ContinuousBatcher:
                 slot = self._free_slot()
                 if slot is None:
                     break
-                # admission control: need pages for prompt + first tokens
-                head = self._queue.queue[0][2]
+                # admission control: POP the head first (a concurrent
+                # higher-priority submit could otherwise slip between a
+                # peek and the get with an unchecked page need), check its
+                # pages, and re-queue it unchanged if they don't fit —
+                # (priority, seq) ordering puts it back in the same place
+                try:
+                    entry = self._queue.get_nowait()
+                except queue.Empty:
+                    break
+                head = entry[2]
                 # pages for the prompt plus its first generated token,
                 # minus any published prefix it can adopt
                 need = (head.input_ids.numel() + PAGE) // PAGE
@@ -465,6 +490,7 @@ class ContinuousBatcher:
                         head.input_ids.tolist()) // PAGE
                 if self.cache.available_pages() \
                         < need + self._committed_pages():
+                    self._queue.put(entry)
                     # page starvation: swap out a young running slot so
                     # the queue keeps moving (at most one per iteration,
                     # never below two live slots, and not while others
@@ -476,7 +502,7 @@ class ContinuousBatcher:
                         starved = 1
                         continue
                     break
-                req = self._queue.get_nowait()[2]
+                req = head
                 if req.cancelled:
                     req.tokens.put(None)
                     req.done.set()
@@ -521,18 +547,7 @@ class ContinuousBatcher:
                         self.slot_prompt[slot] = None
                         self.cache.release_slot(slot)
                     break
-            spec_slot = None
-            if self.speculative:
-                decoding = [i for i, r in enumerate(self.slots)
-                            if r is not None and not self._prefilling(i)
-                            and self.slot_filled[i] > 0]
-                if (len(decoding) == 1
-                        and self.slots[decoding[0]].temperature <= 0
-                        and self.slot_ctx[decoding[0]] is not None):
-                    spec_slot = decoding[0]
-            if spec_slot is not None:
-                busy = self._spec_step(spec_slot)
-            elif self.tracer is not None:
+            if self.tracer is not None:
                 with self.tracer.span("decode_step",
                                       slots=sum(x is not None
                                                 for x in self.slots)):
@@ -718,7 +733,12 @@ class PPContinuousBatcher(ContinuousBatcher):
                 slot = self._free_slot()
                 if slot is None:
                     break
-                head = self._queue.queue[0][2]
+                # pop-check-requeue (see ContinuousBatcher._loop)
+                try:
+                    entry = self._queue.get_nowait()
+                except queue.Empty:
+                    break
+                head = entry[2]
                 ids_list = head.input_ids.tolist()
                 # pages for the prompt plus its first generated token,
                 # minus any published prefix it can adopt
@@ -727,8 +747,9 @@ class PPContinuousBatcher(ContinuousBatcher):
                            if self.prefix_caching else 0)
                 if self.cache.available_pages() < (need - matched // PAGE
                                                    + self._committed_pages()):
+                    self._queue.put(entry)
                     break
-                req = self._queue.get_nowait()[2]
+                req = head
                 if req.cancelled:
                     req.tokens.put(None)
                     req.done.set()

@@ -280,8 +280,14 @@ class InferenceEngine:
         args = normalize_generate_args(request, job.tokenizer)
         ids = job.tokenizer(prompt, return_tensors="pt")["input_ids"]
         max_ctx = job.runner.config.max_position_embeddings
+        # clamp max_new_tokens first (formatter caps can exceed a small
+        # model's context, e.g. GPT-2 1024 vs the 2048 default) and keep
+        # at least one prompt token — as batcher.submit does
+        args["max_new_tokens"] = max(1, min(args["max_new_tokens"],
+                                            max_ctx - 1))
         if ids.shape[1] + args["max_new_tokens"] > max_ctx:
-            ids = ids[:, -(max_ctx - args["max_new_tokens"]):]
+            keep = max(1, max_ctx - args["max_new_tokens"])
+            ids = ids[:, -keep:]
         sp = SamplingParams(
             temperature=args["temperature"] if args["do_sample"] else 0.0,
             top_p=args["top_p"], top_k=int(args.get("top_k", 0) or 0),

@@ -178,7 +178,13 @@ class Attention(nn.Module):
             # prefill with cache write; rotated k / raw v are read back from
             # the cache. off > 0 = chunked prefill continuing an existing
             # sequence (the chunk's queries sit at global rows off..off+S-1
-            # and attend to all off+S cached keys).
+            # and attend to all off+S cached keys). A single q_off serves
+            # the whole view, so cached lengths must agree per row (ragged
+            # multi-token work goes through the decode path instead).
+            if B > 1:
+                assert int(kv_cache.seq_lens.min()) == \
+                    int(kv_cache.seq_lens.max()), \
+                    "ragged cached prefill needs per-row q_off"
             off = int(kv_cache.seq_lens.max())
             kv_len = off + S
             if table is None:

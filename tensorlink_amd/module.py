@@ -285,11 +285,9 @@ class DistributedModel(nn.Module):
         if labels is not None:
             from tensorlink_amd import ops
             loss = ops.causal_lm_loss(logits, labels.to(logits.device))
-            logits = logits if isinstance(logits, torch.Tensor) else logits
-            try:
-                logits.loss = loss
-            except Exception:
-                pass
+            # torch.Tensor supports instance attributes; fail loudly if a
+            # subclass ever forbids it rather than silently dropping loss
+            logits.loss = loss
         return logits
 
     def _pipeline_backward(self, grad_logits):

@@ -240,7 +240,9 @@ class _VocabParallelCE(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, logits, labels, v0, group):
-        # logits [N, Vl] fp32; labels [N] global ids
+        # logits [N, Vl] fp32; labels [N] global ids (-100 = ignored,
+        # matching the non-vp chunked/plain CE paths)
+        valid = labels != -100
         lmax = logits.max(-1).values.contiguous()
         dist.all_reduce(lmax, op=dist.ReduceOp.MAX, group=group)
         z = (logits - lmax.unsqueeze(1)).exp()
@@ -253,20 +255,21 @@ class _VocabParallelCE(torch.autograd.Function):
                - lmax) * in_shard
         tgt = tgt.contiguous()
         dist.all_reduce(tgt, group=group)
-        loss = denom.log() - tgt
-        ctx.save_for_backward(z, denom, idx, in_shard)
+        loss = (denom.log() - tgt) * valid
+        n_valid = valid.sum().clamp(min=1)
+        ctx.save_for_backward(z, denom, idx, in_shard, valid, n_valid)
         ctx.group = group
-        return loss.mean()
+        return loss.sum() / n_valid
 
     @staticmethod
     def backward(ctx, dloss):
-        z, denom, idx, in_shard = ctx.saved_tensors
-        N = z.shape[0]
+        z, denom, idx, in_shard, valid, n_valid = ctx.saved_tensors
         dlogits = z / denom.unsqueeze(1)
         dlogits.scatter_add_(
             1, idx.unsqueeze(1),
             -in_shard.to(dlogits.dtype).unsqueeze(1))
-        dlogits.mul_(dloss / N)
+        dlogits.mul_(valid.to(dlogits.dtype).unsqueeze(1))
+        dlogits.mul_(dloss / n_valid)
         return dlogits, None, None, None
 
 
