@@ -281,37 +281,43 @@ def attention_decode(q, k_cache, v_cache, seq_lens,
 
 
 # ---------------------------------------------------------------------------
-# Linear (skinny-M decode GEMM)
+# Linear (hand-written serving GEMM family)
 # ---------------------------------------------------------------------------
-def _use_skinny(M: int, N: int, K: int) -> bool:
-    """Measured dispatch policy (profiles/skinny_gemm_ab.md, MI355X):
-    the hand-written glds streaming GEMM beats hipBLASLt at small M
-    (M=16: 2.50 vs 1.33 TB/s) and on deep-K narrow-N shapes (down-proj
-    256x3584x18944: 72.7 vs 102 µs); hipBLASLt's tuned stream-K wins the
-    wide-N shapes (gate/up at 2.95 TB/s)."""
-    if K % 32 != 0 or N % 64 != 0 or M > 256:
-        return False
-    if M <= 64:
-        return True
-    # deep-K won the isolated microbench (down-proj 72.7 vs 102 µs) but
-    # measured ~4% SLOWER decode in the real pipeline (L2 state differs);
-    # keep it off at large M until in-pipeline numbers say otherwise.
-    return False
+import os as _os
+
+# Inference GEMMs at M <= this route to the deterministic MFMA family
+# (streaming kernel at M<=64, tiled all-glds kernel above). The family
+# guarantees bitwise M-INDEPENDENT per-row results (same K chunk order,
+# same (N,K)-only split policy), which is what makes chunked prefill /
+# speculative verify / ragged batch decode / plain generate emit
+# identical greedy tokens. Above the threshold (large prefill) hipBLASLt
+# wins on throughput and exact cross-path equality is not claimed.
+GEMM_M_MAX = int(_os.environ.get("TL_GEMM_M_MAX", "512"))
+
+# incremented whenever the hand-written family handles a linear — GPU
+# tests assert on it so a silent hipBLASLt fallback cannot pass as
+# kernel coverage
+gemm_dispatch_count = 0
+
+
+def _use_tl_gemm(M: int, N: int, K: int) -> bool:
+    return K % 32 == 0 and N % 64 == 0 and M <= GEMM_M_MAX
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor,
            bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """x @ W^T + bias. Routes decode shapes where it wins to the
-    hand-written MFMA streaming kernel; everything else (prefill M,
-    training) to torch/hipBLASLt."""
-    import os as _os
+    """x @ W^T + bias. Serving-M shapes route to the hand-written MFMA
+    family (skinny_gemm.hip / gemm_tiled.hip); large prefill M and
+    training go to torch/hipBLASLt."""
+    global gemm_dispatch_count
     if (x.is_cuda and not torch.is_grad_enabled()
             and x.dtype == torch.bfloat16
             and not _os.environ.get("TL_NO_SKINNY")):
         K = x.shape[-1]
         M = x.numel() // K
         N = weight.shape[0]
-        if _use_skinny(M, N, K):
+        if _use_tl_gemm(M, N, K):
+            gemm_dispatch_count += 1
             return _require_ext().skinny_gemm(
                 x.contiguous(), weight,
                 bias.to(torch.bfloat16) if bias is not None else None)

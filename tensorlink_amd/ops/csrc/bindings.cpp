@@ -42,16 +42,27 @@ void tl_decode_attn_mfma(const void* q, const void* k_cache,
                          const void* block_table, void* out,
                          void* partial, void* partial_ml, int B, int Hq,
                          int Hkv, int Smax, int D, float scale, int n_split,
-                         int bt_stride, hipStream_t stream);
+                         int bt_stride, int per_row, hipStream_t stream);
 void tl_skinny_gemm(const void* x, const void* w, const void* bias,
                     void* out, void* partial, int M, int N, int K,
                     int n_split, hipStream_t stream);
+void tl_gemm_tiled(const void* x, const void* w, const void* bias, void* out,
+                   void* partial, int M, int N, int K, int n_split,
+                   hipStream_t stream);
 }
 
 namespace {
 
 hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
+}
+
+// host copy of tl_split_for_len (common.hpp — keep in sync; this file is
+// compiled by plain g++ so it cannot include the device header)
+int tl_split_for_len(int L) {
+  int ns = 1;
+  while (ns < 16 && L > 512 * ns) ns <<= 1;
+  return ns;
 }
 
 #define CHECK_IN(t, d)                                              \
@@ -186,11 +197,15 @@ Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
                    (float)scale, cur_stream());
     return out;
   }
+  int per_row = 0;
   if (n_split <= 0) {
-    // flash-decode split heuristic: target >= ~512 blocks (2 per CU)
-    int ns = 1;
-    while (B * Hkv * ns < 512 && ns < 16) ns <<= 1;
-    n_split = ns;
+    // auto: per-ROW split from each row's own length (tl_split_for_len,
+    // common.hpp) so a sequence's numerics are batch-independent; the
+    // launch z-width comes from a batch-shape-free host bound on L
+    // (cache capacity), no device sync needed.
+    per_row = 1;
+    const int upper = bt ? bt_stride * 128 : Smax;
+    n_split = tl_split_for_len(upper);
   }
   TORCH_CHECK(n_split <= 64, "n_split too large");
   Tensor partial, partial_ml;
@@ -206,7 +221,7 @@ Tensor decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, Tensor seq_lens,
   tl_decode_attn_mfma(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
                       seq_lens.data_ptr(), bt, out.data_ptr(), pp, pml, B,
                       Hq, Hkv, Smax, D, (float)scale, (int)n_split,
-                      bt_stride, cur_stream());
+                      bt_stride, per_row, cur_stream());
   return out;
 }
 
@@ -278,6 +293,19 @@ Tensor swiglu_fused(Tensor gu) {
   return out;
 }
 
+// Split-K policy SHARED by the streaming (M<=64) and tiled (M>64) GEMM
+// kernels. It depends only on (N, K) — never M — so a row's accumulation
+// (split boundaries + s-ordered reduce) is bitwise identical at every M:
+// the serving engine's exact-greedy guarantee rests on this.
+static int gemm_n_split(int N, int K) {
+  // split K so (N/64)*n_split lands near 1 block/CU (guide: ~0.5-1x CUs)
+  int n_split = 1;
+  while ((N / 64) * n_split < 256 && n_split < 8 &&
+         (K / 32) / (n_split * 2) >= 2)
+    n_split <<= 1;
+  return n_split;
+}
+
 Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
   CHECK_IN(x, torch::kBFloat16);
   CHECK_IN(w, torch::kBFloat16);
@@ -286,7 +314,6 @@ Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
   const int N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "K mismatch");
   TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "unsupported shape");
-  TORCH_CHECK(M <= 256, "skinny_gemm supports M <= 256");
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto out = torch::empty(sizes, x.options());
@@ -295,11 +322,7 @@ Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
     CHECK_IN(bias.value(), torch::kBFloat16);
     bp = bias->data_ptr();
   }
-  // split K so (N/64)*n_split lands near 1 block/CU (guide: ~0.5-1x CUs)
-  int n_split = 1;
-  while ((N / 64) * n_split < 256 && n_split < 8 &&
-         (K / 32) / (n_split * 2) >= 2)
-    n_split <<= 1;
+  const int n_split = gemm_n_split(N, K);
   Tensor partial;
   void* pp = nullptr;
   if (n_split > 1) {
@@ -307,8 +330,12 @@ Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
                            x.options().dtype(torch::kFloat));
     pp = partial.data_ptr();
   }
-  tl_skinny_gemm(x.data_ptr(), w.data_ptr(), bp, out.data_ptr(), pp, (int)M,
-                 N, K, n_split, cur_stream());
+  if (M <= 64)
+    tl_skinny_gemm(x.data_ptr(), w.data_ptr(), bp, out.data_ptr(), pp,
+                   (int)M, N, K, n_split, cur_stream());
+  else
+    tl_gemm_tiled(x.data_ptr(), w.data_ptr(), bp, out.data_ptr(), pp,
+                  (int)M, N, K, n_split, cur_stream());
   return out;
 }
 

@@ -69,6 +69,17 @@ DEVINLINE float block_reduce_sum(float x, float* lds) {
 }
 
 __host__ __device__ inline int cdiv(int a, int b) { return (a + b - 1) / b; }
+
+// Flash-decode split count for ONE row of length L. Depends only on L —
+// never on batch size or launch shape — so a sequence's attention
+// numerics are identical whatever batch it is decoded in (the serving
+// engine's exact-greedy guarantee; see decode_attn_mfma.hip). Monotone
+// in L, so a host-side length upper bound gives a valid launch width.
+__host__ __device__ inline int tl_split_for_len(int L) {
+  int ns = 1;
+  while (ns < 16 && L > 512 * ns) ns <<= 1;
+  return ns;
+}
 __host__ __device__ inline int64_t cdiv64(int64_t a, int64_t b) {
   return (a + b - 1) / b;
 }

@@ -43,7 +43,8 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
     bf16* __restrict__ out,            // [B, Hq, D]
     float* __restrict__ partial,       // [B, Hq, SPLIT, D]
     float* __restrict__ partial_ml,    // [B, Hq, SPLIT, 2]
-    int Hq, int Hkv, int Smax, float scale, int n_split, int bt_stride) {
+    int Hq, int Hkv, int Smax, float scale, int n_split, int bt_stride,
+    int per_row) {
   constexpr int KCH = D / 32;        // MFMA k-chunks for QK^T
   constexpr int NS = D / 16;         // PV output col tiles
   const int b = blockIdx.x;
@@ -53,8 +54,15 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
   const int h0 = hkv * G;
   const int L = seq_lens[b];
 
+  // per_row: this ROW's split count comes from its own length (batch
+  // composition must not change a row's accumulation — exact-greedy
+  // guarantee); the launch z-width n_split is a host-side upper bound
+  // and surplus blocks exit.
+  const int ns_row = per_row ? tl_split_for_len(L) : n_split;
+  if (split >= ns_row) return;
+
   // this split's key range
-  const int per_split = (L + n_split - 1) / n_split;
+  const int per_split = (L + ns_row - 1) / ns_row;
   const int k_begin = split * per_split;
   const int k_end = min(L, k_begin + per_split);
 
@@ -252,23 +260,28 @@ __global__ __launch_bounds__(BLOCK) void decode_attn_mfma_kernel(
   }
 }
 
-// combine kernel: merge SPLIT partials per (b, h)
+// combine kernel: merge this row's ns_row partials per (b, h). A row
+// whose ns_row == 1 reduces to f = exp(0) = 1 scales — bitwise equal to
+// the direct-write path.
 template <int D>
 __global__ __launch_bounds__(128) void decode_attn_combine_kernel(
     const float* __restrict__ partial, const float* __restrict__ partial_ml,
-    bf16* __restrict__ out, int Hq, int n_split) {
+    const int* __restrict__ seq_lens, bf16* __restrict__ out, int Hq,
+    int n_split, int per_row) {
   const int bh = blockIdx.x;                    // b * Hq + h
   const int d = threadIdx.x;                    // [0, D)
+  const int ns_row =
+      per_row ? tl_split_for_len(seq_lens[bh / Hq]) : n_split;
   __shared__ float sm[64], sl[64];
-  if (threadIdx.x < n_split) {
+  if (threadIdx.x < ns_row) {
     sm[threadIdx.x] = partial_ml[((int64_t)bh * n_split + threadIdx.x) * 2];
     sl[threadIdx.x] = partial_ml[((int64_t)bh * n_split + threadIdx.x) * 2 + 1];
   }
   __syncthreads();
   float m_tot = -1e30f;
-  for (int s = 0; s < n_split; ++s) m_tot = fmaxf(m_tot, sm[s]);
+  for (int s = 0; s < ns_row; ++s) m_tot = fmaxf(m_tot, sm[s]);
   float l_tot = 0.f, o = 0.f;
-  for (int s = 0; s < n_split; ++s) {
+  for (int s = 0; s < ns_row; ++s) {
     const float f = __expf(sm[s] - m_tot);
     l_tot += sl[s] * f;
     o += partial[((int64_t)bh * n_split + s) * D + d] * f;
@@ -287,7 +300,7 @@ void tl_decode_attn_mfma(const void* q, const void* k_cache,
                          const void* block_table, void* out,
                          void* partial, void* partial_ml, int B, int Hq,
                          int Hkv, int Smax, int D, float scale, int n_split,
-                         int bt_stride, hipStream_t stream) {
+                         int bt_stride, int per_row, hipStream_t stream) {
   dim3 grid(B, Hkv, n_split), block(BLOCK);
 #define LAUNCH(DD, SM, PG)                                                   \
   hipLaunchKernelGGL((decode_attn_mfma_kernel<DD, SM, PG>), grid, block, 0,  \
@@ -295,7 +308,7 @@ void tl_decode_attn_mfma(const void* q, const void* k_cache,
                      (const bf16*)v_cache, (const int*)seq_lens,             \
                      (const int*)block_table, (bf16*)out,                    \
                      (float*)partial, (float*)partial_ml, Hq, Hkv, Smax,     \
-                     scale, n_split, bt_stride)
+                     scale, n_split, bt_stride, per_row)
 #define PICK(DD)                                                             \
   do {                                                                       \
     if (block_table) {                                                       \
@@ -313,11 +326,13 @@ void tl_decode_attn_mfma(const void* q, const void* k_cache,
     if (D == 128)
       hipLaunchKernelGGL((decode_attn_combine_kernel<128>), cgrid, cblock, 0,
                          stream, (const float*)partial,
-                         (const float*)partial_ml, (bf16*)out, Hq, n_split);
+                         (const float*)partial_ml, (const int*)seq_lens,
+                         (bf16*)out, Hq, n_split, per_row);
     else
       hipLaunchKernelGGL((decode_attn_combine_kernel<64>), cgrid, cblock, 0,
                          stream, (const float*)partial,
-                         (const float*)partial_ml, (bf16*)out, Hq, n_split);
+                         (const float*)partial_ml, (const int*)seq_lens,
+                         (bf16*)out, Hq, n_split, per_row);
   }
 }
 

@@ -277,20 +277,68 @@ def test_decode_attn_mfma(B, Hq, Hkv, D, L, ns):
 
 
 @pytest.mark.parametrize("M,N,K,bias", [
-    (256, 3584, 3584, False),     # o_proj
-    (256, 4608, 3584, True),      # fused-qkv shape w/ bias
-    (256, 18944, 3584, False),    # gate/up
-    (256, 3584, 18944, False),    # down
+    (256, 3584, 3584, False),     # o_proj (tiled kernel)
+    (256, 4608, 3584, True),      # fused-qkv shape w/ bias (tiled)
+    (256, 18944, 3584, False),    # gate/up (tiled)
+    (256, 3584, 18944, False),    # down (tiled)
+    (150, 1024, 256, False),      # mid-M, MT=16 partial tail
+    (96, 1024, 256, True),        # MT=8 template
+    (300, 1024, 256, False),      # blockIdx.z m-tiling (2 blocks)
+    (512, 4608, 3584, True),      # dispatch ceiling, real shape
+    (64, 1024, 256, True),        # streaming-kernel boundary
+    (16, 3584, 3584, False),      # small-batch decode (streaming)
     (7, 1024, 256, True),         # tiny M, M tail
     (33, 512, 96, False),         # K%64==32 tail path
 ])
-def test_skinny_gemm(M, N, K, bias):
+def test_tl_gemm(M, N, K, bias):
     ops = _ext()
     torch.manual_seed(12)
     x = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
     w = torch.randn(N, K, device=DEV, dtype=torch.bfloat16) / (K ** 0.5)
     b = torch.randn(N, device=DEV, dtype=torch.bfloat16) if bias else None
+    before = ops.gemm_dispatch_count
     out = ops.linear(x, w, b)
+    # every parametrization must actually hit the hand-written family
+    assert ops.gemm_dispatch_count == before + 1
     ref = torch.nn.functional.linear(x.float(), w.float(),
                                      b.float() if bias else None)
     torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_decode_attn_batch_independence():
+    """A row's decode attention is bitwise identical whatever batch it is
+    in (per-row flash-decode split from its own length — common.hpp
+    tl_split_for_len). Mixed short/long rows vs each row alone."""
+    ops = _ext()
+    torch.manual_seed(14)
+    B, Hq, Hkv, D, Smax = 5, 8, 4, 128, 1536
+    q = torch.randn(B, Hq, D, device=DEV, dtype=torch.bfloat16)
+    kc = torch.randn(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
+    vc = torch.randn(B, Hkv, Smax, D, device=DEV, dtype=torch.bfloat16)
+    # lengths straddling the 512/1024 split boundaries
+    lens = torch.tensor([100, 511, 513, 1024, 1500], device=DEV,
+                        dtype=torch.int32)
+    batched = ops.attention_decode(q, kc, vc, lens)
+    for b in range(B):
+        solo = ops.attention_decode(q[b:b + 1], kc[b:b + 1], vc[b:b + 1],
+                                    lens[b:b + 1])
+        assert torch.equal(batched[b], solo[0]), f"row {b} diverged"
+
+
+def test_tl_gemm_row_m_independence():
+    """The determinism contract: a row's GEMM result is bitwise identical
+    whatever batch it is computed in (streaming M<=64 kernel, tiled
+    kernel, any M, with or without the rows around it). Chunked prefill /
+    speculative verify / ragged decode equality all rest on this."""
+    ops = _ext()
+    torch.manual_seed(13)
+    K, N = 3584, 4608
+    x = torch.randn(300, K, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(N, K, device=DEV, dtype=torch.bfloat16) / (K ** 0.5)
+    full = ops.linear(x, w)                        # tiled, 2 m-blocks
+    one = ops.linear(x[7:8], w)                    # streaming M=1
+    chunk = ops.linear(x[:64], w)                  # streaming M=64
+    mid = ops.linear(x[:150], w)                   # tiled MT=16
+    assert torch.equal(full[7], one[0])
+    assert torch.equal(full[:64], chunk)
+    assert torch.equal(full[:150], mid)
