@@ -61,7 +61,7 @@ hipStream_t cur_stream() {
 // compiled by plain g++ so it cannot include the device header)
 int tl_split_for_len(int L) {
   int ns = 1;
-  while (ns < 16 && L > 512 * ns) ns <<= 1;
+  while (ns < 16 && L > 1024 * ns) ns <<= 1;
   return ns;
 }
 

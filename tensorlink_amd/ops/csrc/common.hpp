@@ -77,7 +77,7 @@ __host__ __device__ inline int cdiv(int a, int b) { return (a + b - 1) / b; }
 // in L, so a host-side length upper bound gives a valid launch width.
 __host__ __device__ inline int tl_split_for_len(int L) {
   int ns = 1;
-  while (ns < 16 && L > 512 * ns) ns <<= 1;
+  while (ns < 16 && L > 1024 * ns) ns <<= 1;
   return ns;
 }
 __host__ __device__ inline int64_t cdiv64(int64_t a, int64_t b) {

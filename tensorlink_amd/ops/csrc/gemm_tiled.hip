@@ -31,6 +31,14 @@ constexpr int BLOCK = 256;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 
+// counted VMEM wait with a compile-time immediate — invisible to hipcc's
+// waitcnt pass, so the glds pipeline is not drained at barriers (guide
+// §5: plain __syncthreads with a glds in flight emits vmcnt(0))
+template <int N>
+DEVINLINE void waitcnt_vm() {
+  asm volatile("s_waitcnt vmcnt(%0)" ::"n"(N) : "memory");
+}
+
 template <int MT, int HAS_BIAS, int SPLIT>
 __global__ __launch_bounds__(BLOCK, 2) void gemm_tiled_kernel(
     const bf16* __restrict__ x,     // [M, K]
@@ -39,11 +47,12 @@ __global__ __launch_bounds__(BLOCK, 2) void gemm_tiled_kernel(
     bf16* __restrict__ out,         // [M, N]
     float* __restrict__ partial,    // [SPLITK, M, N] when SPLIT
     int M, int N, int K, int n_split) {
+  constexpr int MPW = MT / 4;                 // m-tiles per wave
+  static_assert(MT % 4 == 0, "quadrant layout needs MT % 4 == 0");
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int col = lane & 15;
   const int quad = lane >> 4;
-  const int oc = blockIdx.x * 64 + wave * 16 + col;  // output column
   const int m_base = blockIdx.z * (MT * 16);
 
   const int split = SPLIT ? blockIdx.y : 0;
@@ -93,46 +102,205 @@ __global__ __launch_bounds__(BLOCK, 2) void gemm_tiled_kernel(
     }                                                                       \
   }
 
+  // Counted-vmcnt double-buffer pipeline: tile t+1's glds stay in flight
+  // across the barrier (each wave issues a fixed LPT loads per tile —
+  // MT/2 x-chunks + 2 W-chunks, wave-uniform for MT in {8,16} — waits
+  // for its OWN tile-t loads with vmcnt(LPT), then the barrier aligns
+  // the workgroup, so tile t is fully in LDS without draining t+1).
+  constexpr int LPT = MT / 2 + 2;
+  static_assert(MT == 8 || MT == 16, "LPT assumes wave-uniform chunks");
   GT_GLDS_TILE(0, k_begin);
   int buf = 0;
   for (int k0 = k_begin; k0 < k_end; k0 += 64) {
-    __syncthreads();                          // drains this tile's glds
-    if (k0 + 64 < k_end)
+    if (k0 + 64 < k_end) {
       GT_GLDS_TILE(buf ^ 1, k0 + 64);         // in flight under compute
+      waitcnt_vm<LPT>();                      // tile t landed; t+1 out
+    } else {
+      waitcnt_vm<0>();                        // last tile: drain
+    }
+    __builtin_amdgcn_s_barrier();
     const int tile_k = min(64, k_end - k0);
-    const int wrow = wave * 16 + col;         // this lane's w slab row
+    // 2D fragment reuse: this wave owns a (MT/4 x 4)-tile quadrant —
+    // rows [wave*MPW*16, +MPW*16), all 64 panel columns — so each kk
+    // reads MPW a-frags + 4 b-frags and issues 4*MPW mfmas (the 1D
+    // col-strip layout read one a-frag PER mfma and was LDS-issue
+    // bound at ~1.9 TB/s on wide-N shapes).
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       if (kk * 32 < tile_k) {
-        const int wslot = (kk * 4 + quad) ^ (wrow & 7);
-        bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-            &lds[buf][MT * 16 + wrow][wslot * 8]);
+        bf16x8_t bfrag[4], afrag[MPW];
 #pragma unroll
-        for (int m = 0; m < MT; ++m) {
-          const int row = m * 16 + col;
-          const int slot = (kk * 4 + quad) ^ (row & 7);
-          bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-              &lds[buf][row][slot * 8]);
-          acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, bfrag, acc[m], 0, 0, 0);
+        for (int nt = 0; nt < 4; ++nt) {
+          const int wrow = nt * 16 + col;
+          const int wslot = (kk * 4 + quad) ^ (wrow & 7);
+          bfrag[nt] = *reinterpret_cast<const bf16x8_t*>(
+              &lds[buf][MT * 16 + wrow][wslot * 8]);
         }
+#pragma unroll
+        for (int mt = 0; mt < MPW; ++mt) {
+          const int row = wave * (MPW * 16) + mt * 16 + col;
+          const int slot = (kk * 4 + quad) ^ (row & 7);
+          afrag[mt] = *reinterpret_cast<const bf16x8_t*>(
+              &lds[buf][row][slot * 8]);
+        }
+#pragma unroll
+        for (int mt = 0; mt < MPW; ++mt)
+#pragma unroll
+          for (int nt = 0; nt < 4; ++nt)
+            acc[mt * 4 + nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[mt], bfrag[nt], acc[mt * 4 + nt], 0, 0, 0);
       }
     }
+    // all waves done reading buf before the next iteration's glds
+    // overwrite it (hipcc has already counted-waited the ds_reads into
+    // the mfmas; lgkmcnt(0) makes that explicit before the raw barrier)
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
     buf ^= 1;
   }
 #undef GT_GLDS_TILE
 
-  const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
 #pragma unroll
-  for (int m = 0; m < MT; ++m) {
+  for (int mt = 0; mt < MPW; ++mt) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int orow = m_base + m * 16 + quad * 4 + r;
-      if (orow < M) {
-        if (SPLIT)
-          partial[((int64_t)split * M + orow) * N + oc] = acc[m][r];
-        else
-          out[(int64_t)orow * N + oc] = f2bf(acc[m][r] + b);
+    for (int nt = 0; nt < 4; ++nt) {
+      const int oc = blockIdx.x * 64 + nt * 16 + col;
+      const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int orow =
+            m_base + wave * (MPW * 16) + mt * 16 + quad * 4 + r;
+        if (orow < M) {
+          if (SPLIT)
+            partial[((int64_t)split * M + orow) * N + oc] =
+                acc[mt * 4 + nt][r];
+          else
+            out[(int64_t)orow * N + oc] = f2bf(acc[mt * 4 + nt][r] + b);
+        }
+      }
+    }
+  }
+}
+
+// Wide-N variant: 256x128 block tile, 8 waves (4M x 2N quadrants of
+// 64x64), 96 KB LDS, 1 block/CU. For N >= ~16k (gate_up, lm_head) the
+// x slab re-read per output panel dominates on-chip traffic in the
+// BN=64 kernel (4:1 x:W bytes); BN=128 halves it. Same per-row K
+// accumulation semantics as the rest of the family.
+template <int HAS_BIAS, int SPLIT>
+__global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
+    const bf16* __restrict__ x,     // [M, K]
+    const bf16* __restrict__ w,     // [N, K]
+    const bf16* __restrict__ bias,  // [N] or null
+    bf16* __restrict__ out,         // [M, N]
+    float* __restrict__ partial,    // [SPLITK, M, N] when SPLIT
+    int M, int N, int K, int n_split) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;          // 0..7
+  const int col = lane & 15;
+  const int quad = lane >> 4;
+  const int wm = wave >> 1;                   // 0..3: 64-row band
+  const int wn = wave & 1;                    // 0..1: 64-col band
+  const int m_base = blockIdx.z * 256;
+
+  const int split = SPLIT ? blockIdx.y : 0;
+  const int k_per = SPLIT ? ((K / 64 + n_split - 1) / n_split) * 64 : K;
+  const int k_begin = split * k_per;
+  const int k_end = min(K, k_begin + k_per);
+
+  __shared__ bf16 lds[2][256 + 128][64];
+
+  f32x4 acc[16];
+#pragma unroll
+  for (int m = 0; m < 16; ++m) acc[m] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int l8 = lane >> 3, c8 = lane & 7;
+
+  // x: 32 chunks of 8 rows -> 4 per wave; W: 16 chunks -> 2 per wave
+#define GW_GLDS_TILE(bufi, k0)                                              \
+  if ((k0) < k_end) {                                                       \
+    _Pragma("unroll") for (int g = 0; g < 4; ++g) {                         \
+      const int chunk = wave * 4 + g;                                       \
+      const int row = chunk * 8 + l8;                                       \
+      const int xr = min(m_base + row, M - 1);                              \
+      const int kc = min((k0) + (c8 ^ (row & 7)) * 8, K - 8);               \
+      auto gsrc = (const __attribute__((address_space(1))) void*)(          \
+          x + (int64_t)xr * K + kc);                                        \
+      auto ldst = (__attribute__((address_space(3))) void*)(                \
+          &lds[bufi][chunk * 8][0]);                                        \
+      __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);               \
+    }                                                                       \
+    _Pragma("unroll") for (int g = 0; g < 2; ++g) {                        \
+      const int wchunk = wave * 2 + g;                                      \
+      const int wr = wchunk * 8 + l8;                                       \
+      const int kc = min((k0) + (c8 ^ (wr & 7)) * 8, K - 8);                \
+      auto gsrc = (const __attribute__((address_space(1))) void*)(          \
+          w + ((int64_t)blockIdx.x * 128 + wr) * K + kc);                   \
+      auto ldst = (__attribute__((address_space(3))) void*)(                \
+          &lds[bufi][256 + wchunk * 8][0]);                                 \
+      __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);               \
+    }                                                                       \
+  }
+
+  GW_GLDS_TILE(0, k_begin);
+  int buf = 0;
+  for (int k0 = k_begin; k0 < k_end; k0 += 64) {
+    if (k0 + 64 < k_end) {
+      GW_GLDS_TILE(buf ^ 1, k0 + 64);
+      waitcnt_vm<6>();                        // tile t landed; t+1 out
+    } else {
+      waitcnt_vm<0>();
+    }
+    __builtin_amdgcn_s_barrier();
+    const int tile_k = min(64, k_end - k0);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      if (kk * 32 < tile_k) {
+        bf16x8_t bfrag[4], afrag[4];
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          const int wrow = wn * 64 + nt * 16 + col;
+          const int wslot = (kk * 4 + quad) ^ (wrow & 7);
+          bfrag[nt] = *reinterpret_cast<const bf16x8_t*>(
+              &lds[buf][256 + wrow][wslot * 8]);
+        }
+#pragma unroll
+        for (int mt = 0; mt < 4; ++mt) {
+          const int row = wm * 64 + mt * 16 + col;
+          const int slot = (kk * 4 + quad) ^ (row & 7);
+          afrag[mt] = *reinterpret_cast<const bf16x8_t*>(
+              &lds[buf][row][slot * 8]);
+        }
+#pragma unroll
+        for (int mt = 0; mt < 4; ++mt)
+#pragma unroll
+          for (int nt = 0; nt < 4; ++nt)
+            acc[mt * 4 + nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[mt], bfrag[nt], acc[mt * 4 + nt], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+#undef GW_GLDS_TILE
+
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt) {
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int oc = blockIdx.x * 128 + wn * 64 + nt * 16 + col;
+      const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int orow = m_base + wm * 64 + mt * 16 + quad * 4 + r;
+        if (orow < M) {
+          if (SPLIT)
+            partial[((int64_t)split * M + orow) * N + oc] =
+                acc[mt * 4 + nt][r];
+          else
+            out[(int64_t)orow * N + oc] = f2bf(acc[mt * 4 + nt][r] + b);
+        }
       }
     }
   }
@@ -161,6 +329,21 @@ void tl_gemm_tiled(const void* x, const void* w, const void* bias, void* out,
                    hipStream_t stream) {
   const int m_tiles = (M + 15) / 16;
   const int split = n_split > 1 ? 1 : 0;
+  // very-wide-N shapes (lm_head) take the 256x128 8-wave kernel; at
+  // gate_up width (N/128 = 296) its 1-block/CU tail imbalance loses to
+  // the BN=64 kernel. Dispatch is by (N, K) only, so it is M-independent
+  if (N >= 65536 && N % 128 == 0) {
+    dim3 wgrid(N / 128, n_split, cdiv(M, 256)), wblock(512);
+#define WDISPATCH(HB, SP)                                                   \
+  hipLaunchKernelGGL((gemm_tiled_wide_kernel<HB, SP>), wgrid, wblock, 0,    \
+                     stream, (const bf16*)x, (const bf16*)w,                \
+                     (const bf16*)bias, (bf16*)out, (float*)partial, M, N,  \
+                     K, n_split)
+    if (split) WDISPATCH(0, 1);
+    else if (bias) WDISPATCH(1, 0);
+    else WDISPATCH(0, 0);
+#undef WDISPATCH
+  } else {
   const int MT = m_tiles <= 8 ? 8 : 16;
   const int n_mblk = cdiv(m_tiles, MT);
   dim3 grid(N / 64, n_split, n_mblk), block(BLOCK);
@@ -184,6 +367,7 @@ void tl_gemm_tiled(const void* x, const void* w, const void* bias, void* out,
   if (MT == 8) DISPATCH(8);
   else DISPATCH(16);
 #undef DISPATCH
+  }
   if (split) {
     const int64_t MN = (int64_t)M * N;
     dim3 rgrid((uint32_t)((MN + 255) / 256)), rblock(256);

@@ -290,6 +290,7 @@ def test_decode_attn_mfma(B, Hq, Hkv, D, L, ns):
     (7, 1024, 256, True),         # tiny M, M tail
     (33, 512, 96, False),         # K%64==32 tail path
 ])
+@torch.no_grad()
 def test_tl_gemm(M, N, K, bias):
     ops = _ext()
     torch.manual_seed(12)
@@ -325,6 +326,7 @@ def test_decode_attn_batch_independence():
         assert torch.equal(batched[b], solo[0]), f"row {b} diverged"
 
 
+@torch.no_grad()
 def test_tl_gemm_row_m_independence():
     """The determinism contract: a row's GEMM result is bitwise identical
     whatever batch it is computed in (streaming M<=64 kernel, tiled
