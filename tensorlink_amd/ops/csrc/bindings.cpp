@@ -298,6 +298,15 @@ Tensor swiglu_fused(Tensor gu) {
 // (split boundaries + s-ordered reduce) is bitwise identical at every M:
 // the serving engine's exact-greedy guarantee rests on this.
 static int gemm_n_split(int N, int K) {
+  if (N >= 65536) {
+    // 256x256-tile class (gemm_tiled_sq, lm_head widths): enough
+    // column panels to fill the chip without splitting
+    int n_split = 1;
+    while (((N + 255) / 256) * n_split < 256 && n_split < 16 &&
+           (K / 64) / (n_split * 2) >= 16)
+      n_split <<= 1;
+    return n_split;
+  }
   // split K so (N/64)*n_split lands near 1 block/CU (guide: ~0.5-1x CUs)
   int n_split = 1;
   while ((N / 64) * n_split < 256 && n_split < 8 &&

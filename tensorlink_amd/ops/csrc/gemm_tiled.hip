@@ -182,13 +182,17 @@ __global__ __launch_bounds__(BLOCK, 2) void gemm_tiled_kernel(
   }
 }
 
-// Wide-N variant: 256x128 block tile, 8 waves (4M x 2N quadrants of
-// 64x64), 96 KB LDS, 1 block/CU. For N >= ~16k (gate_up, lm_head) the
-// x slab re-read per output panel dominates on-chip traffic in the
-// BN=64 kernel (4:1 x:W bytes); BN=128 halves it. Same per-row K
-// accumulation semantics as the rest of the family.
+// Big-tile variant: 256x256 block tile, 8 waves (2M x 4N quadrants of
+// 128x64), 128 KB LDS, 1 block/CU. For wide-N (gate_up, lm_head) and
+// deep-K (down_proj) shapes the dominant cost in the BN=64 kernel is
+// re-reading the x slab once per output panel (N/BN * M*K*2 bytes, L3-
+// served at ~10 TB/s); BN=256 makes x:W traffic 1:1. Per-iteration MFMA
+// work (256 mfma/block) is also long enough to hide the glds latency
+// that starves the small-tile kernels. Same per-row K accumulation
+// semantics as the rest of the family. Handles N % 256 != 0 (lm_head
+// 151936) with clamped W loads + guarded stores.
 template <int HAS_BIAS, int SPLIT>
-__global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
+__global__ __launch_bounds__(512, 1) void gemm_tiled_sq_kernel(
     const bf16* __restrict__ x,     // [M, K]
     const bf16* __restrict__ w,     // [N, K]
     const bf16* __restrict__ bias,  // [N] or null
@@ -199,8 +203,8 @@ __global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
   const int wave = threadIdx.x >> 6;          // 0..7
   const int col = lane & 15;
   const int quad = lane >> 4;
-  const int wm = wave >> 1;                   // 0..3: 64-row band
-  const int wn = wave & 1;                    // 0..1: 64-col band
+  const int wm = wave >> 2;                   // 0..1: 128-row band
+  const int wn = wave & 3;                    // 0..3: 64-col band
   const int m_base = blockIdx.z * 256;
 
   const int split = SPLIT ? blockIdx.y : 0;
@@ -208,16 +212,16 @@ __global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
   const int k_begin = split * k_per;
   const int k_end = min(K, k_begin + k_per);
 
-  __shared__ bf16 lds[2][256 + 128][64];
+  __shared__ bf16 lds[2][256 + 256][64];
 
-  f32x4 acc[16];
+  f32x4 acc[32];
 #pragma unroll
-  for (int m = 0; m < 16; ++m) acc[m] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  for (int m = 0; m < 32; ++m) acc[m] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
   const int l8 = lane >> 3, c8 = lane & 7;
 
-  // x: 32 chunks of 8 rows -> 4 per wave; W: 16 chunks -> 2 per wave
-#define GW_GLDS_TILE(bufi, k0)                                              \
+  // x: 32 chunks of 8 rows -> 4 per wave; W: 32 chunks -> 4 per wave
+#define GS_GLDS_TILE(bufi, k0)                                              \
   if ((k0) < k_end) {                                                       \
     _Pragma("unroll") for (int g = 0; g < 4; ++g) {                         \
       const int chunk = wave * 4 + g;                                       \
@@ -230,24 +234,25 @@ __global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
           &lds[bufi][chunk * 8][0]);                                        \
       __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);               \
     }                                                                       \
-    _Pragma("unroll") for (int g = 0; g < 2; ++g) {                        \
-      const int wchunk = wave * 2 + g;                                      \
+    _Pragma("unroll") for (int g = 0; g < 4; ++g) {                         \
+      const int wchunk = wave * 4 + g;                                      \
       const int wr = wchunk * 8 + l8;                                       \
+      const int gr = min((int)blockIdx.x * 256 + wr, N - 1);                \
       const int kc = min((k0) + (c8 ^ (wr & 7)) * 8, K - 8);                \
       auto gsrc = (const __attribute__((address_space(1))) void*)(          \
-          w + ((int64_t)blockIdx.x * 128 + wr) * K + kc);                   \
+          w + (int64_t)gr * K + kc);                                        \
       auto ldst = (__attribute__((address_space(3))) void*)(                \
           &lds[bufi][256 + wchunk * 8][0]);                                 \
       __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);               \
     }                                                                       \
   }
 
-  GW_GLDS_TILE(0, k_begin);
+  GS_GLDS_TILE(0, k_begin);
   int buf = 0;
   for (int k0 = k_begin; k0 < k_end; k0 += 64) {
     if (k0 + 64 < k_end) {
-      GW_GLDS_TILE(buf ^ 1, k0 + 64);
-      waitcnt_vm<6>();                        // tile t landed; t+1 out
+      GS_GLDS_TILE(buf ^ 1, k0 + 64);
+      waitcnt_vm<8>();                        // tile t landed; t+1 out
     } else {
       waitcnt_vm<0>();
     }
@@ -256,7 +261,7 @@ __global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       if (kk * 32 < tile_k) {
-        bf16x8_t bfrag[4], afrag[4];
+        bf16x8_t bfrag[4], afrag[8];
 #pragma unroll
         for (int nt = 0; nt < 4; ++nt) {
           const int wrow = wn * 64 + nt * 16 + col;
@@ -265,14 +270,14 @@ __global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
               &lds[buf][256 + wrow][wslot * 8]);
         }
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
-          const int row = wm * 64 + mt * 16 + col;
+        for (int mt = 0; mt < 8; ++mt) {
+          const int row = wm * 128 + mt * 16 + col;
           const int slot = (kk * 4 + quad) ^ (row & 7);
           afrag[mt] = *reinterpret_cast<const bf16x8_t*>(
               &lds[buf][row][slot * 8]);
         }
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt)
+        for (int mt = 0; mt < 8; ++mt)
 #pragma unroll
           for (int nt = 0; nt < 4; ++nt)
             acc[mt * 4 + nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -283,23 +288,25 @@ __global__ __launch_bounds__(512, 1) void gemm_tiled_wide_kernel(
     __builtin_amdgcn_s_barrier();
     buf ^= 1;
   }
-#undef GW_GLDS_TILE
+#undef GS_GLDS_TILE
 
 #pragma unroll
-  for (int mt = 0; mt < 4; ++mt) {
+  for (int mt = 0; mt < 8; ++mt) {
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      const int oc = blockIdx.x * 128 + wn * 64 + nt * 16 + col;
-      const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
+      const int oc = blockIdx.x * 256 + wn * 64 + nt * 16 + col;
+      if (oc < N) {
+        const float b = HAS_BIAS ? bf2f(bias[oc]) : 0.f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int orow = m_base + wm * 64 + mt * 16 + quad * 4 + r;
-        if (orow < M) {
-          if (SPLIT)
-            partial[((int64_t)split * M + orow) * N + oc] =
-                acc[mt * 4 + nt][r];
-          else
-            out[(int64_t)orow * N + oc] = f2bf(acc[mt * 4 + nt][r] + b);
+        for (int r = 0; r < 4; ++r) {
+          const int orow = m_base + wm * 128 + mt * 16 + quad * 4 + r;
+          if (orow < M) {
+            if (SPLIT)
+              partial[((int64_t)split * M + orow) * N + oc] =
+                  acc[mt * 4 + nt][r];
+            else
+              out[(int64_t)orow * N + oc] = f2bf(acc[mt * 4 + nt][r] + b);
+          }
         }
       }
     }
@@ -329,13 +336,14 @@ void tl_gemm_tiled(const void* x, const void* w, const void* bias, void* out,
                    hipStream_t stream) {
   const int m_tiles = (M + 15) / 16;
   const int split = n_split > 1 ? 1 : 0;
-  // very-wide-N shapes (lm_head) take the 256x128 8-wave kernel; at
-  // gate_up width (N/128 = 296) its 1-block/CU tail imbalance loses to
-  // the BN=64 kernel. Dispatch is by (N, K) only, so it is M-independent
-  if (N >= 65536 && N % 128 == 0) {
-    dim3 wgrid(N / 128, n_split, cdiv(M, 256)), wblock(512);
+  // very-wide-N shapes (lm_head) take the 256x256 8-wave kernel; at
+  // gate_up/down widths the BN=64 kernel measures faster (see
+  // profiles/gemm_family_ab.md). Dispatch is by (N, K) only -> the
+  // choice is M-independent
+  if (N >= 65536) {
+    dim3 wgrid(cdiv(N, 256), n_split, cdiv(M, 256)), wblock(512);
 #define WDISPATCH(HB, SP)                                                   \
-  hipLaunchKernelGGL((gemm_tiled_wide_kernel<HB, SP>), wgrid, wblock, 0,    \
+  hipLaunchKernelGGL((gemm_tiled_sq_kernel<HB, SP>), wgrid, wblock, 0,      \
                      stream, (const bf16*)x, (const bf16*)w,                \
                      (const bf16*)bias, (bf16*)out, (float*)partial, M, N,  \
                      K, n_split)
