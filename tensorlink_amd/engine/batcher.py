@@ -160,6 +160,8 @@ class ContinuousBatcher:
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self.steps = 0
+        import random as _random
+        self._rng = _random.Random()     # unseeded-request draw seeds
         from tensorlink_amd.utils.tracing import tracer_from_env
         self.tracer = tracer_from_env(runner.rank)  # TL_TRACE=<prefix>
 
@@ -322,8 +324,47 @@ class ContinuousBatcher:
         self._maybe_finish(slot, tok)
 
     def _sample(self, logits, reqs, slots=None):
-        toks = []
+        """Sample one token per row. On the GPU path, penalty-free
+        sampled rows go through the fused sampling kernel in ONE launch
+        (ops/csrc/sampling.hip — per-request splitmix seeds keep draws
+        batch-independent); greedy / penalized / logprobs rows take the
+        torch reference path row-wise."""
+        toks: list = [None] * len(reqs)
+        fused = []
+        if (logits.is_cuda and logits.dtype == torch.bfloat16
+                and ops.extension_loaded()):
+            fused = [i for i, r in enumerate(reqs)
+                     if r.temperature > 0 and not r.logprobs
+                     and r.presence_penalty == 0.0
+                     and r.frequency_penalty == 0.0]
+        if fused:
+            dev = logits.device
+            n = len(fused)
+            temps = torch.tensor([reqs[i].temperature for i in fused],
+                                 device=dev, dtype=torch.float32)
+            tps = torch.tensor([reqs[i].top_p for i in fused],
+                               device=dev, dtype=torch.float32)
+            tks = torch.tensor([reqs[i].top_k for i in fused],
+                               device=dev, dtype=torch.int32)
+            zero = torch.zeros(n, device=dev, dtype=torch.float32)
+            seeds = []
+            for i in fused:
+                req = reqs[i]
+                req._draws = getattr(req, "_draws", 0) + 1
+                base = (req.seed if req.seed is not None
+                        else self._rng.getrandbits(62))
+                seeds.append(ops.request_seed(base, req._draws))
+            seeds_t = torch.tensor(seeds, device=dev, dtype=torch.int64)
+            rows = (logits if n == len(reqs)
+                    else logits[torch.tensor(fused, device=dev)])
+            got = ops.sample_tokens(rows, temps=temps, top_ps=tps,
+                                    top_ks=tks, pres=zero, freqs=zero,
+                                    seeds=seeds_t).tolist()
+            for j, i in enumerate(fused):
+                toks[i] = int(got[j])
         for i, req in enumerate(reqs):
+            if toks[i] is not None:
+                continue
             counts = (self.slot_counts[slots[i]]
                       if slots is not None else None)
             gen = None
@@ -341,7 +382,7 @@ class ContinuousBatcher:
             if req.logprobs:
                 lp = torch.log_softmax(logits[i].float(), -1)[tok]
                 req.logprob_values.append(float(lp))
-            toks.append(tok)
+            toks[i] = tok
         return toks
 
     def _maybe_finish(self, slot: int, tok: int):

@@ -341,11 +341,44 @@ def adamw_(param: torch.Tensor, grad: torch.Tensor, exp_avg: torch.Tensor,
 
 
 # ---------------------------------------------------------------------------
-# Sampling (torch-composed on GPU for now; fused kernel is a later milestone)
+# Sampling
 # ---------------------------------------------------------------------------
+def _splitmix64(x: int) -> int:
+    x = (x + 0x9E3779B97F4A7C15) & (2 ** 64 - 1)
+    x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & (2 ** 64 - 1)
+    x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & (2 ** 64 - 1)
+    return x ^ (x >> 31)
+
+
+def request_seed(seed: int, step: int) -> int:
+    """Per-draw seed for a seeded request: depends only on (seed, step),
+    so a request's samples are reproducible wherever it is batched.
+    Returned as a signed 63-bit value (torch.long seeds tensor)."""
+    return _splitmix64((seed & (2 ** 63 - 1)) ^ (step * 0xA24BAED4963EE407)) \
+        & (2 ** 63 - 1)
+
+
+def sample_tokens(logits: torch.Tensor, *, temps, top_ps, top_ks, pres,
+                  freqs, counts: Optional[torch.Tensor] = None,
+                  seeds: Optional[torch.Tensor] = None,
+                  counter: Optional[torch.Tensor] = None,
+                  seed_base: int = 0) -> torch.Tensor:
+    """Batched fused sampling (ops/csrc/sampling.hip): penalties +
+    temperature + top-k + top-p + draw, one kernel, graph-capture-safe.
+    logits [B, V] bf16 on GPU; per-row param tensors; counts [B, V] i32
+    is read for penalties and the chosen token's count is incremented."""
+    C = _require_ext()
+    return C.sample_tokens(logits.contiguous(), temps, top_ps, top_ks,
+                           pres, freqs, counts, seeds, counter, seed_base)
+
+
 def sample_token(logits, *, temperature=1.0, top_p=1.0, top_k=0,
                  generator=None, token_counts=None,
                  presence_penalty=0.0, frequency_penalty=0.0):
+    """Single-request sampling with the reference (torch) semantics.
+    The serving engine's hot path uses :func:`sample_tokens`; this stays
+    torch-composed for API parity (dict token_counts, torch.Generator)
+    and for greedy argmax."""
     return ref.sample_token(logits, temperature=temperature, top_p=top_p,
                             top_k=top_k, generator=generator,
                             token_counts=token_counts,

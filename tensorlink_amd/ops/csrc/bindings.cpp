@@ -49,6 +49,11 @@ void tl_skinny_gemm(const void* x, const void* w, const void* bias,
 void tl_gemm_tiled(const void* x, const void* w, const void* bias, void* out,
                    void* partial, int M, int N, int K, int n_split,
                    hipStream_t stream);
+void tl_sample(const void* logits, const void* temps, const void* top_ps,
+               const void* top_ks, const void* pres, const void* freqs,
+               void* counts, const void* seeds, const void* ctr, void* out,
+               uint64_t seed_base, int B, int V, hipStream_t stream);
+void tl_bump_counter(void* ctr, hipStream_t stream);
 }
 
 namespace {
@@ -348,6 +353,48 @@ Tensor skinny_gemm(Tensor x, Tensor w, c10::optional<Tensor> bias) {
   return out;
 }
 
+Tensor sample_tokens(Tensor logits, Tensor temps, Tensor top_ps,
+                     Tensor top_ks, Tensor pres, Tensor freqs,
+                     c10::optional<Tensor> counts,
+                     c10::optional<Tensor> seeds,
+                     c10::optional<Tensor> counter, int64_t seed_base) {
+  CHECK_IN(logits, torch::kBFloat16);
+  CHECK_IN(temps, torch::kFloat);
+  CHECK_IN(top_ps, torch::kFloat);
+  CHECK_IN(top_ks, torch::kInt);
+  CHECK_IN(pres, torch::kFloat);
+  CHECK_IN(freqs, torch::kFloat);
+  const int B = logits.size(0);
+  const int V = logits.size(1);
+  void* cnt = nullptr;
+  if (counts.has_value()) {
+    CHECK_IN(counts.value(), torch::kInt);
+    TORCH_CHECK(counts->size(0) == B && counts->size(1) == V,
+                "counts must be [B, V]");
+    cnt = counts->data_ptr();
+  }
+  const void* sd = nullptr;
+  if (seeds.has_value()) {
+    CHECK_IN(seeds.value(), torch::kLong);
+    sd = seeds->data_ptr();
+  }
+  const void* ct = nullptr;
+  if (counter.has_value()) {
+    CHECK_IN(counter.value(), torch::kLong);
+    ct = counter->data_ptr();
+  }
+  auto out = torch::empty({B}, logits.options().dtype(torch::kLong));
+  tl_sample(logits.data_ptr(), temps.data_ptr(), top_ps.data_ptr(),
+            top_ks.data_ptr(), pres.data_ptr(), freqs.data_ptr(), cnt, sd,
+            ct, out.data_ptr(), (uint64_t)seed_base, B, V, cur_stream());
+  return out;
+}
+
+void bump_sample_counter(Tensor ctr) {
+  CHECK_IN(ctr, torch::kLong);
+  tl_bump_counter(ctr.data_ptr(), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -363,4 +410,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_", &adamw_, "fused AdamW step");
   mod.def("decode_attn", &decode_attn, "GQA decode attention over KV cache");
   mod.def("prefill_attn", &prefill_attn, "causal GQA prefill attention");
+  mod.def("sample_tokens", &sample_tokens,
+          "fused penalties/temperature/top-k/top-p sampling, one block/row");
+  mod.def("bump_sample_counter", &bump_sample_counter,
+          "advance the graph-safe sampling RNG counter");
 }

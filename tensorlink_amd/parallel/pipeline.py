@@ -240,17 +240,22 @@ class PipelineRunner:
             finished = (cur == eos) if eos is not None else None
             n_out = T
             import os as _os
-            use_graph = (self.device.type == "cuda" and sp.temperature <= 0
+            from tensorlink_amd import ops as _ops
+            use_graph = (self.device.type == "cuda"
+                         and (sp.temperature <= 0
+                              or _ops.extension_loaded())
                          and T > 4 and not _os.environ.get("TL_NO_GRAPH")
                          and not self._no_graph
                          and self.kv_mode == "contiguous"
                          and on_token is None and eos is None)
             if use_graph:
                 # hipGraph-captured decode: the whole per-token step (all
-                # layers + head + argmax) replays as one graph launch,
-                # eliminating ~10 kernel-launch gaps per layer.
+                # layers + head + sample) replays as one graph launch,
+                # eliminating ~10 kernel-launch gaps per layer. Sampled
+                # decode uses the fused sampling kernel with a device RNG
+                # counter bumped inside the capture (graph-safe draws).
                 try:
-                    self._graph_decode(cur, positions, out_tokens, T)
+                    self._graph_decode(cur, positions, out_tokens, T, sp)
                 except RuntimeError:
                     # capture-unsupported op somewhere in the stage: fall
                     # back to the eager loop permanently for this runner
@@ -486,25 +491,55 @@ class PipelineRunner:
         return tokens[idx]
 
     def _graph_decode(self, cur: torch.Tensor, positions: torch.Tensor,
-                      out_tokens: torch.Tensor, T: int) -> None:
-        """Greedy decode via a captured hipGraph (world==1).
+                      out_tokens: torch.Tensor, T: int,
+                      sp: "SamplingParams") -> None:
+        """Decode via a captured hipGraph (world==1), greedy or sampled.
 
         The KV cache must already hold the prefill; `positions` holds the
         next position per sequence. State mutated by the captured step:
-        tok/pos static buffers, cache contents and seq_lens (all by device
-        pointer, so replays see fresh values).
+        tok/pos static buffers, cache contents, seq_lens and the RNG
+        counter (all by device pointer, so replays see fresh values).
+        Sampled capture: the fused sampling kernel draws from
+        splitmix(seed_base, counter, row) — the counter advances inside
+        the graph, so every replay gets fresh randomness with no host RNG.
         """
+        from tensorlink_amd import ops
         B = cur.shape[0]
+        skey = (float(sp.temperature), float(sp.top_p), int(sp.top_k),
+                sp.seed)
+        if self._decode_graph is not None and self._decode_graph[3] != skey:
+            self._decode_graph = None
         if self._decode_graph is None:
             tok_buf = cur.clone()
             pos_buf = positions.clone()
             cache = self.kv_cache
             saved_lens = cache.seq_lens.clone()
+            dev = self.device
+            if sp.temperature > 0:
+                C = ops._require_ext()
+                ctr = torch.zeros(1, device=dev, dtype=torch.int64)
+                temps = torch.full((B,), float(sp.temperature), device=dev)
+                tps = torch.full((B,), float(sp.top_p), device=dev)
+                tks = torch.full((B,), int(sp.top_k), device=dev,
+                                 dtype=torch.int32)
+                zero = torch.zeros(B, device=dev)
+                seed_base = (sp.seed if sp.seed is not None
+                             else int(torch.seed()) & (2 ** 62 - 1))
+
+                def pick(logits):
+                    C.bump_sample_counter(ctr)
+                    return ops.sample_tokens(
+                        logits, temps=temps, top_ps=tps, top_ks=tks,
+                        pres=zero, freqs=zero, counter=ctr,
+                        seed_base=seed_base)
+            else:
+                def pick(logits):
+                    return logits.argmax(-1)
 
             def step():
                 logits = self.stage(tok_buf.unsqueeze(1),
                                     pos_buf.unsqueeze(1), kv_cache=cache)
-                tok_buf.copy_(logits.squeeze(1).argmax(-1))
+                tok_buf.copy_(pick(logits.squeeze(1)))
                 pos_buf.add_(1)
 
             s = torch.cuda.Stream()
@@ -517,8 +552,8 @@ class PipelineRunner:
             with torch.cuda.graph(graph):
                 step()
             cache.seq_lens.copy_(saved_lens)
-            self._decode_graph = (graph, tok_buf, pos_buf)
-        graph, tok_buf, pos_buf = self._decode_graph
+            self._decode_graph = (graph, tok_buf, pos_buf, skey)
+        graph, tok_buf, pos_buf = self._decode_graph[:3]
         tok_buf.copy_(cur)
         pos_buf.copy_(positions)
         for t in range(1, T):
