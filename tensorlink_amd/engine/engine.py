@@ -100,7 +100,8 @@ class InferenceEngine:
                  default_init: str = "random"):
         self.rank = rank
         self.world = world
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         self.default_init = default_init
         self.jobs: Dict[str, ModelJob] = {}
         self.metrics = EngineMetrics()

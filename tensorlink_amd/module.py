@@ -121,7 +121,8 @@ class DistributedModel(nn.Module):
         self.tp = tp
         self._procs = []
         self._p2p = None
-        self.device = device
+        self.device = (torch.device(device)
+                       if device is not None else device)
 
         launched_world = int(os.environ.get("WORLD_SIZE", "1"))
         if world_size > 1 and (mode == "torchrun" or

@@ -81,7 +81,8 @@ class CPRunner:
 
     def __init__(self, model, rank: int, cp: int, device=None, seed: int = 0):
         self.rank, self.cp = rank, cp
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         dtype = (torch.bfloat16 if self.device.type == "cuda"
                  else torch.float32)
         config = (model if isinstance(model, ModelConfig)
@@ -319,7 +320,8 @@ class CPTrainer:
                  lr: float = 1e-3, weight_decay: float = 0.01):
         from tensorlink_amd.optim import FusedAdamW
         self.rank, self.cp = rank, cp
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         self.dtype = (torch.bfloat16 if self.device.type == "cuda"
                       else torch.float32)
         config = (model if isinstance(model, ModelConfig)

@@ -194,7 +194,8 @@ class EPRunner:
     def __init__(self, model, rank: int, ep: int, device=None, seed: int = 0,
                  quantize: Optional[str] = None, mode: str = "allreduce"):
         self.rank, self.ep = rank, ep
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         self.stage = build_ep_model(model, rank, ep, device=self.device,
                                     seed=seed, quantize=quantize, mode=mode)
         self.config = self.stage.config
@@ -246,7 +247,8 @@ class EPTrainer:
                  lr: float = 1e-3, weight_decay: float = 0.01):
         from tensorlink_amd.optim import FusedAdamW
         self.rank, self.ep = rank, ep
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         self.stage = build_ep_model(model, rank, ep, device=self.device,
                                     seed=seed)
         self.stage.train()

@@ -69,7 +69,8 @@ class PipelineRunner:
         self.config: ModelConfig = plan.config
         self.rank = rank
         self.world = world
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         self.dtype = dtype or (torch.bfloat16 if self.device.type == "cuda"
                                else torch.float32)
         self.spec = plan.stage_for_rank(rank)

@@ -199,7 +199,8 @@ class TPRunner:
 
     def __init__(self, model, rank: int, tp: int, device=None, seed: int = 0):
         self.rank, self.tp = rank, tp
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         self.stage = build_tp_model(model, rank, tp, device=self.device,
                                     seed=seed)
         self.config = self.stage.config
@@ -296,7 +297,8 @@ class TPTrainer:
         from tensorlink_amd.optim import FusedAdamW
         self.rank, self.tp = rank, tp
         self.vocab_parallel = vocab_parallel
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         self.stage = build_tp_model(model, rank, tp, device=self.device,
                                     seed=seed,
                                     vocab_parallel=vocab_parallel)
@@ -393,7 +395,8 @@ class TPPPRunner:
         pp = world // tp
         self.pp, self.tp = pp, tp
         self.stage_idx, self.tp_rank = rank % pp, rank // pp
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         dtype = dtype or (torch.bfloat16 if self.device.type == "cuda"
                           else torch.float32)
         # same group-creation order on every rank (dist.new_group is
@@ -473,7 +476,8 @@ class TPPPTrainer:
         pp = world // tp
         self.pp, self.tp = pp, tp
         self.stage_idx, self.tp_rank = rank % pp, rank // pp
-        self.device = device if device is not None else device_for_rank()
+        self.device = (torch.device(device) if device is not None
+                       else device_for_rank())
         dtype = dtype or (torch.bfloat16 if self.device.type == "cuda"
                           else torch.float32)
         if pp_groups is None:
