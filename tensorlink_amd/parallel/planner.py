@@ -102,7 +102,8 @@ class ModelParser:
     # -- cost model ---------------------------------------------------------
     def _stage_cost(self, config: ModelConfig, n_layers: int, *,
                     embed: bool, head: bool, batch: int, seq: int,
-                    training: bool, dtype: str, micro_batches: int) -> int:
+                    training: bool, dtype: str, micro_batches: int,
+                    num_stages: int = 1) -> int:
         db = DTYPE_BYTES[dtype]
         params = n_layers * layer_param_bytes(config, db)
         if embed:
@@ -113,10 +114,15 @@ class ModelParser:
         if training:
             total += params                      # grads
             total += 2 * params * (4 // db)      # fp32 Adam moments
-            # 1F1B keeps ≤ num_stages in-flight micro-batches of activations
-            act = n_layers * activation_bytes_per_layer(
-                config, batch, seq, db, True)
-            total += act  # per full batch already (batch arg is global batch)
+            # 1F1B holds <= min(micro_batches, num_stages) in-flight
+            # micro-batches of activations per stage (a stage's backward
+            # for micro-batch m frees its stash before forward m+depth)
+            mb = max(1, micro_batches)
+            micro = (batch + mb - 1) // mb
+            in_flight = min(mb, max(1, num_stages))
+            act = (n_layers * activation_bytes_per_layer(
+                config, micro, seq, db, True) * in_flight)
+            total += act
         else:
             total += activation_bytes_per_layer(config, batch, seq, db, False)
             total += n_layers * kv_cache_bytes_per_layer(config, batch, seq, db)
@@ -195,7 +201,7 @@ class ModelParser:
                     config, spec.num_layers, embed=spec.has_embedding,
                     head=spec.has_head, batch=batch, seq=seq,
                     training=training, dtype=dtype,
-                    micro_batches=micro_batches)
+                    micro_batches=micro_batches, num_stages=pp)
                 stages.append(spec)
             return stages
 

@@ -103,3 +103,30 @@ def test_all_baseline_configs_plan():
     p5 = plan_for_world("mistralai/Mixtral-8x7B-v0.1", 2, batch_size=8,
                         seq_len=4096)
     assert p5.config.is_moe and p5.num_stages == 2
+
+
+def test_training_estimate_counts_1f1b_in_flight():
+    """The training activation term follows min(micro_batches, stages)
+    micro-batches in flight: splitting a big batch
+    into more micro-batches than stages shrinks the estimate, and the
+    in-flight count caps at the stage count."""
+    from tensorlink_amd.models.configs import get_config
+    from tensorlink_amd.parallel.planner import ModelParser
+    cfg = get_config("meta-llama/Llama-3-70B")
+    p = ModelParser(n_workers=8)
+
+    def est(mb):
+        return p._stage_cost(cfg, 10, embed=False, head=False, batch=64,
+                             seq=2048, training=True, dtype="bfloat16",
+                             micro_batches=mb, num_stages=8)
+
+    # mb == stages: 8 in-flight micro-batches == one full batch
+    assert abs(est(8) - est(1)) / est(1) < 0.05
+    # mb >> stages: only stages/mb of the batch is in flight
+    assert est(64) < est(8)
+    # and with enough micro-batches PP=8 training of the 70B config at
+    # batch 64 / seq 2048 now fits the plan (it over-estimated before)
+    plan = p.create_distributed_config(cfg, batch_size=64, seq_len=2048,
+                                       training=True, micro_batches=64,
+                                       num_stages=8)
+    assert plan.num_stages == 8
