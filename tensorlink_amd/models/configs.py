@@ -43,6 +43,9 @@ class ModelConfig:
     expert_intermediate_size: Optional[int] = None
     # HF rope_scaling dict (Llama-3.1 "llama3" frequency scaling)
     rope_scaling: Optional[dict] = None
+    # GPT-NeoX: fraction of each head rotated by RoPE; parallel residual
+    rotary_pct: float = 1.0
+    use_parallel_residual: bool = True
     # architecture tag for checkpoint key mapping
     architecture: str = "llama"
 
@@ -55,6 +58,12 @@ class ModelConfig:
     @property
     def is_moe(self) -> bool:
         return self.num_local_experts > 0
+
+    @property
+    def gated_mlp(self) -> bool:
+        """SwiGLU-style 3-matrix MLP (llama/qwen/mixtral) vs the plain
+        2-matrix GELU MLP (gpt2, neox)."""
+        return self.architecture not in ("gpt2", "neox")
 
     @property
     def q_size(self) -> int:
@@ -76,7 +85,7 @@ class ModelConfig:
                    * self.expert_intermediate_size)
             mlp += h * self.num_local_experts  # router
         else:
-            mlp = 3 * h * self.intermediate_size
+            mlp = (3 if self.gated_mlp else 2) * h * self.intermediate_size
         norms = 2 * h
         per_layer = attn + mlp + norms
         total = self.num_hidden_layers * per_layer + h  # final norm
@@ -112,6 +121,8 @@ class ModelConfig:
             arch = "qwen3"
         if any("Mixtral" in a for a in archs):
             arch = "mixtral"
+        if any("GPTNeoX" in a for a in archs):
+            arch = "neox"
         get = cfg.get
         return cls(
             name=name,
@@ -125,7 +136,8 @@ class ModelConfig:
             head_dim=get("head_dim", None),
             max_position_embeddings=get("max_position_embeddings", 4096),
             rope_theta=get("rope_theta", 10000.0),
-            rms_norm_eps=get("rms_norm_eps", 1e-6),
+            rms_norm_eps=get("rms_norm_eps",
+                             get("layer_norm_eps", 1e-6)),
             tie_word_embeddings=get("tie_word_embeddings", False),
             qkv_bias=(arch == "qwen2" and "Qwen3" not in str(archs)),
             num_local_experts=(get("num_local_experts", 0)
@@ -133,6 +145,8 @@ class ModelConfig:
             num_experts_per_tok=get("num_experts_per_tok", 2) or 2,
             expert_intermediate_size=get("moe_intermediate_size", None),
             rope_scaling=get("rope_scaling", None),
+            rotary_pct=get("rotary_pct", 1.0) or 1.0,
+            use_parallel_residual=get("use_parallel_residual", True),
             architecture=arch,
         )
 
@@ -245,6 +259,33 @@ def _tiny_bigvocab() -> ModelConfig:
         max_position_embeddings=512)
 
 
+def _tiny_neox() -> ModelConfig:
+    # LayerNorm + parallel residual + partial rotary + ungated GELU MLP
+    return ModelConfig(
+        name="tiny-neox", vocab_size=1024, hidden_size=256,
+        intermediate_size=1024, num_hidden_layers=4, num_attention_heads=4,
+        num_key_value_heads=4, head_dim=64, max_position_embeddings=512,
+        rotary_pct=0.25, rms_norm_eps=1e-5, architecture="neox")
+
+
+def _pythia_28b() -> ModelConfig:
+    return ModelConfig(
+        name="EleutherAI/pythia-2.8b", vocab_size=50304, hidden_size=2560,
+        intermediate_size=10240, num_hidden_layers=32,
+        num_attention_heads=32, num_key_value_heads=32, head_dim=80,
+        max_position_embeddings=2048, rotary_pct=0.25, rms_norm_eps=1e-5,
+        architecture="neox")
+
+
+def _pythia_69b() -> ModelConfig:
+    return ModelConfig(
+        name="EleutherAI/pythia-6.9b", vocab_size=50432, hidden_size=4096,
+        intermediate_size=16384, num_hidden_layers=32,
+        num_attention_heads=32, num_key_value_heads=32, head_dim=128,
+        max_position_embeddings=2048, rotary_pct=0.25, rms_norm_eps=1e-5,
+        architecture="neox")
+
+
 def _llama31_8b() -> ModelConfig:
     return ModelConfig(
         name="meta-llama/Llama-3.1-8B", vocab_size=128256,
@@ -274,6 +315,9 @@ PRESETS = {
     "tiny-moe": _tiny_moe,
     "tiny-qwen3-moe": _tiny_qwen3_moe,
     "tiny-bigvocab": _tiny_bigvocab,
+    "tiny-neox": _tiny_neox,
+    "EleutherAI/pythia-2.8b": _pythia_28b,
+    "EleutherAI/pythia-6.9b": _pythia_69b,
     "meta-llama/Llama-3.1-8B": _llama31_8b,
     "Qwen/Qwen3-30B-A3B": _qwen3_30b_a3b,
 }

@@ -374,4 +374,7 @@ def _stage_class(config: ModelConfig):
     if config.architecture == "gpt2":
         from tensorlink_amd.models.gpt2 import Gpt2StageModel
         return Gpt2StageModel
+    if config.architecture == "neox":
+        from tensorlink_amd.models.neox import NeoxStageModel
+        return NeoxStageModel
     return StageModel

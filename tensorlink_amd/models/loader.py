@@ -111,6 +111,17 @@ def load_stage_from_checkpoint(stage: StageModel, ckpt_dir: str,
                                device="cpu", dtype=torch.bfloat16) -> int:
     """Load only this stage's parameters from a local HF checkpoint dir.
     Returns the number of tensors loaded."""
+    arch = stage.config.architecture
+    if arch == "gpt2":
+        from tensorlink_amd.models.gpt2 import load_gpt2_hf_weights
+        n = load_gpt2_hf_weights(stage, ckpt_dir, dtype=dtype)
+        stage.to(device=device, dtype=dtype)
+        return n
+    if arch == "neox":
+        from tensorlink_amd.models.neox import load_neox_hf_weights
+        n = load_neox_hf_weights(stage, ckpt_dir, dtype=dtype)
+        stage.to(device=device, dtype=dtype)
+        return n
     from safetensors import safe_open
 
     index_path = os.path.join(ckpt_dir, "model.safetensors.index.json")

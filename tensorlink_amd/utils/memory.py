@@ -54,7 +54,8 @@ def layer_param_bytes(config, dtype_bytes: int) -> int:
                * config.expert_intermediate_size)
         mlp += h * config.num_local_experts
     else:
-        mlp = 3 * h * config.intermediate_size
+        gated = getattr(config, "gated_mlp", True)
+        mlp = (3 if gated else 2) * h * config.intermediate_size
     return (attn + mlp + 2 * h) * dtype_bytes
 
 

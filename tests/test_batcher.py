@@ -532,7 +532,8 @@ def test_gpt2_through_batcher():
 
 
 @pytest.mark.parametrize("name", ["tiny-moe", "tiny-qwen3",
-                                  "tiny-qwen3-moe", "gpt2-small"])
+                                  "tiny-qwen3-moe", "gpt2-small",
+                                  "tiny-neox"])
 def test_all_families_through_batcher(name):
     """Every model family serves through the batcher (chunked prefill +
     prefix caching) and matches serial greedy; speculation is exact."""
