@@ -234,11 +234,81 @@ class MoEMLP(nn.Module):
         self.experts = nn.ModuleList([MLP(expert_cfg)
                                       for _ in range(self.num_experts)])
 
+    def _fused_kind(self):
+        """"bf16" / "fp8" when every expert is uniform and the grouped
+        kernel (ops/csrc/moe_gemm.hip) can serve this block; None falls
+        back to the per-expert loop (CPU, training, fp4)."""
+        from tensorlink_amd.models.quant import Fp8Linear
+        e0 = self.experts[0]
+        if isinstance(e0.gate_up_proj, Fp8Linear):
+            return "fp8"
+        if isinstance(e0.gate_up_proj, nn.Linear) \
+                and e0.gate_up_proj.bias is None:
+            return "bf16"
+        return None
+
+    def _tables(self, dev, fp8):
+        """Device pointer tables into the per-expert weights (no stacked
+        copy); rebuilt if weights moved."""
+        gus = [(e.gate_up_proj.weight_fp8 if fp8 else e.gate_up_proj.weight)
+               for e in self.experts]
+        key = (str(dev), fp8, gus[0].data_ptr())
+        if getattr(self, "_tbl_key", None) != key:
+            downs = [(e.down_proj.weight_fp8 if fp8 else e.down_proj.weight)
+                     for e in self.experts]
+            mk = lambda ts: torch.tensor([t.data_ptr() for t in ts],
+                                         dtype=torch.int64, device=dev)
+            self._gu_ptrs = mk(gus)
+            self._down_ptrs = mk(downs)
+            if fp8:
+                self._gu_scales = mk([e.gate_up_proj.scale
+                                      for e in self.experts])
+                self._down_scales = mk([e.down_proj.scale
+                                        for e in self.experts])
+            else:
+                self._gu_scales = self._down_scales = None
+            self._tbl_key = key
+        return (self._gu_ptrs, self._down_ptrs, self._gu_scales,
+                self._down_scales)
+
+    def _fused_forward(self, flat, rw, idx, fp8):
+        """Grouped expert GEMMs over expert-sorted (token, slot) pairs;
+        deterministic combine: inverse-permutation gather then a fixed
+        slot-order sum (no atomics — batcher greedy equality holds for
+        MoE models too). hipGraph-capture-safe end to end."""
+        C = ops._require_ext()
+        E, k = self.num_experts, self.top_k
+        T, H = flat.shape
+        P = T * k
+        dev = flat.device
+        gu_p, down_p, gu_s, down_s = self._tables(dev, fp8)
+        fi = idx.reshape(-1)
+        order = fi.argsort(stable=True)
+        seg = torch.zeros(E + 1, device=dev, dtype=torch.int32)
+        seg[1:] = torch.bincount(fi, minlength=E).cumsum(0)
+        pair_tok = (order // k).to(torch.int32)
+        Ie = self.experts[0].inter
+        gu = C.moe_gemm(flat.contiguous(), pair_tok, seg, gu_p, gu_s,
+                        2 * Ie, fp8)
+        act = ops.swiglu_fused(gu)
+        y_pairs = C.moe_gemm(act, None, seg, down_p, down_s, H, fp8)
+        y_pairs = y_pairs * rw.reshape(-1)[order].unsqueeze(1)
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(P, device=dev, dtype=order.dtype)
+        return y_pairs[inv].view(T, k, H).sum(1)
+
     def forward(self, x):
         B, S, H = x.shape
         flat = x.reshape(-1, H)
         weights, idx = ops.moe_topk_router(self.gate(flat), self.top_k)
         weights = weights.to(x.dtype)
+        if (flat.is_cuda and flat.dtype == torch.bfloat16
+                and not torch.is_grad_enabled() and ops.extension_loaded()):
+            kind = self._fused_kind()
+            if kind is not None:
+                out = self._fused_forward(flat, weights, idx,
+                                          kind == "fp8")
+                return out.reshape(B, S, H)
         out = torch.zeros_like(flat)
         for e in range(self.num_experts):
             mask = (idx == e)

@@ -54,6 +54,9 @@ void tl_sample(const void* logits, const void* temps, const void* top_ps,
                void* counts, const void* seeds, const void* ctr, void* out,
                uint64_t seed_base, int B, int V, hipStream_t stream);
 void tl_bump_counter(void* ctr, hipStream_t stream);
+void tl_moe_gemm(const void* x, const void* tok_idx, const void* seg_off,
+                 const void* w_ptrs, const void* s_ptrs, void* out, int E,
+                 int N, int K, int fp8, hipStream_t stream);
 }
 
 namespace {
@@ -395,6 +398,37 @@ void bump_sample_counter(Tensor ctr) {
   tl_bump_counter(ctr.data_ptr(), cur_stream());
 }
 
+Tensor moe_gemm(Tensor x, c10::optional<Tensor> tok_idx, Tensor seg_off,
+                Tensor w_ptrs, c10::optional<Tensor> s_ptrs, int64_t N,
+                bool fp8) {
+  CHECK_IN(x, torch::kBFloat16);
+  CHECK_IN(seg_off, torch::kInt);
+  CHECK_IN(w_ptrs, torch::kLong);
+  const int K = x.size(-1);
+  TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "unsupported MoE shape");
+  const int E = w_ptrs.size(0);
+  TORCH_CHECK(seg_off.size(0) == E + 1, "seg_off must be [E+1]");
+  const void* ti = nullptr;
+  int64_t P;
+  if (tok_idx.has_value()) {
+    CHECK_IN(tok_idx.value(), torch::kInt);
+    ti = tok_idx->data_ptr();
+    P = tok_idx->size(0);
+  } else {
+    P = x.size(0);
+  }
+  const void* sp = nullptr;
+  if (fp8) {
+    TORCH_CHECK(s_ptrs.has_value(), "fp8 needs scale pointers");
+    CHECK_IN(s_ptrs.value(), torch::kLong);
+    sp = s_ptrs->data_ptr();
+  }
+  auto out = torch::empty({P, N}, x.options());
+  tl_moe_gemm(x.data_ptr(), ti, seg_off.data_ptr(), w_ptrs.data_ptr(), sp,
+              out.data_ptr(), E, (int)N, K, fp8 ? 1 : 0, cur_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -414,4 +448,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused penalties/temperature/top-k/top-p sampling, one block/row");
   mod.def("bump_sample_counter", &bump_sample_counter,
           "advance the graph-safe sampling RNG counter");
+  mod.def("moe_gemm", &moe_gemm,
+          "grouped expert GEMM over expert-sorted pair rows");
 }

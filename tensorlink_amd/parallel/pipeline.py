@@ -99,10 +99,14 @@ class PipelineRunner:
                                                   "contiguous")
         self._no_graph = False
         if quantize == "fp8":
+            from tensorlink_amd import ops as _ops
             from tensorlink_amd.models.quant import quantize_experts_fp8
             quantize_experts_fp8(self.stage)
-            # torch._scaled_mm is not hipGraph-capture-safe on ROCm 7.2
-            self._no_graph = True
+            # fp8 experts run the grouped moe_gemm kernel (in-kernel
+            # dequant, capture-safe); only without the extension would
+            # they fall to torch._scaled_mm, which cannot be captured
+            if not _ops.extension_loaded():
+                self._no_graph = True
         elif quantize == "fp8-dense":
             from tensorlink_amd.models.quant import quantize_dense_fp8
             quantize_dense_fp8(self.stage)

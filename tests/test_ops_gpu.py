@@ -463,3 +463,90 @@ def test_graph_sampled_decode():
     assert r._decode_graph is not None      # captured, not eager fallback
     greedy = r.generate(ids, SamplingParams(max_new_tokens=12))
     assert not torch.equal(o1, greedy)
+
+
+# ---------------------------------------------------------------------------
+# Grouped MoE expert GEMM (ops/csrc/moe_gemm.hip)
+# ---------------------------------------------------------------------------
+@torch.no_grad()
+@pytest.mark.parametrize("fp8", [False, True])
+def test_moe_gemm_grouped_vs_loop(fp8):
+    """One grouped launch over expert-sorted pairs equals the per-expert
+    GEMM loop (bf16 exact vs the family kernel; fp8 vs dequant mm)."""
+    ops = _ext()
+    C = ops._require_ext()
+    torch.manual_seed(50)
+    E, K, N, T, k = 4, 256, 512, 33, 2
+    x = torch.randn(T, K, device=DEV, dtype=torch.bfloat16)
+    ws = [torch.randn(N, K, device=DEV, dtype=torch.bfloat16) / K ** 0.5
+          for _ in range(E)]
+    idx = torch.randint(0, E, (T, k), device=DEV)
+    fi = idx.reshape(-1)
+    order = fi.argsort(stable=True)
+    seg = torch.zeros(E + 1, device=DEV, dtype=torch.int32)
+    seg[1:] = torch.bincount(fi, minlength=E).cumsum(0)
+    pair_tok = (order // k).to(torch.int32)
+    if fp8:
+        from tensorlink_amd.models.quant import quantize_fp8_per_channel
+        q = [quantize_fp8_per_channel(w) for w in ws]
+        wp = torch.tensor([a.data_ptr() for a, _ in q], dtype=torch.int64,
+                          device=DEV)
+        sp = torch.tensor([s.data_ptr() for _, s in q], dtype=torch.int64,
+                          device=DEV)
+        out = C.moe_gemm(x, pair_tok, seg, wp, sp, N, True)
+        for p in range(T * k):
+            e = int(fi[order[p]])
+            wq, sc = q[e]
+            ref = (x[pair_tok[p].long()].float()
+                   @ (wq.float() * sc[:, None]).t())
+            torch.testing.assert_close(out[p].float(), ref, atol=5e-2,
+                                       rtol=5e-2)
+    else:
+        wp = torch.tensor([w.data_ptr() for w in ws], dtype=torch.int64,
+                          device=DEV)
+        out = C.moe_gemm(x, pair_tok, seg, wp, None, N, False)
+        for p in range(T * k):
+            e = int(fi[order[p]])
+            # bitwise vs the family streaming kernel on the same row
+            row = C.skinny_gemm(x[pair_tok[p].long()].unsqueeze(0),
+                                ws[e], None)
+            assert torch.equal(out[p], row[0]), p
+
+
+@torch.no_grad()
+def test_moe_fused_block_matches_loop():
+    """MoEMLP's fused grouped path equals the per-expert loop."""
+    from tensorlink_amd.models.configs import get_config
+    from tensorlink_amd.models.dense import MoEMLP
+    _ext()
+    torch.manual_seed(51)
+    cfg = get_config("tiny-moe")
+    mlp = MoEMLP(cfg).to(DEV, torch.bfloat16).eval()
+    x = torch.randn(2, 9, cfg.hidden_size, device=DEV,
+                    dtype=torch.bfloat16)
+    fused = mlp(x)
+    assert mlp._fused_kind() == "bf16"
+    # reference: the python loop (force by pretending no uniform experts)
+    import tensorlink_amd.models.dense as dense_mod
+    orig = MoEMLP._fused_kind
+    MoEMLP._fused_kind = lambda self: None
+    try:
+        loop = mlp(x)
+    finally:
+        MoEMLP._fused_kind = orig
+    torch.testing.assert_close(fused.float(), loop.float(), atol=3e-2,
+                               rtol=3e-2)
+
+
+def test_moe_fp8_graph_decode():
+    """Quantized-MoE decode is hipGraph-captured with the grouped kernel
+    (round 1 disabled graphs for fp8 because of _scaled_mm)."""
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    from tensorlink_amd.parallel.planner import plan_for_world
+    r = PipelineRunner(plan_for_world("tiny-moe", 1), 0, 1, device=DEV,
+                       dtype=torch.bfloat16, seed=6, quantize="fp8")
+    torch.manual_seed(52)
+    ids = torch.randint(0, 1024, (2, 12))
+    out = r.generate(ids, SamplingParams(max_new_tokens=8))
+    assert out.shape == (2, 8)
+    assert r._decode_graph is not None, "fp8 MoE decode fell back to eager"
