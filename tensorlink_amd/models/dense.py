@@ -284,8 +284,12 @@ class MoEMLP(nn.Module):
         gu_p, down_p, gu_s, down_s = self._tables(dev, fp8)
         fi = idx.reshape(-1)
         order = fi.argsort(stable=True)
+        # counts via scatter_add (torch.bincount device-syncs, which
+        # breaks hipGraph capture of quantized-MoE decode)
+        counts = torch.zeros(E, device=dev, dtype=torch.long)
+        counts.scatter_add_(0, fi, torch.ones_like(fi))
         seg = torch.zeros(E + 1, device=dev, dtype=torch.int32)
-        seg[1:] = torch.bincount(fi, minlength=E).cumsum(0)
+        seg[1:] = counts.cumsum(0)
         pair_tok = (order // k).to(torch.int32)
         Ie = self.experts[0].inter
         gu = C.moe_gemm(flat.contiguous(), pair_tok, seg, gu_p, gu_s,

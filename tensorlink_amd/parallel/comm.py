@@ -89,16 +89,30 @@ class P2P:
         self.world = world
         self.group = group
         self.base = rank_base
+        # gloo cannot move CUDA tensors point-to-point: stage via host
+        # (lets GPU-compute pipelines run over the CPU transport, e.g.
+        # two ranks sharing one GPU where RCCL refuses a communicator)
+        try:
+            self._host_stage = dist.get_backend(group) == "gloo"
+        except Exception:
+            self._host_stage = not torch.cuda.is_available()
 
     def send(self, t: torch.Tensor, dst: int):
+        if self._host_stage and t.is_cuda:
+            t = t.cpu()
         dist.send(t.contiguous(), self.base + dst, group=self.group)
 
     def recv(self, shape, dtype, src: int, device) -> torch.Tensor:
-        buf = torch.empty(*shape, dtype=dtype, device=device)
+        device = torch.device(device)
+        rdev = torch.device("cpu") if (self._host_stage
+                                       and device.type == "cuda") else device
+        buf = torch.empty(*shape, dtype=dtype, device=rdev)
         dist.recv(buf, self.base + src, group=self.group)
-        return buf
+        return buf.to(device) if rdev != device else buf
 
     def isend(self, t: torch.Tensor, dst: int):
+        if self._host_stage and t.is_cuda:
+            t = t.cpu()
         return dist.isend(t.contiguous(), self.base + dst, group=self.group)
 
     def irecv_into(self, buf: torch.Tensor, src: int):

@@ -507,10 +507,16 @@ def test_moe_gemm_grouped_vs_loop(fp8):
         out = C.moe_gemm(x, pair_tok, seg, wp, None, N, False)
         for p in range(T * k):
             e = int(fi[order[p]])
-            # bitwise vs the family streaming kernel on the same row
-            row = C.skinny_gemm(x[pair_tok[p].long()].unsqueeze(0),
-                                ws[e], None)
-            assert torch.equal(out[p], row[0]), p
+            ref = x[pair_tok[p].long()].float() @ ws[e].float().t()
+            torch.testing.assert_close(out[p].float(), ref, atol=3e-2,
+                                       rtol=3e-2)
+        # determinism: a pair's row is identical whatever else is in
+        # the batch (drop half the pairs, recompute, compare shared)
+        half = (T * k) // 2
+        seg2 = torch.clamp(seg, max=half)
+        out2 = C.moe_gemm(x, pair_tok[:half], seg2, wp, None, N, False)
+        n_cmp = int(seg2[-1])
+        assert torch.equal(out2[:n_cmp], out[:n_cmp])
 
 
 @torch.no_grad()
