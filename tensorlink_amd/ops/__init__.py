@@ -300,8 +300,26 @@ GEMM_M_MAX = int(_os.environ.get("TL_GEMM_M_MAX", "512"))
 gemm_dispatch_count = 0
 
 
+# (N,K)-pure shape classes; TL_GEMM_LIB_CLASSES routes whole classes to
+# hipBLASLt for in-pipeline A/B (a class flips for ALL M at once, so the
+# bitwise M-independence guarantee is preserved per process)
+_LIB_CLASSES = frozenset(
+    c for c in _os.environ.get("TL_GEMM_LIB_CLASSES", "").split(",") if c)
+
+
+def _gemm_class(N: int, K: int) -> str:
+    if N >= 65536:
+        return "wide"
+    if N >= 16384:
+        return "gateup"
+    if K >= 8192:
+        return "deepk"
+    return "narrow"
+
+
 def _use_tl_gemm(M: int, N: int, K: int) -> bool:
-    return K % 32 == 0 and N % 64 == 0 and M <= GEMM_M_MAX
+    return (K % 32 == 0 and N % 64 == 0 and M <= GEMM_M_MAX
+            and _gemm_class(N, K) not in _LIB_CLASSES)
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor,

@@ -305,19 +305,29 @@ Tensor swiglu_fused(Tensor gu) {
 // kernels. It depends only on (N, K) — never M — so a row's accumulation
 // (split boundaries + s-ordered reduce) is bitwise identical at every M:
 // the serving engine's exact-greedy guarantee rests on this.
+// split-block target (~blocks before M/z tiling); env-tunable for
+// in-pipeline A/B — read ONCE so the policy stays (N,K)-pure in-process
+static int gemm_split_target() {
+  static int t = [] {
+    const char* e = getenv("TL_GEMM_SPLIT_TARGET");
+    return e ? atoi(e) : 256;
+  }();
+  return t;
+}
+
 static int gemm_n_split(int N, int K) {
   if (N >= 65536) {
     // 256x256-tile class (gemm_tiled_sq, lm_head widths): enough
     // column panels to fill the chip without splitting
     int n_split = 1;
-    while (((N + 255) / 256) * n_split < 256 && n_split < 16 &&
-           (K / 64) / (n_split * 2) >= 16)
+    while (((N + 255) / 256) * n_split < gemm_split_target() &&
+           n_split < 16 && (K / 64) / (n_split * 2) >= 16)
       n_split <<= 1;
     return n_split;
   }
-  // split K so (N/64)*n_split lands near 1 block/CU (guide: ~0.5-1x CUs)
+  // split K so (N/64)*n_split lands near the target block count
   int n_split = 1;
-  while ((N / 64) * n_split < 256 && n_split < 8 &&
+  while ((N / 64) * n_split < gemm_split_target() && n_split < 8 &&
          (K / 32) / (n_split * 2) >= 2)
     n_split <<= 1;
   return n_split;
