@@ -303,8 +303,13 @@ gemm_dispatch_count = 0
 # (N,K)-pure shape classes; TL_GEMM_LIB_CLASSES routes whole classes to
 # hipBLASLt for in-pipeline A/B (a class flips for ALL M at once, so the
 # bitwise M-independence guarantee is preserved per process)
+# default: the gateup class (16k <= N < 64k) routes to hipBLASLt — the
+# BN=64/256 custom structures both lose ~40 us there in-pipeline
+# (profiles/gemm_family_ab.md); all other classes are custom, keeping
+# bitwise M-independence by construction where the tests assert it
 _LIB_CLASSES = frozenset(
-    c for c in _os.environ.get("TL_GEMM_LIB_CLASSES", "").split(",") if c)
+    c for c in _os.environ.get("TL_GEMM_LIB_CLASSES",
+                               "gateup").split(",") if c)
 
 
 def _gemm_class(N: int, K: int) -> str:

@@ -169,6 +169,9 @@ class PipelineRunner:
         Returns generated tokens [B, max_new] on the FIRST rank, else None.
         """
         sp = sampling or SamplingParams()
+        if sp.seed is not None:
+            # a seeded call reproduces regardless of what ran before
+            self._gen = None
         # agree on shapes AND micro-batch split (rank 0 is authoritative)
         if self.world > 1:
             meta = None
@@ -522,7 +525,9 @@ class PipelineRunner:
             dev = self.device
             if sp.temperature > 0:
                 C = ops._require_ext()
-                ctr = torch.zeros(1, device=dev, dtype=torch.int64)
+                self._sample_ctr = torch.zeros(1, device=dev,
+                                               dtype=torch.int64)
+                ctr = self._sample_ctr
                 temps = torch.full((B,), float(sp.temperature), device=dev)
                 tps = torch.full((B,), float(sp.top_p), device=dev)
                 tks = torch.full((B,), int(sp.top_k), device=dev,
@@ -559,6 +564,10 @@ class PipelineRunner:
             cache.seq_lens.copy_(saved_lens)
             self._decode_graph = (graph, tok_buf, pos_buf, skey)
         graph, tok_buf, pos_buf = self._decode_graph[:3]
+        if sp.temperature > 0 and getattr(self, "_sample_ctr",
+                                          None) is not None:
+            # reset the RNG counter so a seeded generate() reproduces
+            self._sample_ctr.zero_()
         tok_buf.copy_(cur)
         pos_buf.copy_(positions)
         for t in range(1, T):
