@@ -221,13 +221,46 @@ def attention_prefill(q, k, v, causal: bool = True,
                                        q_off)
 
 
+class _FlashAttnFn(torch.autograd.Function):
+    """Hand-written flash attention for training (gfx950): forward =
+    prefill kernel saving the logsumexp; backward = the two MFMA kernels
+    in ops/csrc/flash_bwd.hip (dKV over key tiles, dQ over query tiles)
+    with Delta = rowsum(dO*O) precomputed in fp32."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        C = _require_ext()
+        out, lse = C.prefill_attn_lse(q.contiguous(), k.contiguous(),
+                                      v.contiguous(), scale, causal)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        C = _require_ext()
+        q, k, v, out, lse = ctx.saved_tensors
+        delta = (dout.float() * out.float()).sum(-1)          # [B,S,Hq]
+        delta = delta.permute(0, 2, 1).contiguous()           # [B,Hq,S]
+        dq, dk, dv = C.attn_bwd(q.contiguous(), k.contiguous(),
+                                v.contiguous(), dout.contiguous(), lse,
+                                delta, ctx.scale, ctx.causal)
+        return dq, dk, dv, None, None
+
+
 def attention_train(q, k, v, causal: bool = True,
                     scale: Optional[float] = None) -> torch.Tensor:
-    """Training attention with autograd: composed torch SDPA (backed by
-    hipBLASLt GEMMs + ROCm flash backward). Hand-written backward kernels
-    are a later milestone; forward inference never uses this."""
+    """Training attention with autograd. GPU bf16 runs the hand-written
+    flash forward+backward kernels; CPU (and unsupported head dims)
+    composes torch SDPA."""
     B, S, Hq, D = q.shape
     Hkv = k.shape[2]
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if (q.is_cuda and q.dtype == torch.bfloat16 and _C is not None
+            and D in (64, 128) and k.shape[1] == S):
+        return _FlashAttnFn.apply(q, k, v, causal, scale)
     qt = q.transpose(1, 2)
     kt = k.transpose(1, 2)
     vt = v.transpose(1, 2)

@@ -57,6 +57,15 @@ void tl_bump_counter(void* ctr, hipStream_t stream);
 void tl_moe_gemm(const void* x, const void* tok_idx, const void* seg_off,
                  const void* w_ptrs, const void* s_ptrs, void* out, int E,
                  int N, int K, int fp8, hipStream_t stream);
+void tl_prefill_attn_lse(const void* q, const void* k, const void* v,
+                         void* out, void* lse, int B, int Sq, int Skv,
+                         int q_off, int Hq, int Hkv, int D, float scale,
+                         int causal, hipStream_t stream);
+void tl_attn_bwd(const void* q, const void* k, const void* v,
+                 const void* dout, const void* lse, const void* delta,
+                 void* dq, void* dk, void* dv, int B, int S, int Hq,
+                 int Hkv, int D, float scale, int causal,
+                 hipStream_t stream);
 }
 
 namespace {
@@ -248,6 +257,46 @@ Tensor prefill_attn(Tensor q, Tensor k, Tensor v, double scale, bool causal,
                   B, Sq, Skv, (int)q_off, Hq, Hkv, D, (float)scale,
                   causal ? 1 : 0, cur_stream());
   return out;
+}
+
+std::vector<Tensor> prefill_attn_lse(Tensor q, Tensor k, Tensor v,
+                                     double scale, bool causal) {
+  CHECK_IN(q, torch::kBFloat16);
+  CHECK_IN(k, torch::kBFloat16);
+  CHECK_IN(v, torch::kBFloat16);
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  auto out = torch::empty_like(q);
+  auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat));
+  tl_prefill_attn_lse(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      out.data_ptr(), lse.data_ptr(), B, Sq, Skv, 0, Hq,
+                      Hkv, D, (float)scale, causal ? 1 : 0, cur_stream());
+  return {out, lse};
+}
+
+std::vector<Tensor> attn_bwd(Tensor q, Tensor k, Tensor v, Tensor dout,
+                             Tensor lse, Tensor delta, double scale,
+                             bool causal) {
+  CHECK_IN(q, torch::kBFloat16);
+  CHECK_IN(k, torch::kBFloat16);
+  CHECK_IN(v, torch::kBFloat16);
+  CHECK_IN(dout, torch::kBFloat16);
+  CHECK_IN(lse, torch::kFloat);
+  CHECK_IN(delta, torch::kFloat);
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(k.size(1) == S, "training attention needs Sq == Skv");
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  tl_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
+              lse.data_ptr(), delta.data_ptr(), dq.data_ptr(),
+              dk.data_ptr(), dv.data_ptr(), B, S, Hq, Hkv, D, (float)scale,
+              causal ? 1 : 0, cur_stream());
+  return {dq, dk, dv};
 }
 
 // qkv: [T, row_stride] fused projection output (q | k | v per row).
@@ -460,4 +509,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "advance the graph-safe sampling RNG counter");
   mod.def("moe_gemm", &moe_gemm,
           "grouped expert GEMM over expert-sorted pair rows");
+  mod.def("prefill_attn_lse", &prefill_attn_lse,
+          "causal GQA prefill attention returning (out, logsumexp)");
+  mod.def("attn_bwd", &attn_bwd,
+          "flash-attention backward: (dq, dk, dv) from saved lse/delta");
 }

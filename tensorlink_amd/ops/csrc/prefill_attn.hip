@@ -35,8 +35,9 @@ typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 template <int D>
 __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
-    const bf16* __restrict__ v, bf16* __restrict__ out, int B, int Sq,
-    int Skv, int q_off, int Hq, int Hkv, float scale, int causal) {
+    const bf16* __restrict__ v, bf16* __restrict__ out,
+    float* __restrict__ lse, int B, int Sq, int Skv, int q_off, int Hq,
+    int Hkv, float scale, int causal) {
   static_assert(D == 64 || D == 128);
   constexpr int KCH = D / 32;      // QK^T k-chunks
   constexpr int NSUB = BN / 16;    // score col tiles = 4
@@ -191,7 +192,7 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
     __syncthreads();  // before next tile overwrites K/VT
   }
 
-  // ---- epilogue: divide by l, store ----
+  // ---- epilogue: divide by l, store (+ logsumexp for backward) ----
   {
     const int64_t obase = ((int64_t)b * Sq) * Hq * D + (int64_t)h * D;
 #pragma unroll
@@ -203,6 +204,9 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
         for (int n = 0; n < NS_PV; ++n)
           out[obase + (int64_t)(q0 + row) * Hq * D + n * 16 + col] =
               f2bf(oacc[n][r] * linv);
+        if (lse && col == 0)
+          lse[((int64_t)b * Hq + h) * Sq + q0 + row] =
+              l_run[r] > 0.f ? m_run[r] + __logf(l_run[r]) : -1e30f;
       }
     }
   }
@@ -212,20 +216,33 @@ __global__ __launch_bounds__(BLOCK) void prefill_attn_kernel(
 
 extern "C" {
 
+void tl_prefill_attn_lse(const void* q, const void* k, const void* v,
+                         void* out, void* lse, int B, int Sq, int Skv,
+                         int q_off, int Hq, int Hkv, int D, float scale,
+                         int causal, hipStream_t stream);
+
 void tl_prefill_attn(const void* q, const void* k, const void* v, void* out,
                      int B, int Sq, int Skv, int q_off, int Hq, int Hkv,
                      int D, float scale, int causal, hipStream_t stream) {
+  tl_prefill_attn_lse(q, k, v, out, nullptr, B, Sq, Skv, q_off, Hq, Hkv, D,
+                      scale, causal, stream);
+}
+
+void tl_prefill_attn_lse(const void* q, const void* k, const void* v,
+                         void* out, void* lse, int B, int Sq, int Skv,
+                         int q_off, int Hq, int Hkv, int D, float scale,
+                         int causal, hipStream_t stream) {
   dim3 grid((Sq + BM - 1) / BM, Hq, B), block(BLOCK);
   if (D == 128)
     hipLaunchKernelGGL((prefill_attn_kernel<128>), grid, block, 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (bf16*)out, B, Sq, Skv, q_off, Hq, Hkv, scale,
-                       causal);
+                       (bf16*)out, (float*)lse, B, Sq, Skv, q_off, Hq, Hkv,
+                       scale, causal);
   else if (D == 64)
     hipLaunchKernelGGL((prefill_attn_kernel<64>), grid, block, 0, stream,
                        (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                       (bf16*)out, B, Sq, Skv, q_off, Hq, Hkv, scale,
-                       causal);
+                       (bf16*)out, (float*)lse, B, Sq, Skv, q_off, Hq, Hkv,
+                       scale, causal);
 }
 
 }  // extern "C"

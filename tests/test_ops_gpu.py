@@ -556,3 +556,60 @@ def test_moe_fp8_graph_decode():
     out = r.generate(ids, SamplingParams(max_new_tokens=8))
     assert out.shape == (2, 8)
     assert r._decode_graph is not None, "fp8 MoE decode fell back to eager"
+
+
+# ---------------------------------------------------------------------------
+# Flash-attention backward (ops/csrc/flash_bwd.hip)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("B,S,Hq,Hkv,D,causal", [
+    (2, 128, 8, 2, 128, True),    # GQA G=4
+    (1, 200, 4, 4, 64, True),     # MHA, ragged S, D=64
+    (2, 64, 28, 4, 128, True),    # Qwen2.5-7B head shape
+    (1, 96, 8, 8, 128, False),    # non-causal
+])
+def test_flash_attn_backward(B, S, Hq, Hkv, D, causal):
+    """Hand-written flash fwd+bwd vs fp32 autograd reference."""
+    ops = _ext()
+    torch.manual_seed(60)
+    mk = lambda *s: (torch.randn(*s, device=DEV, dtype=torch.bfloat16)
+                     .requires_grad_(True))
+    q, k, v = mk(B, S, Hq, D), mk(B, S, Hkv, D), mk(B, S, Hkv, D)
+    out = ops.attention_train(q, k, v, causal=causal)
+    # the custom path must actually be active
+    assert out.grad_fn is not None \
+        and "FlashAttn" in type(out.grad_fn).__name__
+    dout = torch.randn_like(out)
+    out.backward(dout)
+
+    q32 = q.detach().float().requires_grad_(True)
+    k32 = k.detach().float().requires_grad_(True)
+    v32 = v.detach().float().requires_grad_(True)
+    rep = Hq // Hkv
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q32.transpose(1, 2),
+        k32.transpose(1, 2).repeat_interleave(rep, dim=1),
+        v32.transpose(1, 2).repeat_interleave(rep, dim=1),
+        is_causal=causal, scale=1.0 / math.sqrt(D)).transpose(1, 2)
+    torch.testing.assert_close(out.float(), ref.detach(), atol=2e-2,
+                               rtol=2e-2)
+    ref.backward(dout.float())
+    torch.testing.assert_close(q.grad.float(), q32.grad, atol=7e-2,
+                               rtol=7e-2)
+    torch.testing.assert_close(k.grad.float(), k32.grad, atol=7e-2,
+                               rtol=7e-2)
+    torch.testing.assert_close(v.grad.float(), v32.grad, atol=7e-2,
+                               rtol=7e-2)
+
+
+def test_training_step_uses_flash_bwd():
+    """A PipelineTrainer step runs with the flash kernels on the path
+    and the loss decreases."""
+    from tensorlink_amd.parallel.pipeline import PipelineTrainer
+    from tensorlink_amd.parallel.planner import plan_for_world
+    t = PipelineTrainer(plan_for_world("tiny", 1, training=True), 0, 1,
+                        device=DEV, seed=1, lr=5e-3)
+    torch.manual_seed(61)
+    ids = torch.randint(0, 1024, (2, 64))
+    losses = [t.train_step(ids, labels=ids) for _ in range(4)]
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0], losses
