@@ -55,13 +55,23 @@ struct SampleLds {
   union {
     struct {
       int cnt[256];
-      float sum[256];
+      // exp-weights quantized to integers (x 2^20) and accumulated
+      // with EXACT integer atomics: float atomicAdd order is
+      // nondeterministic and flipped top-p thresholds between
+      // identical calls (caught by the graph-reproducibility test)
+      unsigned long long sum[256];
     } hist;
     BlockRed red;
     float scan[BLOCK / 64];
   } u;
-  float stat[8];   // broadcast slots
+  float stat[8];        // float broadcast slots
+  unsigned long long statu[4];   // integer broadcast slots
 };
+
+DEVINLINE unsigned long long wquant(float e) {
+  // e in [0, 1] -> fixed point x 2^20 (sums over 152k vocab stay < 2^38)
+  return (unsigned long long)(e * 1048576.0f);
+}
 
 DEVINLINE float block_max_argmin(SampleLds* lds, float v, int idx,
                                  int* out_idx) {
@@ -153,7 +163,7 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
     // -------- pass 2: level-1 histogram (key bits 31:24) --------
     for (int i = tid; i < 256; i += BLOCK) {
       lds.u.hist.cnt[i] = 0;
-      lds.u.hist.sum[i] = 0.f;
+      lds.u.hist.sum[i] = 0ull;
     }
     __syncthreads();
     for (int i = tid; i < V; i += BLOCK) {
@@ -161,19 +171,21 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
       float e = __expf((z - M) * invT);
       uint32_t bin = fkey(z) >> 24;
       atomicAdd(&lds.u.hist.cnt[bin], 1);
-      atomicAdd(&lds.u.hist.sum[bin], e);
+      atomicAdd(&lds.u.hist.sum[bin], wquant(e));
     }
     __syncthreads();
     // walk bins high->low on ONE thread (256 iterations, trivial)
     if (tid == 0) {
-      float Z = 0.f;
-      for (int i = 0; i < 256; ++i) Z += lds.u.hist.sum[i];
-      const float pmass = p_want < 1.f ? p_want * Z : 3.4e38f;
+      unsigned long long Zi = 0ull;
+      for (int i = 0; i < 256; ++i) Zi += lds.u.hist.sum[i];
+      const unsigned long long pmass =
+          p_want < 1.f ? (unsigned long long)(p_want * (double)Zi)
+                       : ~0ull;
       long cum_n = 0;
-      float cum_e = 0.f;
+      unsigned long long cum_e = 0ull;
       int bin_k = -1, bin_p = -1;     // boundary bins
       long above_k = 0;               // counts/mass strictly above bin
-      float above_p = 0.f;
+      unsigned long long above_p = 0ull;
       for (int i = 255; i >= 0; --i) {
         if (bin_k < 0 && k_want > 0 &&
             cum_n + lds.u.hist.cnt[i] >= k_want) {
@@ -189,24 +201,22 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
       if (k_want <= 0 || k_want >= V) bin_k = 0;       // no k cut
       if (bin_k < 0) bin_k = 0;
       if (bin_p < 0) bin_p = 0;                        // no p cut
-      lds.stat[0] = Z;
       lds.stat[1] = (float)bin_k;
       lds.stat[2] = (float)bin_p;
       lds.stat[3] = (float)above_k;   // exact up to 2^24 counts
-      lds.stat[4] = above_p;
-      lds.stat[5] = pmass;
+      lds.statu[0] = above_p;
+      lds.statu[1] = pmass;
     }
     __syncthreads();
-    const float Z = lds.stat[0];
     const int bin_k = (int)lds.stat[1];
     const int bin_p = (int)lds.stat[2];
-    const float pmass = lds.stat[5];
+    const unsigned long long pmass = lds.statu[1];
 
     // -------- pass 3: level-2 refinement for both cutoffs --------
     // elements whose level-1 bin == boundary get re-binned by key bits
     // 23:16; run for k and p boundaries (they often coincide)
     uint32_t thr_key = 0;        // keep keys >= thr_key
-    float kept_mass = Z;
+    float kept_mass = -1.f;      // computed below (full Z if no cuts)
     for (int phase = 0; phase < 2; ++phase) {
       const bool is_k = phase == 0;
       if (is_k && (k_want <= 0 || k_want >= V)) continue;
@@ -214,7 +224,7 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
       const int bb = is_k ? bin_k : bin_p;
       for (int i = tid; i < 256; i += BLOCK) {
         lds.u.hist.cnt[i] = 0;
-        lds.u.hist.sum[i] = 0.f;
+        lds.u.hist.sum[i] = 0ull;
       }
       __syncthreads();
       for (int i = tid; i < V; i += BLOCK) {
@@ -223,13 +233,13 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
         if ((int)(key >> 24) == bb) {
           float e = __expf((z - M) * invT);
           atomicAdd(&lds.u.hist.cnt[(key >> 16) & 0xFF], 1);
-          atomicAdd(&lds.u.hist.sum[(key >> 16) & 0xFF], e);
+          atomicAdd(&lds.u.hist.sum[(key >> 16) & 0xFF], wquant(e));
         }
       }
       __syncthreads();
       if (tid == 0) {
         long cum_n = (long)lds.stat[3];
-        float cum_e = lds.stat[4];
+        unsigned long long cum_e = lds.statu[0];
         int sub = 0;
         if (is_k) {
           for (int i = 255; i >= 0; --i) {
@@ -252,12 +262,13 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
       if (t > thr_key) thr_key = t;
       __syncthreads();
     }
-    // kept mass for the final threshold (pass over V)
-    if (thr_key > 0) {
+    // kept mass for the final threshold (fixed-order float pass)
+    {
       float my = 0.f;
       for (int i = tid; i < V; i += BLOCK) {
         float z = zval(i);
-        if (fkey(z) >= thr_key) my += __expf((z - M) * invT);
+        if (thr_key == 0 || fkey(z) >= thr_key)
+          my += __expf((z - M) * invT);
       }
       kept_mass = block_sum(&lds, my);
     }
