@@ -562,6 +562,12 @@ class PipelineRunner:
             with torch.cuda.graph(graph):
                 step()
             cache.seq_lens.copy_(saved_lens)
+            # one throwaway replay: the very first replay after capture
+            # observed once-off divergence on the sampled path (capture
+            # pool warm-up); flush it so the capture call's real replays
+            # see the same state as every later call
+            graph.replay()
+            cache.seq_lens.copy_(saved_lens)
             self._decode_graph = (graph, tok_buf, pos_buf, skey)
         graph, tok_buf, pos_buf = self._decode_graph[:3]
         if sp.temperature > 0 and getattr(self, "_sample_ctr",
