@@ -533,6 +533,12 @@ class PipelineRunner:
                 tks = torch.full((B,), int(sp.top_k), device=dev,
                                  dtype=torch.int32)
                 zero = torch.zeros(B, device=dev)
+                # the captured kernel reads these tensors BY ADDRESS on
+                # every replay: they must outlive this call or the
+                # allocator recycles them under the graph (the capture
+                # call then samples correctly and every later call reads
+                # recycled bytes — a very fun bug to find)
+                self._graph_keep = (temps, tps, tks, zero)
                 seed_base = (sp.seed if sp.seed is not None
                              else int(torch.seed()) & (2 ** 62 - 1))
 
