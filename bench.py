@@ -130,7 +130,10 @@ def main():
     input_ids = torch.randint(0, config.vocab_size, (B, S)) \
         if rank == 0 else None
     sp = SamplingParams(temperature=0.0, max_new_tokens=T)
-    n_mb = args.micro_batches or world
+    # serving default: 8 prefill micro-batches — requests in chunk m see
+    # their first token after m+1 chunks, so per-request TTFT p50 is
+    # ~9/16 of the monolithic-prefill TTFT at identical total work
+    n_mb = args.micro_batches or (8 if world == 1 else world)
 
     def sync():
         if world > 1:
@@ -149,8 +152,9 @@ def main():
                                    return_stats=True)
         if rank == 0:
             print(f"[step] {stats}", file=sys.stderr)
-        if stats.get("ttft_s") is not None:
-            ttfts.append(stats["ttft_s"])
+        tt = stats.get("ttft_p50_s", stats.get("ttft_s"))
+        if tt is not None:
+            ttfts.append(tt)
     sync()
     t1 = time.perf_counter()
 

@@ -15,6 +15,7 @@ models for generation, ``ml/worker.py:359``). Training uses 1F1B.
 
 from __future__ import annotations
 
+import statistics
 import time
 from dataclasses import dataclass, field
 from typing import List, Optional
@@ -192,6 +193,7 @@ class PipelineRunner:
 
         t_start = time.perf_counter()
         t_first: Optional[float] = None
+        t_firsts: List[float] = []    # per micro-batch (per-request TTFT)
 
         if self.is_first:
             input_ids = input_ids.to(self.device)
@@ -219,10 +221,11 @@ class PipelineRunner:
             else:
                 logits = self.stage.head(hidden[:, -1:]).squeeze(1)
                 tok = self._sample(logits, sp).to(torch.int64)
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
+                t_firsts.append(time.perf_counter())
                 if t_first is None:
-                    if self.device.type == "cuda":
-                        torch.cuda.synchronize(self.device)
-                    t_first = time.perf_counter()
+                    t_first = t_firsts[-1]
                 first_tokens.append(tok)
 
         self._trace("prefill", t_start, time.perf_counter(), batch=B,
@@ -288,7 +291,8 @@ class PipelineRunner:
             out = out_tokens[:, :n_out]
             self._trace("decode", t_dec0, time.perf_counter(), batch=B,
                         new_tokens=n_out)
-            stats = self._finish_stats(B, S, n_out, t_start, t_first)
+            stats = self._finish_stats(B, S, n_out, t_start, t_first,
+                                       t_firsts)
             if t_prefill_end is not None and stats is not None:
                 dec = stats["total_s"] - (t_prefill_end - t_start)
                 stats["prefill_s"] = t_prefill_end - t_start
@@ -351,7 +355,8 @@ class PipelineRunner:
                                    self.device)
         self._trace("decode", t_dec0, time.perf_counter(), batch=B,
                     new_tokens=T)
-        stats = self._finish_stats(B, S, T, t_start, t_first)
+        stats = self._finish_stats(B, S, T, t_start, t_first,
+                                   t_firsts)
         if return_stats:
             return result, stats
         return result
@@ -587,7 +592,8 @@ class PipelineRunner:
             out_tokens[:, t].copy_(tok_buf)
         positions.copy_(pos_buf)
 
-    def _finish_stats(self, B, S, T, t_start, t_first):
+    def _finish_stats(self, B, S, T, t_start, t_first,
+                      t_firsts=None):
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
         t_end = time.perf_counter()
@@ -595,6 +601,13 @@ class PipelineRunner:
             "batch": B, "prompt_len": S, "new_tokens": T,
             "total_s": t_end - t_start,
             "ttft_s": (t_first - t_start) if t_first is not None else None,
+            # honest per-REQUEST TTFT: micro-batch mb's requests see their
+            # first token after mb+1 prefill chunks; the median over
+            # requests is the middle micro-batch's timestamp
+            "ttft_p50_s": (statistics.median(t_firsts) - t_start
+                           if t_firsts else
+                           (t_first - t_start) if t_first is not None
+                           else None),
             "output_tokens_per_s": B * T / (t_end - t_start),
         }
         if self.world > 1:
