@@ -630,6 +630,26 @@ class PPContinuousBatcher(ContinuousBatcher):
 
     # ------------------------- shared execution -----------------------
     def _sample_rows(self, logits, samps):
+        """Sample the last rank's rows. GPU bf16 rows with sampling go
+        through the fused kernel in one launch; greedy rows and the CPU
+        tier use the torch path."""
+        if (logits.is_cuda and logits.dtype == torch.bfloat16
+                and ops.extension_loaded()
+                and any(t > 0 for t, _, _ in samps)):
+            dev = logits.device
+            temps = torch.tensor([t for t, _, _ in samps], device=dev,
+                                 dtype=torch.float32)
+            tps = torch.tensor([p for _, p, _ in samps], device=dev,
+                               dtype=torch.float32)
+            tks = torch.tensor([k for _, _, k in samps], device=dev,
+                               dtype=torch.int32)
+            zero = torch.zeros(len(samps), device=dev)
+            seeds = torch.tensor(
+                [self._rng.getrandbits(62) for _ in samps], device=dev,
+                dtype=torch.int64)
+            return ops.sample_tokens(logits.contiguous(), temps=temps,
+                                     top_ps=tps, top_ks=tks, pres=zero,
+                                     freqs=zero, seeds=seeds).tolist()
         return [int(ops.sample_token(logits[i:i + 1], temperature=t,
                                      top_p=p, top_k=k)[0])
                 for i, (t, p, k) in enumerate(samps)]
