@@ -69,7 +69,9 @@ class Fp8Linear(nn.Module):
                 pass
         # dequant fallback (CPU tier / missing fp8 GEMM support)
         w = self.weight_fp8.float() * self.scale[:, None]
-        out = torch.nn.functional.linear(x2.float(), w, self.bias)
+        out = torch.nn.functional.linear(
+            x2.float(), w,
+            self.bias.float() if self.bias is not None else None)
         return out.to(x.dtype).reshape(*shape, self.out_features)
 
 

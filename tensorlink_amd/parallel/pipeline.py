@@ -261,7 +261,9 @@ class PipelineRunner:
                 out_tokens[mb * b:(mb + 1) * b, 0] = first_tokens[mb]
 
         if self.world == 1:
-            t_prefill_end = t_first
+            # end of the WHOLE prefill (all micro-batches), not the
+            # first chunk — keeps the prefill/decode phase split honest
+            t_prefill_end = (t_firsts[-1] if t_firsts else t_first)
             cur = first_tokens[0] if n_mb == 1 else torch.cat(first_tokens)
             if on_token is not None:
                 on_token(0, cur)
