@@ -707,7 +707,11 @@ class PPContinuousBatcher(ContinuousBatcher):
         # decode: split the batch into up to `world` groups so successive
         # groups overlap across stages (rank s works on group g while
         # rank s-1 already runs group g+1 — all inter-stage transfers are
-        # isends, only recv blocks)
+        # isends, only recv blocks). A decode entry is
+        # (slot, len, tok, samp, prop): a speculating slot contributes
+        # 1+len(prop) ragged rows with per-row seq_lens over its page
+        # table — the same decode-kernel verify as the single-rank
+        # scheduler, executed in lockstep on every rank.
         groups = []
         if decodes:
             n_groups = min(r.world, len(decodes))
@@ -715,16 +719,19 @@ class PPContinuousBatcher(ContinuousBatcher):
             groups = [decodes[i:i + gsz]
                       for i in range(0, len(decodes), gsz)]
             for d in decodes:
-                self.cache.ensure(d[0], d[1] + 2)
+                self.cache.ensure(d[0], d[1] + len(d[4]) + 2)
         for sub in groups:
-            B = len(sub)
-            slots = [d[0] for d in sub]
-            lens = torch.tensor([d[1] for d in sub], device=dev,
-                                dtype=torch.int32)
-            view = _SlotView(self.cache, slots, lens)
+            rows_tok, rows_len, rows_slot = [], [], []
+            for slot, ln, tok, samp, prop in sub:
+                rows_tok += [tok] + list(prop)
+                rows_len += [ln + j for j in range(len(prop) + 1)]
+                rows_slot += [slot] * (len(prop) + 1)
+            B = len(rows_tok)
+            lens = torch.tensor(rows_len, device=dev, dtype=torch.int32)
+            view = _SlotView(self.cache, rows_slot, lens)
             pos = lens.unsqueeze(1)
             if r.is_first:
-                toks = torch.tensor([d[2] for d in sub], device=dev,
+                toks = torch.tensor(rows_tok, device=dev,
                                     dtype=torch.int64).unsqueeze(1)
                 hidden = self.stage(toks, pos, kv_cache=view,
                                     return_logits=False)
@@ -736,7 +743,18 @@ class PPContinuousBatcher(ContinuousBatcher):
                 self._isend(hidden, r.next_rank)
             else:
                 logits = self.stage.head(hidden).squeeze(1)
-                new = self._sample_rows(logits, [d[3] for d in sub])
+                # spec rows return raw argmax targets (rank 0 accepts);
+                # plain rows are sampled here
+                new = [0] * B
+                off = 0
+                for slot, ln, tok, samp, prop in sub:
+                    m = len(prop)
+                    if m > 0:
+                        new[off:off + m + 1] =                             logits[off:off + m + 1].argmax(-1).tolist()
+                    else:
+                        new[off] = self._sample_rows(
+                            logits[off:off + 1], [samp])[0]
+                    off += m + 1
                 self._isend(torch.tensor(new, dtype=torch.int64,
                                          device=dev), 0)
         # rank 0 collects the sampled tokens (isend on the last rank
@@ -750,7 +768,8 @@ class PPContinuousBatcher(ContinuousBatcher):
             if groups:
                 new_tokens = []
                 for sub in groups:
-                    new_tokens += r.p2p.recv((len(sub),), torch.int64,
+                    rows = sum(1 + len(d[4]) for d in sub)
+                    new_tokens += r.p2p.recv((rows,), torch.int64,
                                              r.world - 1, dev).tolist()
         return chunk_tok, new_tokens
 
@@ -780,6 +799,7 @@ class PPContinuousBatcher(ContinuousBatcher):
             req.done.set()
             self.slots[slot] = None
             self.slot_prompt[slot] = None
+            self.slot_ctx[slot] = None
             # defer the page release so every rank frees on the same step
             self._pending_releases.append(slot)
 
@@ -819,6 +839,7 @@ class PPContinuousBatcher(ContinuousBatcher):
                 self.slot_prompt[slot] = req.input_ids
                 self.slot_filled[slot] = matched
                 self.slot_len[slot] = 0
+                self.slot_ctx[slot] = None
                 self.slot_samp[slot] = (req.temperature, req.top_p,
                                         req.top_k)
                 admissions.append((slot, ids_list))
@@ -846,13 +867,23 @@ class PPContinuousBatcher(ContinuousBatcher):
                              prompt[start:end].tolist())
                     chunk_slot = slot
                     break
-            decodes = [(i, self.slot_len[i], self.slot_last[i],
-                        self.slot_samp[i])
-                       for i in range(self.max_slots)
-                       if self.slots[i] is not None
-                       and not self._prefilling(i)
-                       and self.slot_filled[i] > 0
-                       and i != chunk_slot]
+            from tensorlink_amd.parallel.pipeline import PipelineRunner
+            decodes = []
+            for i in range(self.max_slots):
+                if (self.slots[i] is None or self._prefilling(i)
+                        or self.slot_filled[i] == 0 or i == chunk_slot):
+                    continue
+                prop = []
+                if (self.speculative and self.slot_ctx[i] is not None
+                        and self.slot_samp[i][0] <= 0):
+                    req = self.slots[i]
+                    room = min(req.max_new_tokens - self.slot_emitted[i],
+                               self.max_ctx - 1 - self.slot_len[i]) - 1
+                    prop = PipelineRunner._lookup_propose(
+                        self.slot_ctx[i], self.lookup_n,
+                        self.spec_k)[:max(0, room)]
+                decodes.append((i, self.slot_len[i], self.slot_last[i],
+                                self.slot_samp[i], prop))
             if chunk is None and not decodes and not admissions \
                     and not self._pending_releases:
                 self._wake.wait(0.05)
@@ -871,15 +902,42 @@ class PPContinuousBatcher(ContinuousBatcher):
                     self.slot_len[slot] = s_total
                     self.slot_last[slot] = chunk_tok
                     self.slot_emitted[slot] = 1
+                    if self.speculative:
+                        self.slot_ctx[slot] =                             self.slot_prompt[slot].tolist() + [chunk_tok]
                     req.tokens.put(chunk_tok)
                     self._finish0(slot, chunk_tok)
             if new_tokens:
-                for (slot, _, _, _), tok in zip(decodes, new_tokens):
-                    self.slot_len[slot] += 1
-                    self.slot_emitted[slot] += 1
-                    self.slots[slot].tokens.put(tok)
-                    self.slot_last[slot] = tok
-                    self._finish0(slot, tok)
+                off = 0
+                for (slot, _, _, _, prop) in decodes:
+                    m = len(prop)
+                    seg = new_tokens[off:off + m + 1]
+                    off += m + 1
+                    if self.slots[slot] is None:
+                        continue
+                    if m > 0:
+                        a = 0
+                        while a < m and prop[a] == seg[a]:
+                            a += 1
+                        accepted = list(prop[:a]) + [seg[a]]
+                        self.spec_accepted += a
+                        self.slot_len[slot] += 1 + a
+                        for tok in accepted:
+                            self.slot_emitted[slot] += 1
+                            self.slots[slot].tokens.put(tok)
+                            self.slot_ctx[slot].append(tok)
+                            self.slot_last[slot] = tok
+                            self._finish0(slot, tok)
+                            if self.slots[slot] is None:
+                                break
+                    else:
+                        tok = seg[0]
+                        self.slot_len[slot] += 1
+                        self.slot_emitted[slot] += 1
+                        self.slots[slot].tokens.put(tok)
+                        if self.slot_ctx[slot] is not None:
+                            self.slot_ctx[slot].append(tok)
+                        self.slot_last[slot] = tok
+                        self._finish0(slot, tok)
                 self.steps += 1
         r.p2p.broadcast_obj(("batch_stop",), src=0)
         for w, _ in self._send_keep:
