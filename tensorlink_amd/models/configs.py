@@ -114,6 +114,16 @@ class ModelConfig:
             known = {f.name for f in dataclasses.fields(cls)}
             return cls(**{k: v for k, v in cfg.items() if k in known})
         archs = cfg.get("architectures") or []
+        # llama-SHAPED architectures map onto the llama family; anything
+        # not recognized raises loudly instead of silently mis-mapping
+        # (a Gemma/Phi config would otherwise run with wrong structure)
+        _known = ("Llama", "Mistral", "Qwen", "Mixtral", "GPTNeoX",
+                  "GPT2", "LlamaForCausalLM")
+        if archs and not any(k in a for a in archs for k in _known):
+            raise KeyError(
+                f"unsupported architecture(s) {archs}: the native zoo "
+                "covers Llama/Mistral-shaped, Qwen2, Qwen3(-MoE), "
+                "Mixtral, GPT-2 and GPT-NeoX/Pythia families")
         arch = "llama"
         if any("Qwen" in a for a in archs):
             arch = "qwen2"

@@ -853,3 +853,21 @@ def test_neox_family():
         assert n == len(state), (n, len(state))
         for name, p in full.state_dict().items():
             assert torch.equal(p, stage.state_dict()[name]), name
+
+
+def test_from_hf_config_rejects_unknown_architecture():
+    """An architecture the zoo does not implement raises loudly instead
+    of silently mapping onto the llama shape."""
+    import pytest as _pytest
+
+    from tensorlink_amd.models.configs import ModelConfig
+    with _pytest.raises(KeyError):
+        ModelConfig.from_hf_config(
+            {"architectures": ["Gemma2ForCausalLM"], "hidden_size": 256})
+    # Mistral is llama-shaped and maps fine
+    cfg = ModelConfig.from_hf_config(
+        {"architectures": ["MistralForCausalLM"], "hidden_size": 512,
+         "num_attention_heads": 8, "num_hidden_layers": 4,
+         "num_key_value_heads": 2, "intermediate_size": 1024,
+         "vocab_size": 32000})
+    assert cfg.architecture == "llama" and cfg.num_key_value_heads == 2
