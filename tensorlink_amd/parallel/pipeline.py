@@ -564,8 +564,9 @@ class PipelineRunner:
                 # call then samples correctly and every later call reads
                 # recycled bytes — a very fun bug to find)
                 self._graph_keep = (temps, tps, tks, zero)
+                import random as _random
                 seed_base = (sp.seed if sp.seed is not None
-                             else int(torch.seed()) & (2 ** 62 - 1))
+                             else _random.getrandbits(62))
 
                 def pick(logits):
                     C.bump_sample_counter(ctr)
@@ -601,9 +602,10 @@ class PipelineRunner:
             cache.seq_lens.copy_(saved_lens)
             self._decode_graph = (graph, tok_buf, pos_buf, skey)
         graph, tok_buf, pos_buf = self._decode_graph[:3]
-        if sp.temperature > 0 and getattr(self, "_sample_ctr",
-                                          None) is not None:
-            # reset the RNG counter so a seeded generate() reproduces
+        if (sp.temperature > 0 and sp.seed is not None
+                and getattr(self, "_sample_ctr", None) is not None):
+            # reset the RNG counter so a SEEDED generate() reproduces;
+            # unseeded calls keep advancing it (fresh draws per call)
             self._sample_ctr.zero_()
         tok_buf.copy_(cur)
         pos_buf.copy_(positions)
