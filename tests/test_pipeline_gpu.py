@@ -444,3 +444,33 @@ def test_serving_feature_composition_gpu():
         assert o1 == ref[0].cpu().tolist()
     finally:
         b.stop()
+
+
+def test_neox_family_gpu():
+    """GPT-NeoX family on the HIP path: partial-rotary rope kernel
+    (D_rot=16), LayerNorm blocks, parallel residual — generate is
+    well-formed and the batcher (chunked prefill + paged cache)
+    reproduces serial greedy exactly."""
+    from tensorlink_amd.engine.batcher import ContinuousBatcher
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner, SamplingParams
+    r = PipelineRunner(plan_for_world("tiny-neox", 1), 0, 1, device=DEV,
+                       dtype=torch.bfloat16, seed=5)
+    torch.manual_seed(71)
+    ids = torch.randint(0, 1024, (2, 20))
+    out = r.generate(ids, SamplingParams(max_new_tokens=8))
+    assert out.shape == (2, 8)
+    b = ContinuousBatcher(r, max_slots=2, max_ctx=256,
+                          prefill_chunk=16).start()
+    try:
+        p = torch.randint(0, 1024, (40,))
+        got = b.submit(p.clone(), max_new_tokens=6).result(timeout=120)
+        import os
+        os.environ["TL_NO_GRAPH"] = "1"
+        try:
+            ref = r.generate(p.unsqueeze(0), SamplingParams(max_new_tokens=6))
+        finally:
+            del os.environ["TL_NO_GRAPH"]
+        assert got == ref[0].cpu().tolist()
+    finally:
+        b.stop()
