@@ -1337,3 +1337,68 @@ def test_pp2_speculative_batching_matches_serial():
             toks.append(t)
             cur = torch.cat([cur, torch.tensor([[t]])], 1)
         assert o == toks, (o, toks)
+
+
+def _pp_compose_worker(rank, world, port, q):
+    sys.path.insert(0, REPO)
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    from tensorlink_amd.engine.batcher import PPContinuousBatcher
+    from tensorlink_amd.parallel.comm import init_distributed
+    from tensorlink_amd.parallel.planner import plan_for_world
+    from tensorlink_amd.parallel.pipeline import PipelineRunner
+    init_distributed(backend="gloo")
+    plan = plan_for_world("tiny", world)
+    r = PipelineRunner(plan, rank, world, device=torch.device("cpu"),
+                       seed=10)
+    b = PPContinuousBatcher(r, max_slots=2, max_ctx=512, prefill_chunk=64,
+                            prefix_caching=True, speculative=True)
+    if rank != 0:
+        b.serve_follower()
+        dist.destroy_process_group()
+        return
+    b.start()
+    torch.manual_seed(19)
+    p = torch.randint(0, 1024, (150,))
+    o1 = b.submit(p.clone(), max_new_tokens=30).result(timeout=120)
+    o2 = b.submit(p.clone(), max_new_tokens=30).result(timeout=120)
+    hits, acc = b.cache.hits, b.spec_accepted
+    b.stop()
+    q.put((rank, (p.tolist(), o1, o2, hits, acc)))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_pp2_spec_prefix_chunked_composition():
+    """The round-1 failing composition (chunked prefill + prefix caching
+    + speculation) on the PP batcher: repeat prompt reuses pages, the
+    lockstep ragged verify accepts proposals, and output equals serial
+    greedy."""
+    outs = dict(_run_collect(_pp_compose_worker, 2, _port(27), n_results=1))
+    prompt, o1, o2, hits, acc = outs[0]
+    # (acceptance count is sequence-dependent — this prompt's output has
+    # no 3-gram repeats; acc > 0 is pinned by the dedicated spec test)
+    assert o1 == o2 and hits == 128
+
+    import torch
+    from tensorlink_amd.models.dense import build_stage
+    from tensorlink_amd.models.loader import init_random_stage
+    from tensorlink_amd.parallel.planner import plan_for_world
+    plan = plan_for_world("tiny", 2)
+    stages = []
+    for rk in range(2):
+        st = build_stage(plan.config, plan.stage_for_rank(rk))
+        init_random_stage(st, dtype=torch.float32, seed=10 + rk)
+        stages.append(st)
+    cur = torch.tensor(prompt, dtype=torch.int64).unsqueeze(0)
+    toks = []
+    for _ in range(30):
+        pos = torch.arange(cur.shape[1]).unsqueeze(0)
+        h = stages[0](cur, pos.contiguous(), return_logits=False)
+        lg = stages[1](h, pos.contiguous())
+        t = int(lg[0, -1].argmax())
+        toks.append(t)
+        cur = torch.cat([cur, torch.tensor([[t]])], 1)
+    assert o1 == toks, (o1, toks)
