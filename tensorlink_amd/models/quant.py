@@ -47,9 +47,44 @@ class Fp8Linear(nn.Module):
         return cls(w_fp8, scale,
                    linear.bias.detach() if linear.bias is not None else None)
 
+    def _kernel_tables(self):
+        if getattr(self, "_ptr_key", None) != self.weight_fp8.data_ptr():
+            dev = self.weight_fp8.device
+            self._wp = torch.tensor([self.weight_fp8.data_ptr()],
+                                    dtype=torch.int64, device=dev)
+            self._sp = torch.tensor([self.scale.data_ptr()],
+                                    dtype=torch.int64, device=dev)
+            self._segs = {}
+            self._ptr_key = self.weight_fp8.data_ptr()
+        return self._wp, self._sp
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         shape = x.shape[:-1]
         x2 = x.reshape(-1, self.in_features)
+        from tensorlink_amd import ops
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and ops.extension_loaded()
+                and self.out_features % 64 == 0
+                and self.in_features % 32 == 0):
+            # weight-only fp8 through the grouped-GEMM kernel as a
+            # single-expert group: in-kernel e4m3 dequant + epilogue
+            # per-channel scales, hipGraph-capture-safe, bf16 x (more
+            # accurate than _scaled_mm's fp8-quantized activations)
+            M = x2.shape[0]
+            wp, sp = self._kernel_tables()
+            seg = self._segs.get(M)
+            if seg is None:
+                # per-M segment tensors persist (captured graphs read
+                # them by address; no device sync on the hot path)
+                seg = torch.tensor([0, M], dtype=torch.int32,
+                                   device=x.device)
+                self._segs[M] = seg
+            out = ops._require_ext().moe_gemm(
+                x2.contiguous(), None, seg, wp, sp,
+                self.out_features, True)
+            if self.bias is not None:
+                out = out + self.bias
+            return out.reshape(*shape, self.out_features)
         if x.is_cuda and hasattr(torch, "_scaled_mm"):
             try:
                 # dynamic per-tensor activation scale

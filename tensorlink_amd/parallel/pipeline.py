@@ -109,9 +109,15 @@ class PipelineRunner:
             if not _ops.extension_loaded():
                 self._no_graph = True
         elif quantize == "fp8-dense":
+            from tensorlink_amd import ops as _ops
             from tensorlink_amd.models.quant import quantize_dense_fp8
             quantize_dense_fp8(self.stage)
-            self._no_graph = True
+            # with the extension, fp8 projections run the grouped-GEMM
+            # in-kernel-dequant path (capture-safe); a capture attempt
+            # that still hits _scaled_mm falls back to eager via the
+            # existing RuntimeError catch
+            if not _ops.extension_loaded():
+                self._no_graph = True
         elif quantize == "fp4-dense":
             from tensorlink_amd.models.quant import quantize_dense_fp4
             quantize_dense_fp4(self.stage)
