@@ -150,3 +150,45 @@ def test_mxfp4_fuzz_bounds(N, groups, scale):
     assert bool((err <= grp.abs().amax(-1) * 0.26 + 1e-9).all())
     p2, e2 = quantize_mxfp4(wq)
     torch.testing.assert_close(dequantize_mxfp4(p2, e2), wq)
+
+
+@settings(max_examples=10, deadline=None)
+@given(lens=st.lists(st.integers(3, 160), min_size=1, max_size=4),
+       chunk=st.sampled_from([None, 16, 64]),
+       prefix=st.booleans(), new=st.integers(1, 16),
+       mix=st.booleans())
+def test_batcher_fuzz_speculative_ragged(lens, chunk, prefix, new, mix):
+    """The ragged speculative decode batch (greedy slots contribute 1+k
+    verify rows, optionally mixed with concurrent SAMPLED requests in
+    the same batch) always reproduces serial greedy for the greedy
+    requests, for any prompt mix / chunking / prefix caching."""
+    from tensorlink_amd.engine.batcher import ContinuousBatcher
+    from tensorlink_amd.parallel.pipeline import (PipelineRunner,
+                                                  SamplingParams)
+    r = PipelineRunner(plan_for_world("tiny", 1), 0, 1,
+                       device=torch.device("cpu"), seed=10)
+    b = ContinuousBatcher(r, max_slots=4, max_ctx=512,
+                          prefill_chunk=chunk, prefix_caching=prefix,
+                          speculative=True).start()
+    try:
+        g = torch.Generator().manual_seed(sum(lens) * 31 + new)
+        prompts = [torch.randint(0, 1024, (n,), generator=g)
+                   for n in lens]
+        reqs = [b.submit(p.clone(), max_new_tokens=new) for p in prompts]
+        noise = []
+        if mix:
+            # concurrent sampled + penalized requests share the ragged
+            # batch but must not perturb the greedy rows
+            noise = [b.submit(torch.randint(0, 1024, (24,), generator=g),
+                              max_new_tokens=new, temperature=0.9,
+                              seed=7, presence_penalty=0.3)
+                     for _ in range(2)]
+        outs = [rq.result(timeout=120) for rq in reqs]
+        for rq in noise:
+            assert len(rq.result(timeout=120)) == new
+        for p, o in zip(prompts, outs):
+            ref = r.generate(p.unsqueeze(0),
+                             SamplingParams(max_new_tokens=new))
+            assert o == ref[0].tolist()
+    finally:
+        b.stop()
