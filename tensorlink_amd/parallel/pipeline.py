@@ -341,10 +341,21 @@ class PipelineRunner:
                 w, _ = inflight.pop(0)
                 w.wait()
 
+        # hoist per-micro-batch cache views out of the tick loop: the
+        # contiguous views alias stable parent storage (seq_lens slices
+        # advance through the parent), and building them per tick was
+        # measurable host overhead at world>1 eager decode
+        from tensorlink_amd.models.paged import PagedKVCache
+        if isinstance(self.kv_cache, PagedKVCache):
+            mb_caches = None
+        else:
+            mb_caches = [_kv_slice(self.kv_cache, mb * b, (mb + 1) * b)
+                         for mb in range(n_mb)]
         for step in range(T - 1):
             for mb in range(n_mb):
                 s, e = mb * b, (mb + 1) * b
-                cache = _kv_slice(self.kv_cache, s, e)
+                cache = (mb_caches[mb] if mb_caches is not None
+                         else _kv_slice(self.kv_cache, s, e))
                 pos = positions[s:e].unsqueeze(1)
                 if self.is_first:
                     tok = self.p2p.recv((b,), torch.int64, self.world - 1,
