@@ -312,13 +312,14 @@ class DistributedModel(nn.Module):
     def generate(self, input_ids=None, max_new_tokens: int = 64,
                  temperature: float = 0.0, top_p: float = 1.0,
                  top_k: int = 0, do_sample: bool = False,
-                 eos_token_id: Optional[int] = None, **kw) -> torch.Tensor:
+                 eos_token_id: Optional[int] = None,
+                 seed: Optional[int] = None, **kw) -> torch.Tensor:
         if isinstance(input_ids, dict):
             input_ids = input_ids["input_ids"]
         sp = SamplingParams(
             temperature=temperature if do_sample else 0.0, top_p=top_p,
             top_k=top_k, max_new_tokens=max_new_tokens,
-            eos_token_id=eos_token_id)
+            eos_token_id=eos_token_id, seed=seed)
         if self.tp > 1 and self.world_size > 1:
             import torch.distributed as dist
             box = [("generate", input_ids.tolist(), sp)]
