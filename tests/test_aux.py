@@ -479,3 +479,27 @@ def test_serve_checkpoint_directory(tmp_path):
         cur = torch.cat([cur, lg[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(cur[:, 8:], out)
     eng.unload_model(d)
+
+
+def test_engine_two_models_concurrently():
+    """Two hosted jobs serve interleaved requests (the reference keeps a
+    job table of multiple hosted models — validator job lifecycle)."""
+    import torch
+
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(device=torch.device("cpu"))
+    eng.load_model("tiny")
+    eng.load_model("tiny-neox")
+    try:
+        torch.manual_seed(9)
+        for name in ("tiny", "tiny-neox", "tiny", "tiny-neox"):
+            out = eng.generate({"hf_name": name, "message": "ab",
+                                "max_new_tokens": 4,
+                                "output_format": "simple"})
+            assert isinstance(out.get("response", out.get("text", "")),
+                              str) or out
+        assert set(eng.jobs) >= {"tiny", "tiny-neox"}
+    finally:
+        eng.unload_model("tiny")
+        eng.unload_model("tiny-neox")
+        eng.shutdown()
