@@ -64,8 +64,13 @@ class Fp8Linear(nn.Module):
         from tensorlink_amd import ops
         if (x.is_cuda and x.dtype == torch.bfloat16
                 and ops.extension_loaded()
+                and x2.shape[0] <= 512
                 and self.out_features % 64 == 0
                 and self.in_features % 32 == 0):
+            # serving-M rows only: the grouped kernel fills the chip by
+            # column panels (N/64 blocks), fine for decode batches but
+            # pathological at prefill M — big-M stays on _scaled_mm /
+            # dequant below
             # weight-only fp8 through the grouped-GEMM kernel as a
             # single-expert group: in-kernel e4m3 dequant + epilogue
             # per-channel scales, hipGraph-capture-safe, bf16 x (more
