@@ -219,3 +219,33 @@ def test_reference_request_aliases():
                         "is_chat_completion": True})
     assert out.get("object") == "chat.completion"
     eng.unload_model("tiny")
+
+
+def test_http_speculative_continuous_with_seed():
+    """REST over the speculative continuous batcher: greedy requests
+    come back consistently (spec is exact); seeded sampled requests
+    reproduce across calls."""
+    import torch
+    from fastapi.testclient import TestClient
+
+    from tensorlink_amd.api.server import create_app
+    from tensorlink_amd.engine.engine import InferenceEngine
+    eng = InferenceEngine(rank=0, world=1, device=torch.device("cpu"))
+    eng.load_model("tiny", continuous=True, max_slots=4, max_ctx=256,
+                   prefill_chunk=32, speculative=True)
+    client = TestClient(create_app(eng))
+    try:
+        greedy = [client.post("/v1/generate", json={
+            "hf_name": "tiny", "message": "spec spec spec spec",
+            "max_new_tokens": 6, "do_sample": False,
+            "output_format": "simple"}).json() for _ in range(2)]
+        assert greedy[0]["response"] == greedy[1]["response"]
+        seeded = [client.post("/v1/generate", json={
+            "hf_name": "tiny", "message": "sample me",
+            "max_new_tokens": 6, "do_sample": True, "temperature": 0.9,
+            "seed": 123, "output_format": "simple"}).json()
+            for _ in range(2)]
+        assert seeded[0]["response"] == seeded[1]["response"]
+    finally:
+        eng.unload_model("tiny")
+        eng.shutdown()
