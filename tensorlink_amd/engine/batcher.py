@@ -464,6 +464,16 @@ class ContinuousBatcher:
         view = _SlotView(self.cache, rows_slot, lens)
         pos = lens.unsqueeze(1)
         logits = self.stage(toks.unsqueeze(1), pos, kv_cache=view).squeeze(1)
+        # sample all plain rows in ONE batched call (penalties/seeds per
+        # row inside _sample); speculative slots accept below
+        plain = [i for i in active if i not in props]
+        plain_toks = {}
+        if plain:
+            rows = logits[torch.tensor([seg[i][0] for i in plain],
+                                       device=logits.device)]
+            got = self._sample(rows, [self.slots[i] for i in plain],
+                               slots=plain)
+            plain_toks = dict(zip(plain, got))
         for i in active:
             off, m = seg[i]
             req = self.slots[i]
@@ -489,8 +499,7 @@ class ContinuousBatcher:
                     if self.slots[i] is None:
                         break
             else:
-                tok = self._sample(logits[off:off + 1], [req],
-                                   slots=[i])[0]
+                tok = plain_toks[i]
                 self.slot_len[i] += 1
                 self.slot_emitted[i] += 1
                 self.slot_counts[i][tok] = \
